@@ -1,0 +1,276 @@
+// Python bindings for megba_amd.
+// The Python layer supplies the raw problem (numpy arrays) and, for
+// multi-process CPU runs, an in-place allreduce callback (torch.distributed
+// gloo); the GPU path uses RCCL natively (gpu/gpu_engine.hip).
+#include <pybind11/functional.h>
+#include <pybind11/numpy.h>
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <memory>
+
+#include "megba/common.hpp"
+#include "megba/cpu_engine.hpp"
+#include "megba/engine.hpp"
+#include "megba/lm.hpp"
+#include "megba/problem.hpp"
+
+#ifdef MEGBA_WITH_GPU
+#include "megba/gpu/gpu_engine.hpp"
+#endif
+
+namespace py = pybind11;
+using namespace megba;
+
+namespace {
+
+template <typename T>
+HostAllreduce<T> wrapAllreduce(py::object fn) {
+  if (fn.is_none()) return nullptr;
+  // Keep the callable alive in the closure; reacquire the GIL per call.
+  auto holder = std::make_shared<py::object>(std::move(fn));
+  return [holder](T* data, std::size_t n) {
+    py::gil_scoped_acquire gil;
+    py::array arr(py::dtype::of<T>(), {(py::ssize_t)n}, {(py::ssize_t)sizeof(T)},
+                  data, py::none());
+    (*holder)(arr);
+  };
+}
+
+struct PyProblem {
+  BAProblemHost prob;
+  ProblemIndex ix;
+  ProblemOption opt;
+  std::unique_ptr<Engine<double>> engD;
+  std::unique_ptr<Engine<float>> engF;
+  bool isDouble = true;
+
+  PyProblem(py::array_t<double, py::array::c_style | py::array::forcecast> cams,
+            py::array_t<double, py::array::c_style | py::array::forcecast> pts,
+            py::array_t<int, py::array::c_style | py::array::forcecast> camIdx,
+            py::array_t<int, py::array::c_style | py::array::forcecast> ptIdx,
+            py::array_t<double, py::array::c_style | py::array::forcecast> meas,
+            py::object info) {
+    MEGBA_CHECK(cams.ndim() == 2 && cams.shape(1) == 9, "cams must be (ncam,9)");
+    MEGBA_CHECK(pts.ndim() == 2 && pts.shape(1) == 3, "pts must be (npt,3)");
+    MEGBA_CHECK(meas.ndim() == 2 && meas.shape(1) == 2, "meas must be (nobs,2)");
+    prob.ncam = (int)cams.shape(0);
+    prob.npt = (int)pts.shape(0);
+    prob.nobs = (int64_t)meas.shape(0);
+    prob.cams.assign(cams.data(), cams.data() + cams.size());
+    prob.pts.assign(pts.data(), pts.data() + pts.size());
+    prob.camIdx.assign(camIdx.data(), camIdx.data() + camIdx.size());
+    prob.ptIdx.assign(ptIdx.data(), ptIdx.data() + ptIdx.size());
+    prob.meas.assign(meas.data(), meas.data() + meas.size());
+    if (!info.is_none()) {
+      auto infoArr = py::cast<
+          py::array_t<double, py::array::c_style | py::array::forcecast>>(info);
+      MEGBA_CHECK(infoArr.ndim() == 2 && infoArr.shape(1) == 3 &&
+                      infoArr.shape(0) == prob.nobs,
+                  "info must be (nobs,3): w00,w01,w11");
+      prob.info.assign(infoArr.data(), infoArr.data() + infoArr.size());
+    }
+  }
+
+  void build(const std::string& device, const std::string& dtype, int rank,
+             int worldSize, int deviceIndex, const std::string& diff,
+             const std::string& schur, py::object allreduce,
+             py::object rcclId) {
+    opt.rank = rank;
+    opt.worldSize = worldSize;
+    opt.deviceIndex = deviceIndex;
+    opt.device = device == "gpu" ? Device::GPU : Device::CPU;
+    opt.diff = diff == "analytical" ? DiffMode::ANALYTICAL : DiffMode::AUTO;
+    opt.schur = schur == "implicit" ? SchurMode::IMPLICIT : SchurMode::EXPLICIT;
+    isDouble = dtype != "float32";
+    ix = buildIndex(prob, worldSize);
+    if (opt.device == Device::CPU) {
+      if (isDouble)
+        engD = makeCpuEngine<double>(prob, ix, opt, wrapAllreduce<double>(allreduce));
+      else
+        engF = makeCpuEngine<float>(prob, ix, opt, wrapAllreduce<float>(allreduce));
+    } else {
+#ifdef MEGBA_WITH_GPU
+      std::string id;
+      if (!rcclId.is_none()) id = py::cast<std::string>(rcclId);
+      if (isDouble)
+        engD = makeGpuEngine<double>(prob, ix, opt, id);
+      else
+        engF = makeGpuEngine<float>(prob, ix, opt, id);
+#else
+      MEGBA_CHECK(false, "built without GPU support");
+#endif
+    }
+  }
+
+  template <typename F>
+  auto withEngine(F&& f) {
+    MEGBA_CHECK(engD || engF, "call build() first");
+    if (isDouble) return f(*engD);
+    return f(*engF);
+  }
+
+  py::dict solve(int maxIter, double tau, double eps1, double eps2,
+                 int solverMaxIter, double solverTol, double refuseRatio,
+                 bool forceIterations, bool verbose) {
+    AlgoOptionLM a;
+    a.maxIter = maxIter;
+    a.initialRegion = tau;
+    a.epsilon1 = eps1;
+    a.epsilon2 = eps2;
+    a.forceIterations = forceIterations;
+    a.verbose = verbose;
+    SolverOptionPCG s;
+    s.maxIter = solverMaxIter;
+    s.tol = solverTol;
+    s.refuseRatio = refuseRatio;
+    LMReport rep;
+    {
+      py::gil_scoped_release rel;
+      if (isDouble)
+        rep = runLM<double>(*engD, a, s);
+      else
+        rep = runLM<float>(*engF, a, s);
+    }
+    py::dict d;
+    d["final_chi2"] = rep.finalChi2;
+    d["accepted"] = rep.acceptedSteps;
+    d["rejected"] = rep.rejectedSteps;
+    d["total_ms"] = rep.totalMs;
+    py::list iters;
+    for (const auto& it : rep.iters) {
+      py::dict e;
+      e["iter"] = it.iter;
+      e["accepted"] = it.accepted;
+      e["chi2"] = it.chi2;
+      e["elapsed_ms"] = it.elapsedMs;
+      e["pcg_iters"] = it.pcgIters;
+      iters.append(e);
+    }
+    d["iters"] = iters;
+    return d;
+  }
+
+  // Fine-grained steps for tests.
+  double forward() {
+    return withEngine([&](auto& e) { return e.forward(); });
+  }
+  void acceptForward() {
+    withEngine([&](auto& e) { e.acceptForward(); return 0; });
+  }
+  void buildLinearSystem() {
+    py::gil_scoped_release rel;
+    withEngine([&](auto& e) { e.buildLinearSystem(); return 0; });
+  }
+  void processDiag(double region) {
+    withEngine([&](auto& e) { e.processDiag(region); return 0; });
+  }
+  int solveLinear(int maxIter, double tol, double refuseRatio) {
+    SolverOptionPCG s;
+    s.maxIter = maxIter;
+    s.tol = tol;
+    s.refuseRatio = refuseRatio;
+    py::gil_scoped_release rel;
+    return withEngine([&](auto& e) { return e.solveLinear(s); });
+  }
+  void updateParams() {
+    withEngine([&](auto& e) { e.updateParams(); return 0; });
+  }
+  double rhoDenominator(double chi2) {
+    return withEngine([&](auto& e) { return e.rhoDenominator(chi2); });
+  }
+  double deltaXL2() { return withEngine([&](auto& e) { return e.deltaXL2(); }); }
+  double xL2() { return withEngine([&](auto& e) { return e.xL2(); }); }
+  double gInf() { return withEngine([&](auto& e) { return e.gInf(); }); }
+
+  py::tuple getParams() {
+    py::array_t<double> cams({prob.ncam, 9});
+    py::array_t<double> pts({prob.npt, 3});
+    withEngine([&](auto& e) {
+      e.getParams(cams.mutable_data(), pts.mutable_data());
+      return 0;
+    });
+    return py::make_tuple(cams, pts);
+  }
+
+  py::dict dump() {
+    DenseDump dd = withEngine([&](auto& e) { return e.dump(); });
+    py::dict d;
+    auto toArr = [](std::vector<double>& v) {
+      return py::array_t<double>((py::ssize_t)v.size(), v.data());
+    };
+    d["e0"] = dd.e0;
+    d["e1"] = dd.e1;
+    d["r"] = toArr(dd.r);
+    d["Jc"] = toArr(dd.Jc);
+    d["Jp"] = toArr(dd.Jp);
+    d["Hpp"] = toArr(dd.Hpp);
+    d["Hll"] = toArr(dd.Hll);
+    d["Hpl"] = toArr(dd.Hpl);
+    d["g"] = toArr(dd.g);
+    d["deltaX"] = toArr(dd.deltaX);
+    return d;
+  }
+
+  py::dict indexInfo() {
+    py::dict d;
+    d["ncam"] = ix.ncam;
+    d["npt"] = ix.npt;
+    d["nobs"] = ix.nobs;
+    d["cam_of"] = py::array_t<int>((py::ssize_t)ix.camOf.size(), ix.camOf.data());
+    d["pt_of"] = py::array_t<int>((py::ssize_t)ix.ptOf.size(), ix.ptOf.data());
+    d["cam_rowptr"] =
+        py::array_t<int64_t>((py::ssize_t)ix.camRowPtr.size(), ix.camRowPtr.data());
+    d["split"] = py::array_t<int64_t>((py::ssize_t)ix.split.size(), ix.split.data());
+    d["perm"] = py::array_t<int64_t>((py::ssize_t)ix.perm.size(), ix.perm.data());
+    return d;
+  }
+};
+
+}  // namespace
+
+PYBIND11_MODULE(_core, m) {
+  m.doc() = "megba_amd core (MI355X-native distributed bundle adjustment)";
+
+  py::class_<PyProblem>(m, "Problem")
+      .def(py::init<py::array_t<double, py::array::c_style | py::array::forcecast>,
+                    py::array_t<double, py::array::c_style | py::array::forcecast>,
+                    py::array_t<int, py::array::c_style | py::array::forcecast>,
+                    py::array_t<int, py::array::c_style | py::array::forcecast>,
+                    py::array_t<double, py::array::c_style | py::array::forcecast>,
+                    py::object>(),
+           py::arg("cams"), py::arg("pts"), py::arg("cam_idx"),
+           py::arg("pt_idx"), py::arg("meas"), py::arg("info") = py::none())
+      .def("build", &PyProblem::build, py::arg("device") = "cpu",
+           py::arg("dtype") = "float64", py::arg("rank") = 0,
+           py::arg("world_size") = 1, py::arg("device_index") = 0,
+           py::arg("diff") = "auto", py::arg("schur") = "explicit",
+           py::arg("allreduce") = py::none(), py::arg("rccl_id") = py::none())
+      .def("solve", &PyProblem::solve, py::arg("max_iter") = 20,
+           py::arg("tau") = 1e4, py::arg("epsilon1") = 1.0,
+           py::arg("epsilon2") = 1e-10, py::arg("solver_max_iter") = 100,
+           py::arg("solver_tol") = 1e-1, py::arg("solver_refuse_ratio") = 1.0,
+           py::arg("force_iterations") = false, py::arg("verbose") = true)
+      .def("forward", &PyProblem::forward)
+      .def("accept_forward", &PyProblem::acceptForward)
+      .def("build_linear_system", &PyProblem::buildLinearSystem)
+      .def("process_diag", &PyProblem::processDiag)
+      .def("solve_linear", &PyProblem::solveLinear, py::arg("max_iter") = 100,
+           py::arg("tol") = 1e-1, py::arg("refuse_ratio") = 1.0)
+      .def("update_params", &PyProblem::updateParams)
+      .def("rho_denominator", &PyProblem::rhoDenominator)
+      .def("delta_x_l2", &PyProblem::deltaXL2)
+      .def("x_l2", &PyProblem::xL2)
+      .def("g_inf", &PyProblem::gInf)
+      .def("get_params", &PyProblem::getParams)
+      .def("dump", &PyProblem::dump)
+      .def("index_info", &PyProblem::indexInfo);
+
+#ifdef MEGBA_WITH_GPU
+  m.attr("has_gpu_support") = true;
+  m.def("rccl_unique_id", []() { return py::bytes(rcclUniqueIdString()); });
+  m.def("hip_device_count", []() { return hipDeviceCountSafe(); });
+#else
+  m.attr("has_gpu_support") = false;
+#endif
+}

@@ -1,0 +1,28 @@
+// CPU engine: full LM + Schur-complement PCG on the host (OpenMP).
+// Serves as (a) the BASELINE config-1 "CPU/Eigen reference path" equivalent,
+// (b) the numerics oracle every HIP kernel is tested against, and (c) the
+// world_size>1 correctness testbed (the allreduce hook is pluggable, so
+// multi-process CPU tests over gloo exercise the same distributed math the
+// RCCL path uses).
+#pragma once
+
+#include <functional>
+#include <memory>
+
+#include "common.hpp"
+#include "engine.hpp"
+#include "problem.hpp"
+
+namespace megba {
+
+// In-place sum-allreduce over all ranks (no-op when null / worldSize==1).
+template <typename T>
+using HostAllreduce = std::function<void(T*, std::size_t)>;
+
+template <typename T>
+std::unique_ptr<Engine<T>> makeCpuEngine(const BAProblemHost& prob,
+                                         const ProblemIndex& ix,
+                                         const ProblemOption& opt,
+                                         HostAllreduce<T> allreduce);
+
+}  // namespace megba

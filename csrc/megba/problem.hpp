@@ -1,0 +1,121 @@
+// Host-side problem container and index construction.
+//
+// Capability anchor: the reference's BaseProblem/HessianEntrance/EdgeVector
+// index build (/root/reference/src/problem/base_problem.cpp:96-214,
+// src/linear_system/schur_LM_linear_system.cpp:20-84).  Redesign: observations
+// are globally counting-sorted by (camera, point) once on the host; the sorted
+// order simultaneously provides
+//   * the contiguous per-rank edge partition (data parallelism axis),
+//   * the block-CSR row structure of Hpl (camera block-rows = sorted runs),
+//   * conflict-free per-edge Hpl slots (one block per observation).
+// No per-device HessianEntrance sets / lower_bound passes are needed.
+#pragma once
+
+#include <algorithm>
+#include <cstdint>
+#include <vector>
+
+#include "common.hpp"
+
+namespace megba {
+
+// Raw problem as handed over from Python (original observation order).
+struct BAProblemHost {
+  int ncam = 0;
+  int npt = 0;
+  int64_t nobs = 0;
+  std::vector<double> cams;  // ncam*9
+  std::vector<double> pts;   // npt*3
+  std::vector<int> camIdx;   // nobs
+  std::vector<int> ptIdx;    // nobs
+  std::vector<double> meas;  // nobs*2 ([obs][2])
+  std::vector<double> info;  // optional nobs*3 (2x2 sym weights w00,w01,w11); empty = identity
+};
+
+struct ProblemIndex {
+  int ncam = 0, npt = 0;
+  int64_t nobs = 0;
+  // Arrays in (camera, point)-sorted order:
+  std::vector<int> camOf, ptOf;     // nobs
+  std::vector<double> measSorted;   // nobs*2
+  std::vector<double> infoSorted;   // nobs*3 or empty
+  std::vector<int64_t> camRowPtr;   // ncam+1: edge range of each camera
+  std::vector<int64_t> split;      // worldSize+1 contiguous edge partition
+  std::vector<int64_t> perm;       // sorted position -> original observation id
+};
+
+inline ProblemIndex buildIndex(const BAProblemHost& p, int worldSize) {
+  MEGBA_CHECK(p.ncam > 0 && p.npt > 0 && p.nobs > 0, "empty problem");
+  MEGBA_CHECK((int64_t)p.camIdx.size() == p.nobs && (int64_t)p.ptIdx.size() == p.nobs,
+              "index array size mismatch");
+  ProblemIndex ix;
+  ix.ncam = p.ncam;
+  ix.npt = p.npt;
+  ix.nobs = p.nobs;
+  const int64_t n = p.nobs;
+
+  // Stable counting sort: by point, then by camera -> (cam, pt) order.
+  std::vector<int64_t> tmpPerm(n), cnt;
+  {
+    cnt.assign((size_t)p.npt + 1, 0);
+    for (int64_t e = 0; e < n; ++e) {
+      const int pt = p.ptIdx[e];
+      MEGBA_CHECK(pt >= 0 && pt < p.npt, "point index out of range");
+      cnt[pt + 1]++;
+    }
+    for (int v = 0; v < p.npt; ++v) cnt[v + 1] += cnt[v];
+    for (int64_t e = 0; e < n; ++e) tmpPerm[cnt[p.ptIdx[e]]++] = e;
+  }
+  ix.perm.resize(n);
+  {
+    cnt.assign((size_t)p.ncam + 1, 0);
+    for (int64_t e = 0; e < n; ++e) {
+      const int c = p.camIdx[e];
+      MEGBA_CHECK(c >= 0 && c < p.ncam, "camera index out of range");
+      cnt[c + 1]++;
+    }
+    for (int v = 0; v < p.ncam; ++v) cnt[v + 1] += cnt[v];
+    ix.camRowPtr.assign(cnt.begin(), cnt.end());  // prefix before scatter
+    for (int64_t k = 0; k < n; ++k) {
+      const int64_t e = tmpPerm[k];
+      ix.perm[cnt[p.camIdx[e]]++] = e;
+    }
+  }
+
+  ix.camOf.resize(n);
+  ix.ptOf.resize(n);
+  ix.measSorted.resize(n * 2);
+  const bool hasInfo = !p.info.empty();
+  if (hasInfo) ix.infoSorted.resize(n * 3);
+  for (int64_t k = 0; k < n; ++k) {
+    const int64_t e = ix.perm[k];
+    ix.camOf[k] = p.camIdx[e];
+    ix.ptOf[k] = p.ptIdx[e];
+    ix.measSorted[2 * k] = p.meas[2 * e];
+    ix.measSorted[2 * k + 1] = p.meas[2 * e + 1];
+    if (hasInfo) {
+      ix.infoSorted[3 * k] = p.info[3 * e];
+      ix.infoSorted[3 * k + 1] = p.info[3 * e + 1];
+      ix.infoSorted[3 * k + 2] = p.info[3 * e + 2];
+    }
+  }
+
+  // Every vertex must be observed (else its Hessian block is singular).
+  {
+    std::vector<char> seenPt((size_t)p.npt, 0);
+    for (int64_t k = 0; k < n; ++k) seenPt[ix.ptOf[k]] = 1;
+    for (int v = 0; v < p.npt; ++v)
+      MEGBA_CHECK(seenPt[v], "point with no observations");
+    for (int c = 0; c < p.ncam; ++c)
+      MEGBA_CHECK(ix.camRowPtr[c + 1] > ix.camRowPtr[c],
+                  "camera with no observations");
+  }
+
+  // Balanced contiguous partition over sorted edges.
+  ix.split.resize(worldSize + 1);
+  for (int r = 0; r <= worldSize; ++r)
+    ix.split[r] = (n * r) / worldSize;
+  return ix;
+}
+
+}  // namespace megba
