@@ -1,0 +1,148 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: distributed LM+PCG bundle adjustment on a synthetic
+BAL Venice-1778-shaped problem (BASELINE.json headline config: 1778 cameras,
+993923 points, ~5M observations, fp64).
+
+Contract (driver): `python bench.py --gpus N --steps K --warmup W`, launched
+via torch.distributed.run for N>1 (one rank per GPU over RCCL).  One "step" =
+one full LM iteration (forward + Jacobians, Hessian assembly + allreduce,
+damped Schur-complement PCG with its per-iteration allreduces, parameter
+update, gain-ratio accept/reject) — the unit the reference logs per
+iteration (MegBA lm_algo.cu "Iter k ... elapsed ms").  Data is synthetic
+(no network for datasets); parameters are ground truth + noise so LM runs a
+realistic trajectory; force_iterations keeps the step count exact.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+MODELS = {
+    # BASELINE.json configs (synthetic, same shapes as the BAL datasets).
+    "venice1778": dict(ncam=1778, npt=993923, nobs=5_000_000),
+    "trafalgar257": dict(ncam=257, npt=65132, nobs=225_911),
+    "ladybug49": dict(ncam=49, npt=7776, nobs=31_843),
+    "final13682": dict(ncam=13682, npt=4_456_117, nobs=28_987_644),
+    "synth20k": dict(ncam=20_000, npt=10_000_000, nobs=50_000_000),
+    "tiny": dict(ncam=30, npt=400, nobs=3_000),
+}
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--model", default="venice1778", choices=sorted(MODELS))
+    ap.add_argument("--device", default="gpu", choices=["gpu", "cpu"])
+    ap.add_argument("--dtype", default="float64",
+                    choices=["float64", "float32"])
+    ap.add_argument("--diff", default="auto", choices=["auto", "analytical"])
+    ap.add_argument("--schur", default="explicit",
+                    choices=["explicit", "implicit"])
+    ap.add_argument("--verbose", action="store_true")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+
+    import megba_amd as mb
+
+    shape = MODELS[args.model]
+    t0 = time.time()
+    cams, pts, ci, pi, meas = mb.synthesize_bal(
+        shape["ncam"], shape["npt"], shape["nobs"], seed=7)
+    if rank == 0:
+        print(f"# synthesized {args.model} in {time.time()-t0:.1f}s",
+              file=sys.stderr)
+
+    dist = None
+    rccl_id = None
+    allreduce = None
+    if world > 1:
+        import torch.distributed as tdist
+        dist = tdist
+        # gloo for bootstrap only; the GPU engine runs RCCL natively.
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        if args.device == "gpu":
+            from megba_amd.dist import broadcast_rccl_id
+            rccl_id = broadcast_rccl_id(rank)
+        else:
+            from megba_amd.dist import gloo_allreduce_callback
+            allreduce = gloo_allreduce_callback()
+
+    p = mb.BAProblem(cams, pts, ci, pi, meas)
+    p.build(device=args.device, dtype=args.dtype, rank=rank,
+            world_size=world, device_index=local_rank, diff=args.diff,
+            schur=args.schur, allreduce=allreduce, rccl_id=rccl_id)
+
+    # Reference demo solver flags (MegBA README.md:54-67).
+    p.lm_init(tau=1e4, epsilon1=1.0, epsilon2=1e-10, solver_max_iter=100,
+              solver_tol=1e-1, solver_refuse_ratio=1.0,
+              force_iterations=True, verbose=args.verbose and rank == 0)
+
+    def sync():
+        if dist is not None:
+            dist.barrier()
+        try:
+            import torch
+            if torch.cuda.is_available():
+                torch.cuda.synchronize()
+        except Exception:
+            pass
+
+    for _ in range(args.warmup):
+        p.lm_step()
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        log = p.lm_step()
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks.
+    if dist is not None:
+        import torch
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    if rank == 0:
+        ms_per_step = elapsed / args.steps * 1000.0
+        line = {
+            "metric": "lm_iterations_per_s",
+            "value": args.steps / elapsed,
+            "unit": "iters/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "fp64" if args.dtype == "float64" else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": f"BAL-{args.model}-synthetic",
+                "cameras": shape["ncam"],
+                "points": shape["npt"],
+                "observations": shape["nobs"],
+                "diff": args.diff,
+                "schur": args.schur,
+                "parallelism": f"edge-dp{world}",
+                "solver": {"tau": 1e4, "solver_tol": 0.1,
+                           "solver_refuse_ratio": 1.0,
+                           "solver_max_iter": 100},
+                "final_chi2": log["chi2"],
+            },
+        }
+        print(json.dumps(line), flush=True)
+
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -151,6 +151,48 @@ struct PyProblem {
     return d;
   }
 
+  // Step-level LM control (bench.py times exact step counts with this).
+  std::unique_ptr<LMSession<double>> sessD;
+  std::unique_ptr<LMSession<float>> sessF;
+  double lmInit(int maxIter, double tau, double eps1, double eps2,
+                int solverMaxIter, double solverTol, double refuseRatio,
+                bool forceIterations, bool verbose) {
+    AlgoOptionLM a;
+    a.maxIter = maxIter;
+    a.initialRegion = tau;
+    a.epsilon1 = eps1;
+    a.epsilon2 = eps2;
+    a.forceIterations = forceIterations;
+    a.verbose = verbose;
+    SolverOptionPCG so;
+    so.maxIter = solverMaxIter;
+    so.tol = solverTol;
+    so.refuseRatio = refuseRatio;
+    py::gil_scoped_release rel;
+    if (isDouble) {
+      sessD = std::make_unique<LMSession<double>>(*engD, a, so);
+      return sessD->init();
+    }
+    sessF = std::make_unique<LMSession<float>>(*engF, a, so);
+    return sessF->init();
+  }
+  py::dict lmStep() {
+    MEGBA_CHECK(sessD || sessF, "lm_init first");
+    IterLog log;
+    {
+      py::gil_scoped_release rel;
+      log = isDouble ? sessD->step() : sessF->step();
+    }
+    py::dict e;
+    e["iter"] = log.iter;
+    e["accepted"] = log.accepted;
+    e["chi2"] = log.chi2;
+    e["elapsed_ms"] = log.elapsedMs;
+    e["pcg_iters"] = log.pcgIters;
+    e["stopped"] = isDouble ? sessD->stopped() : sessF->stopped();
+    return e;
+  }
+
   // Fine-grained steps for tests.
   double forward() {
     return withEngine([&](auto& e) { return e.forward(); });
@@ -251,6 +293,12 @@ PYBIND11_MODULE(_core, m) {
            py::arg("epsilon2") = 1e-10, py::arg("solver_max_iter") = 100,
            py::arg("solver_tol") = 1e-1, py::arg("solver_refuse_ratio") = 1.0,
            py::arg("force_iterations") = false, py::arg("verbose") = true)
+      .def("lm_init", &PyProblem::lmInit, py::arg("max_iter") = 1000000,
+           py::arg("tau") = 1e4, py::arg("epsilon1") = 1.0,
+           py::arg("epsilon2") = 1e-10, py::arg("solver_max_iter") = 100,
+           py::arg("solver_tol") = 1e-1, py::arg("solver_refuse_ratio") = 1.0,
+           py::arg("force_iterations") = false, py::arg("verbose") = false)
+      .def("lm_step", &PyProblem::lmStep)
       .def("forward", &PyProblem::forward)
       .def("accept_forward", &PyProblem::acceptForward)
       .def("build_linear_system", &PyProblem::buildLinearSystem)

@@ -120,7 +120,8 @@ __global__ void kRedFinal(const double* part, int nb, double* out) {
 // Forward (fused register autodiff)
 // ---------------------------------------------------------------------------
 template <typename T>
-__global__ void kForward(int64_t nL, const int* __restrict__ camOf,
+__global__ __launch_bounds__(256, 2) void kForward(
+                         int64_t nL, const int* __restrict__ camOf,
                          const int* __restrict__ ptOf,
                          const T* __restrict__ params, int ncam,
                          const T* __restrict__ meas, T* __restrict__ rOut,
@@ -259,10 +260,23 @@ __global__ void kInvert(int nBlk, const T* __restrict__ Hd, T* __restrict__ Hinv
                         int* fail) {
   for (int64_t b = blockIdx.x * (int64_t)kBlk + threadIdx.x; b < nBlk;
        b += (int64_t)gridDim.x * kBlk) {
+    if (!spdInvertPacked<T, D>(Hd + b * D * D, Hinv + b * D * D))
+      atomicOr(fail, 1);  // rare: retried by kInvertJitter
+  }
+}
+
+// Rare path: re-run EVERY block, retrying semi-definite ones with a growing
+// relative diagonal jitter (matches the CPU oracle).  Launched only when
+// kInvert reported a failure, so its register/scratch cost is off the hot
+// path.
+template <typename T, int D>
+__global__ void kInvertJitter(int nBlk, const T* __restrict__ Hd,
+                              T* __restrict__ Hinv, int* fail) {
+  for (int64_t b = blockIdx.x * (int64_t)kBlk + threadIdx.x; b < nBlk;
+       b += (int64_t)gridDim.x * kBlk) {
     const T* a = Hd + b * D * D;
     T* out = Hinv + b * D * D;
     if (spdInvertPacked<T, D>(a, out)) continue;
-    // Semi-definite block: relative jitter retry (matches CPU oracle).
     T buf[D * D];
     T mx = T(0);
     for (int i = 0; i < D; ++i) {
@@ -277,7 +291,7 @@ __global__ void kInvert(int nBlk, const T* __restrict__ Hd, T* __restrict__ Hinv
       for (int i = 0; i < D; ++i) buf[i * D + i] += jit;
       ok = spdInvertPacked<T, D>(buf, out);
     }
-    if (!ok) atomicOr(fail, 1);
+    if (!ok) atomicOr(fail + 1, 1);
   }
 }
 
@@ -529,7 +543,7 @@ class GpuEngine final : public Engine<T> {
     dTemp_ = dalloc<T>(np_);
 
     dPart_ = dalloc<double>(kRedBlocks + 4);
-    dFail_ = dalloc<int>(1);
+    dFail_ = dalloc<int>(2);
 
     // Chunk table for the E*w block-row kernel.
     std::vector<int> cCam, cLo, cHi;
@@ -621,17 +635,29 @@ class GpuEngine final : public Engine<T> {
   }
 
   int solveLinear(const SolverOptionPCG& opt) override {
-    // Block inverses (preconditioner + Cinv).
-    HIP_CHECK(hipMemsetAsync(dFail_, 0, sizeof(int), stream_));
+    // Block inverses (preconditioner + Cinv).  Fast path first; the jitter
+    // retry kernel runs only if some block was numerically semi-definite.
+    HIP_CHECK(hipMemsetAsync(dFail_, 0, 2 * sizeof(int), stream_));
     hipLaunchKernelGGL((kInvert<T, 9>), dim3(gridFor(ncam_)), dim3(kBlk), 0,
                        stream_, ncam_, dHppD_, dHppInv_, dFail_);
     hipLaunchKernelGGL((kInvert<T, 3>), dim3(gridFor(npt_)), dim3(kBlk), 0,
                        stream_, npt_, dHllD_, dHllInv_, dFail_);
-    int fail = 0;
-    HIP_CHECK(hipMemcpyAsync(&fail, dFail_, sizeof(int), hipMemcpyDeviceToHost,
-                             stream_));
+    int fail[2] = {0, 0};
+    HIP_CHECK(hipMemcpyAsync(fail, dFail_, 2 * sizeof(int),
+                             hipMemcpyDeviceToHost, stream_));
     sync();
-    MEGBA_CHECK(!fail, "singular Hessian block");
+    if (fail[0]) {
+      hipLaunchKernelGGL((kInvertJitter<T, 9>), dim3(gridFor(ncam_)),
+                         dim3(kBlk), 0, stream_, ncam_, dHppD_, dHppInv_,
+                         dFail_);
+      hipLaunchKernelGGL((kInvertJitter<T, 3>), dim3(gridFor(npt_)),
+                         dim3(kBlk), 0, stream_, npt_, dHllD_, dHllInv_,
+                         dFail_);
+      HIP_CHECK(hipMemcpyAsync(fail, dFail_, 2 * sizeof(int),
+                               hipMemcpyDeviceToHost, stream_));
+      sync();
+      MEGBA_CHECK(!fail[1], "singular Hessian block");
+    }
 
     const T* gc = dG_;
     const T* gp = dG_ + nc_;

@@ -8,6 +8,9 @@
 //   stop on ||dx|| <= eps2*(||x||+eps1) or ||g||inf <= eps1.
 // The per-iteration log line format ("Iter k error: E, log error: L, elapsed
 // M ms") is kept as the de-facto UX (lm_algo.cu:190-220).
+//
+// LMSession exposes the loop one iteration at a time so bench.py can time an
+// exact number of steps; runLM() is the batch driver on top of it.
 #pragma once
 
 #include <chrono>
@@ -20,87 +23,124 @@
 namespace megba {
 
 template <typename T>
-LMReport runLM(Engine<T>& eng, const AlgoOptionLM& opt,
-               const SolverOptionPCG& sopt) {
-  using Clock = std::chrono::steady_clock;
-  const auto t0 = Clock::now();
-  auto elapsedMs = [&]() {
-    return std::chrono::duration<double, std::milli>(Clock::now() - t0).count();
-  };
+class LMSession {
+ public:
+  LMSession(Engine<T>& eng, const AlgoOptionLM& opt,
+            const SolverOptionPCG& sopt)
+      : eng_(eng), opt_(opt), sopt_(sopt) {}
 
-  LMReport rep;
-  double chi2New = eng.forward();
-  eng.acceptForward();
-  eng.buildLinearSystem();
-  eng.backupGDx();
-  eng.backupParams();
-  double chi2 = chi2New;
-  if (opt.verbose) {
-    std::printf("Start with error: %.10g, log error: %.6f, elapsed %.1f ms\n",
-                chi2New / 2, std::log10(chi2New / 2), elapsedMs());
-    std::fflush(stdout);
-  }
-  rep.iters.push_back({0, true, chi2New, elapsedMs(), 0});
-
-  bool stop = false;
-  double v = 2.0;
-  double region = opt.initialRegion;
-  int k = 0;
-  while ((!stop || opt.forceIterations) && k < opt.maxIter) {
-    ++k;
-    eng.processDiag(region);
-    const int pcgIters = eng.solveLinear(sopt);
-    const double deltaXL2 = eng.deltaXL2();
-    const double xL2 = eng.xL2();
-    if (!opt.forceIterations &&
-        deltaXL2 <= opt.epsilon2 * (xL2 + opt.epsilon1)) {
-      break;
+  // Initial forward + assembly (iteration 0). Returns initial chi2.
+  double init() {
+    t0_ = Clock::now();
+    chi2New_ = eng_.forward();
+    eng_.acceptForward();
+    eng_.buildLinearSystem();
+    eng_.backupGDx();
+    eng_.backupParams();
+    region_ = opt_.initialRegion;
+    v_ = 2.0;
+    k_ = 0;
+    stop_ = false;
+    if (opt_.verbose) {
+      std::printf("Start with error: %.10g, log error: %.6f, elapsed %.1f ms\n",
+                  chi2New_ / 2, std::log10(chi2New_ / 2), elapsedMs());
+      std::fflush(stdout);
     }
-    eng.updateParams();
-    const double rhoDenominator = eng.rhoDenominator(chi2New);
-    const double chi2Old = chi2New;
-    chi2New = eng.forward();
-    const double rho = -(chi2Old - chi2New) / rhoDenominator;
+    inited_ = true;
+    return chi2New_;
+  }
+
+  bool stopped() const { return stop_; }
+  int iter() const { return k_; }
+  double chi2() const { return chi2New_; }
+
+  // One LM iteration. Returns the iteration log entry.
+  IterLog step() {
+    MEGBA_CHECK(inited_, "LMSession::init() first");
+    ++k_;
     IterLog log;
-    log.iter = k;
-    log.pcgIters = pcgIters;
-    if (chi2Old > chi2New) {
+    log.iter = k_;
+    eng_.processDiag(region_);
+    log.pcgIters = eng_.solveLinear(sopt_);
+    const double deltaXL2 = eng_.deltaXL2();
+    const double xL2 = eng_.xL2();
+    if (!opt_.forceIterations &&
+        deltaXL2 <= opt_.epsilon2 * (xL2 + opt_.epsilon1)) {
+      stop_ = true;
+      log.accepted = false;
+      log.chi2 = chi2New_;
+      log.elapsedMs = elapsedMs();
+      return log;
+    }
+    eng_.updateParams();
+    const double rhoDenominator = eng_.rhoDenominator(chi2New_);
+    const double chi2Old = chi2New_;
+    chi2New_ = eng_.forward();
+    const double rho = -(chi2Old - chi2New_) / rhoDenominator;
+    if (chi2Old > chi2New_) {
       log.accepted = true;
-      log.chi2 = chi2New;
-      eng.acceptForward();
-      eng.buildLinearSystem();
-      eng.backupGDx();
-      eng.backupParams();
-      chi2 = chi2New;
-      region /= std::max(1.0 / 3.0, 1.0 - std::pow(2.0 * rho - 1.0, 3.0));
-      v = 2.0;
-      rep.acceptedSteps++;
-      const double gnorm = eng.gInf();
-      stop = gnorm <= opt.epsilon1;
-      if (opt.verbose) {
+      log.chi2 = chi2New_;
+      eng_.acceptForward();
+      eng_.buildLinearSystem();
+      eng_.backupGDx();
+      eng_.backupParams();
+      region_ /= std::max(1.0 / 3.0, 1.0 - std::pow(2.0 * rho - 1.0, 3.0));
+      v_ = 2.0;
+      const double gnorm = eng_.gInf();
+      if (!opt_.forceIterations && gnorm <= opt_.epsilon1) stop_ = true;
+      if (opt_.verbose) {
         std::printf("Iter %d error: %.10g, log error: %.6f, elapsed %.1f ms\n",
-                    k, chi2New / 2, std::log10(chi2New / 2), elapsedMs());
+                    k_, chi2New_ / 2, std::log10(chi2New_ / 2), elapsedMs());
         std::fflush(stdout);
       }
     } else {
       log.accepted = false;
       log.chi2 = chi2Old;
-      eng.rollbackParams();
-      eng.rollbackGDx();
-      chi2New = chi2Old;
-      region /= v;
-      v *= 2.0;
-      rep.rejectedSteps++;
-      if (opt.verbose) {
-        std::printf("Iter %d failed, elapsed %.1f ms\n", k, elapsedMs());
+      eng_.rollbackParams();
+      eng_.rollbackGDx();
+      chi2New_ = chi2Old;
+      region_ /= v_;
+      v_ *= 2.0;
+      if (opt_.verbose) {
+        std::printf("Iter %d failed, elapsed %.1f ms\n", k_, elapsedMs());
         std::fflush(stdout);
       }
     }
     log.elapsedMs = elapsedMs();
-    rep.iters.push_back(log);
+    return log;
   }
-  rep.finalChi2 = chi2;
-  rep.totalMs = elapsedMs();
+
+ private:
+  using Clock = std::chrono::steady_clock;
+  double elapsedMs() const {
+    return std::chrono::duration<double, std::milli>(Clock::now() - t0_).count();
+  }
+  Engine<T>& eng_;
+  AlgoOptionLM opt_;
+  SolverOptionPCG sopt_;
+  Clock::time_point t0_;
+  double chi2New_ = 0, region_ = 0, v_ = 2.0;
+  int k_ = 0;
+  bool stop_ = false, inited_ = false;
+};
+
+template <typename T>
+LMReport runLM(Engine<T>& eng, const AlgoOptionLM& opt,
+               const SolverOptionPCG& sopt) {
+  LMSession<T> s(eng, opt, sopt);
+  LMReport rep;
+  const double chi0 = s.init();
+  rep.iters.push_back({0, true, chi0, 0.0, 0});
+  while (!s.stopped() && s.iter() < opt.maxIter) {
+    IterLog log = s.step();
+    rep.iters.push_back(log);
+    if (log.accepted)
+      rep.acceptedSteps++;
+    else if (!s.stopped())
+      rep.rejectedSteps++;
+    rep.totalMs = log.elapsedMs;
+  }
+  rep.finalChi2 = s.chi2();
   if (opt.verbose) {
     std::printf("Finished\n");
     std::fflush(stdout);
