@@ -89,19 +89,29 @@ def synthesize_bal(ncam, npt, nobs, seed=0, pixel_noise=1.0,
             cam_idx[slot] = c
             slot += 1
 
-    # Project with ground truth.
-    P = np.einsum('eij,ej->ei', Rs[cam_idx], pts_gt[pt_idx]) \
-        + cams_gt[cam_idx, 3:6]
-    # Guard: keep points in front of the camera (z<0 in BAL convention).
-    bad = P[:, 2] > -0.5
-    if bad.any():
-        # Re-aim bad observations at points near the center (always visible).
-        nbad = int(bad.sum())
-        pt_idx[bad] = rng.integers(0, npt, size=nbad)
-        # project again; clip z
-        P = np.einsum('eij,ej->ei', Rs[cam_idx], pts_gt[pt_idx]) \
-            + cams_gt[cam_idx, 3:6]
-        P[:, 2] = np.minimum(P[:, 2], -0.5)
+    # Project with ground truth; observations behind the camera (z >= -0.5,
+    # BAL looks down -z) get their CAMERA resampled so point degrees stay
+    # intact; a tiny unlucky remainder is depth-clamped.
+    def project(idx=None):
+        if idx is None:
+            return np.einsum('eij,ej->ei', Rs[cam_idx], pts_gt[pt_idx]) \
+                + cams_gt[cam_idx, 3:6]
+        return np.einsum('eij,ej->ei', Rs[cam_idx[idx]], pts_gt[pt_idx[idx]]) \
+            + cams_gt[cam_idx[idx], 3:6]
+    P = project()
+    for _ in range(8):
+        bad = np.where(P[:, 2] > -0.5)[0]
+        if len(bad) == 0:
+            break
+        cam_idx[bad] = rng.integers(0, ncam, size=len(bad))
+        P[bad] = project(bad)
+    # Re-assert camera degree floor (resampling may have starved a camera).
+    counts = np.bincount(cam_idx, minlength=ncam)
+    for c in np.where(counts < 2)[0]:
+        take = rng.integers(0, nobs, size=2 - counts[c])
+        cam_idx[take] = c
+        P[take] = project(take)
+    P[:, 2] = np.minimum(P[:, 2], -0.5)
     p = -P[:, :2] / P[:, 2:3]
     r2 = (p ** 2).sum(axis=1)
     k1 = cams_gt[cam_idx, 7]
