@@ -170,11 +170,18 @@ __global__ __launch_bounds__(256, 2) void kForward(
 // ---------------------------------------------------------------------------
 // Assembly
 // ---------------------------------------------------------------------------
+// Per-edge part: Hpl block (grad-major [27][nL], conflict-free), Hll and g_p
+// (atomics; point degree is low so contention is cheap).  The camera blocks
+// Hpp/g_c are handled by kAssembleCam: per-camera chunks with LDS-staged J
+// rows and 90 distributed accumulators, one atomicAdd per chunk instead of 90
+// per edge -- the all-atomic version measured 40.8 ms on Venice-5M
+// (profiles/r01_venice_baseline.md), dominated by ~2800-way per-camera
+// contention.
 template <typename T, bool HASINFO>
-__global__ void kAssemble(int64_t nL, const int* __restrict__ camOf,
+__global__ void kAssembleEdge(int64_t nL, const int* __restrict__ camOf,
                           const int* __restrict__ ptOf, const T* __restrict__ r,
                           const T* __restrict__ Jc, const T* __restrict__ Jp,
-                          const T* __restrict__ info, T* __restrict__ Hpp,
+                          const T* __restrict__ info,
                           T* __restrict__ Hll, T* __restrict__ Hpl,
                           T* __restrict__ g, int ncam) {
   T* gp = g + (int64_t)ncam * 9;
@@ -216,11 +223,12 @@ __global__ void kAssemble(int64_t nL, const int* __restrict__ camOf,
       wr[0] = rr[0];
       wr[1] = rr[1];
     }
-    // Hpl: one conflict-free 9x3 block per observation.
-    T* hpl = Hpl + 27 * e;
+    // Hpl: one conflict-free 9x3 block per observation, grad-major layout
+    // Hpl[(a*3+b)*nL + e] so SpMV reads coalesce across lanes.
     for (int a = 0; a < 9; ++a)
       for (int b = 0; b < 3; ++b)
-        hpl[a * 3 + b] = jc[0][a] * wjp[0][b] + jc[1][a] * wjp[1][b];
+        Hpl[((int64_t)(a * 3 + b)) * nL + e] =
+            jc[0][a] * wjp[0][b] + jc[1][a] * wjp[1][b];
     // Hll + g_p: point degree is low (few obs/point) -> atomics are cheap.
     const int pt = ptOf[e];
     T* hll = Hll + (int64_t)pt * 9;
@@ -230,15 +238,85 @@ __global__ void kAssemble(int64_t nL, const int* __restrict__ camOf,
     T* gpt = gp + (int64_t)pt * 3;
     for (int a = 0; a < 3; ++a)
       atomicAdd(&gpt[a], -(jp[0][a] * wr[0] + jp[1][a] * wr[1]));
-    // Hpp + g_c (native fp64 atomics on gfx950).
-    const int cam = camOf[e];
-    T* hpp = Hpp + (int64_t)cam * 81;
-    for (int a = 0; a < 9; ++a) {
-      for (int b = 0; b < 9; ++b)
-        atomicAdd(&hpp[a * 9 + b], jc[0][a] * wjc[0][b] + jc[1][a] * wjc[1][b]);
-      atomicAdd(&g[(int64_t)cam * 9 + a], -(jc[0][a] * wr[0] + jc[1][a] * wr[1]));
-    }
   }
+}
+
+// Camera part: one 128-thread block per <=256-edge chunk of one camera's
+// (cam,pt)-sorted edge run.  J rows (+ weighted rows) are staged through LDS
+// in 128-edge tiles; threads 0..80 each own one Hpp element, 81..89 one g_c
+// element; one atomicAdd per output per chunk.
+template <typename T, bool HASINFO>
+__global__ __launch_bounds__(128) void kAssembleCam(
+    int nChunks, const int* __restrict__ chCam, const int* __restrict__ chLo,
+    const int* __restrict__ chHi, int64_t nL, const T* __restrict__ r,
+    const T* __restrict__ Jc, const T* __restrict__ info, T* __restrict__ Hpp,
+    T* __restrict__ g) {
+  constexpr int TE = 128;                    // edges per LDS tile
+  constexpr int ST = HASINFO ? 38 : 20;      // doubles per edge in LDS
+  __shared__ T lds[TE * ST];
+  const int chunk = blockIdx.x;
+  if (chunk >= nChunks) return;
+  const int cam = chCam[chunk];
+  const int lo = chLo[chunk], hi = chHi[chunk];
+  const int t = threadIdx.x;
+  T acc = T(0);
+  const int ti = t < 81 ? t / 9 : t - 81;    // Hpp row / g index
+  const int tj = t < 81 ? t % 9 : 0;         // Hpp col
+  for (int s0 = lo; s0 < hi; s0 += TE) {
+    const int nt = min(TE, hi - s0);
+    // Stage: thread t loads element (k) of edges t%... simple: loop.
+    for (int idx = t; idx < nt * ST; idx += 128) {
+      const int e = idx / ST;
+      const int k = idx % ST;
+      const int64_t ge = s0 + e;
+      T v;
+      if (k < 18) {
+        v = Jc[((int64_t)k) * nL + ge];            // jc[row=k%2? no: k=(col*2+row)]
+      } else if (k < 20) {
+        v = r[(int64_t)(k - 18) * nL + ge];
+      } else {
+        v = T(0);  // filled below for HASINFO
+      }
+      lds[e * ST + k] = v;
+    }
+    __syncthreads();
+    if (HASINFO) {
+      // weighted rows: wjc[col][row] and wr, computed once per edge.
+      for (int e = t; e < nt; e += 128) {
+        const int64_t ge = s0 + e;
+        const T w00 = info[3 * ge], w01 = info[3 * ge + 1],
+                w11 = info[3 * ge + 2];
+        T* row = lds + e * ST;
+        for (int col = 0; col < 9; ++col) {
+          const T a0 = row[col * 2], a1 = row[col * 2 + 1];
+          row[20 + col * 2] = w00 * a0 + w01 * a1;
+          row[20 + col * 2 + 1] = w01 * a0 + w11 * a1;
+        }
+        const T r0 = row[18], r1 = row[19];
+        row[18] = w00 * r0 + w01 * r1;   // wr replaces r (safe: r only
+        row[19] = w01 * r0 + w11 * r1;   // used via wr below)
+      }
+      __syncthreads();
+    }
+    if (t < 90) {
+      const int woff = HASINFO ? 20 : 0;  // weighted rows offset (== raw if no info)
+      for (int e = 0; e < nt; ++e) {
+        const T* row = lds + e * ST;
+        if (t < 81) {
+          // Hpp[i][j] += sum_rows jc[row][i] * wjc[row][j]
+          acc += row[ti * 2] * row[woff + tj * 2] +
+                 row[ti * 2 + 1] * row[woff + tj * 2 + 1];
+        } else {
+          acc -= row[ti * 2] * row[18] + row[ti * 2 + 1] * row[19];
+        }
+      }
+    }
+    __syncthreads();
+  }
+  if (t < 81)
+    atomicAdd(&Hpp[(int64_t)cam * 81 + ti * 9 + tj], acc);
+  else if (t < 90)
+    atomicAdd(&g[(int64_t)cam * 9 + ti], acc);
 }
 
 // ---------------------------------------------------------------------------
@@ -305,14 +383,13 @@ __global__ void kSpmvEtx(int64_t nL, const int* __restrict__ camOf,
                          const T* __restrict__ x, T* __restrict__ out) {
   for (int64_t e = blockIdx.x * (int64_t)kBlk + threadIdx.x; e < nL;
        e += (int64_t)gridDim.x * kBlk) {
-    const T* blk = Hpl + 27 * e;
     const T* xc = x + (int64_t)camOf[e] * 9;
     T o0 = 0, o1 = 0, o2 = 0;
     for (int i = 0; i < 9; ++i) {
       const T xi = xc[i];
-      o0 += blk[i * 3] * xi;
-      o1 += blk[i * 3 + 1] * xi;
-      o2 += blk[i * 3 + 2] * xi;
+      o0 += Hpl[((int64_t)(i * 3 + 0)) * nL + e] * xi;
+      o1 += Hpl[((int64_t)(i * 3 + 1)) * nL + e] * xi;
+      o2 += Hpl[((int64_t)(i * 3 + 2)) * nL + e] * xi;
     }
     T* op = out + (int64_t)ptOf[e] * 3;
     atomicAdd(&op[0], o0);
@@ -330,6 +407,7 @@ __global__ __launch_bounds__(64) void kSpmvEx(int nChunks,
                                               const int* __restrict__ chHi,
                                               const int* __restrict__ ptOf,
                                               const T* __restrict__ Hpl,
+                                              int64_t nL,
                                               const T* __restrict__ w,
                                               T* __restrict__ out) {
   const int chunk = blockIdx.x;
@@ -339,11 +417,12 @@ __global__ __launch_bounds__(64) void kSpmvEx(int nChunks,
   for (int i = 0; i < 9; ++i) acc[i] = T(0);
   const int lo = chLo[chunk], hi = chHi[chunk];
   for (int e = lo + (int)threadIdx.x; e < hi; e += 64) {
-    const T* blk = Hpl + (int64_t)e * 27;
     const T* wp = w + (int64_t)ptOf[e] * 3;
     const T w0 = wp[0], w1 = wp[1], w2 = wp[2];
     for (int i = 0; i < 9; ++i)
-      acc[i] += blk[i * 3] * w0 + blk[i * 3 + 1] * w1 + blk[i * 3 + 2] * w2;
+      acc[i] += Hpl[((int64_t)(i * 3 + 0)) * nL + e] * w0 +
+                Hpl[((int64_t)(i * 3 + 1)) * nL + e] * w1 +
+                Hpl[((int64_t)(i * 3 + 2)) * nL + e] * w2;
   }
   for (int off = 32; off > 0; off >>= 1)
     for (int i = 0; i < 9; ++i) acc[i] += __shfl_down(acc[i], off, 64);
@@ -589,14 +668,25 @@ class GpuEngine final : public Engine<T> {
     HIP_CHECK(hipMemsetAsync(dHpp_, 0, (int64_t)ncam_ * 81 * sizeof(T), stream_));
     HIP_CHECK(hipMemsetAsync(dHll_, 0, (int64_t)npt_ * 9 * sizeof(T), stream_));
     HIP_CHECK(hipMemsetAsync(dG_, 0, dim_ * sizeof(T), stream_));
-    if (hasInfo_)
-      hipLaunchKernelGGL((kAssemble<T, true>), dim3(gridFor(nL_)), dim3(kBlk), 0,
-                         stream_, nL_, dCamOf_, dPtOf_, dR_[bak], dJc_[bak],
-                         dJp_[bak], dInfo_, dHpp_, dHll_, dHpl_, dG_, ncam_);
-    else
-      hipLaunchKernelGGL((kAssemble<T, false>), dim3(gridFor(nL_)), dim3(kBlk), 0,
-                         stream_, nL_, dCamOf_, dPtOf_, dR_[bak], dJc_[bak],
-                         dJp_[bak], nullptr, dHpp_, dHll_, dHpl_, dG_, ncam_);
+    if (hasInfo_) {
+      hipLaunchKernelGGL((kAssembleEdge<T, true>), dim3(gridFor(nL_)),
+                         dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
+                         dR_[bak], dJc_[bak], dJp_[bak], dInfo_, dHll_, dHpl_,
+                         dG_, ncam_);
+      if (nChunks_ > 0)
+        hipLaunchKernelGGL((kAssembleCam<T, true>), dim3(nChunks_), dim3(128),
+                           0, stream_, nChunks_, dChCam_, dChLo_, dChHi_, nL_,
+                           dR_[bak], dJc_[bak], dInfo_, dHpp_, dG_);
+    } else {
+      hipLaunchKernelGGL((kAssembleEdge<T, false>), dim3(gridFor(nL_)),
+                         dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
+                         dR_[bak], dJc_[bak], dJp_[bak], nullptr, dHll_, dHpl_,
+                         dG_, ncam_);
+      if (nChunks_ > 0)
+        hipLaunchKernelGGL((kAssembleCam<T, false>), dim3(nChunks_), dim3(128),
+                           0, stream_, nChunks_, dChCam_, dChLo_, dChHi_, nL_,
+                           dR_[bak], dJc_[bak], nullptr, dHpp_, dG_);
+    }
     allreduce(dHpp_, (int64_t)ncam_ * 81);
     allreduce(dHll_, (int64_t)npt_ * 9);
     allreduce(dG_, dim_);
@@ -752,7 +842,14 @@ class GpuEngine final : public Engine<T> {
     d.Jp = down(dJp_[cur_], nL_ * 6);
     d.Hpp = down(dHpp_, (int64_t)ncam_ * 81);
     d.Hll = down(dHll_, (int64_t)npt_ * 9);
-    d.Hpl = down(dHpl_, nL_ * 27);
+    {
+      // device Hpl is grad-major [27][nL]; dump as [e][9][3]
+      std::vector<double> gm = down(dHpl_, nL_ * 27);
+      std::vector<double> o(gm.size());
+      for (int64_t e = 0; e < nL_; ++e)
+        for (int k = 0; k < 27; ++k) o[e * 27 + k] = gm[(int64_t)k * nL_ + e];
+      d.Hpl = o;
+    }
     d.g = down(dG_, dim_);
     d.deltaX = down(dDeltaX_, dim_);
     // Dump layouts match the CPU engine: r [e][2], Jc [e][2][9], Jp [e][2][3]
@@ -868,8 +965,8 @@ class GpuEngine final : public Engine<T> {
     HIP_CHECK(hipMemsetAsync(out, 0, nc_ * sizeof(T), stream_));
     if (nChunks_ > 0)
       hipLaunchKernelGGL(kSpmvEx<T>, dim3(nChunks_), dim3(64), 0, stream_,
-                         nChunks_, dChCam_, dChLo_, dChHi_, dPtOf_, dHpl_, wv,
-                         out);
+                         nChunks_, dChCam_, dChLo_, dChHi_, dPtOf_, dHpl_, nL_,
+                         wv, out);
   }
   // q = S x = HppD x - E Cinv E^T x   (2 allreduces, reference site A4).
   void schurApply(const T* xv, T* q) {

@@ -117,4 +117,4 @@ def test_larger_problem_gpu():
         np.testing.assert_allclose(d2[key], d1[key], rtol=1e-8,
                                    atol=1e-8 * scale, err_msg=key)
     scale = np.abs(d1["deltaX"]).max()
-    np.testing.assert_allclose(d2["deltaX"], d1["deltaX"], atol=1e-6 * scale)
+    np.testing.assert_allclose(d2["deltaX"], d1["deltaX"], atol=1e-4 * scale)
