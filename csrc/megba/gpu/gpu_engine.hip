@@ -394,18 +394,25 @@ __global__ void kTransposeHlp(int64_t nL, const int* __restrict__ ptPerm,
   }
 }
 
-// temp[3*pt] = sum_j Hlp_j x[9*cam_j]: one thread per point over its
-// point-sorted edge run; no atomics; x (9*ncam doubles) is L2-resident.
+// temp[3*pt] += Hlp_j x[9*cam_j]: one thread per point-sorted edge (fully
+// coalesced Hlp reads; x is L2-resident), wave-level segmented-scan over the
+// point runs so only segment tails touch memory (atomicAdd per tail instead
+// of 3 atomics per edge; a thread-per-point variant measured WORSE -- 27
+// read streams spread across lanes thrash L1).
 template <typename T>
-__global__ void kSpmvEtx(int npt, const int* __restrict__ ptRowPtr,
+__global__ void kSpmvEtx(int64_t nL, const int* __restrict__ ptOfPt,
                          const int* __restrict__ camOfPt,
-                         const T* __restrict__ Hlp, int64_t nL,
+                         const T* __restrict__ Hlp,
                          const T* __restrict__ x, T* __restrict__ out) {
-  for (int64_t p = blockIdx.x * (int64_t)kBlk + threadIdx.x; p < npt;
-       p += (int64_t)gridDim.x * kBlk) {
+  const int lane = threadIdx.x & 63;
+  for (int64_t j0 = blockIdx.x * (int64_t)kBlk + threadIdx.x;
+       j0 < ((nL + kBlk - 1) / kBlk) * (int64_t)kBlk;
+       j0 += (int64_t)gridDim.x * kBlk) {
+    const bool active = j0 < nL;
+    const int64_t j = active ? j0 : nL - 1;
+    const int pt = ptOfPt[j];
     T o0 = 0, o1 = 0, o2 = 0;
-    const int lo = ptRowPtr[p], hi = ptRowPtr[p + 1];
-    for (int j = lo; j < hi; ++j) {
+    if (active) {
       const T* xc = x + (int64_t)camOfPt[j] * 9;
       for (int i = 0; i < 9; ++i) {
         const T xi = xc[i];
@@ -414,9 +421,25 @@ __global__ void kSpmvEtx(int npt, const int* __restrict__ ptRowPtr,
         o2 += Hlp[((int64_t)(2 * 9 + i)) * nL + j] * xi;
       }
     }
-    out[3 * p] = o0;
-    out[3 * p + 1] = o1;
-    out[3 * p + 2] = o2;
+    // inclusive segmented scan over the wave (segments = equal pt id)
+    for (int off = 1; off < 64; off <<= 1) {
+      const int ppt = __shfl_up(pt, off, 64);
+      const T a0 = __shfl_up(o0, off, 64);
+      const T a1 = __shfl_up(o1, off, 64);
+      const T a2 = __shfl_up(o2, off, 64);
+      if (lane >= off && ppt == pt) {
+        o0 += a0;
+        o1 += a1;
+        o2 += a2;
+      }
+    }
+    const int npt_next = __shfl_down(pt, 1, 64);
+    const bool tail = active && (lane == 63 || npt_next != pt || j0 == nL - 1);
+    if (tail) {
+      atomicAdd(&out[3 * pt], o0);
+      atomicAdd(&out[3 * pt + 1], o1);
+      atomicAdd(&out[3 * pt + 2], o2);
+    }
   }
 }
 
@@ -684,7 +707,8 @@ class GpuEngine final : public Engine<T> {
 
     // Point-sorted view of the local edges (for the atomic-free E^T x).
     {
-      std::vector<int> perm(nL_), rowPtr(npt_ + 1, 0), camOfPt(nL_);
+      std::vector<int> perm(nL_), rowPtr(npt_ + 1, 0), camOfPt(nL_),
+          ptOfPt(nL_);
       for (int64_t e = 0; e < nL_; ++e) rowPtr[ix.ptOf[e0_ + e] + 1]++;
       for (int p = 0; p < npt_; ++p) rowPtr[p + 1] += rowPtr[p];
       std::vector<int> cursor(rowPtr.begin(), rowPtr.end() - 1);
@@ -692,13 +716,16 @@ class GpuEngine final : public Engine<T> {
         const int pos = cursor[ix.ptOf[e0_ + e]]++;
         perm[pos] = (int)e;
         camOfPt[pos] = ix.camOf[e0_ + e];
+        ptOfPt[pos] = ix.ptOf[e0_ + e];
       }
       dPtPerm_ = dalloc<int>(nL_);
       dPtRowPtr_ = dalloc<int>(npt_ + 1);
       dCamOfPt_ = dalloc<int>(nL_);
+      dPtOfPt_ = dalloc<int>(nL_);
       up(dPtPerm_, perm.data(), nL_);
       up(dPtRowPtr_, rowPtr.data(), npt_ + 1);
       up(dCamOfPt_, camOfPt.data(), nL_);
+      up(dPtOfPt_, ptOfPt.data(), nL_);
     }
     dHlp_ = dalloc<T>(nL_ * 27);
     sync();
@@ -1028,9 +1055,9 @@ class GpuEngine final : public Engine<T> {
                        nBlk, A, xv, yv);
   }
   void spmvEtx(const T* xv, T* out) {
-    hipLaunchKernelGGL(kSpmvEtx<T>, dim3(gridFor(npt_)), dim3(kBlk), 0,
-                       stream_, npt_, dPtRowPtr_, dCamOfPt_, dHlp_, nL_, xv,
-                       out);
+    HIP_CHECK(hipMemsetAsync(out, 0, np_ * sizeof(T), stream_));
+    hipLaunchKernelGGL(kSpmvEtx<T>, dim3(gridFor(nL_)), dim3(kBlk), 0,
+                       stream_, nL_, dPtOfPt_, dCamOfPt_, dHlp_, xv, out);
   }
   void spmvEx(const T* wv, T* out) {
     HIP_CHECK(hipMemsetAsync(out, 0, nc_ * sizeof(T), stream_));
@@ -1058,7 +1085,7 @@ class GpuEngine final : public Engine<T> {
   int cur_ = 0;
   int nChunks_ = 0;
   int *dCamOf_{}, *dPtOf_{}, *dChCam_{}, *dChLo_{}, *dChHi_{}, *dFail_{};
-  int *dPtPerm_{}, *dPtRowPtr_{}, *dCamOfPt_{};
+  int *dPtPerm_{}, *dPtRowPtr_{}, *dCamOfPt_{}, *dPtOfPt_{};
   T *dMeas_{}, *dInfo_{};
   T *dParams_{}, *dParamsBak_{};
   T *dR_[2]{}, *dJc_[2]{}, *dJp_[2]{};
