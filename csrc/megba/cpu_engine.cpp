@@ -3,6 +3,7 @@
 #include <cmath>
 #include <cstring>
 
+#include "analytical.hpp"
 #include "bal_functor.hpp"
 #include "lm.hpp"
 #include "smallmat.hpp"
@@ -18,7 +19,8 @@ class CpuEngine final : public Engine<T> {
         rank_(opt.rank),
         world_(opt.worldSize),
         ncam_(ix.ncam),
-        npt_(ix.npt) {
+        npt_(ix.npt),
+        analytical_(opt.diff == DiffMode::ANALYTICAL) {
     e0_ = ix.split[rank_];
     e1_ = ix.split[rank_ + 1];
     nL_ = e1_ - e0_;
@@ -73,9 +75,20 @@ class CpuEngine final : public Engine<T> {
     T chi2 = T(0);
 #pragma omp parallel for schedule(static) reduction(+ : chi2)
     for (int64_t e = 0; e < nL_; ++e) {
-      J cam[9], pt[3], res[2];
       const T* cp = &cams_[(size_t)camOf_[e] * 9];
       const T* pp = &pts_[(size_t)ptOf_[e] * 3];
+      if (analytical_) {
+        T res[2], jc[2][9], jp[2][3];
+        balAnalytical<T>(cp, pp, &meas_[2 * e], res, jc, jp);
+        for (int row = 0; row < 2; ++row) {
+          rCur_[2 * e + row] = res[row];
+          chi2 += res[row] * res[row];
+          for (int i = 0; i < 9; ++i) JcCur_[18 * e + 9 * row + i] = jc[row][i];
+          for (int i = 0; i < 3; ++i) JpCur_[6 * e + 3 * row + i] = jp[row][i];
+        }
+        continue;
+      }
+      J cam[9], pt[3], res[2];
       for (int i = 0; i < 9; ++i) cam[i] = J::leaf(cp[i], i);
       for (int i = 0; i < 3; ++i) pt[i] = J::leaf(pp[i], 9 + i);
       balReprojectionError<T, J>(cam, pt, &meas_[2 * e], res);
@@ -458,6 +471,7 @@ class CpuEngine final : public Engine<T> {
 
   HostAllreduce<T> ar_;
   int rank_, world_, ncam_, npt_;
+  bool analytical_ = false;
   int camLo_ = 0, camHi_ = 0;
   int64_t e0_ = 0, e1_ = 0, nL_ = 0, dim_ = 0;
   std::vector<int> camOf_, ptOf_;

@@ -29,6 +29,7 @@
 #include <cstring>
 #include <vector>
 
+#include "../analytical.hpp"
 #include "../bal_functor.hpp"
 #include "../smallmat.hpp"
 #include "gpu_engine.hpp"
@@ -156,6 +157,44 @@ __global__ __launch_bounds__(256, 2) void kForward(
         rOut[(int64_t)row * nL + e] = res[row].v;
         chi2 += (double)res[row].v * (double)res[row].v;
       }
+    }
+  }
+  sm[threadIdx.x] = chi2;
+  __syncthreads();
+  for (int s = kBlk / 2; s > 0; s >>= 1) {
+    if (threadIdx.x < s) sm[threadIdx.x] += sm[threadIdx.x + s];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) atomicAdd(chi2Acc, sm[0]);
+}
+
+// Analytical-derivative forward: one thread per edge, closed-form residual +
+// 2x12 Jacobian (reference C11, src/geo/analytical_derivatives.cu) -- no
+// dual-number redundancy, ~70 fp64 values live per lane.
+template <typename T>
+__global__ void kForwardAnalytical(int64_t nL, const int* __restrict__ camOf,
+                                   const int* __restrict__ ptOf,
+                                   const T* __restrict__ params, int ncam,
+                                   const T* __restrict__ meas,
+                                   T* __restrict__ rOut, T* __restrict__ Jc,
+                                   T* __restrict__ Jp, double* chi2Acc) {
+  __shared__ double sm[kBlk];
+  double chi2 = 0.0;
+  const T* ptsBase = params + (int64_t)ncam * 9;
+  for (int64_t e = blockIdx.x * (int64_t)kBlk + threadIdx.x; e < nL;
+       e += (int64_t)gridDim.x * kBlk) {
+    const T* cp = params + (int64_t)camOf[e] * 9;
+    const T* pp = ptsBase + (int64_t)ptOf[e] * 3;
+    const T m[2] = {meas[2 * e], meas[2 * e + 1]};
+    T res[2], jc[2][9], jp[2][3];
+    balAnalytical<T>(cp, pp, m, res, jc, jp);
+    for (int row = 0; row < 2; ++row) {
+      rOut[(int64_t)row * nL + e] = res[row];
+      chi2 += (double)res[row] * (double)res[row];
+      for (int col = 0; col < 9; ++col)
+        Jc[((int64_t)(col * 2 + row)) * nL + e] = jc[row][col];
+      for (int col = 0; col < 3; ++col)
+        Jp[((int64_t)(col * 2 + row)) * nL + e] = jp[row][col];
     }
   }
   sm[threadIdx.x] = chi2;
@@ -605,7 +644,8 @@ class GpuEngine final : public Engine<T> {
  public:
   GpuEngine(const BAProblemHost& prob, const ProblemIndex& ix,
             const ProblemOption& opt, const std::string& rcclId)
-      : rank_(opt.rank), world_(opt.worldSize), ncam_(ix.ncam), npt_(ix.npt) {
+      : rank_(opt.rank), world_(opt.worldSize), ncam_(ix.ncam), npt_(ix.npt),
+        analytical_(opt.diff == DiffMode::ANALYTICAL) {
     HIP_CHECK(hipSetDevice(opt.deviceIndex));
     HIP_CHECK(hipStreamCreate(&stream_));
     e0_ = ix.split[rank_];
@@ -739,9 +779,16 @@ class GpuEngine final : public Engine<T> {
 
   double forward() override {
     zeroScalar();
-    hipLaunchKernelGGL(kForward<T>, dim3(gridFor(nL_ * 4)), dim3(kBlk), 0,
-                       stream_, nL_, dCamOf_, dPtOf_, dParams_, ncam_, dMeas_,
-                       dR_[cur_], dJc_[cur_], dJp_[cur_], scalarPtr());
+    if (analytical_)
+      hipLaunchKernelGGL(kForwardAnalytical<T>, dim3(gridFor(nL_)), dim3(kBlk),
+                         0, stream_, nL_, dCamOf_, dPtOf_, dParams_, ncam_,
+                         dMeas_, dR_[cur_], dJc_[cur_], dJp_[cur_],
+                         scalarPtr());
+    else
+      hipLaunchKernelGGL(kForward<T>, dim3(gridFor(nL_ * 4)), dim3(kBlk), 0,
+                         stream_, nL_, dCamOf_, dPtOf_, dParams_, ncam_,
+                         dMeas_, dR_[cur_], dJc_[cur_], dJp_[cur_],
+                         scalarPtr());
     return globalScalar();
   }
 
@@ -1082,6 +1129,7 @@ class GpuEngine final : public Engine<T> {
   int rank_, world_, ncam_, npt_;
   int64_t e0_ = 0, e1_ = 0, nL_ = 0, nc_ = 0, np_ = 0, dim_ = 0;
   bool hasInfo_ = false;
+  bool analytical_ = false;
   int cur_ = 0;
   int nChunks_ = 0;
   int *dCamOf_{}, *dPtOf_{}, *dChCam_{}, *dChLo_{}, *dChHi_{}, *dFail_{};
