@@ -20,7 +20,8 @@ class CpuEngine final : public Engine<T> {
         world_(opt.worldSize),
         ncam_(ix.ncam),
         npt_(ix.npt),
-        analytical_(opt.diff == DiffMode::ANALYTICAL) {
+        analytical_(opt.diff == DiffMode::ANALYTICAL),
+        implicit_(opt.schur == SchurMode::IMPLICIT) {
     e0_ = ix.split[rank_];
     e1_ = ix.split[rank_ + 1];
     nL_ = e1_ - e0_;
@@ -58,7 +59,7 @@ class CpuEngine final : public Engine<T> {
 
     Hpp_.assign((size_t)ncam_ * 81, T(0));
     Hll_.assign((size_t)npt_ * 9, T(0));
-    Hpl_.assign((size_t)nL_ * 27, T(0));
+    Hpl_.assign(implicit_ ? (size_t)0 : (size_t)nL_ * 27, T(0));
     dim_ = (int64_t)ncam_ * 9 + (int64_t)npt_ * 3;
     g_.assign(dim_, T(0));
     HppD_.assign(Hpp_.size(), T(0));
@@ -119,11 +120,13 @@ class CpuEngine final : public Engine<T> {
       T wJc[2][9], wJp[2][3], wr[2];
       weightedRows(e, wJc, wJp, wr);
       const T* Jp = &JpBak_[6 * e];
-      T* hpl = &Hpl_[27 * e];
-      for (int i = 0; i < 9; ++i)
-        for (int j = 0; j < 3; ++j)
-          hpl[i * 3 + j] = JcBak_[18 * e + i] * wJp[0][j] +
-                           JcBak_[18 * e + 9 + i] * wJp[1][j];
+      if (!implicit_) {
+        T* hpl = &Hpl_[27 * e];
+        for (int i = 0; i < 9; ++i)
+          for (int j = 0; j < 3; ++j)
+            hpl[i * 3 + j] = JcBak_[18 * e + i] * wJp[0][j] +
+                             JcBak_[18 * e + 9 + i] * wJp[1][j];
+      }
       T* hll = &Hll_[(size_t)ptOf_[e] * 9];
       for (int i = 0; i < 3; ++i)
         for (int j = 0; j < 3; ++j) {
@@ -401,6 +404,29 @@ class CpuEngine final : public Engine<T> {
   // temp[3npt] = sum over local edges of Hpl_e^T * x[cam]; caller allreduces.
   void spmvEtx(const T* x, T* temp) {
     std::fill(temp, temp + (size_t)npt_ * 3, T(0));
+    if (implicit_) {
+      // matrix-free: Hpl_e^T x = Jp^T W (Jc x)  (reference C23,
+      // implicit_schur_pcg_solver.cu:57-90)
+#pragma omp parallel for schedule(static)
+      for (int64_t e = 0; e < nL_; ++e) {
+        const T* Jc = &JcBak_[18 * e];
+        const T* Jp = &JpBak_[6 * e];
+        const T* xc = &x[(size_t)camOf_[e] * 9];
+        T u0 = T(0), u1 = T(0);
+        for (int i = 0; i < 9; ++i) {
+          u0 += Jc[i] * xc[i];
+          u1 += Jc[9 + i] * xc[i];
+        }
+        applyInfo(e, u0, u1);
+        T* out = &temp[(size_t)ptOf_[e] * 3];
+        for (int j = 0; j < 3; ++j) {
+          const T v = Jp[j] * u0 + Jp[3 + j] * u1;
+#pragma omp atomic
+          out[j] += v;
+        }
+      }
+      return;
+    }
 #pragma omp parallel for schedule(static)
     for (int64_t e = 0; e < nL_; ++e) {
       const T* blk = &Hpl_[27 * e];
@@ -415,9 +441,37 @@ class CpuEngine final : public Engine<T> {
     }
   }
 
+  inline void applyInfo(int64_t e, T& u0, T& u1) {
+    if (hasInfo_) {
+      const T w00 = info_[3 * e], w01 = info_[3 * e + 1],
+              w11 = info_[3 * e + 2];
+      const T a = w00 * u0 + w01 * u1;
+      u1 = w01 * u0 + w11 * u1;
+      u0 = a;
+    }
+  }
+
   // out[9ncam] = sum over local edges of Hpl_e * w[pt]  (no allreduce here).
   void spmvEx(const T* w, T* out) {
     std::fill(out, out + (size_t)ncam_ * 9, T(0));
+    if (implicit_) {
+#pragma omp parallel for schedule(dynamic, 8)
+      for (int c = camLo_; c < camHi_; ++c) {
+        const int64_t lo = std::max(camRowPtr_[c], e0_) - e0_;
+        const int64_t hi = std::min(camRowPtr_[c + 1], e1_) - e0_;
+        T* oc = &out[(size_t)c * 9];
+        for (int64_t e = lo; e < hi; ++e) {
+          const T* Jc = &JcBak_[18 * e];
+          const T* Jp = &JpBak_[6 * e];
+          const T* wp = &w[(size_t)ptOf_[e] * 3];
+          T u0 = Jp[0] * wp[0] + Jp[1] * wp[1] + Jp[2] * wp[2];
+          T u1 = Jp[3] * wp[0] + Jp[4] * wp[1] + Jp[5] * wp[2];
+          applyInfo(e, u0, u1);
+          for (int i = 0; i < 9; ++i) oc[i] += Jc[i] * u0 + Jc[9 + i] * u1;
+        }
+      }
+      return;
+    }
 #pragma omp parallel for schedule(dynamic, 8)
     for (int c = camLo_; c < camHi_; ++c) {
       const int64_t lo = std::max(camRowPtr_[c], e0_) - e0_;
@@ -472,6 +526,7 @@ class CpuEngine final : public Engine<T> {
   HostAllreduce<T> ar_;
   int rank_, world_, ncam_, npt_;
   bool analytical_ = false;
+  bool implicit_ = false;
   int camLo_ = 0, camHi_ = 0;
   int64_t e0_ = 0, e1_ = 0, nL_ = 0, dim_ = 0;
   std::vector<int> camOf_, ptOf_;

@@ -263,11 +263,13 @@ __global__ void kAssembleEdge(int64_t nL, const int* __restrict__ camOf,
       wr[1] = rr[1];
     }
     // Hpl: one conflict-free 9x3 block per observation, grad-major layout
-    // Hpl[(a*3+b)*nL + e] so SpMV reads coalesce across lanes.
-    for (int a = 0; a < 9; ++a)
-      for (int b = 0; b < 3; ++b)
-        Hpl[((int64_t)(a * 3 + b)) * nL + e] =
-            jc[0][a] * wjp[0][b] + jc[1][a] * wjp[1][b];
+    // Hpl[(a*3+b)*nL + e] so SpMV reads coalesce across lanes.  Null in
+    // implicit-Schur mode (diagonal blocks only, reference C15).
+    if (Hpl != nullptr)
+      for (int a = 0; a < 9; ++a)
+        for (int b = 0; b < 3; ++b)
+          Hpl[((int64_t)(a * 3 + b)) * nL + e] =
+              jc[0][a] * wjp[0][b] + jc[1][a] * wjp[1][b];
     // Hll + g_p: point degree is low (few obs/point) -> atomics are cheap.
     const int pt = ptOf[e];
     T* hll = Hll + (int64_t)pt * 9;
@@ -516,6 +518,76 @@ __global__ __launch_bounds__(64) void kSpmvEx(int nChunks,
   }
 }
 
+// Implicit (matrix-free) Schur products from the accepted Jacobians
+// (reference C23, implicit_schur_pcg_solver.cu:21-90): E^T x = Jp^T W (Jc x)
+// per edge with atomic scatter (point degree is low); E w = Jc^T W (Jp w)
+// chunked per camera block-row with a wave reduction.
+template <typename T, bool HASINFO>
+__global__ void kSpmvEtxImp(int64_t nL, const int* __restrict__ camOf,
+                            const int* __restrict__ ptOf,
+                            const T* __restrict__ Jc, const T* __restrict__ Jp,
+                            const T* __restrict__ info,
+                            const T* __restrict__ x, T* __restrict__ out) {
+  for (int64_t e = blockIdx.x * (int64_t)kBlk + threadIdx.x; e < nL;
+       e += (int64_t)gridDim.x * kBlk) {
+    const T* xc = x + (int64_t)camOf[e] * 9;
+    T u0 = T(0), u1 = T(0);
+    for (int i = 0; i < 9; ++i) {
+      const T xi = xc[i];
+      u0 += Jc[((int64_t)(i * 2 + 0)) * nL + e] * xi;
+      u1 += Jc[((int64_t)(i * 2 + 1)) * nL + e] * xi;
+    }
+    if (HASINFO) {
+      const T w00 = info[3 * e], w01 = info[3 * e + 1], w11 = info[3 * e + 2];
+      const T a = w00 * u0 + w01 * u1;
+      u1 = w01 * u0 + w11 * u1;
+      u0 = a;
+    }
+    T* op = out + (int64_t)ptOf[e] * 3;
+    for (int j = 0; j < 3; ++j)
+      atomicAdd(&op[j], Jp[((int64_t)(j * 2 + 0)) * nL + e] * u0 +
+                            Jp[((int64_t)(j * 2 + 1)) * nL + e] * u1);
+  }
+}
+
+template <typename T, bool HASINFO>
+__global__ __launch_bounds__(64) void kSpmvExImp(
+    int nChunks, const int* __restrict__ chCam, const int* __restrict__ chLo,
+    const int* __restrict__ chHi, const int* __restrict__ ptOf,
+    const T* __restrict__ Jc, const T* __restrict__ Jp,
+    const T* __restrict__ info, int64_t nL, const T* __restrict__ w,
+    T* __restrict__ out) {
+  const int chunk = blockIdx.x;
+  if (chunk >= nChunks) return;
+  const int cam = chCam[chunk];
+  T acc[9];
+  for (int i = 0; i < 9; ++i) acc[i] = T(0);
+  const int lo = chLo[chunk], hi = chHi[chunk];
+  for (int e = lo + (int)threadIdx.x; e < hi; e += 64) {
+    const T* wp = w + (int64_t)ptOf[e] * 3;
+    T u0 = T(0), u1 = T(0);
+    for (int j = 0; j < 3; ++j) {
+      u0 += Jp[((int64_t)(j * 2 + 0)) * nL + e] * wp[j];
+      u1 += Jp[((int64_t)(j * 2 + 1)) * nL + e] * wp[j];
+    }
+    if (HASINFO) {
+      const T w00 = info[3 * e], w01 = info[3 * e + 1], w11 = info[3 * e + 2];
+      const T a = w00 * u0 + w01 * u1;
+      u1 = w01 * u0 + w11 * u1;
+      u0 = a;
+    }
+    for (int i = 0; i < 9; ++i)
+      acc[i] += Jc[((int64_t)(i * 2 + 0)) * nL + e] * u0 +
+                Jc[((int64_t)(i * 2 + 1)) * nL + e] * u1;
+  }
+  for (int off = 32; off > 0; off >>= 1)
+    for (int i = 0; i < 9; ++i) acc[i] += __shfl_down(acc[i], off, 64);
+  if (threadIdx.x == 0) {
+    T* oc = out + (int64_t)cam * 9;
+    for (int i = 0; i < 9; ++i) atomicAdd(&oc[i], acc[i]);
+  }
+}
+
 // Block-diagonal matvec, one thread per output row.
 // MODE 0: y = A x;  MODE 1: y = A x - y  (the reference's rw=1,dw=-1 gemv).
 template <typename T, int D, int MODE>
@@ -645,7 +717,8 @@ class GpuEngine final : public Engine<T> {
   GpuEngine(const BAProblemHost& prob, const ProblemIndex& ix,
             const ProblemOption& opt, const std::string& rcclId)
       : rank_(opt.rank), world_(opt.worldSize), ncam_(ix.ncam), npt_(ix.npt),
-        analytical_(opt.diff == DiffMode::ANALYTICAL) {
+        analytical_(opt.diff == DiffMode::ANALYTICAL),
+        implicit_(opt.schur == SchurMode::IMPLICIT) {
     HIP_CHECK(hipSetDevice(opt.deviceIndex));
     HIP_CHECK(hipStreamCreate(&stream_));
     e0_ = ix.split[rank_];
@@ -699,7 +772,7 @@ class GpuEngine final : public Engine<T> {
     // Linear system.
     dHpp_ = dalloc<T>((int64_t)ncam_ * 81);
     dHll_ = dalloc<T>((int64_t)npt_ * 9);
-    dHpl_ = dalloc<T>(nL_ * 27);
+    if (!implicit_) dHpl_ = dalloc<T>(nL_ * 27);
     dG_ = dalloc<T>(dim_);
     dGBak_ = dalloc<T>(dim_);
     dHppD_ = dalloc<T>((int64_t)ncam_ * 81);
@@ -767,7 +840,7 @@ class GpuEngine final : public Engine<T> {
       up(dCamOfPt_, camOfPt.data(), nL_);
       up(dPtOfPt_, ptOfPt.data(), nL_);
     }
-    dHlp_ = dalloc<T>(nL_ * 27);
+    if (!implicit_) dHlp_ = dalloc<T>(nL_ * 27);
     sync();
   }
 
@@ -818,8 +891,9 @@ class GpuEngine final : public Engine<T> {
                            0, stream_, nChunks_, dChCam_, dChLo_, dChHi_, nL_,
                            dR_[bak], dJc_[bak], nullptr, dHpp_, dG_);
     }
-    hipLaunchKernelGGL(kTransposeHlp<T>, dim3(gridFor(nL_)), dim3(kBlk), 0,
-                       stream_, nL_, dPtPerm_, dHpl_, dHlp_);
+    if (!implicit_)
+      hipLaunchKernelGGL(kTransposeHlp<T>, dim3(gridFor(nL_)), dim3(kBlk), 0,
+                         stream_, nL_, dPtPerm_, dHpl_, dHlp_);
     allreduce(dHpp_, (int64_t)ncam_ * 81);
     allreduce(dHll_, (int64_t)npt_ * 9);
     allreduce(dG_, dim_);
@@ -978,7 +1052,7 @@ class GpuEngine final : public Engine<T> {
     d.Jp = down(dJp_[cur_], nL_ * 6);
     d.Hpp = down(dHpp_, (int64_t)ncam_ * 81);
     d.Hll = down(dHll_, (int64_t)npt_ * 9);
-    {
+    if (!implicit_) {
       // device Hpl is grad-major [27][nL]; dump as [e][9][3]
       std::vector<double> gm = down(dHpl_, nL_ * 27);
       std::vector<double> o(gm.size());
@@ -1103,15 +1177,39 @@ class GpuEngine final : public Engine<T> {
   }
   void spmvEtx(const T* xv, T* out) {
     HIP_CHECK(hipMemsetAsync(out, 0, np_ * sizeof(T), stream_));
+    if (implicit_) {
+      const int bak = cur_ ^ 1;
+      if (hasInfo_)
+        hipLaunchKernelGGL((kSpmvEtxImp<T, true>), dim3(gridFor(nL_)),
+                           dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
+                           dJc_[bak], dJp_[bak], dInfo_, xv, out);
+      else
+        hipLaunchKernelGGL((kSpmvEtxImp<T, false>), dim3(gridFor(nL_)),
+                           dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
+                           dJc_[bak], dJp_[bak], nullptr, xv, out);
+      return;
+    }
     hipLaunchKernelGGL(kSpmvEtx<T>, dim3(gridFor(nL_)), dim3(kBlk), 0,
                        stream_, nL_, dPtOfPt_, dCamOfPt_, dHlp_, xv, out);
   }
   void spmvEx(const T* wv, T* out) {
     HIP_CHECK(hipMemsetAsync(out, 0, nc_ * sizeof(T), stream_));
-    if (nChunks_ > 0)
-      hipLaunchKernelGGL(kSpmvEx<T>, dim3(nChunks_), dim3(64), 0, stream_,
-                         nChunks_, dChCam_, dChLo_, dChHi_, dPtOf_, dHpl_, nL_,
-                         wv, out);
+    if (nChunks_ == 0) return;
+    if (implicit_) {
+      const int bak = cur_ ^ 1;
+      if (hasInfo_)
+        hipLaunchKernelGGL((kSpmvExImp<T, true>), dim3(nChunks_), dim3(64), 0,
+                           stream_, nChunks_, dChCam_, dChLo_, dChHi_, dPtOf_,
+                           dJc_[bak], dJp_[bak], dInfo_, nL_, wv, out);
+      else
+        hipLaunchKernelGGL((kSpmvExImp<T, false>), dim3(nChunks_), dim3(64), 0,
+                           stream_, nChunks_, dChCam_, dChLo_, dChHi_, dPtOf_,
+                           dJc_[bak], dJp_[bak], nullptr, nL_, wv, out);
+      return;
+    }
+    hipLaunchKernelGGL(kSpmvEx<T>, dim3(nChunks_), dim3(64), 0, stream_,
+                       nChunks_, dChCam_, dChLo_, dChHi_, dPtOf_, dHpl_, nL_,
+                       wv, out);
   }
   // q = S x = HppD x - E Cinv E^T x   (2 allreduces, reference site A4).
   void schurApply(const T* xv, T* q) {
@@ -1130,6 +1228,7 @@ class GpuEngine final : public Engine<T> {
   int64_t e0_ = 0, e1_ = 0, nL_ = 0, nc_ = 0, np_ = 0, dim_ = 0;
   bool hasInfo_ = false;
   bool analytical_ = false;
+  bool implicit_ = false;
   int cur_ = 0;
   int nChunks_ = 0;
   int *dCamOf_{}, *dPtOf_{}, *dChCam_{}, *dChLo_{}, *dChHi_{}, *dFail_{};
