@@ -32,6 +32,7 @@ SOURCES = [
     # (path, is_device_code)
     ("megba/cpu_engine.cpp", False),
     ("megba/gpu/gpu_engine.hip", True),
+    ("megba/jv/jetvector.hip", True),
     ("bindings.cpp", False),
 ]
 

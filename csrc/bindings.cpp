@@ -10,6 +10,7 @@
 #include <memory>
 
 #include "megba/common.hpp"
+#include "megba/jv/jetvector.hpp"
 #include "megba/cpu_engine.hpp"
 #include "megba/engine.hpp"
 #include "megba/lm.hpp"
@@ -271,8 +272,74 @@ struct PyProblem {
 
 }  // namespace
 
+namespace {
+
+// Python-facing JetVector (fp64).  The C++ layer is templated on
+// double/float; Python exposes double, which is what tests and the
+// user-defined-edge path use.
+struct PyJetVec {
+  JetVec<double> v;
+};
+
+PyJetVec mkJv(py::array_t<double, py::array::c_style | py::array::forcecast> value,
+              py::object grad, int N, int grad_pos, bool gpu) {
+  const int64_t n = value.size();
+  const double* gptr = nullptr;
+  py::array_t<double, py::array::c_style | py::array::forcecast> garr;
+  if (!grad.is_none()) {
+    garr = py::cast<py::array_t<double, py::array::c_style | py::array::forcecast>>(grad);
+    MEGBA_CHECK(garr.size() == (py::ssize_t)(N * n), "grad must be (N, nItem)");
+    gptr = garr.data();
+  }
+  return PyJetVec{jvFromHost<double>(value.data(), gptr, n, N, grad_pos, gpu)};
+}
+
+py::tuple jvDownload(const PyJetVec& a) {
+  py::array_t<double> value((py::ssize_t)a.v.nItem);
+  py::array_t<double> grad({(py::ssize_t)a.v.N, (py::ssize_t)a.v.nItem});
+  jvToHost<double>(a.v, value.mutable_data(), grad.mutable_data());
+  return py::make_tuple(value, grad);
+}
+
+std::vector<JetVec<double>> jvList(py::sequence seq) {
+  std::vector<JetVec<double>> out;
+  for (auto h : seq) out.push_back(py::cast<PyJetVec&>(h).v);
+  return out;
+}
+
+py::list jvWrap(const std::vector<JetVec<double>>& vs) {
+  py::list out;
+  for (const auto& v : vs) out.append(PyJetVec{v});
+  return out;
+}
+
+}  // namespace
+
 PYBIND11_MODULE(_core, m) {
   m.doc() = "megba_amd core (MI355X-native distributed bundle adjustment)";
+
+  py::class_<PyJetVec>(m, "JetVector")
+      .def(py::init(&mkJv), py::arg("value"), py::arg("grad") = py::none(),
+           py::arg("N") = 0, py::arg("grad_pos") = -1, py::arg("gpu") = false)
+      .def_property_readonly("n_item", [](const PyJetVec& a) { return a.v.nItem; })
+      .def_property_readonly("N", [](const PyJetVec& a) { return a.v.N; })
+      .def("to_numpy", &jvDownload);
+  m.def("jv_scalar", [](double s, int N) { return PyJetVec{jvScalar<double>(s, N)}; });
+  auto binop = [](const char* name, JvOp op, py::module_& m) {};
+  m.def("jv_add", [](const PyJetVec& a, const PyJetVec& b) { return PyJetVec{jvBinary(JvOp::Add, a.v, b.v)}; });
+  m.def("jv_sub", [](const PyJetVec& a, const PyJetVec& b) { return PyJetVec{jvBinary(JvOp::Sub, a.v, b.v)}; });
+  m.def("jv_mul", [](const PyJetVec& a, const PyJetVec& b) { return PyJetVec{jvBinary(JvOp::Mul, a.v, b.v)}; });
+  m.def("jv_div", [](const PyJetVec& a, const PyJetVec& b) { return PyJetVec{jvBinary(JvOp::Div, a.v, b.v)}; });
+  m.def("jv_neg", [](const PyJetVec& a) { return PyJetVec{jvUnary(JvUnary::Neg, a.v)}; });
+  m.def("jv_abs", [](const PyJetVec& a) { return PyJetVec{jvUnary(JvUnary::Abs, a.v)}; });
+  m.def("jv_sin", [](const PyJetVec& a) { return PyJetVec{jvUnary(JvUnary::Sin, a.v)}; });
+  m.def("jv_cos", [](const PyJetVec& a) { return PyJetVec{jvUnary(JvUnary::Cos, a.v)}; });
+  m.def("jv_sqrt", [](const PyJetVec& a) { return PyJetVec{jvUnary(JvUnary::Sqrt, a.v)}; });
+  m.def("jv_angle_axis_to_rotation", [](py::sequence aa) { return jvWrap(jvAngleAxisToRotation(jvList(aa))); });
+  m.def("jv_rotation2d", [](const PyJetVec& t) { return jvWrap(jvRotation2D(t.v)); });
+  m.def("jv_quaternion_to_rotation", [](py::sequence q) { return jvWrap(jvQuaternionToRotation(jvList(q))); });
+  m.def("jv_normalize_quaternion", [](py::sequence q) { return jvWrap(jvNormalizeQuaternion(jvList(q))); });
+  m.def("jv_radial_distortion", [](py::sequence p, py::sequence intr) { return PyJetVec{jvRadialDistortion(jvList(p), jvList(intr))}; });
 
   py::class_<PyProblem>(m, "Problem")
       .def(py::init<py::array_t<double, py::array::c_style | py::array::forcecast>,

@@ -1,0 +1,440 @@
+// JetVector elementwise + geometry ops: HIP kernels (gfx950) and the CPU
+// (OpenMP) backend, generated from one template over operand kinds.
+// Reference anchor: src/operator/jet_vector_math_impl.cu (39 hand-written
+// kernels) and src/operator/jet_vector_math_impl.cpp (host backend).
+#include <hip/hip_runtime.h>
+
+#include <cmath>
+#include <cstring>
+
+#include "../jet.hpp"  // MEGBA_HD
+#include "jetvector.hpp"
+
+namespace megba {
+
+#define JV_HIP_CHECK(expr)                                                   \
+  do {                                                                       \
+    hipError_t _e = (expr);                                                  \
+    MEGBA_CHECK(_e == hipSuccess,                                            \
+                std::string("HIP error: ") + hipGetErrorString(_e));         \
+  } while (0)
+
+template <typename T>
+DeviceBuf<T>::~DeviceBuf() {
+  if (!ptr) return;
+  if (onGpu)
+    (void)hipFree(ptr);
+  else
+    free(ptr);
+}
+template struct DeviceBuf<double>;
+template struct DeviceBuf<float>;
+
+namespace {
+
+template <typename T>
+std::shared_ptr<DeviceBuf<T>> makeBuf(int64_t n, bool onGpu) {
+  auto b = std::make_shared<DeviceBuf<T>>();
+  b->n = n;
+  b->onGpu = onGpu;
+  if (onGpu) {
+    void* p = nullptr;
+    JV_HIP_CHECK(hipMalloc(&p, (n > 0 ? n : 1) * sizeof(T)));
+    b->ptr = (T*)p;
+  } else {
+    b->ptr = (T*)malloc((n > 0 ? n : 1) * sizeof(T));
+    MEGBA_CHECK(b->ptr, "malloc failed");
+  }
+  return b;
+}
+
+// Operand access generic over kind, usable on host and device.
+template <typename T, int K>  // 0=DENSE 1=JPV 2=SCALAR
+struct Operand {
+  const T* value;
+  const T* grad;
+  T scalar;
+  int gradPos;
+  int64_t n;
+  MEGBA_HD inline T val(int64_t i) const {
+    return K == 2 ? scalar : value[i];
+  }
+  MEGBA_HD inline T der(int g, int64_t i) const {
+    if (K == 0) return grad[(int64_t)g * n + i];
+    if (K == 1) return g == gradPos ? T(1) : T(0);
+    return T(0);
+  }
+};
+
+template <typename T, int OP, int KA, int KB>
+__global__ void kJvBinary(Operand<T, KA> a, Operand<T, KB> b, int64_t n, int N,
+                          T* ov, T* og) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const T av = a.val(i), bv = b.val(i);
+    T v;
+    if (OP == 0) v = av + bv;
+    if (OP == 1) v = av - bv;
+    if (OP == 2) v = av * bv;
+    if (OP == 3) v = av / bv;
+    ov[i] = v;
+    for (int g = 0; g < N; ++g) {
+      const T ag = a.der(g, i), bg = b.der(g, i);
+      T d;
+      if (OP == 0) d = ag + bg;
+      if (OP == 1) d = ag - bg;
+      if (OP == 2) d = ag * bv + av * bg;
+      if (OP == 3) d = (ag - v * bg) / bv;
+      og[(int64_t)g * n + i] = d;
+    }
+  }
+}
+
+template <typename T, int OP>
+__global__ void kJvUnary(const T* av, int64_t n, int N, const T* ag_, int agPos,
+                         int agKind, T* ov, T* og) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const T v = av[i];
+    T outv, scale;  // d out = scale * d in  (except Neg/Abs sign logic)
+    if (OP == 0) { outv = -v; scale = T(-1); }
+    if (OP == 1) { outv = v < T(0) ? -v : v; scale = v < T(0) ? T(-1) : T(1); }
+    if (OP == 2) { outv = ::sin(v); scale = ::cos(v); }
+    if (OP == 3) { outv = ::cos(v); scale = -::sin(v); }
+    if (OP == 4) { outv = ::sqrt(v); scale = T(0.5) / outv; }
+    ov[i] = outv;
+    for (int g = 0; g < N; ++g) {
+      T ing;
+      if (agKind == 0) ing = ag_[(int64_t)g * n + i];
+      else if (agKind == 1) ing = g == agPos ? T(1) : T(0);
+      else ing = T(0);
+      og[(int64_t)g * n + i] = scale * ing;
+    }
+  }
+}
+
+// ---- CPU versions ----
+template <typename T, int OP, int KA, int KB>
+void cpuJvBinary(Operand<T, KA> a, Operand<T, KB> b, int64_t n, int N, T* ov,
+                 T* og) {
+#pragma omp parallel for schedule(static)
+  for (int64_t i = 0; i < n; ++i) {
+    const T av = a.val(i), bv = b.val(i);
+    T v{};
+    if (OP == 0) v = av + bv;
+    if (OP == 1) v = av - bv;
+    if (OP == 2) v = av * bv;
+    if (OP == 3) v = av / bv;
+    ov[i] = v;
+    for (int g = 0; g < N; ++g) {
+      const T ag = a.der(g, i), bg = b.der(g, i);
+      T d{};
+      if (OP == 0) d = ag + bg;
+      if (OP == 1) d = ag - bg;
+      if (OP == 2) d = ag * bv + av * bg;
+      if (OP == 3) d = (ag - v * bg) / bv;
+      og[(int64_t)g * n + i] = d;
+    }
+  }
+}
+
+template <typename T, int OP>
+void cpuJvUnary(const T* av, int64_t n, int N, const T* ag_, int agPos,
+                int agKind, T* ov, T* og) {
+#pragma omp parallel for schedule(static)
+  for (int64_t i = 0; i < n; ++i) {
+    const T v = av[i];
+    T outv{}, scale{};
+    if (OP == 0) { outv = -v; scale = T(-1); }
+    if (OP == 1) { outv = v < T(0) ? -v : v; scale = v < T(0) ? T(-1) : T(1); }
+    if (OP == 2) { outv = std::sin(v); scale = std::cos(v); }
+    if (OP == 3) { outv = std::cos(v); scale = -std::sin(v); }
+    if (OP == 4) { outv = std::sqrt(v); scale = T(0.5) / outv; }
+    ov[i] = outv;
+    for (int g = 0; g < N; ++g) {
+      T ing;
+      if (agKind == 0) ing = ag_[(int64_t)g * n + i];
+      else if (agKind == 1) ing = g == agPos ? T(1) : T(0);
+      else ing = T(0);
+      og[(int64_t)g * n + i] = scale * ing;
+    }
+  }
+}
+
+template <typename T, int K>
+Operand<T, K> makeOperand(const JetVec<T>& a) {
+  Operand<T, K> o;
+  o.value = a.value ? a.value->ptr : nullptr;
+  o.grad = a.grad ? a.grad->ptr : nullptr;
+  o.scalar = a.scalarVal;
+  o.gradPos = a.gradPos;
+  o.n = a.nItem;
+  return o;
+}
+
+inline int jvGrid(int64_t n) {
+  int64_t g = (n + 255) / 256;
+  return (int)(g < 1 ? 1 : (g > 4096 ? 4096 : g));
+}
+
+template <typename T, int OP, int KA, int KB>
+void runBinary(const JetVec<T>& a, const JetVec<T>& b, JetVec<T>& out) {
+  auto oa = makeOperand<T, KA>(a);
+  auto ob = makeOperand<T, KB>(b);
+  if (out.onGpu) {
+    hipLaunchKernelGGL((kJvBinary<T, OP, KA, KB>), dim3(jvGrid(out.nItem)),
+                       dim3(256), 0, 0, oa, ob, out.nItem, out.N,
+                       out.value->ptr, out.grad->ptr);
+    JV_HIP_CHECK(hipGetLastError());
+  } else {
+    cpuJvBinary<T, OP, KA, KB>(oa, ob, out.nItem, out.N, out.value->ptr,
+                               out.grad->ptr);
+  }
+}
+
+template <typename T, int OP>
+void dispatchBinary(const JetVec<T>& a, const JetVec<T>& b, JetVec<T>& out) {
+  const int ka = (int)a.kind(), kb = (int)b.kind();
+  // JvKind: DENSE=0, JPV=1, SCALAR=2 (enum order)
+  if (ka == 0 && kb == 0) runBinary<T, OP, 0, 0>(a, b, out);
+  else if (ka == 0 && kb == 1) runBinary<T, OP, 0, 1>(a, b, out);
+  else if (ka == 0 && kb == 2) runBinary<T, OP, 0, 2>(a, b, out);
+  else if (ka == 1 && kb == 0) runBinary<T, OP, 1, 0>(a, b, out);
+  else if (ka == 1 && kb == 1) runBinary<T, OP, 1, 1>(a, b, out);
+  else if (ka == 1 && kb == 2) runBinary<T, OP, 1, 2>(a, b, out);
+  else if (ka == 2 && kb == 0) runBinary<T, OP, 2, 0>(a, b, out);
+  else if (ka == 2 && kb == 1) runBinary<T, OP, 2, 1>(a, b, out);
+  else MEGBA_CHECK(false, "scalar op scalar is not a JetVector op");
+}
+
+}  // namespace
+
+template <typename T>
+JetVec<T> jvFromHost(const T* value, const T* grad, int64_t nItem, int N,
+                     int gradPos, bool onGpu) {
+  JetVec<T> v;
+  v.nItem = nItem;
+  v.N = N;
+  v.gradPos = gradPos;
+  v.onGpu = onGpu;
+  v.value = makeBuf<T>(nItem, onGpu);
+  if (onGpu)
+    JV_HIP_CHECK(hipMemcpy(v.value->ptr, value, nItem * sizeof(T),
+                           hipMemcpyHostToDevice));
+  else
+    std::memcpy(v.value->ptr, value, nItem * sizeof(T));
+  if (grad != nullptr && gradPos < 0) {
+    v.grad = makeBuf<T>((int64_t)N * nItem, onGpu);
+    if (onGpu)
+      JV_HIP_CHECK(hipMemcpy(v.grad->ptr, grad, (int64_t)N * nItem * sizeof(T),
+                             hipMemcpyHostToDevice));
+    else
+      std::memcpy(v.grad->ptr, grad, (int64_t)N * nItem * sizeof(T));
+  }
+  return v;
+}
+
+template <typename T>
+JetVec<T> jvScalar(T s, int N) {
+  JetVec<T> v;
+  v.isScalar = true;
+  v.scalarVal = s;
+  v.N = N;
+  return v;
+}
+
+template <typename T>
+void jvToHost(const JetVec<T>& a, T* value, T* grad) {
+  MEGBA_CHECK(!a.isScalar, "cannot download a scalar JetVector");
+  if (a.onGpu)
+    JV_HIP_CHECK(hipMemcpy(value, a.value->ptr, a.nItem * sizeof(T),
+                           hipMemcpyDeviceToHost));
+  else
+    std::memcpy(value, a.value->ptr, a.nItem * sizeof(T));
+  if (!grad) return;
+  if (a.grad) {
+    if (a.onGpu)
+      JV_HIP_CHECK(hipMemcpy(grad, a.grad->ptr,
+                             (int64_t)a.N * a.nItem * sizeof(T),
+                             hipMemcpyDeviceToHost));
+    else
+      std::memcpy(grad, a.grad->ptr, (int64_t)a.N * a.nItem * sizeof(T));
+  } else {
+    for (int g = 0; g < a.N; ++g)
+      for (int64_t i = 0; i < a.nItem; ++i)
+        grad[(int64_t)g * a.nItem + i] = (g == a.gradPos) ? T(1) : T(0);
+  }
+}
+
+template <typename T>
+static JetVec<T> denseLike(const JetVec<T>& a, const JetVec<T>& b) {
+  const JetVec<T>& ref = a.isScalar ? b : a;
+  MEGBA_CHECK(!ref.isScalar, "need at least one vector operand");
+  if (!a.isScalar && !b.isScalar) {
+    MEGBA_CHECK(a.nItem == b.nItem, "JetVector item-count mismatch");
+    MEGBA_CHECK(a.N == b.N, "JetVector gradient-width mismatch");
+    MEGBA_CHECK(a.onGpu == b.onGpu, "JetVector device mismatch");
+  }
+  JetVec<T> out;
+  out.nItem = ref.nItem;
+  out.N = ref.N;
+  out.onGpu = ref.onGpu;
+  out.value = makeBuf<T>(out.nItem, out.onGpu);
+  out.grad = makeBuf<T>((int64_t)out.N * out.nItem, out.onGpu);
+  return out;
+}
+
+template <typename T>
+JetVec<T> jvBinary(JvOp op, const JetVec<T>& a, const JetVec<T>& b) {
+  JetVec<T> out = denseLike(a, b);
+  switch (op) {
+    case JvOp::Add: dispatchBinary<T, 0>(a, b, out); break;
+    case JvOp::Sub: dispatchBinary<T, 1>(a, b, out); break;
+    case JvOp::Mul: dispatchBinary<T, 2>(a, b, out); break;
+    case JvOp::Div: dispatchBinary<T, 3>(a, b, out); break;
+  }
+  return out;
+}
+
+template <typename T>
+JetVec<T> jvUnary(JvUnary op, const JetVec<T>& a) {
+  MEGBA_CHECK(!a.isScalar, "unary op needs a vector operand");
+  JetVec<T> out = denseLike(a, a);
+  const int agKind = (int)a.kind();
+  const T* ag = a.grad ? a.grad->ptr : nullptr;
+  const int OP = (int)op;
+  if (out.onGpu) {
+    switch (OP) {
+      case 0: hipLaunchKernelGGL((kJvUnary<T, 0>), dim3(jvGrid(a.nItem)), dim3(256), 0, 0, a.value->ptr, a.nItem, a.N, ag, a.gradPos, agKind, out.value->ptr, out.grad->ptr); break;
+      case 1: hipLaunchKernelGGL((kJvUnary<T, 1>), dim3(jvGrid(a.nItem)), dim3(256), 0, 0, a.value->ptr, a.nItem, a.N, ag, a.gradPos, agKind, out.value->ptr, out.grad->ptr); break;
+      case 2: hipLaunchKernelGGL((kJvUnary<T, 2>), dim3(jvGrid(a.nItem)), dim3(256), 0, 0, a.value->ptr, a.nItem, a.N, ag, a.gradPos, agKind, out.value->ptr, out.grad->ptr); break;
+      case 3: hipLaunchKernelGGL((kJvUnary<T, 3>), dim3(jvGrid(a.nItem)), dim3(256), 0, 0, a.value->ptr, a.nItem, a.N, ag, a.gradPos, agKind, out.value->ptr, out.grad->ptr); break;
+      case 4: hipLaunchKernelGGL((kJvUnary<T, 4>), dim3(jvGrid(a.nItem)), dim3(256), 0, 0, a.value->ptr, a.nItem, a.N, ag, a.gradPos, agKind, out.value->ptr, out.grad->ptr); break;
+    }
+    JV_HIP_CHECK(hipGetLastError());
+  } else {
+    switch (OP) {
+      case 0: cpuJvUnary<T, 0>(a.value->ptr, a.nItem, a.N, ag, a.gradPos, agKind, out.value->ptr, out.grad->ptr); break;
+      case 1: cpuJvUnary<T, 1>(a.value->ptr, a.nItem, a.N, ag, a.gradPos, agKind, out.value->ptr, out.grad->ptr); break;
+      case 2: cpuJvUnary<T, 2>(a.value->ptr, a.nItem, a.N, ag, a.gradPos, agKind, out.value->ptr, out.grad->ptr); break;
+      case 3: cpuJvUnary<T, 3>(a.value->ptr, a.nItem, a.N, ag, a.gradPos, agKind, out.value->ptr, out.grad->ptr); break;
+      case 4: cpuJvUnary<T, 4>(a.value->ptr, a.nItem, a.N, ag, a.gradPos, agKind, out.value->ptr, out.grad->ptr); break;
+    }
+  }
+  return out;
+}
+
+// ---------------------------------------------------------------------------
+// Geometry (composed from elementwise ops; reference include/geo/geo.cuh)
+// ---------------------------------------------------------------------------
+namespace {
+template <typename T>
+JetVec<T> operator+(const JetVec<T>& a, const JetVec<T>& b) { return jvBinary(JvOp::Add, a, b); }
+template <typename T>
+JetVec<T> operator-(const JetVec<T>& a, const JetVec<T>& b) { return jvBinary(JvOp::Sub, a, b); }
+template <typename T>
+JetVec<T> operator*(const JetVec<T>& a, const JetVec<T>& b) { return jvBinary(JvOp::Mul, a, b); }
+template <typename T>
+JetVec<T> operator/(const JetVec<T>& a, const JetVec<T>& b) { return jvBinary(JvOp::Div, a, b); }
+}  // namespace
+
+template <typename T>
+std::vector<JetVec<T>> jvAngleAxisToRotation(const std::vector<JetVec<T>>& aa) {
+  MEGBA_CHECK(aa.size() == 3, "angle-axis needs 3 components");
+  const int N = aa[0].N;
+  auto S = [&](double v) { return jvScalar<T>((T)v, N); };
+  JetVec<T> t2 = aa[0] * aa[0] + aa[1] * aa[1] + aa[2] * aa[2];
+  // NOTE: vectorised expression form has no per-item branch; near-zero angles
+  // are handled by the tiny epsilon regulariser below (matches the
+  // reference's vectorised evaluation, which likewise divides by theta).
+  JetVec<T> theta = jvUnary(JvUnary::Sqrt, t2 + S(1e-30));
+  JetVec<T> c = jvUnary(JvUnary::Cos, theta);
+  JetVec<T> s = jvUnary(JvUnary::Sin, theta);
+  JetVec<T> omc = S(1.0) - c;
+  std::vector<JetVec<T>> w;
+  for (int i = 0; i < 3; ++i) w.push_back(aa[i] / theta);
+  std::vector<JetVec<T>> R;
+  R.reserve(9);
+  // R = c I + s [w]x + (1-c) w w^T  (row-major)
+  R.push_back(c + omc * (w[0] * w[0]));
+  R.push_back(omc * (w[0] * w[1]) - s * w[2]);
+  R.push_back(omc * (w[0] * w[2]) + s * w[1]);
+  R.push_back(omc * (w[1] * w[0]) + s * w[2]);
+  R.push_back(c + omc * (w[1] * w[1]));
+  R.push_back(omc * (w[1] * w[2]) - s * w[0]);
+  R.push_back(omc * (w[2] * w[0]) - s * w[1]);
+  R.push_back(omc * (w[2] * w[1]) + s * w[0]);
+  R.push_back(c + omc * (w[2] * w[2]));
+  return R;
+}
+
+template <typename T>
+std::vector<JetVec<T>> jvRotation2D(const JetVec<T>& theta) {
+  JetVec<T> c = jvUnary(JvUnary::Cos, theta);
+  JetVec<T> s = jvUnary(JvUnary::Sin, theta);
+  return {c, jvUnary(JvUnary::Neg, s), s, c};
+}
+
+template <typename T>
+std::vector<JetVec<T>> jvQuaternionToRotation(const std::vector<JetVec<T>>& q) {
+  MEGBA_CHECK(q.size() == 4, "quaternion needs 4 components");
+  const int N = q[0].N;
+  auto S = [&](double v) { return jvScalar<T>((T)v, N); };
+  const JetVec<T>&w = q[0], &x = q[1], &y = q[2], &z = q[3];
+  JetVec<T> two = S(2.0);
+  std::vector<JetVec<T>> R;
+  R.push_back(S(1.0) - two * (y * y + z * z));
+  R.push_back(two * (x * y - w * z));
+  R.push_back(two * (x * z + w * y));
+  R.push_back(two * (x * y + w * z));
+  R.push_back(S(1.0) - two * (x * x + z * z));
+  R.push_back(two * (y * z - w * x));
+  R.push_back(two * (x * z - w * y));
+  R.push_back(two * (y * z + w * x));
+  R.push_back(S(1.0) - two * (x * x + y * y));
+  return R;
+}
+
+template <typename T>
+std::vector<JetVec<T>> jvNormalizeQuaternion(const std::vector<JetVec<T>>& q) {
+  MEGBA_CHECK(q.size() == 4, "quaternion needs 4 components");
+  JetVec<T> n = jvUnary(
+      JvUnary::Sqrt, q[0] * q[0] + q[1] * q[1] + q[2] * q[2] + q[3] * q[3]);
+  std::vector<JetVec<T>> out;
+  for (int i = 0; i < 4; ++i) out.push_back(q[i] / n);
+  return out;
+}
+
+template <typename T>
+JetVec<T> jvRadialDistortion(const std::vector<JetVec<T>>& p,
+                             const std::vector<JetVec<T>>& intr) {
+  MEGBA_CHECK(p.size() >= 2 && intr.size() == 3,
+              "radial distortion needs p[>=2], intr[3]");
+  const int N = intr[0].N;
+  JetVec<T> r2 = p[0] * p[0] + p[1] * p[1];
+  JetVec<T> d = jvScalar<T>(T(1), N) + r2 * (intr[1] + intr[2] * r2);
+  return intr[0] * d;
+}
+
+// Explicit instantiations.
+#define JV_INST(T)                                                            \
+  template JetVec<T> jvFromHost<T>(const T*, const T*, int64_t, int, int,     \
+                                   bool);                                     \
+  template JetVec<T> jvScalar<T>(T, int);                                     \
+  template void jvToHost<T>(const JetVec<T>&, T*, T*);                        \
+  template JetVec<T> jvBinary<T>(JvOp, const JetVec<T>&, const JetVec<T>&);   \
+  template JetVec<T> jvUnary<T>(JvUnary, const JetVec<T>&);                   \
+  template std::vector<JetVec<T>> jvAngleAxisToRotation<T>(                   \
+      const std::vector<JetVec<T>>&);                                         \
+  template std::vector<JetVec<T>> jvRotation2D<T>(const JetVec<T>&);          \
+  template std::vector<JetVec<T>> jvQuaternionToRotation<T>(                  \
+      const std::vector<JetVec<T>>&);                                         \
+  template std::vector<JetVec<T>> jvNormalizeQuaternion<T>(                   \
+      const std::vector<JetVec<T>>&);                                         \
+  template JetVec<T> jvRadialDistortion<T>(const std::vector<JetVec<T>>&,     \
+                                           const std::vector<JetVec<T>>&);
+JV_INST(double)
+JV_INST(float)
+#undef JV_INST
+
+}  // namespace megba

@@ -1,0 +1,213 @@
+"""JetVector op layer: values/gradients vs a numpy dual-number oracle,
+JPV and scalar operand kinds, geometry ops (CPU backend here; the GPU
+backend is the same kernels compiled for gfx950, tested under -m gpu)."""
+import numpy as np
+import pytest
+
+from megba_amd import _core
+
+
+N = 4
+NI = 257
+rng = np.random.default_rng(0)
+
+
+class Ref:
+    """numpy forward-mode dual numbers, grad shape (N, nItem)."""
+
+    def __init__(self, v, g=None):
+        self.v = np.asarray(v, dtype=np.float64)
+        self.g = np.zeros((N, len(self.v))) if g is None else g
+
+    @staticmethod
+    def leaf(v, pos):
+        r = Ref(v)
+        r.g[pos] = 1.0
+        return r
+
+    def __add__(self, o):
+        return Ref(self.v + o.v, self.g + o.g)
+
+    def __sub__(self, o):
+        return Ref(self.v - o.v, self.g - o.g)
+
+    def __mul__(self, o):
+        return Ref(self.v * o.v, self.g * o.v + self.v * o.g)
+
+    def __truediv__(self, o):
+        q = self.v / o.v
+        return Ref(q, (self.g - q * o.g) / o.v)
+
+
+def dense(gpu=False):
+    v = rng.normal(size=NI) + 3.0
+    g = rng.normal(size=(N, NI))
+    return _core.JetVector(v, g, N=N, gpu=gpu), Ref(v, g.copy())
+
+
+def jpv(pos, gpu=False):
+    v = rng.normal(size=NI) + 3.0
+    jv = _core.JetVector(v, None, N=N, grad_pos=pos, gpu=gpu)
+    return jv, Ref.leaf(v, pos)
+
+
+def check(jv, ref, rtol=1e-12):
+    v, g = jv.to_numpy()
+    np.testing.assert_allclose(v, ref.v, rtol=rtol, atol=1e-13)
+    np.testing.assert_allclose(g, ref.g, rtol=rtol, atol=1e-13)
+
+
+OPS = [( _core.jv_add, lambda a, b: a + b),
+       (_core.jv_sub, lambda a, b: a - b),
+       (_core.jv_mul, lambda a, b: a * b),
+       (_core.jv_div, lambda a, b: a / b)]
+
+
+def run_binary_suite(gpu):
+    for op, ref_op in OPS:
+        a, ra = dense(gpu)
+        b, rb = dense(gpu)
+        check(op(a, b), ref_op(ra, rb))
+        # JPV operands
+        c, rc = jpv(1, gpu)
+        check(op(a, c), ref_op(ra, rc))
+        check(op(c, b), ref_op(rc, rb))
+        d, rd = jpv(3, gpu)
+        check(op(c, d), ref_op(rc, rd))
+        # scalar operands
+        s = _core.jv_scalar(1.75, N)
+        rs = Ref(np.full(NI, 1.75))
+        check(op(a, s), ref_op(ra, rs))
+        check(op(s, b), ref_op(rs, rb))
+
+
+def run_unary_suite(gpu):
+    a, ra = dense(gpu)
+    v, g = _core.jv_neg(a).to_numpy()
+    np.testing.assert_allclose(v, -ra.v)
+    np.testing.assert_allclose(g, -ra.g)
+    v, g = _core.jv_sin(a).to_numpy()
+    np.testing.assert_allclose(v, np.sin(ra.v))
+    np.testing.assert_allclose(g, np.cos(ra.v) * ra.g)
+    v, g = _core.jv_cos(a).to_numpy()
+    np.testing.assert_allclose(v, np.cos(ra.v))
+    np.testing.assert_allclose(g, -np.sin(ra.v) * ra.g)
+    v, g = _core.jv_sqrt(a).to_numpy()
+    np.testing.assert_allclose(v, np.sqrt(ra.v))
+    np.testing.assert_allclose(g, 0.5 / np.sqrt(ra.v) * ra.g)
+    b, rb = jpv(0, gpu)
+    v, g = _core.jv_abs(b).to_numpy()
+    np.testing.assert_allclose(v, np.abs(rb.v))
+    np.testing.assert_allclose(g, np.sign(rb.v) * rb.g)
+
+
+def test_binary_cpu():
+    run_binary_suite(False)
+
+
+def test_unary_cpu():
+    run_unary_suite(False)
+
+
+def test_angle_axis_geo_cpu():
+    # R(aa) from the JetVector composition vs scipy-style rotation matrices,
+    # and gradients vs finite differences.
+    ni = 40
+    aa_val = rng.normal(scale=0.6, size=(3, ni))
+    aa = [_core.JetVector(aa_val[i], None, N=3, grad_pos=i) for i in range(3)]
+    R = _core.jv_angle_axis_to_rotation(aa)
+    assert len(R) == 9
+
+    def rot(a):
+        th = np.linalg.norm(a)
+        w = a / th
+        K = np.array([[0, -w[2], w[1]], [w[2], 0, -w[0]], [-w[1], w[0], 0]])
+        return np.eye(3) + np.sin(th) * K + (1 - np.cos(th)) * K @ K
+
+    for k in range(9):
+        v, g = R[k].to_numpy()
+        i, j = divmod(k, 3)
+        for item in range(0, ni, 7):
+            a = aa_val[:, item]
+            np.testing.assert_allclose(v[item], rot(a)[i, j], rtol=1e-9)
+            for d in range(3):
+                eps = 1e-7
+                ap, am = a.copy(), a.copy()
+                ap[d] += eps
+                am[d] -= eps
+                fd = (rot(ap)[i, j] - rot(am)[i, j]) / (2 * eps)
+                np.testing.assert_allclose(g[d, item], fd, atol=1e-6)
+
+
+def test_quaternion_geo_cpu():
+    ni = 30
+    q_val = rng.normal(size=(4, ni))
+    q = [_core.JetVector(q_val[i], None, N=4, grad_pos=i) for i in range(4)]
+    qn = _core.jv_normalize_quaternion(q)
+    R = _core.jv_quaternion_to_rotation(qn)
+    for item in range(0, ni, 5):
+        qq = q_val[:, item]
+        qq = qq / np.linalg.norm(qq)
+        w, x, y, z = qq
+        Rref = np.array([
+            [1 - 2 * (y * y + z * z), 2 * (x * y - w * z), 2 * (x * z + w * y)],
+            [2 * (x * y + w * z), 1 - 2 * (x * x + z * z), 2 * (y * z - w * x)],
+            [2 * (x * z - w * y), 2 * (y * z + w * x), 1 - 2 * (x * x + y * y)]])
+        for k in range(9):
+            v, _ = R[k].to_numpy()
+            np.testing.assert_allclose(v[item], Rref.reshape(-1)[k], rtol=1e-9)
+        # orthonormality
+        Rm = np.array([R[k].to_numpy()[0][item] for k in range(9)]).reshape(3, 3)
+        np.testing.assert_allclose(Rm @ Rm.T, np.eye(3), atol=1e-9)
+
+
+def test_rotation2d_cpu():
+    th_val = rng.normal(size=20)
+    th = _core.JetVector(th_val, None, N=1, grad_pos=0)
+    R = _core.jv_rotation2d(th)
+    v0, g0 = R[0].to_numpy()
+    v1, g1 = R[1].to_numpy()
+    np.testing.assert_allclose(v0, np.cos(th_val))
+    np.testing.assert_allclose(v1, -np.sin(th_val))
+    np.testing.assert_allclose(g0[0], -np.sin(th_val))
+
+
+def test_radial_distortion_cpu():
+    ni = 25
+    p = [_core.JetVector(rng.normal(size=ni), None, N=2, grad_pos=i)
+         for i in range(2)]
+    intr_val = np.stack([np.full(ni, 500.0), np.full(ni, 1e-3),
+                         np.full(ni, 1e-6)])
+    intr = [_core.JetVector(intr_val[i], None, N=2) for i in range(3)]
+    fr = _core.jv_radial_distortion(p, intr)
+    v, g = fr.to_numpy()
+    pv = np.stack([p[0].to_numpy()[0], p[1].to_numpy()[0]])
+    r2 = (pv ** 2).sum(axis=0)
+    np.testing.assert_allclose(
+        v, intr_val[0] * (1 + intr_val[1] * r2 + intr_val[2] * r2 ** 2))
+
+
+@pytest.mark.gpu
+def test_binary_gpu():
+    run_binary_suite(True)
+
+
+@pytest.mark.gpu
+def test_unary_gpu():
+    run_unary_suite(True)
+
+
+@pytest.mark.gpu
+def test_geo_gpu_matches_cpu():
+    ni = 64
+    aa_val = rng.normal(scale=0.6, size=(3, ni))
+    for gpu in (False, True):
+        aa = [_core.JetVector(aa_val[i], None, N=3, grad_pos=i, gpu=gpu)
+              for i in range(3)]
+        R = _core.jv_angle_axis_to_rotation(aa)
+        out = [R[k].to_numpy() for k in range(9)]
+        if not gpu:
+            ref = out
+    for k in range(9):
+        np.testing.assert_allclose(out[k][0], ref[k][0], rtol=1e-12)
+        np.testing.assert_allclose(out[k][1], ref[k][1], rtol=1e-12)
