@@ -51,7 +51,7 @@ struct PyProblem {
             py::array_t<int, py::array::c_style | py::array::forcecast> camIdx,
             py::array_t<int, py::array::c_style | py::array::forcecast> ptIdx,
             py::array_t<double, py::array::c_style | py::array::forcecast> meas,
-            py::object info) {
+            py::object info, py::object camFixed, py::object ptFixed) {
     MEGBA_CHECK(cams.ndim() == 2 && cams.shape(1) == 9, "cams must be (ncam,9)");
     MEGBA_CHECK(pts.ndim() == 2 && pts.shape(1) == 3, "pts must be (npt,3)");
     MEGBA_CHECK(meas.ndim() == 2 && meas.shape(1) == 2, "meas must be (nobs,2)");
@@ -71,6 +71,15 @@ struct PyProblem {
                   "info must be (nobs,3): w00,w01,w11");
       prob.info.assign(infoArr.data(), infoArr.data() + infoArr.size());
     }
+    auto readMask = [](py::object o, int n, std::vector<uint8_t>& dst) {
+      if (o.is_none()) return;
+      auto arr = py::cast<
+          py::array_t<uint8_t, py::array::c_style | py::array::forcecast>>(o);
+      MEGBA_CHECK((int)arr.size() == n, "fixed mask size mismatch");
+      dst.assign(arr.data(), arr.data() + arr.size());
+    };
+    readMask(camFixed, prob.ncam, prob.camFixed);
+    readMask(ptFixed, prob.npt, prob.ptFixed);
   }
 
   void build(const std::string& device, const std::string& dtype, int rank,
@@ -347,9 +356,10 @@ PYBIND11_MODULE(_core, m) {
                     py::array_t<int, py::array::c_style | py::array::forcecast>,
                     py::array_t<int, py::array::c_style | py::array::forcecast>,
                     py::array_t<double, py::array::c_style | py::array::forcecast>,
-                    py::object>(),
+                    py::object, py::object, py::object>(),
            py::arg("cams"), py::arg("pts"), py::arg("cam_idx"),
-           py::arg("pt_idx"), py::arg("meas"), py::arg("info") = py::none())
+           py::arg("pt_idx"), py::arg("meas"), py::arg("info") = py::none(),
+           py::arg("cam_fixed") = py::none(), py::arg("pt_fixed") = py::none())
       .def("build", &PyProblem::build, py::arg("device") = "cpu",
            py::arg("dtype") = "float64", py::arg("rank") = 0,
            py::arg("world_size") = 1, py::arg("device_index") = 0,

@@ -43,6 +43,10 @@ class CpuEngine final : public Engine<T> {
     camHi_ = camOf_.empty() ? 0 : camOf_.back() + 1;
     camRowPtr_.assign(ix.camRowPtr.begin(), ix.camRowPtr.end());
 
+    camFixed_ = prob.camFixed;
+    ptFixed_ = prob.ptFixed;
+    if (camFixed_.empty()) camFixed_.assign(ncam_, 0);
+    if (ptFixed_.empty()) ptFixed_.assign(npt_, 0);
     cams_.resize((size_t)ncam_ * 9);
     pts_.resize((size_t)npt_ * 3);
     for (size_t i = 0; i < cams_.size(); ++i) cams_[i] = (T)prob.cams[i];
@@ -87,6 +91,7 @@ class CpuEngine final : public Engine<T> {
           for (int i = 0; i < 9; ++i) JcCur_[18 * e + 9 * row + i] = jc[row][i];
           for (int i = 0; i < 3; ++i) JpCur_[6 * e + 3 * row + i] = jp[row][i];
         }
+        zeroFixed(e);
         continue;
       }
       J cam[9], pt[3], res[2];
@@ -99,6 +104,7 @@ class CpuEngine final : public Engine<T> {
         for (int i = 0; i < 9; ++i) JcCur_[18 * e + 9 * row + i] = res[row].d[i];
         for (int i = 0; i < 3; ++i) JpCur_[6 * e + 3 * row + i] = res[row].d[9 + i];
       }
+      zeroFixed(e);
     }
     T buf = chi2;
     if (ar_) ar_(&buf, 1);
@@ -200,11 +206,23 @@ class CpuEngine final : public Engine<T> {
     HppD_ = Hpp_;
     HllD_ = Hll_;
 #pragma omp parallel for schedule(static)
-    for (int c = 0; c < ncam_; ++c)
+    for (int c = 0; c < ncam_; ++c) {
+      if (camFixed_[c]) {
+        for (int i = 0; i < 81; ++i) HppD_[(size_t)c * 81 + i] = T(0);
+        for (int i = 0; i < 9; ++i) HppD_[(size_t)c * 81 + i * 10] = T(1);
+        continue;
+      }
       for (int i = 0; i < 9; ++i) HppD_[(size_t)c * 81 + i * 10] *= f;
+    }
 #pragma omp parallel for schedule(static)
-    for (int p = 0; p < npt_; ++p)
+    for (int p = 0; p < npt_; ++p) {
+      if (ptFixed_[p]) {
+        for (int i = 0; i < 9; ++i) HllD_[(size_t)p * 9 + i] = T(0);
+        for (int i = 0; i < 3; ++i) HllD_[(size_t)p * 9 + i * 4] = T(1);
+        continue;
+      }
       for (int i = 0; i < 3; ++i) HllD_[(size_t)p * 9 + i * 4] *= f;
+    }
   }
 
   int solveLinear(const SolverOptionPCG& opt) override {
@@ -341,6 +359,17 @@ class CpuEngine final : public Engine<T> {
   }
 
  private:
+  // Fixed vertices (g2o parity, reference base_vertex.h `fixed`): their J
+  // columns are zeroed after each forward, so all their H blocks and g
+  // entries vanish; processDiag then writes an identity diagonal block so
+  // the solve is well-posed with deltaX = 0 for them.
+  inline void zeroFixed(int64_t e) {
+    if (camFixed_[camOf_[e]])
+      for (int i = 0; i < 18; ++i) JcCur_[18 * e + i] = T(0);
+    if (ptFixed_[ptOf_[e]])
+      for (int i = 0; i < 6; ++i) JpCur_[6 * e + i] = T(0);
+  }
+
   // Weighted rows of the ACCEPTED (post-acceptForward) jacobian set.
   inline void weightedRows(int64_t e, T wJc[2][9], T wJp[2][3], T wr[2]) {
     const T* Jc = &JcBak_[18 * e];
@@ -530,6 +559,7 @@ class CpuEngine final : public Engine<T> {
   int camLo_ = 0, camHi_ = 0;
   int64_t e0_ = 0, e1_ = 0, nL_ = 0, dim_ = 0;
   std::vector<int> camOf_, ptOf_;
+  std::vector<uint8_t> camFixed_, ptFixed_;
   std::vector<int64_t> camRowPtr_;
   std::vector<T> meas_, info_;
   bool hasInfo_ = false;

@@ -125,7 +125,10 @@ __global__ __launch_bounds__(256, 2) void kForward(
                          int64_t nL, const int* __restrict__ camOf,
                          const int* __restrict__ ptOf,
                          const T* __restrict__ params, int ncam,
-                         const T* __restrict__ meas, T* __restrict__ rOut,
+                         const T* __restrict__ meas,
+                         const unsigned char* __restrict__ camFixed,
+                         const unsigned char* __restrict__ ptFixed,
+                         T* __restrict__ rOut,
                          T* __restrict__ Jc, T* __restrict__ Jp,
                          double* chi2Acc) {
   using J3 = Jet<T, 3>;
@@ -145,13 +148,16 @@ __global__ __launch_bounds__(256, 2) void kForward(
     for (int k = 0; k < 3; ++k) pt[k] = J3::leaf(pp[k], 9 + k - base);
     const T m[2] = {meas[2 * e], meas[2 * e + 1]};
     balReprojectionError<T, J3>(cam, pt, m, res);
+    const bool cfix = camFixed && camFixed[camOf[e]];
+    const bool pfix = ptFixed && ptFixed[ptOf[e]];
     for (int row = 0; row < 2; ++row) {
       for (int j = 0; j < 3; ++j) {
         const int col = base + j;
         if (col < 9)
-          Jc[((int64_t)(col * 2 + row)) * nL + e] = res[row].d[j];
+          Jc[((int64_t)(col * 2 + row)) * nL + e] = cfix ? T(0) : res[row].d[j];
         else
-          Jp[((int64_t)((col - 9) * 2 + row)) * nL + e] = res[row].d[j];
+          Jp[((int64_t)((col - 9) * 2 + row)) * nL + e] =
+              pfix ? T(0) : res[row].d[j];
       }
       if (sub == 0) {
         rOut[(int64_t)row * nL + e] = res[row].v;
@@ -176,6 +182,8 @@ __global__ void kForwardAnalytical(int64_t nL, const int* __restrict__ camOf,
                                    const int* __restrict__ ptOf,
                                    const T* __restrict__ params, int ncam,
                                    const T* __restrict__ meas,
+                                   const unsigned char* __restrict__ camFixed,
+                                   const unsigned char* __restrict__ ptFixed,
                                    T* __restrict__ rOut, T* __restrict__ Jc,
                                    T* __restrict__ Jp, double* chi2Acc) {
   __shared__ double sm[kBlk];
@@ -188,13 +196,15 @@ __global__ void kForwardAnalytical(int64_t nL, const int* __restrict__ camOf,
     const T m[2] = {meas[2 * e], meas[2 * e + 1]};
     T res[2], jc[2][9], jp[2][3];
     balAnalytical<T>(cp, pp, m, res, jc, jp);
+    const bool cfix = camFixed && camFixed[camOf[e]];
+    const bool pfix = ptFixed && ptFixed[ptOf[e]];
     for (int row = 0; row < 2; ++row) {
       rOut[(int64_t)row * nL + e] = res[row];
       chi2 += (double)res[row] * (double)res[row];
       for (int col = 0; col < 9; ++col)
-        Jc[((int64_t)(col * 2 + row)) * nL + e] = jc[row][col];
+        Jc[((int64_t)(col * 2 + row)) * nL + e] = cfix ? T(0) : jc[row][col];
       for (int col = 0; col < 3; ++col)
-        Jp[((int64_t)(col * 2 + row)) * nL + e] = jp[row][col];
+        Jp[((int64_t)(col * 2 + row)) * nL + e] = pfix ? T(0) : jp[row][col];
     }
   }
   sm[threadIdx.x] = chi2;
@@ -365,12 +375,19 @@ __global__ __launch_bounds__(128) void kAssembleCam(
 // ---------------------------------------------------------------------------
 template <typename T, int D>
 __global__ void kDamp(int64_t nElem, const T* __restrict__ H, T* __restrict__ Hd,
-                      T f) {
+                      T f, const unsigned char* __restrict__ fixed) {
   for (int64_t i = blockIdx.x * (int64_t)kBlk + threadIdx.x; i < nElem;
        i += (int64_t)gridDim.x * kBlk) {
     const int within = (int)(i % (D * D));
+    const bool diag = within / D == within % D;
+    if (fixed && fixed[i / (D * D)]) {
+      // fixed vertex: identity block => deltaX = 0 for it (J columns are
+      // already zeroed by the forward kernels)
+      Hd[i] = diag ? T(1) : T(0);
+      continue;
+    }
     const T v = H[i];
-    Hd[i] = (within / D == within % D) ? v * f : v;
+    Hd[i] = diag ? v * f : v;
   }
 }
 
@@ -743,6 +760,14 @@ class GpuEngine final : public Engine<T> {
     dPtOf_ = dalloc<int>(nL_);
     up(dCamOf_, ix.camOf.data() + e0_, nL_);
     up(dPtOf_, ix.ptOf.data() + e0_, nL_);
+    if (!prob.camFixed.empty()) {
+      dCamFixed_ = dalloc<unsigned char>(ncam_);
+      up(dCamFixed_, prob.camFixed.data(), ncam_);
+    }
+    if (!prob.ptFixed.empty()) {
+      dPtFixed_ = dalloc<unsigned char>(npt_);
+      up(dPtFixed_, prob.ptFixed.data(), npt_);
+    }
     dMeas_ = dalloc<T>(nL_ * 2);
     upCast(dMeas_, ix.measSorted.data() + 2 * e0_, nL_ * 2);
     if (hasInfo_) {
@@ -855,13 +880,13 @@ class GpuEngine final : public Engine<T> {
     if (analytical_)
       hipLaunchKernelGGL(kForwardAnalytical<T>, dim3(gridFor(nL_)), dim3(kBlk),
                          0, stream_, nL_, dCamOf_, dPtOf_, dParams_, ncam_,
-                         dMeas_, dR_[cur_], dJc_[cur_], dJp_[cur_],
-                         scalarPtr());
+                         dMeas_, dCamFixed_, dPtFixed_, dR_[cur_], dJc_[cur_],
+                         dJp_[cur_], scalarPtr());
     else
       hipLaunchKernelGGL(kForward<T>, dim3(gridFor(nL_ * 4)), dim3(kBlk), 0,
                          stream_, nL_, dCamOf_, dPtOf_, dParams_, ncam_,
-                         dMeas_, dR_[cur_], dJc_[cur_], dJp_[cur_],
-                         scalarPtr());
+                         dMeas_, dCamFixed_, dPtFixed_, dR_[cur_], dJc_[cur_],
+                         dJp_[cur_], scalarPtr());
     return globalScalar();
   }
 
@@ -925,10 +950,10 @@ class GpuEngine final : public Engine<T> {
     const T f = T(1) + T(1) / (T)region;
     hipLaunchKernelGGL((kDamp<T, 9>), dim3(gridFor((int64_t)ncam_ * 81)),
                        dim3(kBlk), 0, stream_, (int64_t)ncam_ * 81, dHpp_,
-                       dHppD_, f);
+                       dHppD_, f, dCamFixed_);
     hipLaunchKernelGGL((kDamp<T, 3>), dim3(gridFor((int64_t)npt_ * 9)),
                        dim3(kBlk), 0, stream_, (int64_t)npt_ * 9, dHll_, dHllD_,
-                       f);
+                       f, dPtFixed_);
   }
 
   int solveLinear(const SolverOptionPCG& opt) override {
@@ -1234,6 +1259,7 @@ class GpuEngine final : public Engine<T> {
   int *dCamOf_{}, *dPtOf_{}, *dChCam_{}, *dChLo_{}, *dChHi_{}, *dFail_{};
   int *dPtPerm_{}, *dPtRowPtr_{}, *dCamOfPt_{}, *dPtOfPt_{};
   T *dMeas_{}, *dInfo_{};
+  unsigned char *dCamFixed_{}, *dPtFixed_{};
   T *dParams_{}, *dParamsBak_{};
   T *dR_[2]{}, *dJc_[2]{}, *dJp_[2]{};
   T *dHpp_{}, *dHll_{}, *dHpl_{}, *dHlp_{}, *dG_{}, *dGBak_{};

@@ -30,6 +30,8 @@ struct BAProblemHost {
   std::vector<int> ptIdx;    // nobs
   std::vector<double> meas;  // nobs*2 ([obs][2])
   std::vector<double> info;  // optional nobs*3 (2x2 sym weights w00,w01,w11); empty = identity
+  std::vector<uint8_t> camFixed;  // optional ncam (g2o-style fixed vertices)
+  std::vector<uint8_t> ptFixed;   // optional npt
 };
 
 struct ProblemIndex {

@@ -18,7 +18,8 @@ class BAProblem:
     info: optional (nobs, 3) upper-tri 2x2 information [w00, w01, w11]
     """
 
-    def __init__(self, cams, pts, cam_idx, pt_idx, meas, info=None):
+    def __init__(self, cams, pts, cam_idx, pt_idx, meas, info=None,
+                 cam_fixed=None, pt_fixed=None):
         from . import _core
         self.cams = np.ascontiguousarray(cams, dtype=np.float64)
         self.pts = np.ascontiguousarray(pts, dtype=np.float64)
@@ -27,8 +28,12 @@ class BAProblem:
         self.meas = np.ascontiguousarray(meas, dtype=np.float64)
         self.info = None if info is None else np.ascontiguousarray(
             info, dtype=np.float64)
+        cf = None if cam_fixed is None else np.ascontiguousarray(
+            cam_fixed, dtype=np.uint8)
+        pf = None if pt_fixed is None else np.ascontiguousarray(
+            pt_fixed, dtype=np.uint8)
         self._core = _core.Problem(self.cams, self.pts, self.cam_idx,
-                                   self.pt_idx, self.meas, self.info)
+                                   self.pt_idx, self.meas, self.info, cf, pf)
         self._built = False
 
     # -- build -------------------------------------------------------------
