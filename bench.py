@@ -61,7 +61,8 @@ def main():
     dist = None
     rccl_id = None
     allreduce = None
-    if world > 1:
+    launched_distributed = "TORCHELASTIC_RUN_ID" in os.environ or "MASTER_ADDR" in os.environ
+    if world > 1 or launched_distributed:
         import torch.distributed as tdist
         dist = tdist
         # gloo for bootstrap only; the GPU engine runs RCCL natively.

@@ -10,6 +10,7 @@
 #include <memory>
 
 #include "megba/common.hpp"
+#include "megba/custom.hpp"
 #include "megba/jv/jetvector.hpp"
 #include "megba/cpu_engine.hpp"
 #include "megba/engine.hpp"
@@ -24,6 +25,11 @@ namespace py = pybind11;
 using namespace megba;
 
 namespace {
+
+// Python-facing JetVector (fp64).
+struct PyJetVec {
+  JetVec<double> v;
+};
 
 template <typename T>
 HostAllreduce<T> wrapAllreduce(py::object fn) {
@@ -84,8 +90,8 @@ struct PyProblem {
 
   void build(const std::string& device, const std::string& dtype, int rank,
              int worldSize, int deviceIndex, const std::string& diff,
-             const std::string& schur, py::object allreduce,
-             py::object rcclId) {
+             const std::string& schur, py::object allreduce, py::object rcclId,
+             py::object customForward) {
     opt.rank = rank;
     opt.worldSize = worldSize;
     opt.deviceIndex = deviceIndex;
@@ -93,10 +99,14 @@ struct PyProblem {
     opt.diff = diff == "analytical" ? DiffMode::ANALYTICAL : DiffMode::AUTO;
     opt.schur = schur == "implicit" ? SchurMode::IMPLICIT : SchurMode::EXPLICIT;
     isDouble = dtype != "float32";
+    MEGBA_CHECK(customForward.is_none() || isDouble,
+                "custom_forward requires float64");
+    CustomForward<double> cf = wrapCustomForward(customForward);
     ix = buildIndex(prob, worldSize);
     if (opt.device == Device::CPU) {
       if (isDouble)
-        engD = makeCpuEngine<double>(prob, ix, opt, wrapAllreduce<double>(allreduce));
+        engD = makeCpuEngine<double>(prob, ix, opt,
+                                     wrapAllreduce<double>(allreduce), cf);
       else
         engF = makeCpuEngine<float>(prob, ix, opt, wrapAllreduce<float>(allreduce));
     } else {
@@ -104,7 +114,7 @@ struct PyProblem {
       std::string id;
       if (!rcclId.is_none()) id = py::cast<std::string>(rcclId);
       if (isDouble)
-        engD = makeGpuEngine<double>(prob, ix, opt, id);
+        engD = makeGpuEngine<double>(prob, ix, opt, id, cf);
       else
         engF = makeGpuEngine<float>(prob, ix, opt, id);
 #else
@@ -112,6 +122,8 @@ struct PyProblem {
 #endif
     }
   }
+
+  static CustomForward<double> wrapCustomForward(py::object fn);
 
   template <typename F>
   auto withEngine(F&& f) {
@@ -283,13 +295,6 @@ struct PyProblem {
 
 namespace {
 
-// Python-facing JetVector (fp64).  The C++ layer is templated on
-// double/float; Python exposes double, which is what tests and the
-// user-defined-edge path use.
-struct PyJetVec {
-  JetVec<double> v;
-};
-
 PyJetVec mkJv(py::array_t<double, py::array::c_style | py::array::forcecast> value,
               py::object grad, int N, int grad_pos, bool gpu) {
   const int64_t n = value.size();
@@ -320,6 +325,25 @@ py::list jvWrap(const std::vector<JetVec<double>>& vs) {
   py::list out;
   for (const auto& v : vs) out.append(PyJetVec{v});
   return out;
+}
+
+CustomForward<double> PyProblem::wrapCustomForward(py::object fn) {
+  if (fn.is_none()) return nullptr;
+  auto holder = std::make_shared<py::object>(std::move(fn));
+  return [holder](const std::vector<JetVec<double>>& camL,
+                  const std::vector<JetVec<double>>& ptL,
+                  const std::vector<JetVec<double>>& meas,
+                  std::vector<JetVec<double>>& res) {
+    py::gil_scoped_acquire gil;
+    py::list c, p, m;
+    for (const auto& v : camL) c.append(PyJetVec{v});
+    for (const auto& v : ptL) p.append(PyJetVec{v});
+    for (const auto& v : meas) m.append(PyJetVec{v});
+    py::object out = (*holder)(c, p, m);
+    py::sequence seq = py::cast<py::sequence>(out);
+    res.clear();
+    for (auto h : seq) res.push_back(py::cast<PyJetVec&>(h).v);
+  };
 }
 
 }  // namespace
@@ -364,7 +388,8 @@ PYBIND11_MODULE(_core, m) {
            py::arg("dtype") = "float64", py::arg("rank") = 0,
            py::arg("world_size") = 1, py::arg("device_index") = 0,
            py::arg("diff") = "auto", py::arg("schur") = "explicit",
-           py::arg("allreduce") = py::none(), py::arg("rccl_id") = py::none())
+           py::arg("allreduce") = py::none(), py::arg("rccl_id") = py::none(),
+           py::arg("custom_forward") = py::none())
       .def("solve", &PyProblem::solve, py::arg("max_iter") = 20,
            py::arg("tau") = 1e4, py::arg("epsilon1") = 1.0,
            py::arg("epsilon2") = 1e-10, py::arg("solver_max_iter") = 100,

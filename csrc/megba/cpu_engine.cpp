@@ -14,8 +14,10 @@ template <typename T>
 class CpuEngine final : public Engine<T> {
  public:
   CpuEngine(const BAProblemHost& prob, const ProblemIndex& ix,
-            const ProblemOption& opt, HostAllreduce<T> allreduce)
+            const ProblemOption& opt, HostAllreduce<T> allreduce,
+            CustomForward<T> customForward)
       : ar_(std::move(allreduce)),
+        customFwd_(std::move(customForward)),
         rank_(opt.rank),
         world_(opt.worldSize),
         ncam_(ix.ncam),
@@ -76,6 +78,7 @@ class CpuEngine final : public Engine<T> {
   }
 
   double forward() override {
+    if (customFwd_) return forwardCustom();
     using J = Jet<T, 12>;
     T chi2 = T(0);
 #pragma omp parallel for schedule(static) reduction(+ : chi2)
@@ -359,6 +362,53 @@ class CpuEngine final : public Engine<T> {
   }
 
  private:
+  double forwardCustom() {
+    // Gather the 12 parameter leaves + 2 measurement rows as JetVectors,
+    // run the user expression, repack the residual dual parts.
+    std::vector<T> leaf((size_t)12 * nL_), measRow((size_t)2 * nL_);
+#pragma omp parallel for schedule(static)
+    for (int64_t e = 0; e < nL_; ++e) {
+      const T* cp = &cams_[(size_t)camOf_[e] * 9];
+      const T* pp = &pts_[(size_t)ptOf_[e] * 3];
+      for (int k = 0; k < 9; ++k) leaf[(size_t)k * nL_ + e] = cp[k];
+      for (int k = 0; k < 3; ++k) leaf[(size_t)(9 + k) * nL_ + e] = pp[k];
+      measRow[e] = meas_[2 * e];
+      measRow[nL_ + e] = meas_[2 * e + 1];
+    }
+    std::vector<JetVec<T>> camL, ptL, ms, res;
+    for (int k = 0; k < 9; ++k)
+      camL.push_back(jvView<T>(&leaf[(size_t)k * nL_], nL_, 12, k, false));
+    for (int k = 0; k < 3; ++k)
+      ptL.push_back(jvView<T>(&leaf[(size_t)(9 + k) * nL_], nL_, 12, 9 + k, false));
+    for (int r = 0; r < 2; ++r)
+      ms.push_back(jvView<T>(&measRow[(size_t)r * nL_], nL_, 12, -1, false));
+    customFwd_(camL, ptL, ms, res);
+    MEGBA_CHECK(res.size() == 2, "custom forward must return 2 residuals");
+    for (int r = 0; r < 2; ++r) {
+      MEGBA_CHECK(res[r].kind() == JvKind::DENSE && res[r].nItem == nL_ &&
+                      res[r].N == 12 && !res[r].onGpu,
+                  "custom residual must be a dense CPU JetVector (N=12)");
+    }
+    T chi2 = T(0);
+#pragma omp parallel for schedule(static) reduction(+ : chi2)
+    for (int64_t e = 0; e < nL_; ++e) {
+      for (int row = 0; row < 2; ++row) {
+        const T v = res[row].value->ptr[e];
+        rCur_[2 * e + row] = v;
+        chi2 += v * v;
+        const T* g = res[row].grad->ptr;
+        for (int k = 0; k < 9; ++k)
+          JcCur_[18 * e + 9 * row + k] = g[(size_t)k * nL_ + e];
+        for (int k = 0; k < 3; ++k)
+          JpCur_[6 * e + 3 * row + k] = g[(size_t)(9 + k) * nL_ + e];
+      }
+      zeroFixed(e);
+    }
+    T buf = chi2;
+    if (ar_) ar_(&buf, 1);
+    return (double)buf;
+  }
+
   // Fixed vertices (g2o parity, reference base_vertex.h `fixed`): their J
   // columns are zeroed after each forward, so all their H blocks and g
   // entries vanish; processDiag then writes an identity diagonal block so
@@ -553,6 +603,7 @@ class CpuEngine final : public Engine<T> {
   }
 
   HostAllreduce<T> ar_;
+  CustomForward<T> customFwd_;
   int rank_, world_, ncam_, npt_;
   bool analytical_ = false;
   bool implicit_ = false;
@@ -573,16 +624,18 @@ template <typename T>
 std::unique_ptr<Engine<T>> makeCpuEngine(const BAProblemHost& prob,
                                          const ProblemIndex& ix,
                                          const ProblemOption& opt,
-                                         HostAllreduce<T> allreduce) {
-  return std::make_unique<CpuEngine<T>>(prob, ix, opt, std::move(allreduce));
+                                         HostAllreduce<T> allreduce,
+                                         CustomForward<T> customForward) {
+  return std::make_unique<CpuEngine<T>>(prob, ix, opt, std::move(allreduce),
+                                        std::move(customForward));
 }
 
 template std::unique_ptr<Engine<double>> makeCpuEngine<double>(
     const BAProblemHost&, const ProblemIndex&, const ProblemOption&,
-    HostAllreduce<double>);
+    HostAllreduce<double>, CustomForward<double>);
 template std::unique_ptr<Engine<float>> makeCpuEngine<float>(
     const BAProblemHost&, const ProblemIndex&, const ProblemOption&,
-    HostAllreduce<float>);
+    HostAllreduce<float>, CustomForward<float>);
 
 // Instantiate the LM driver here as well.
 template LMReport runLM<double>(Engine<double>&, const AlgoOptionLM&,

@@ -10,6 +10,7 @@
 #include <memory>
 
 #include "common.hpp"
+#include "custom.hpp"
 #include "engine.hpp"
 #include "problem.hpp"
 
@@ -23,6 +24,7 @@ template <typename T>
 std::unique_ptr<Engine<T>> makeCpuEngine(const BAProblemHost& prob,
                                          const ProblemIndex& ix,
                                          const ProblemOption& opt,
-                                         HostAllreduce<T> allreduce);
+                                         HostAllreduce<T> allreduce,
+                                         CustomForward<T> customForward = nullptr);
 
 }  // namespace megba

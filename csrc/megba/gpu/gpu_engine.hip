@@ -32,6 +32,7 @@
 #include "../analytical.hpp"
 #include "../bal_functor.hpp"
 #include "../smallmat.hpp"
+#include "../jv/jetvector.hpp"
 #include "gpu_engine.hpp"
 
 namespace megba {
@@ -214,6 +215,45 @@ __global__ void kForwardAnalytical(int64_t nL, const int* __restrict__ camOf,
     __syncthreads();
   }
   if (threadIdx.x == 0) atomicAdd(chi2Acc, sm[0]);
+}
+
+// Custom-edge support: gather parameter leaves, repack residual JetVectors.
+template <typename T>
+__global__ void kGatherLeaves(int64_t nL, const int* __restrict__ camOf,
+                              const int* __restrict__ ptOf,
+                              const T* __restrict__ params, int ncam,
+                              T* __restrict__ leaf /*[12][nL]*/) {
+  const T* ptsBase = params + (int64_t)ncam * 9;
+  for (int64_t e = blockIdx.x * (int64_t)kBlk + threadIdx.x; e < nL;
+       e += (int64_t)gridDim.x * kBlk) {
+    const T* cp = params + (int64_t)camOf[e] * 9;
+    const T* pp = ptsBase + (int64_t)ptOf[e] * 3;
+    for (int k = 0; k < 9; ++k) leaf[(int64_t)k * nL + e] = cp[k];
+    for (int k = 0; k < 3; ++k) leaf[(int64_t)(9 + k) * nL + e] = pp[k];
+  }
+}
+
+template <typename T>
+__global__ void kRepackRes(int64_t nL, int row, const T* __restrict__ rv,
+                           const T* __restrict__ rg /*[12][nL]*/,
+                           const int* __restrict__ camOf,
+                           const int* __restrict__ ptOf,
+                           const unsigned char* __restrict__ camFixed,
+                           const unsigned char* __restrict__ ptFixed,
+                           T* __restrict__ rOut, T* __restrict__ Jc,
+                           T* __restrict__ Jp) {
+  for (int64_t e = blockIdx.x * (int64_t)kBlk + threadIdx.x; e < nL;
+       e += (int64_t)gridDim.x * kBlk) {
+    rOut[(int64_t)row * nL + e] = rv[e];
+    const bool cfix = camFixed && camFixed[camOf[e]];
+    const bool pfix = ptFixed && ptFixed[ptOf[e]];
+    for (int k = 0; k < 9; ++k)
+      Jc[((int64_t)(k * 2 + row)) * nL + e] =
+          cfix ? T(0) : rg[(int64_t)k * nL + e];
+    for (int k = 0; k < 3; ++k)
+      Jp[((int64_t)(k * 2 + row)) * nL + e] =
+          pfix ? T(0) : rg[(int64_t)(9 + k) * nL + e];
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -732,8 +772,10 @@ template <typename T>
 class GpuEngine final : public Engine<T> {
  public:
   GpuEngine(const BAProblemHost& prob, const ProblemIndex& ix,
-            const ProblemOption& opt, const std::string& rcclId)
-      : rank_(opt.rank), world_(opt.worldSize), ncam_(ix.ncam), npt_(ix.npt),
+            const ProblemOption& opt, const std::string& rcclId,
+            CustomForward<T> customForward)
+      : customFwd_(std::move(customForward)),
+        rank_(opt.rank), world_(opt.worldSize), ncam_(ix.ncam), npt_(ix.npt),
         analytical_(opt.diff == DiffMode::ANALYTICAL),
         implicit_(opt.schur == SchurMode::IMPLICIT) {
     HIP_CHECK(hipSetDevice(opt.deviceIndex));
@@ -770,6 +812,17 @@ class GpuEngine final : public Engine<T> {
     }
     dMeas_ = dalloc<T>(nL_ * 2);
     upCast(dMeas_, ix.measSorted.data() + 2 * e0_, nL_ * 2);
+    if (customFwd_) {
+      // row-split measurement copy + leaf slab for the user forward()
+      dLeaf_ = dalloc<T>(nL_ * 12);
+      dMeasSplit_ = dalloc<T>(nL_ * 2);
+      std::vector<T> ms(nL_ * 2);
+      for (int64_t e = 0; e < nL_; ++e) {
+        ms[e] = (T)ix.measSorted[2 * (e0_ + e)];
+        ms[nL_ + e] = (T)ix.measSorted[2 * (e0_ + e) + 1];
+      }
+      up(dMeasSplit_, ms.data(), nL_ * 2);
+    }
     if (hasInfo_) {
       dInfo_ = dalloc<T>(nL_ * 3);
       upCast(dInfo_, ix.infoSorted.data() + 3 * e0_, nL_ * 3);
@@ -876,6 +929,7 @@ class GpuEngine final : public Engine<T> {
   }
 
   double forward() override {
+    if (customFwd_) return forwardCustom();
     zeroScalar();
     if (analytical_)
       hipLaunchKernelGGL(kForwardAnalytical<T>, dim3(gridFor(nL_)), dim3(kBlk),
@@ -891,6 +945,34 @@ class GpuEngine final : public Engine<T> {
   }
 
   void acceptForward() override { cur_ ^= 1; }  // accepted set = dX_[cur_^1]
+
+  double forwardCustom() {
+    hipLaunchKernelGGL(kGatherLeaves<T>, dim3(gridFor(nL_)), dim3(kBlk), 0,
+                       stream_, nL_, dCamOf_, dPtOf_, dParams_, ncam_, dLeaf_);
+    sync();  // JetVector ops run on the null stream
+    std::vector<JetVec<T>> camL, ptL, ms, res;
+    for (int k = 0; k < 9; ++k)
+      camL.push_back(jvView<T>(dLeaf_ + (int64_t)k * nL_, nL_, 12, k, true));
+    for (int k = 0; k < 3; ++k)
+      ptL.push_back(
+          jvView<T>(dLeaf_ + (int64_t)(9 + k) * nL_, nL_, 12, 9 + k, true));
+    for (int r = 0; r < 2; ++r)
+      ms.push_back(jvView<T>(dMeasSplit_ + (int64_t)r * nL_, nL_, 12, -1, true));
+    customFwd_(camL, ptL, ms, res);
+    MEGBA_CHECK(res.size() == 2, "custom forward must return 2 residuals");
+    HIP_CHECK(hipDeviceSynchronize());
+    for (int r = 0; r < 2; ++r) {
+      MEGBA_CHECK(res[r].kind() == JvKind::DENSE && res[r].nItem == nL_ &&
+                      res[r].N == 12 && res[r].onGpu,
+                  "custom residual must be a dense GPU JetVector (N=12)");
+      hipLaunchKernelGGL(kRepackRes<T>, dim3(gridFor(nL_)), dim3(kBlk), 0,
+                         stream_, nL_, r, res[r].value->ptr, res[r].grad->ptr,
+                         dCamOf_, dPtOf_, dCamFixed_, dPtFixed_, dR_[cur_],
+                         dJc_[cur_], dJp_[cur_]);
+    }
+    reduceDetAsync(dR_[cur_], dR_[cur_], nL_ * 2, ROp::SumSq, scalarPtr());
+    return globalScalar();
+  }
 
   void buildLinearSystem() override {
     const int bak = cur_ ^ 1;
@@ -1258,7 +1340,8 @@ class GpuEngine final : public Engine<T> {
   int nChunks_ = 0;
   int *dCamOf_{}, *dPtOf_{}, *dChCam_{}, *dChLo_{}, *dChHi_{}, *dFail_{};
   int *dPtPerm_{}, *dPtRowPtr_{}, *dCamOfPt_{}, *dPtOfPt_{};
-  T *dMeas_{}, *dInfo_{};
+  T *dMeas_{}, *dInfo_{}, *dLeaf_{}, *dMeasSplit_{};
+  CustomForward<T> customFwd_;
   unsigned char *dCamFixed_{}, *dPtFixed_{};
   T *dParams_{}, *dParamsBak_{};
   T *dR_[2]{}, *dJc_[2]{}, *dJp_[2]{};
@@ -1274,16 +1357,18 @@ template <typename T>
 std::unique_ptr<Engine<T>> makeGpuEngine(const BAProblemHost& prob,
                                          const ProblemIndex& ix,
                                          const ProblemOption& opt,
-                                         const std::string& rcclId) {
-  return std::make_unique<GpuEngine<T>>(prob, ix, opt, rcclId);
+                                         const std::string& rcclId,
+                                         CustomForward<T> customForward) {
+  return std::make_unique<GpuEngine<T>>(prob, ix, opt, rcclId,
+                                        std::move(customForward));
 }
 
 template std::unique_ptr<Engine<double>> makeGpuEngine<double>(
     const BAProblemHost&, const ProblemIndex&, const ProblemOption&,
-    const std::string&);
+    const std::string&, CustomForward<double>);
 template std::unique_ptr<Engine<float>> makeGpuEngine<float>(
     const BAProblemHost&, const ProblemIndex&, const ProblemOption&,
-    const std::string&);
+    const std::string&, CustomForward<float>);
 
 std::string rcclUniqueIdString() {
   ncclUniqueId id;

@@ -6,6 +6,7 @@
 #include <string>
 
 #include "../common.hpp"
+#include "../custom.hpp"
 #include "../engine.hpp"
 #include "../problem.hpp"
 
@@ -19,7 +20,8 @@ template <typename T>
 std::unique_ptr<Engine<T>> makeGpuEngine(const BAProblemHost& prob,
                                          const ProblemIndex& ix,
                                          const ProblemOption& opt,
-                                         const std::string& rcclId);
+                                         const std::string& rcclId,
+                                         CustomForward<T> customForward = nullptr);
 
 std::string rcclUniqueIdString();
 int hipDeviceCountSafe();

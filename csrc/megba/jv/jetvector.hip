@@ -6,6 +6,8 @@
 
 #include <cmath>
 #include <cstring>
+#include <mutex>
+#include <unordered_map>
 
 #include "../jet.hpp"  // MEGBA_HD
 #include "jetvector.hpp"
@@ -19,16 +21,71 @@ namespace megba {
                 std::string("HIP error: ") + hipGetErrorString(_e));         \
   } while (0)
 
+// Caching device allocator for JetVector temporaries -- the MemoryPool
+// equivalent (reference C4, src/resource/memory_pool.cu): residual
+// expressions allocate/free same-sized blocks every forward, so freed blocks
+// go to a size-keyed free list instead of hipFree.  No LIFO discipline is
+// needed (the reference threw on out-of-order frees, memory_pool.cu:169).
+namespace {
+std::mutex gPoolMu;
+std::unordered_map<size_t, std::vector<void*>> gPool;
+
+void* poolAlloc(size_t bytes) {
+  {
+    std::lock_guard<std::mutex> lk(gPoolMu);
+    auto it = gPool.find(bytes);
+    if (it != gPool.end() && !it->second.empty()) {
+      void* p = it->second.back();
+      it->second.pop_back();
+      return p;
+    }
+  }
+  void* p = nullptr;
+  JV_HIP_CHECK(hipMalloc(&p, bytes));
+  return p;
+}
+
+void poolFree(void* p, size_t bytes) {
+  std::lock_guard<std::mutex> lk(gPoolMu);
+  gPool[bytes].push_back(p);
+}
+}  // namespace
+
+// (DeviceBuf dtor defined after the pool so it can return blocks to it.)
+void jvPoolTrim() {
+  std::lock_guard<std::mutex> lk(gPoolMu);
+  for (auto& kv : gPool)
+    for (void* p : kv.second) (void)hipFree(p);
+  gPool.clear();
+}
+
 template <typename T>
 DeviceBuf<T>::~DeviceBuf() {
-  if (!ptr) return;
+  if (!ptr || !owned) return;
   if (onGpu)
-    (void)hipFree(ptr);
+    poolFree(ptr, (n > 0 ? n : 1) * sizeof(T));
   else
     free(ptr);
 }
 template struct DeviceBuf<double>;
 template struct DeviceBuf<float>;
+
+template <typename T>
+JetVec<T> jvView(T* ptr, int64_t nItem, int N, int gradPos, bool onGpu) {
+  JetVec<T> v;
+  v.nItem = nItem;
+  v.N = N;
+  v.gradPos = gradPos;
+  v.onGpu = onGpu;
+  v.value = std::make_shared<DeviceBuf<T>>();
+  v.value->ptr = ptr;
+  v.value->n = nItem;
+  v.value->onGpu = onGpu;
+  v.value->owned = false;
+  return v;
+}
+template JetVec<double> jvView<double>(double*, int64_t, int, int, bool);
+template JetVec<float> jvView<float>(float*, int64_t, int, int, bool);
 
 namespace {
 
@@ -38,9 +95,7 @@ std::shared_ptr<DeviceBuf<T>> makeBuf(int64_t n, bool onGpu) {
   b->n = n;
   b->onGpu = onGpu;
   if (onGpu) {
-    void* p = nullptr;
-    JV_HIP_CHECK(hipMalloc(&p, (n > 0 ? n : 1) * sizeof(T)));
-    b->ptr = (T*)p;
+    b->ptr = (T*)poolAlloc((n > 0 ? n : 1) * sizeof(T));
   } else {
     b->ptr = (T*)malloc((n > 0 ? n : 1) * sizeof(T));
     MEGBA_CHECK(b->ptr, "malloc failed");

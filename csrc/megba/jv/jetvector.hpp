@@ -27,10 +27,11 @@ namespace megba {
 enum class JvKind { DENSE, JPV, SCALAR };
 
 template <typename T>
-struct DeviceBuf {  // owns device or host memory
+struct DeviceBuf {  // owns (or views) device or host memory
   T* ptr = nullptr;
   int64_t n = 0;
   bool onGpu = false;
+  bool owned = true;
   ~DeviceBuf();
 };
 
@@ -53,6 +54,14 @@ struct JetVec {
 
 enum class JvOp { Add, Sub, Mul, Div };
 enum class JvUnary { Neg, Abs, Sin, Cos, Sqrt };
+
+// Non-owning view over engine-managed memory (used to hand parameter leaves
+// to user forward() callbacks without copies).
+template <typename T>
+JetVec<T> jvView(T* ptr, int64_t nItem, int N, int gradPos, bool onGpu);
+
+// Return all cached device blocks to the runtime.
+void jvPoolTrim();
 
 // Factories -----------------------------------------------------------------
 template <typename T>

@@ -118,3 +118,23 @@ def test_larger_problem_gpu():
                                    atol=1e-8 * scale, err_msg=key)
     scale = np.abs(d1["deltaX"]).max()
     np.testing.assert_allclose(d2["deltaX"], d1["deltaX"], atol=1e-4 * scale)
+
+
+@pytest.mark.gpu
+def test_rccl_world1_via_torchrun(tmp_path):
+    """Exercises the full distributed bootstrap (gloo rendezvous + RCCL
+    ncclCommInitRank + allreduces) with world_size=1 on one GPU -- the same
+    code path the driver's multi-GPU scaling run takes."""
+    import os
+    import subprocess
+    import sys
+    env = dict(os.environ)
+    env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "1", "--master-addr", "127.0.0.1",
+         "--master-port", "29517", "bench.py", "--model", "tiny",
+         "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, timeout=600, env=env)
+    assert r.returncode == 0, r.stderr[-3000:]
+    assert '"lm_iterations_per_s"' in r.stdout
