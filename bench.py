@@ -61,7 +61,10 @@ def main():
     dist = None
     rccl_id = None
     allreduce = None
-    launched_distributed = "TORCHELASTIC_RUN_ID" in os.environ or "MASTER_ADDR" in os.environ
+    launched_distributed = ("TORCHELASTIC_RUN_ID" in os.environ or
+                            ("MASTER_ADDR" in os.environ and
+                             "MASTER_PORT" in os.environ and
+                             "RANK" in os.environ))
     if world > 1 or launched_distributed:
         import torch.distributed as tdist
         dist = tdist
