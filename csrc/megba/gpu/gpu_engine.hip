@@ -266,13 +266,21 @@ __global__ void kRepackRes(int64_t nL, int row, const T* __restrict__ rv,
 // per edge -- the all-atomic version measured 40.8 ms on Venice-5M
 // (profiles/r01_venice_baseline.md), dominated by ~2800-way per-camera
 // contention.
-template <typename T, bool HASINFO>
+// EXPL=true (explicit Schur): writes the Hpl block grad-major (for E*w) and
+// a point-sorted edge-major slab row [40] = {Hlp block transposed (27),
+// Hll upper-sym contribution (6), g_p contribution (3), pad} -- 320 B
+// contiguous per edge, so the scatter costs ~1 line-fetch per 64 B instead
+// of the 8x amplification of a per-element transpose gather.  kFinalizePt
+// then streams the slab in point order with NO atomics.  EXPL=false
+// (implicit): no Hpl/slab; Hll/g_p accumulate via atomics as before.
+template <typename T, bool HASINFO, bool EXPL>
 __global__ void kAssembleEdge(int64_t nL, const int* __restrict__ camOf,
                           const int* __restrict__ ptOf, const T* __restrict__ r,
                           const T* __restrict__ Jc, const T* __restrict__ Jp,
                           const T* __restrict__ info,
                           T* __restrict__ Hll, T* __restrict__ Hpl,
-                          T* __restrict__ g, int ncam) {
+                          T* __restrict__ g, int ncam,
+                          const int* __restrict__ ptPos, T* __restrict__ slab) {
   T* gp = g + (int64_t)ncam * 9;
   for (int64_t e = blockIdx.x * (int64_t)kBlk + threadIdx.x; e < nL;
        e += (int64_t)gridDim.x * kBlk) {
@@ -312,23 +320,70 @@ __global__ void kAssembleEdge(int64_t nL, const int* __restrict__ camOf,
       wr[0] = rr[0];
       wr[1] = rr[1];
     }
-    // Hpl: one conflict-free 9x3 block per observation, grad-major layout
-    // Hpl[(a*3+b)*nL + e] so SpMV reads coalesce across lanes.  Null in
-    // implicit-Schur mode (diagonal blocks only, reference C15).
-    if (Hpl != nullptr)
+    if (EXPL) {
+      // Hpl grad-major (cam-sorted, for E*w)
+      T blk[9][3];
       for (int a = 0; a < 9; ++a)
-        for (int b = 0; b < 3; ++b)
-          Hpl[((int64_t)(a * 3 + b)) * nL + e] =
-              jc[0][a] * wjp[0][b] + jc[1][a] * wjp[1][b];
-    // Hll + g_p: point degree is low (few obs/point) -> atomics are cheap.
-    const int pt = ptOf[e];
-    T* hll = Hll + (int64_t)pt * 9;
-    for (int a = 0; a < 3; ++a)
+        for (int b = 0; b < 3; ++b) {
+          blk[a][b] = jc[0][a] * wjp[0][b] + jc[1][a] * wjp[1][b];
+          Hpl[((int64_t)(a * 3 + b)) * nL + e] = blk[a][b];
+        }
+      // slab row at this edge's point-sorted position
+      T* row = slab + (int64_t)ptPos[e] * 40;
       for (int b = 0; b < 3; ++b)
-        atomicAdd(&hll[a * 3 + b], jp[0][a] * wjp[0][b] + jp[1][a] * wjp[1][b]);
-    T* gpt = gp + (int64_t)pt * 3;
-    for (int a = 0; a < 3; ++a)
-      atomicAdd(&gpt[a], -(jp[0][a] * wr[0] + jp[1][a] * wr[1]));
+        for (int a = 0; a < 9; ++a) row[b * 9 + a] = blk[a][b];
+      row[27] = jp[0][0] * wjp[0][0] + jp[1][0] * wjp[1][0];  // hll 00
+      row[28] = jp[0][0] * wjp[0][1] + jp[1][0] * wjp[1][1];  // 01
+      row[29] = jp[0][0] * wjp[0][2] + jp[1][0] * wjp[1][2];  // 02
+      row[30] = jp[0][1] * wjp[0][1] + jp[1][1] * wjp[1][1];  // 11
+      row[31] = jp[0][1] * wjp[0][2] + jp[1][1] * wjp[1][2];  // 12
+      row[32] = jp[0][2] * wjp[0][2] + jp[1][2] * wjp[1][2];  // 22
+      for (int a = 0; a < 3; ++a)
+        row[33 + a] = -(jp[0][a] * wr[0] + jp[1][a] * wr[1]);  // g_p
+    } else {
+      const int pt = ptOf[e];
+      T* hll = Hll + (int64_t)pt * 9;
+      for (int a = 0; a < 3; ++a)
+        for (int b = 0; b < 3; ++b)
+          atomicAdd(&hll[a * 3 + b],
+                    jp[0][a] * wjp[0][b] + jp[1][a] * wjp[1][b]);
+      T* gpt = gp + (int64_t)pt * 3;
+      for (int a = 0; a < 3; ++a)
+        atomicAdd(&gpt[a], -(jp[0][a] * wr[0] + jp[1][a] * wr[1]));
+    }
+  }
+}
+
+// Explicit-mode finalize: one thread per point streams its contiguous slab
+// rows (point-sorted), writing Hlp grad-major, the Hll block and g_p
+// directly -- no atomics, no transpose gather.
+template <typename T>
+__global__ void kFinalizePt(int npt, const int* __restrict__ ptRowPtr,
+                            const T* __restrict__ slab, int64_t nL,
+                            T* __restrict__ Hlp, T* __restrict__ Hll,
+                            T* __restrict__ gp) {
+  for (int64_t p = blockIdx.x * (int64_t)kBlk + threadIdx.x; p < npt;
+       p += (int64_t)gridDim.x * kBlk) {
+    const int lo = ptRowPtr[p], hi = ptRowPtr[p + 1];
+    T hll[6] = {0, 0, 0, 0, 0, 0};
+    T g3[3] = {0, 0, 0};
+    for (int j = lo; j < hi; ++j) {
+      const T* row = slab + (int64_t)j * 40;
+      for (int k = 0; k < 27; ++k) Hlp[(int64_t)k * nL + j] = row[k];
+      for (int k = 0; k < 6; ++k) hll[k] += row[27 + k];
+      for (int k = 0; k < 3; ++k) g3[k] += row[33 + k];
+    }
+    T* H = Hll + (int64_t)p * 9;
+    H[0] = hll[0];
+    H[1] = hll[1];
+    H[2] = hll[2];
+    H[3] = hll[1];
+    H[4] = hll[3];
+    H[5] = hll[4];
+    H[6] = hll[2];
+    H[7] = hll[4];
+    H[8] = hll[5];
+    for (int k = 0; k < 3; ++k) gp[3 * p + k] = g3[k];
   }
 }
 
@@ -909,6 +964,10 @@ class GpuEngine final : public Engine<T> {
         camOfPt[pos] = ix.camOf[e0_ + e];
         ptOfPt[pos] = ix.ptOf[e0_ + e];
       }
+      std::vector<int> ptPos(nL_);
+      for (int64_t j = 0; j < nL_; ++j) ptPos[perm[j]] = (int)j;
+      dPtPos_ = dalloc<int>(nL_);
+      up(dPtPos_, ptPos.data(), nL_);
       dPtPerm_ = dalloc<int>(nL_);
       dPtRowPtr_ = dalloc<int>(npt_ + 1);
       dCamOfPt_ = dalloc<int>(nL_);
@@ -918,7 +977,10 @@ class GpuEngine final : public Engine<T> {
       up(dCamOfPt_, camOfPt.data(), nL_);
       up(dPtOfPt_, ptOfPt.data(), nL_);
     }
-    if (!implicit_) dHlp_ = dalloc<T>(nL_ * 27);
+    if (!implicit_) {
+      dHlp_ = dalloc<T>(nL_ * 27);
+      dSlab_ = dalloc<T>(nL_ * 40);
+    }
     sync();
   }
 
@@ -980,27 +1042,40 @@ class GpuEngine final : public Engine<T> {
     HIP_CHECK(hipMemsetAsync(dHll_, 0, (int64_t)npt_ * 9 * sizeof(T), stream_));
     HIP_CHECK(hipMemsetAsync(dG_, 0, dim_ * sizeof(T), stream_));
     if (hasInfo_) {
-      hipLaunchKernelGGL((kAssembleEdge<T, true>), dim3(gridFor(nL_)),
-                         dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
-                         dR_[bak], dJc_[bak], dJp_[bak], dInfo_, dHll_, dHpl_,
-                         dG_, ncam_);
+      if (implicit_)
+        hipLaunchKernelGGL((kAssembleEdge<T, true, false>), dim3(gridFor(nL_)),
+                           dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
+                           dR_[bak], dJc_[bak], dJp_[bak], dInfo_, dHll_,
+                           dHpl_, dG_, ncam_, dPtPos_, dSlab_);
+      else
+        hipLaunchKernelGGL((kAssembleEdge<T, true, true>), dim3(gridFor(nL_)),
+                           dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
+                           dR_[bak], dJc_[bak], dJp_[bak], dInfo_, dHll_,
+                           dHpl_, dG_, ncam_, dPtPos_, dSlab_);
       if (nChunks_ > 0)
         hipLaunchKernelGGL((kAssembleCam<T, true>), dim3(nChunks_), dim3(128),
                            0, stream_, nChunks_, dChCam_, dChLo_, dChHi_, nL_,
                            dR_[bak], dJc_[bak], dInfo_, dHpp_, dG_);
     } else {
-      hipLaunchKernelGGL((kAssembleEdge<T, false>), dim3(gridFor(nL_)),
-                         dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
-                         dR_[bak], dJc_[bak], dJp_[bak], nullptr, dHll_, dHpl_,
-                         dG_, ncam_);
+      if (implicit_)
+        hipLaunchKernelGGL((kAssembleEdge<T, false, false>),
+                           dim3(gridFor(nL_)), dim3(kBlk), 0, stream_, nL_,
+                           dCamOf_, dPtOf_, dR_[bak], dJc_[bak], dJp_[bak],
+                           nullptr, dHll_, dHpl_, dG_, ncam_, dPtPos_, dSlab_);
+      else
+        hipLaunchKernelGGL((kAssembleEdge<T, false, true>),
+                           dim3(gridFor(nL_)), dim3(kBlk), 0, stream_, nL_,
+                           dCamOf_, dPtOf_, dR_[bak], dJc_[bak], dJp_[bak],
+                           nullptr, dHll_, dHpl_, dG_, ncam_, dPtPos_, dSlab_);
       if (nChunks_ > 0)
         hipLaunchKernelGGL((kAssembleCam<T, false>), dim3(nChunks_), dim3(128),
                            0, stream_, nChunks_, dChCam_, dChLo_, dChHi_, nL_,
                            dR_[bak], dJc_[bak], nullptr, dHpp_, dG_);
     }
     if (!implicit_)
-      hipLaunchKernelGGL(kTransposeHlp<T>, dim3(gridFor(nL_)), dim3(kBlk), 0,
-                         stream_, nL_, dPtPerm_, dHpl_, dHlp_);
+      hipLaunchKernelGGL(kFinalizePt<T>, dim3(gridFor(npt_)), dim3(kBlk), 0,
+                         stream_, npt_, dPtRowPtr_, dSlab_, nL_, dHlp_, dHll_,
+                         dG_ + nc_);
     allreduce(dHpp_, (int64_t)ncam_ * 81);
     allreduce(dHll_, (int64_t)npt_ * 9);
     allreduce(dG_, dim_);
@@ -1339,13 +1414,13 @@ class GpuEngine final : public Engine<T> {
   int cur_ = 0;
   int nChunks_ = 0;
   int *dCamOf_{}, *dPtOf_{}, *dChCam_{}, *dChLo_{}, *dChHi_{}, *dFail_{};
-  int *dPtPerm_{}, *dPtRowPtr_{}, *dCamOfPt_{}, *dPtOfPt_{};
+  int *dPtPerm_{}, *dPtRowPtr_{}, *dCamOfPt_{}, *dPtOfPt_{}, *dPtPos_{};
   T *dMeas_{}, *dInfo_{}, *dLeaf_{}, *dMeasSplit_{};
   CustomForward<T> customFwd_;
   unsigned char *dCamFixed_{}, *dPtFixed_{};
   T *dParams_{}, *dParamsBak_{};
   T *dR_[2]{}, *dJc_[2]{}, *dJp_[2]{};
-  T *dHpp_{}, *dHll_{}, *dHpl_{}, *dHlp_{}, *dG_{}, *dGBak_{};
+  T *dHpp_{}, *dHll_{}, *dHpl_{}, *dHlp_{}, *dSlab_{}, *dG_{}, *dGBak_{};
   T *dHppD_{}, *dHllD_{}, *dHppInv_{}, *dHllInv_{};
   T *dDeltaX_{}, *dDeltaXBak_{};
   T *dP_{}, *dRr_{}, *dZ_{}, *dQ_{}, *dV_{}, *dW_{}, *dTemp_{}, *dXBak_{};
