@@ -36,11 +36,11 @@ HostAllreduce<T> wrapAllreduce(py::object fn) {
   if (fn.is_none()) return nullptr;
   // Keep the callable alive in the closure; reacquire the GIL per call.
   auto holder = std::make_shared<py::object>(std::move(fn));
-  return [holder](T* data, std::size_t n) {
+  return [holder](T* data, std::size_t n, char op) {
     py::gil_scoped_acquire gil;
     py::array arr(py::dtype::of<T>(), {(py::ssize_t)n}, {(py::ssize_t)sizeof(T)},
                   data, py::none());
-    (*holder)(arr);
+    (*holder)(arr, op == 'm' ? "max" : "sum");
   };
 }
 
@@ -283,8 +283,9 @@ struct PyProblem {
     d["nobs"] = ix.nobs;
     d["cam_of"] = py::array_t<int>((py::ssize_t)ix.camOf.size(), ix.camOf.data());
     d["pt_of"] = py::array_t<int>((py::ssize_t)ix.ptOf.size(), ix.ptOf.data());
-    d["cam_rowptr"] =
-        py::array_t<int64_t>((py::ssize_t)ix.camRowPtr.size(), ix.camRowPtr.data());
+    d["pt_rowptr"] =
+        py::array_t<int64_t>((py::ssize_t)ix.ptRowPtr.size(), ix.ptRowPtr.data());
+    d["pt_split"] = py::array_t<int>((py::ssize_t)ix.ptSplit.size(), ix.ptSplit.data());
     d["split"] = py::array_t<int64_t>((py::ssize_t)ix.split.size(), ix.split.data());
     d["perm"] = py::array_t<int64_t>((py::ssize_t)ix.perm.size(), ix.perm.data());
     return d;

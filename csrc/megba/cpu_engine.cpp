@@ -40,10 +40,11 @@ class CpuEngine final : public Engine<T> {
       for (int64_t e = 0; e < nL_ * 3; ++e)
         info_[e] = (T)ix.infoSorted[3 * e0_ + e];
     }
-    // Local camera range for segment loops.
-    camLo_ = camOf_.empty() ? 0 : camOf_.front();
-    camHi_ = camOf_.empty() ? 0 : camOf_.back() + 1;
-    camRowPtr_.assign(ix.camRowPtr.begin(), ix.camRowPtr.end());
+    // Local point range (partition is point-aligned: every local edge's
+    // point is owned by this rank).
+    ptLo_ = ix.ptSplit[rank_];
+    ptHi_ = ix.ptSplit[rank_ + 1];
+    ptRowPtr_.assign(ix.ptRowPtr.begin(), ix.ptRowPtr.end());
 
     camFixed_ = prob.camFixed;
     ptFixed_ = prob.ptFixed;
@@ -110,7 +111,7 @@ class CpuEngine final : public Engine<T> {
       zeroFixed(e);
     }
     T buf = chi2;
-    if (ar_) ar_(&buf, 1);
+    if (ar_) ar_(&buf, 1, 's');
     return (double)buf;
   }
 
@@ -121,14 +122,12 @@ class CpuEngine final : public Engine<T> {
     T* gc = g_.data();
     T* gp = g_.data() + (size_t)ncam_ * 9;
 
-    // Per-edge: weighted J rows, Hpl block, and scatter into Hll/g_p
-    // (atomics: average point degree is small) -- camera blocks are handled
-    // by the segment loop below to avoid heavy contention.
+    // Per-edge: weighted J rows, Hpl block, camera blocks via atomics
+    // (cameras are the replicated, randomly-interleaved side now).
 #pragma omp parallel for schedule(static)
     for (int64_t e = 0; e < nL_; ++e) {
       T wJc[2][9], wJp[2][3], wr[2];
       weightedRows(e, wJc, wJp, wr);
-      const T* Jp = &JpBak_[6 * e];
       if (!implicit_) {
         T* hpl = &Hpl_[27 * e];
         for (int i = 0; i < 9; ++i)
@@ -136,46 +135,44 @@ class CpuEngine final : public Engine<T> {
             hpl[i * 3 + j] = JcBak_[18 * e + i] * wJp[0][j] +
                              JcBak_[18 * e + 9 + i] * wJp[1][j];
       }
-      T* hll = &Hll_[(size_t)ptOf_[e] * 9];
-      for (int i = 0; i < 3; ++i)
-        for (int j = 0; j < 3; ++j) {
-          const T vInc = Jp[i] * wJp[0][j] + Jp[3 + i] * wJp[1][j];
+      const int c = camOf_[e];
+      T* hpp = &Hpp_[(size_t)c * 81];
+      const T* Jc = &JcBak_[18 * e];
+      for (int i = 0; i < 9; ++i) {
+        for (int j = 0; j < 9; ++j) {
+          const T vInc = Jc[i] * wJc[0][j] + Jc[9 + i] * wJc[1][j];
 #pragma omp atomic
-          hll[i * 3 + j] += vInc;
+          hpp[i * 9 + j] += vInc;
         }
-      T* gpt = &gp[(size_t)ptOf_[e] * 3];
-      for (int i = 0; i < 3; ++i) {
-        const T vInc = -(Jp[i] * wr[0] + Jp[3 + i] * wr[1]);
+        const T gInc = -(Jc[i] * wr[0] + Jc[9 + i] * wr[1]);
 #pragma omp atomic
-        gpt[i] += vInc;
+        gc[(size_t)c * 9 + i] += gInc;
       }
     }
 
-    // Camera segments (edges are (cam,pt)-sorted, so each camera's local
-    // edges are contiguous).
-#pragma omp parallel for schedule(dynamic, 8)
-    for (int c = camLo_; c < camHi_; ++c) {
-      const int64_t lo = std::max(camRowPtr_[c], e0_) - e0_;
-      const int64_t hi = std::min(camRowPtr_[c + 1], e1_) - e0_;
-      if (lo >= hi) continue;
-      T* hpp = &Hpp_[(size_t)c * 81];
-      T* gcam = &gc[(size_t)c * 9];
+    // Point blocks: local point segments (edges are (pt,cam)-sorted).
+#pragma omp parallel for schedule(static)
+    for (int p = ptLo_; p < ptHi_; ++p) {
+      const int64_t lo = ptRowPtr_[p] - e0_;
+      const int64_t hi = ptRowPtr_[p + 1] - e0_;
+      T* hll = &Hll_[(size_t)p * 9];
+      T* gpt = &gp[(size_t)p * 3];
       for (int64_t e = lo; e < hi; ++e) {
         T wJc[2][9], wJp[2][3], wr[2];
         weightedRows(e, wJc, wJp, wr);
-        const T* Jc = &JcBak_[18 * e];
-        for (int i = 0; i < 9; ++i) {
-          for (int j = 0; j < 9; ++j)
-            hpp[i * 9 + j] += Jc[i] * wJc[0][j] + Jc[9 + i] * wJc[1][j];
-          gcam[i] -= Jc[i] * wr[0] + Jc[9 + i] * wr[1];
+        const T* Jp = &JpBak_[6 * e];
+        for (int i = 0; i < 3; ++i) {
+          for (int j = 0; j < 3; ++j)
+            hll[i * 3 + j] += Jp[i] * wJp[0][j] + Jp[3 + i] * wJp[1][j];
+          gpt[i] -= Jp[i] * wr[0] + Jp[3 + i] * wr[1];
         }
       }
     }
 
+    // Only the small camera-side quantities cross ranks.
     if (ar_) {
-      ar_(Hpp_.data(), Hpp_.size());
-      ar_(Hll_.data(), Hll_.size());
-      ar_(g_.data(), g_.size());
+      ar_(Hpp_.data(), Hpp_.size(), 's');
+      ar_(gc, (size_t)ncam_ * 9, 's');
     }
   }
 
@@ -218,7 +215,7 @@ class CpuEngine final : public Engine<T> {
       for (int i = 0; i < 9; ++i) HppD_[(size_t)c * 81 + i * 10] *= f;
     }
 #pragma omp parallel for schedule(static)
-    for (int p = 0; p < npt_; ++p) {
+    for (int p = ptLo_; p < ptHi_; ++p) {
       if (ptFixed_[p]) {
         for (int i = 0; i < 9; ++i) HllD_[(size_t)p * 9 + i] = T(0);
         for (int i = 0; i < 3; ++i) HllD_[(size_t)p * 9 + i * 4] = T(1);
@@ -240,7 +237,7 @@ class CpuEngine final : public Engine<T> {
     applyHllInv(gp, w.data());
     spmvEx(w.data(), v.data());
     for (int64_t i = 0; i < nc; ++i) v[i] = gc[i] / (T)world_ - v[i];
-    if (ar_) ar_(v.data(), nc);
+    if (ar_) ar_(v.data(), nc, 's');
     // Warm start from current deltaX camera part.
     std::memcpy(x.data(), deltaX_.data(), nc * sizeof(T));
     // r = v - S x
@@ -279,11 +276,10 @@ class CpuEngine final : public Engine<T> {
     }
     // Back-substitution: deltaX_p = Cinv * (g_p - E^T x).
     spmvEtx(x.data(), temp.data());
-    if (ar_) ar_(temp.data(), (size_t)npt_ * 3);
     std::memcpy(deltaX_.data(), x.data(), nc * sizeof(T));
     T* dxp = deltaX_.data() + nc;
 #pragma omp parallel for schedule(static)
-    for (int ptI = 0; ptI < npt_; ++ptI) {
+    for (int ptI = ptLo_; ptI < ptHi_; ++ptI) {
       T rhs[3];
       for (int i = 0; i < 3; ++i) rhs[i] = gp[3 * ptI + i] - temp[3 * ptI + i];
       matVec<T, 3>(&HllInv_[(size_t)ptI * 9], rhs, &dxp[3 * ptI]);
@@ -293,19 +289,33 @@ class CpuEngine final : public Engine<T> {
 
   double deltaXL2() override {
     double s = 0;
-    for (int64_t i = 0; i < dim_; ++i) s += (double)deltaX_[i] * deltaX_[i];
-    return std::sqrt(s);
+    const int64_t nc = (int64_t)ncam_ * 9;
+    for (int64_t i = 0; i < nc; ++i) s += (double)deltaX_[i] * deltaX_[i];
+    T sp = T(0);
+    for (int64_t i = nc + (int64_t)ptLo_ * 3; i < nc + (int64_t)ptHi_ * 3; ++i)
+      sp += deltaX_[i] * deltaX_[i];
+    if (ar_) ar_(&sp, 1, 's');
+    return std::sqrt(s + (double)sp);
   }
   double xL2() override {
     double s = 0;
     for (const T v : cams_) s += (double)v * v;
-    for (const T v : pts_) s += (double)v * v;
-    return std::sqrt(s);
+    T sp = T(0);
+    for (int64_t i = (int64_t)ptLo_ * 3; i < (int64_t)ptHi_ * 3; ++i)
+      sp += pts_[i] * pts_[i];
+    if (ar_) ar_(&sp, 1, 's');
+    return std::sqrt(s + (double)sp);
   }
   double gInf() override {
     double m = 0;
-    for (int64_t i = 0; i < dim_; ++i) m = std::max(m, std::abs((double)g_[i]));
-    return m;
+    const int64_t nc = (int64_t)ncam_ * 9;
+    for (int64_t i = 0; i < nc; ++i)
+      m = std::max(m, std::abs((double)g_[i]));
+    T mp = T(0);
+    for (int64_t i = nc + (int64_t)ptLo_ * 3; i < nc + (int64_t)ptHi_ * 3; ++i)
+      mp = std::max(mp, (T)std::abs((double)g_[i]));
+    if (ar_) ar_(&mp, 1, 'm');
+    return std::max(m, (double)mp);
   }
 
   void updateParams() override {
@@ -313,7 +323,8 @@ class CpuEngine final : public Engine<T> {
     for (int64_t i = 0; i < (int64_t)cams_.size(); ++i) cams_[i] += deltaX_[i];
     const T* dxp = deltaX_.data() + (size_t)ncam_ * 9;
 #pragma omp parallel for schedule(static)
-    for (int64_t i = 0; i < (int64_t)pts_.size(); ++i) pts_[i] += dxp[i];
+    for (int64_t i = (int64_t)ptLo_ * 3; i < (int64_t)ptHi_ * 3; ++i)
+      pts_[i] += dxp[i];
   }
 
   double rhoDenominator(double chi2Backup) override {
@@ -334,13 +345,22 @@ class CpuEngine final : public Engine<T> {
       }
     }
     T buf = s;
-    if (ar_) ar_(&buf, 1);
+    if (ar_) ar_(&buf, 1, 's');
     return (double)buf - chi2Backup;
   }
 
   // ---- debug access -------------------------------------------------------
-  void getParams(double* cams, double* pts) const override {
+  void getParams(double* cams, double* pts) override {
     for (size_t i = 0; i < cams_.size(); ++i) cams[i] = (double)cams_[i];
+    if (ar_ && world_ > 1) {
+      // points are sharded: zero non-local entries and sum across ranks
+      std::vector<T> full((size_t)npt_ * 3, T(0));
+      for (int64_t i = (int64_t)ptLo_ * 3; i < (int64_t)ptHi_ * 3; ++i)
+        full[i] = pts_[i];
+      ar_(full.data(), full.size(), 's');
+      for (size_t i = 0; i < full.size(); ++i) pts[i] = (double)full[i];
+      return;
+    }
     for (size_t i = 0; i < pts_.size(); ++i) pts[i] = (double)pts_[i];
   }
   DenseDump dump() const override {
@@ -405,7 +425,7 @@ class CpuEngine final : public Engine<T> {
       zeroFixed(e);
     }
     T buf = chi2;
-    if (ar_) ar_(&buf, 1);
+    if (ar_) ar_(&buf, 1, 's');
     return (double)buf;
   }
 
@@ -458,7 +478,7 @@ class CpuEngine final : public Engine<T> {
         jitterInvert<9>(&HppD_[(size_t)c * 81], &HppInv_[(size_t)c * 81]);
     }
 #pragma omp parallel for schedule(static)
-    for (int p = 0; p < npt_; ++p) {
+    for (int p = ptLo_; p < ptHi_; ++p) {
       if (!spdInvert<T, 3>(&HllD_[(size_t)p * 9], &HllInv_[(size_t)p * 9]))
         jitterInvert<3>(&HllD_[(size_t)p * 9], &HllInv_[(size_t)p * 9]);
     }
@@ -480,43 +500,33 @@ class CpuEngine final : public Engine<T> {
     MEGBA_CHECK(false, "singular Hessian block");
   }
 
-  // temp[3npt] = sum over local edges of Hpl_e^T * x[cam]; caller allreduces.
+  // temp[3*pt] = Hpl^T x over this rank's point segments; fully local (the
+  // point side is sharded -- no communication, unlike reference site A4).
   void spmvEtx(const T* x, T* temp) {
-    std::fill(temp, temp + (size_t)npt_ * 3, T(0));
-    if (implicit_) {
-      // matrix-free: Hpl_e^T x = Jp^T W (Jc x)  (reference C23,
-      // implicit_schur_pcg_solver.cu:57-90)
 #pragma omp parallel for schedule(static)
-      for (int64_t e = 0; e < nL_; ++e) {
-        const T* Jc = &JcBak_[18 * e];
-        const T* Jp = &JpBak_[6 * e];
+    for (int p = ptLo_; p < ptHi_; ++p) {
+      const int64_t lo = ptRowPtr_[p] - e0_;
+      const int64_t hi = ptRowPtr_[p + 1] - e0_;
+      T o[3] = {T(0), T(0), T(0)};
+      for (int64_t e = lo; e < hi; ++e) {
         const T* xc = &x[(size_t)camOf_[e] * 9];
-        T u0 = T(0), u1 = T(0);
-        for (int i = 0; i < 9; ++i) {
-          u0 += Jc[i] * xc[i];
-          u1 += Jc[9 + i] * xc[i];
-        }
-        applyInfo(e, u0, u1);
-        T* out = &temp[(size_t)ptOf_[e] * 3];
-        for (int j = 0; j < 3; ++j) {
-          const T v = Jp[j] * u0 + Jp[3 + j] * u1;
-#pragma omp atomic
-          out[j] += v;
+        if (implicit_) {
+          const T* Jc = &JcBak_[18 * e];
+          const T* Jp = &JpBak_[6 * e];
+          T u0 = T(0), u1 = T(0);
+          for (int i = 0; i < 9; ++i) {
+            u0 += Jc[i] * xc[i];
+            u1 += Jc[9 + i] * xc[i];
+          }
+          applyInfo(e, u0, u1);
+          for (int j = 0; j < 3; ++j) o[j] += Jp[j] * u0 + Jp[3 + j] * u1;
+        } else {
+          const T* blk = &Hpl_[27 * e];
+          for (int j = 0; j < 3; ++j)
+            for (int i = 0; i < 9; ++i) o[j] += blk[i * 3 + j] * xc[i];
         }
       }
-      return;
-    }
-#pragma omp parallel for schedule(static)
-    for (int64_t e = 0; e < nL_; ++e) {
-      const T* blk = &Hpl_[27 * e];
-      const T* xc = &x[(size_t)camOf_[e] * 9];
-      T* out = &temp[(size_t)ptOf_[e] * 3];
-      for (int j = 0; j < 3; ++j) {
-        T s = T(0);
-        for (int i = 0; i < 9; ++i) s += blk[i * 3 + j] * xc[i];
-#pragma omp atomic
-        out[j] += s;
-      }
+      for (int j = 0; j < 3; ++j) temp[(size_t)p * 3 + j] = o[j];
     }
   }
 
@@ -530,45 +540,40 @@ class CpuEngine final : public Engine<T> {
     }
   }
 
-  // out[9ncam] = sum over local edges of Hpl_e * w[pt]  (no allreduce here).
+  // out[9ncam] = partial E w over local edges (caller allreduces 9*ncam --
+  // the ONLY per-iteration collective, 128 KB on Venice).
   void spmvEx(const T* w, T* out) {
     std::fill(out, out + (size_t)ncam_ * 9, T(0));
-    if (implicit_) {
-#pragma omp parallel for schedule(dynamic, 8)
-      for (int c = camLo_; c < camHi_; ++c) {
-        const int64_t lo = std::max(camRowPtr_[c], e0_) - e0_;
-        const int64_t hi = std::min(camRowPtr_[c + 1], e1_) - e0_;
-        T* oc = &out[(size_t)c * 9];
-        for (int64_t e = lo; e < hi; ++e) {
-          const T* Jc = &JcBak_[18 * e];
-          const T* Jp = &JpBak_[6 * e];
-          const T* wp = &w[(size_t)ptOf_[e] * 3];
-          T u0 = Jp[0] * wp[0] + Jp[1] * wp[1] + Jp[2] * wp[2];
-          T u1 = Jp[3] * wp[0] + Jp[4] * wp[1] + Jp[5] * wp[2];
-          applyInfo(e, u0, u1);
-          for (int i = 0; i < 9; ++i) oc[i] += Jc[i] * u0 + Jc[9 + i] * u1;
+#pragma omp parallel for schedule(static)
+    for (int64_t e = 0; e < nL_; ++e) {
+      const T* wp = &w[(size_t)ptOf_[e] * 3];
+      T* oc = &out[(size_t)camOf_[e] * 9];
+      if (implicit_) {
+        const T* Jc = &JcBak_[18 * e];
+        const T* Jp = &JpBak_[6 * e];
+        T u0 = Jp[0] * wp[0] + Jp[1] * wp[1] + Jp[2] * wp[2];
+        T u1 = Jp[3] * wp[0] + Jp[4] * wp[1] + Jp[5] * wp[2];
+        applyInfo(e, u0, u1);
+        for (int i = 0; i < 9; ++i) {
+          const T vInc = Jc[i] * u0 + Jc[9 + i] * u1;
+#pragma omp atomic
+          oc[i] += vInc;
         }
-      }
-      return;
-    }
-#pragma omp parallel for schedule(dynamic, 8)
-    for (int c = camLo_; c < camHi_; ++c) {
-      const int64_t lo = std::max(camRowPtr_[c], e0_) - e0_;
-      const int64_t hi = std::min(camRowPtr_[c + 1], e1_) - e0_;
-      T* oc = &out[(size_t)c * 9];
-      for (int64_t e = lo; e < hi; ++e) {
+      } else {
         const T* blk = &Hpl_[27 * e];
-        const T* wp = &w[(size_t)ptOf_[e] * 3];
-        for (int i = 0; i < 9; ++i)
-          oc[i] += blk[i * 3] * wp[0] + blk[i * 3 + 1] * wp[1] +
-                   blk[i * 3 + 2] * wp[2];
+        for (int i = 0; i < 9; ++i) {
+          const T vInc = blk[i * 3] * wp[0] + blk[i * 3 + 1] * wp[1] +
+                         blk[i * 3 + 2] * wp[2];
+#pragma omp atomic
+          oc[i] += vInc;
+        }
       }
     }
   }
 
   void applyHllInv(const T* in, T* out) {
 #pragma omp parallel for schedule(static)
-    for (int p = 0; p < npt_; ++p)
+    for (int p = ptLo_; p < ptHi_; ++p)
       matVec<T, 3>(&HllInv_[(size_t)p * 9], &in[3 * p], &out[3 * p]);
   }
   void applyHppInv(const T* in, T* out) {
@@ -580,10 +585,9 @@ class CpuEngine final : public Engine<T> {
   // q = S x = HppD x - E Cinv E^T x   (2 allreduces, reference site A4).
   void schurApply(const T* x, T* q, T* temp, T* w) {
     spmvEtx(x, temp);
-    if (ar_) ar_(temp, (size_t)npt_ * 3);
     applyHllInv(temp, w);
     spmvEx(w, q);
-    if (ar_) ar_(q, (size_t)ncam_ * 9);
+    if (ar_) ar_(q, (size_t)ncam_ * 9, 's');
 #pragma omp parallel for schedule(static)
     for (int c = 0; c < ncam_; ++c) {
       T bx[9];
@@ -607,11 +611,11 @@ class CpuEngine final : public Engine<T> {
   int rank_, world_, ncam_, npt_;
   bool analytical_ = false;
   bool implicit_ = false;
-  int camLo_ = 0, camHi_ = 0;
+  int ptLo_ = 0, ptHi_ = 0;
   int64_t e0_ = 0, e1_ = 0, nL_ = 0, dim_ = 0;
   std::vector<int> camOf_, ptOf_;
   std::vector<uint8_t> camFixed_, ptFixed_;
-  std::vector<int64_t> camRowPtr_;
+  std::vector<int64_t> ptRowPtr_;
   std::vector<T> meas_, info_;
   bool hasInfo_ = false;
   std::vector<T> cams_, pts_, camsBak_, ptsBak_;

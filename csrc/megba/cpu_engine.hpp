@@ -16,9 +16,10 @@
 
 namespace megba {
 
-// In-place sum-allreduce over all ranks (no-op when null / worldSize==1).
+// In-place allreduce over all ranks (no-op when null / worldSize==1).
+// op: 's' = sum, 'm' = max.
 template <typename T>
-using HostAllreduce = std::function<void(T*, std::size_t)>;
+using HostAllreduce = std::function<void(T*, std::size_t, char)>;
 
 template <typename T>
 std::unique_ptr<Engine<T>> makeCpuEngine(const BAProblemHost& prob,

@@ -67,7 +67,7 @@ class Engine {
   virtual double rhoDenominator(double chi2Backup) = 0;
 
   // Introspection (tests / write-back).
-  virtual void getParams(double* cams, double* pts) const = 0;
+  virtual void getParams(double* cams, double* pts) = 0;
   virtual DenseDump dump() const = 0;
 };
 

@@ -1,38 +1,43 @@
 // MI355X (gfx950) engine: HIP kernels + host orchestration + RCCL collectives.
 //
-// Kernel design notes (MI355X-first redesign, not a port — behavioural
-// anchors cite /root/reference):
+// Kernel / distribution design (MI355X-first redesign, not a port —
+// behavioural anchors cite /root/reference):
+//  * Observations are (point, camera)-sorted and partitioned on POINT
+//    boundaries: each rank owns its points outright, so the whole point side
+//    (Hll, g_p, Cinv, E^T x, back-substitution) is communication-free and
+//    the only per-PCG-iteration collective is an allreduce of the replicated
+//    camera vector (9*ncam = 128 KB on Venice).  The reference replicates
+//    the point side instead and allreduces 3*npt words (24 MB on Venice)
+//    per iteration (its sites A1/A3-A6) — a ~190x traffic reduction sized
+//    for xGMI's per-link ring bandwidth.
 //  * kForward fuses the ENTIRE vectorised-autodiff forward pass into one
 //    kernel: each observation is handled by 4 consecutive lanes, each lane
 //    carrying a Jet<T,3> slice of the 12-wide dual part, so every
 //    intermediate of the reprojection expression lives in VGPRs.  The
-//    reference instead launches ~1 CUDA kernel per elementary op and streams
+//    reference launches ~1 CUDA kernel per elementary op and streams
 //    (N+1)*nItem doubles through HBM each time
-//    (src/operator/jet_vector_math_impl.cu).  Forward-mode gradient slices
-//    never interact, so the lane split needs no cross-lane traffic.
-//  * kAssemble scatters J^T W J blocks with native fp64 atomics (gfx950 has
-//    hardware f64 atomicAdd; the reference needed a CAS shim pre-sm60,
-//    src/edge/build_linear_system.cu:19-43).  Hpl gets one conflict-free
-//    block per observation ((cam,pt)-sorted order = BSR layout).
-//  * The Schur product S*p = B*p - E*Cinv*E^T*p runs as chunked block-row
-//    kernels over the BSR blocks (one wave per <=256-edge chunk of a camera
-//    row, wave-wide shuffle reduction) instead of cuSPARSE scalar CSR SpMV
-//    (reference schur_pcg_solver.cu:206-232).
-//  * Control-flow scalars (rho, p^T q, norms) use a fixed-shape two-pass
-//    deterministic reduction so every rank takes identical PCG branches —
-//    required because ranks run the replicated vector ops independently and
-//    collectives must stay matched.
+//    (src/operator/jet_vector_math_impl.cu).
+//  * Assembly transports the camera-side per-edge data through a cam-sorted
+//    edge-major slab (contiguous ~400 B per edge, ~1 line-fetch per 64 B)
+//    instead of per-element transpose gathers (8x amplification) or
+//    per-element atomics (the all-atomic version measured 40.8 ms on
+//    Venice-5M, profiles/r01_venice_single_gpu.md).  Point-side Hll/g_p
+//    accumulate with a wave-level segmented scan over the point-sorted runs
+//    — atomics only at segment tails.
+//  * Control-flow scalars (rho, p^T q, norms) use fixed-shape two-pass
+//    deterministic reductions so every rank takes identical PCG branches.
 #include <hip/hip_runtime.h>
 #include <rccl/rccl.h>
 
 #include <cmath>
 #include <cstring>
+#include <type_traits>
 #include <vector>
 
 #include "../analytical.hpp"
 #include "../bal_functor.hpp"
-#include "../smallmat.hpp"
 #include "../jv/jetvector.hpp"
+#include "../smallmat.hpp"
 #include "gpu_engine.hpp"
 
 namespace megba {
@@ -123,15 +128,11 @@ __global__ void kRedFinal(const double* part, int nb, double* out) {
 // ---------------------------------------------------------------------------
 template <typename T>
 __global__ __launch_bounds__(256, 2) void kForward(
-                         int64_t nL, const int* __restrict__ camOf,
-                         const int* __restrict__ ptOf,
-                         const T* __restrict__ params, int ncam,
-                         const T* __restrict__ meas,
-                         const unsigned char* __restrict__ camFixed,
-                         const unsigned char* __restrict__ ptFixed,
-                         T* __restrict__ rOut,
-                         T* __restrict__ Jc, T* __restrict__ Jp,
-                         double* chi2Acc) {
+    int64_t nL, const int* __restrict__ camOf, const int* __restrict__ ptOf,
+    const T* __restrict__ params, int ncam, const T* __restrict__ meas,
+    const unsigned char* __restrict__ camFixed,
+    const unsigned char* __restrict__ ptFixed, T* __restrict__ rOut,
+    T* __restrict__ Jc, T* __restrict__ Jp, double* chi2Acc) {
   using J3 = Jet<T, 3>;
   __shared__ double sm[kBlk];
   double chi2 = 0.0;
@@ -141,7 +142,7 @@ __global__ __launch_bounds__(256, 2) void kForward(
        i += (int64_t)gridDim.x * kBlk) {
     const int64_t e = i >> 2;
     const int sub = (int)(i & 3);       // this lane's 3-wide gradient slice
-    const int base = 3 * sub;           // global gradient columns [base,base+3)
+    const int base = 3 * sub;           // gradient columns [base,base+3)
     const T* cp = params + (int64_t)camOf[e] * 9;
     const T* pp = ptsBase + (int64_t)ptOf[e] * 3;
     J3 cam[9], pt[3], res[2];
@@ -176,8 +177,7 @@ __global__ __launch_bounds__(256, 2) void kForward(
 }
 
 // Analytical-derivative forward: one thread per edge, closed-form residual +
-// 2x12 Jacobian (reference C11, src/geo/analytical_derivatives.cu) -- no
-// dual-number redundancy, ~70 fp64 values live per lane.
+// 2x12 Jacobian (reference C11, src/geo/analytical_derivatives.cu).
 template <typename T>
 __global__ void kForwardAnalytical(int64_t nL, const int* __restrict__ camOf,
                                    const int* __restrict__ ptOf,
@@ -259,146 +259,147 @@ __global__ void kRepackRes(int64_t nL, int row, const T* __restrict__ rv,
 // ---------------------------------------------------------------------------
 // Assembly
 // ---------------------------------------------------------------------------
-// Per-edge part: Hpl block (grad-major [27][nL], conflict-free), Hll and g_p
-// (atomics; point degree is low so contention is cheap).  The camera blocks
-// Hpp/g_c are handled by kAssembleCam: per-camera chunks with LDS-staged J
-// rows and 90 distributed accumulators, one atomicAdd per chunk instead of 90
-// per edge -- the all-atomic version measured 40.8 ms on Venice-5M
-// (profiles/r01_venice_baseline.md), dominated by ~2800-way per-camera
-// contention.
-// EXPL=true (explicit Schur): writes the Hpl block grad-major (for E*w) and
-// a point-sorted edge-major slab row [40] = {Hlp block transposed (27),
-// Hll upper-sym contribution (6), g_p contribution (3), pad} -- 320 B
-// contiguous per edge, so the scatter costs ~1 line-fetch per 64 B instead
-// of the 8x amplification of a per-element transpose gather.  kFinalizePt
-// then streams the slab in point order with NO atomics.  EXPL=false
-// (implicit): no Hpl/slab; Hll/g_p accumulate via atomics as before.
+// Slab row layout (doubles), written at the edge's CAM-sorted position:
+//   [0..26]                 Hpl block (EXPL only)
+//   [JCOFF..JCOFF+17]       Jc rows (col-major pairs: jc[col][row])
+//   [JCOFF+18, JCOFF+19]    weighted residual rows
+//   [JCOFF+20..JCOFF+37]    weighted Jc rows (HASINFO only)
+template <bool EXPL, bool HASINFO>
+struct SlabLayout {
+  static constexpr int JCOFF = EXPL ? 27 : 0;
+  static constexpr int WROFF = JCOFF + 18;
+  static constexpr int WJCOFF = WROFF + 2;
+  static constexpr int SW = WJCOFF + (HASINFO ? 18 : 0);
+};
+
+// Per-edge pass, primary ((pt,cam)-sorted) order: Hpl grad-major (for E^T x),
+// the cam-sorted slab row, and the point-side Hll/g_p via a wave-level
+// segmented scan (atomics only at point-run tails).
 template <typename T, bool HASINFO, bool EXPL>
 __global__ void kAssembleEdge(int64_t nL, const int* __restrict__ camOf,
-                          const int* __restrict__ ptOf, const T* __restrict__ r,
-                          const T* __restrict__ Jc, const T* __restrict__ Jp,
-                          const T* __restrict__ info,
-                          T* __restrict__ Hll, T* __restrict__ Hpl,
-                          T* __restrict__ g, int ncam,
-                          const int* __restrict__ ptPos, T* __restrict__ slab) {
+                              const int* __restrict__ ptOf,
+                              const T* __restrict__ r, const T* __restrict__ Jc,
+                              const T* __restrict__ Jp,
+                              const T* __restrict__ info, T* __restrict__ Hll,
+                              T* __restrict__ Hpl, T* __restrict__ g, int ncam,
+                              const int* __restrict__ camPos,
+                              T* __restrict__ slab) {
+  using L = SlabLayout<EXPL, HASINFO>;
   T* gp = g + (int64_t)ncam * 9;
-  for (int64_t e = blockIdx.x * (int64_t)kBlk + threadIdx.x; e < nL;
-       e += (int64_t)gridDim.x * kBlk) {
-    T jc[2][9], jp[2][3], rr[2];
-    for (int k = 0; k < 9; ++k) {
-      jc[0][k] = Jc[((int64_t)(k * 2 + 0)) * nL + e];
-      jc[1][k] = Jc[((int64_t)(k * 2 + 1)) * nL + e];
-    }
-    for (int k = 0; k < 3; ++k) {
-      jp[0][k] = Jp[((int64_t)(k * 2 + 0)) * nL + e];
-      jp[1][k] = Jp[((int64_t)(k * 2 + 1)) * nL + e];
-    }
-    rr[0] = r[e];
-    rr[1] = r[nL + e];
-    T wjc[2][9], wjp[2][3], wr[2];
-    if (HASINFO) {
-      const T w00 = info[3 * e], w01 = info[3 * e + 1], w11 = info[3 * e + 2];
+  const int lane = threadIdx.x & 63;
+  const int64_t nWork = ((nL + kBlk - 1) / kBlk) * (int64_t)kBlk;
+  for (int64_t e0i = blockIdx.x * (int64_t)kBlk + threadIdx.x; e0i < nWork;
+       e0i += (int64_t)gridDim.x * kBlk) {
+    const bool active = e0i < nL;
+    const int64_t e = active ? e0i : nL - 1;
+    const int pt = ptOf[e];
+    T hll6[6] = {0, 0, 0, 0, 0, 0};
+    T gp3[3] = {0, 0, 0};
+    if (active) {
+      T jc[2][9], jp[2][3], rr[2];
       for (int k = 0; k < 9; ++k) {
-        wjc[0][k] = w00 * jc[0][k] + w01 * jc[1][k];
-        wjc[1][k] = w01 * jc[0][k] + w11 * jc[1][k];
+        jc[0][k] = Jc[((int64_t)(k * 2 + 0)) * nL + e];
+        jc[1][k] = Jc[((int64_t)(k * 2 + 1)) * nL + e];
       }
       for (int k = 0; k < 3; ++k) {
-        wjp[0][k] = w00 * jp[0][k] + w01 * jp[1][k];
-        wjp[1][k] = w01 * jp[0][k] + w11 * jp[1][k];
+        jp[0][k] = Jp[((int64_t)(k * 2 + 0)) * nL + e];
+        jp[1][k] = Jp[((int64_t)(k * 2 + 1)) * nL + e];
       }
-      wr[0] = w00 * rr[0] + w01 * rr[1];
-      wr[1] = w01 * rr[0] + w11 * rr[1];
-    } else {
-      for (int k = 0; k < 9; ++k) {
-        wjc[0][k] = jc[0][k];
-        wjc[1][k] = jc[1][k];
-      }
-      for (int k = 0; k < 3; ++k) {
-        wjp[0][k] = jp[0][k];
-        wjp[1][k] = jp[1][k];
-      }
-      wr[0] = rr[0];
-      wr[1] = rr[1];
-    }
-    if (EXPL) {
-      // Hpl grad-major (cam-sorted, for E*w)
-      T blk[9][3];
-      for (int a = 0; a < 9; ++a)
-        for (int b = 0; b < 3; ++b) {
-          blk[a][b] = jc[0][a] * wjp[0][b] + jc[1][a] * wjp[1][b];
-          Hpl[((int64_t)(a * 3 + b)) * nL + e] = blk[a][b];
+      rr[0] = r[e];
+      rr[1] = r[nL + e];
+      T wjp[2][3], wr[2];
+      T wjc0[9], wjc1[9];
+      if (HASINFO) {
+        const T w00 = info[3 * e], w01 = info[3 * e + 1], w11 = info[3 * e + 2];
+        for (int k = 0; k < 9; ++k) {
+          wjc0[k] = w00 * jc[0][k] + w01 * jc[1][k];
+          wjc1[k] = w01 * jc[0][k] + w11 * jc[1][k];
         }
-      // slab row at this edge's point-sorted position
-      T* row = slab + (int64_t)ptPos[e] * 40;
-      for (int b = 0; b < 3; ++b)
-        for (int a = 0; a < 9; ++a) row[b * 9 + a] = blk[a][b];
-      row[27] = jp[0][0] * wjp[0][0] + jp[1][0] * wjp[1][0];  // hll 00
-      row[28] = jp[0][0] * wjp[0][1] + jp[1][0] * wjp[1][1];  // 01
-      row[29] = jp[0][0] * wjp[0][2] + jp[1][0] * wjp[1][2];  // 02
-      row[30] = jp[0][1] * wjp[0][1] + jp[1][1] * wjp[1][1];  // 11
-      row[31] = jp[0][1] * wjp[0][2] + jp[1][1] * wjp[1][2];  // 12
-      row[32] = jp[0][2] * wjp[0][2] + jp[1][2] * wjp[1][2];  // 22
-      for (int a = 0; a < 3; ++a)
-        row[33 + a] = -(jp[0][a] * wr[0] + jp[1][a] * wr[1]);  // g_p
-    } else {
-      const int pt = ptOf[e];
-      T* hll = Hll + (int64_t)pt * 9;
-      for (int a = 0; a < 3; ++a)
-        for (int b = 0; b < 3; ++b)
-          atomicAdd(&hll[a * 3 + b],
-                    jp[0][a] * wjp[0][b] + jp[1][a] * wjp[1][b]);
-      T* gpt = gp + (int64_t)pt * 3;
-      for (int a = 0; a < 3; ++a)
-        atomicAdd(&gpt[a], -(jp[0][a] * wr[0] + jp[1][a] * wr[1]));
+        for (int k = 0; k < 3; ++k) {
+          wjp[0][k] = w00 * jp[0][k] + w01 * jp[1][k];
+          wjp[1][k] = w01 * jp[0][k] + w11 * jp[1][k];
+        }
+        wr[0] = w00 * rr[0] + w01 * rr[1];
+        wr[1] = w01 * rr[0] + w11 * rr[1];
+      } else {
+        for (int k = 0; k < 3; ++k) {
+          wjp[0][k] = jp[0][k];
+          wjp[1][k] = jp[1][k];
+        }
+        wr[0] = rr[0];
+        wr[1] = rr[1];
+      }
+      T* row = slab + (int64_t)camPos[e] * L::SW;
+      if (EXPL) {
+        for (int a = 0; a < 9; ++a)
+          for (int b = 0; b < 3; ++b) {
+            const T v = jc[0][a] * wjp[0][b] + jc[1][a] * wjp[1][b];
+            Hpl[((int64_t)(a * 3 + b)) * nL + e] = v;
+            row[a * 3 + b] = v;
+          }
+      }
+      for (int k = 0; k < 9; ++k) {
+        row[L::JCOFF + k * 2] = jc[0][k];
+        row[L::JCOFF + k * 2 + 1] = jc[1][k];
+      }
+      row[L::WROFF] = wr[0];
+      row[L::WROFF + 1] = wr[1];
+      if (HASINFO)
+        for (int k = 0; k < 9; ++k) {
+          row[L::WJCOFF + k * 2] = wjc0[k];
+          row[L::WJCOFF + k * 2 + 1] = wjc1[k];
+        }
+      // point-side contributions
+      hll6[0] = jp[0][0] * wjp[0][0] + jp[1][0] * wjp[1][0];
+      hll6[1] = jp[0][0] * wjp[0][1] + jp[1][0] * wjp[1][1];
+      hll6[2] = jp[0][0] * wjp[0][2] + jp[1][0] * wjp[1][2];
+      hll6[3] = jp[0][1] * wjp[0][1] + jp[1][1] * wjp[1][1];
+      hll6[4] = jp[0][1] * wjp[0][2] + jp[1][1] * wjp[1][2];
+      hll6[5] = jp[0][2] * wjp[0][2] + jp[1][2] * wjp[1][2];
+      for (int k = 0; k < 3; ++k)
+        gp3[k] = -(jp[0][k] * wr[0] + jp[1][k] * wr[1]);
+    }
+    // segmented scan over the wave's point runs
+    for (int off = 1; off < 64; off <<= 1) {
+      const int ppt = __shfl_up(pt, off, 64);
+      T a6[6], a3[3];
+      for (int k = 0; k < 6; ++k) a6[k] = __shfl_up(hll6[k], off, 64);
+      for (int k = 0; k < 3; ++k) a3[k] = __shfl_up(gp3[k], off, 64);
+      if (lane >= off && ppt == pt) {
+        for (int k = 0; k < 6; ++k) hll6[k] += a6[k];
+        for (int k = 0; k < 3; ++k) gp3[k] += a3[k];
+      }
+    }
+    const int nextPt = __shfl_down(pt, 1, 64);
+    const bool tail = active && (lane == 63 || nextPt != pt || e0i == nL - 1);
+    if (tail) {
+      T* H = Hll + (int64_t)pt * 9;
+      atomicAdd(&H[0], hll6[0]);
+      atomicAdd(&H[1], hll6[1]);
+      atomicAdd(&H[2], hll6[2]);
+      atomicAdd(&H[3], hll6[1]);
+      atomicAdd(&H[4], hll6[3]);
+      atomicAdd(&H[5], hll6[4]);
+      atomicAdd(&H[6], hll6[2]);
+      atomicAdd(&H[7], hll6[4]);
+      atomicAdd(&H[8], hll6[5]);
+      for (int k = 0; k < 3; ++k) atomicAdd(&gp[(int64_t)pt * 3 + k], gp3[k]);
     }
   }
 }
 
-// Explicit-mode finalize: one thread per point streams its contiguous slab
-// rows (point-sorted), writing Hlp grad-major, the Hll block and g_p
-// directly -- no atomics, no transpose gather.
-template <typename T>
-__global__ void kFinalizePt(int npt, const int* __restrict__ ptRowPtr,
-                            const T* __restrict__ slab, int64_t nL,
-                            T* __restrict__ Hlp, T* __restrict__ Hll,
-                            T* __restrict__ gp) {
-  for (int64_t p = blockIdx.x * (int64_t)kBlk + threadIdx.x; p < npt;
-       p += (int64_t)gridDim.x * kBlk) {
-    const int lo = ptRowPtr[p], hi = ptRowPtr[p + 1];
-    T hll[6] = {0, 0, 0, 0, 0, 0};
-    T g3[3] = {0, 0, 0};
-    for (int j = lo; j < hi; ++j) {
-      const T* row = slab + (int64_t)j * 40;
-      for (int k = 0; k < 27; ++k) Hlp[(int64_t)k * nL + j] = row[k];
-      for (int k = 0; k < 6; ++k) hll[k] += row[27 + k];
-      for (int k = 0; k < 3; ++k) g3[k] += row[33 + k];
-    }
-    T* H = Hll + (int64_t)p * 9;
-    H[0] = hll[0];
-    H[1] = hll[1];
-    H[2] = hll[2];
-    H[3] = hll[1];
-    H[4] = hll[3];
-    H[5] = hll[4];
-    H[6] = hll[2];
-    H[7] = hll[4];
-    H[8] = hll[5];
-    for (int k = 0; k < 3; ++k) gp[3 * p + k] = g3[k];
-  }
-}
-
-// Camera part: one 128-thread block per <=256-edge chunk of one camera's
-// (cam,pt)-sorted edge run.  J rows (+ weighted rows) are staged through LDS
-// in 128-edge tiles; threads 0..80 each own one Hpp element, 81..89 one g_c
-// element; one atomicAdd per output per chunk.
-template <typename T, bool HASINFO>
+// Camera blocks: one 128-thread block per <=256-row chunk of one camera's
+// cam-sorted slab rows.  Rows are contiguous, staged through LDS in 128-row
+// tiles; threads 0..80 each own one Hpp element, 81..89 one g_c element;
+// one atomicAdd per output per chunk.
+template <typename T, bool HASINFO, bool EXPL>
 __global__ __launch_bounds__(128) void kAssembleCam(
     int nChunks, const int* __restrict__ chCam, const int* __restrict__ chLo,
-    const int* __restrict__ chHi, int64_t nL, const T* __restrict__ r,
-    const T* __restrict__ Jc, const T* __restrict__ info, T* __restrict__ Hpp,
-    T* __restrict__ g) {
-  constexpr int TE = 128;                    // edges per LDS tile
-  constexpr int ST = HASINFO ? 38 : 20;      // doubles per edge in LDS
+    const int* __restrict__ chHi, const T* __restrict__ slab,
+    T* __restrict__ Hpp, T* __restrict__ g) {
+  using L = SlabLayout<EXPL, HASINFO>;
+  constexpr int TE = 128;
+  constexpr int ST = HASINFO ? 38 : 20;  // jc 18 + wr 2 (+ wjc 18)
   __shared__ T lds[TE * ST];
   const int chunk = blockIdx.x;
   if (chunk >= nChunks) return;
@@ -406,55 +407,26 @@ __global__ __launch_bounds__(128) void kAssembleCam(
   const int lo = chLo[chunk], hi = chHi[chunk];
   const int t = threadIdx.x;
   T acc = T(0);
-  const int ti = t < 81 ? t / 9 : t - 81;    // Hpp row / g index
-  const int tj = t < 81 ? t % 9 : 0;         // Hpp col
+  const int ti = t < 81 ? t / 9 : t - 81;
+  const int tj = t < 81 ? t % 9 : 0;
   for (int s0 = lo; s0 < hi; s0 += TE) {
     const int nt = min(TE, hi - s0);
-    // Stage: thread t loads element (k) of edges t%... simple: loop.
     for (int idx = t; idx < nt * ST; idx += 128) {
-      const int e = idx / ST;
+      const int row = idx / ST;
       const int k = idx % ST;
-      const int64_t ge = s0 + e;
-      T v;
-      if (k < 18) {
-        v = Jc[((int64_t)k) * nL + ge];            // jc[row=k%2? no: k=(col*2+row)]
-      } else if (k < 20) {
-        v = r[(int64_t)(k - 18) * nL + ge];
-      } else {
-        v = T(0);  // filled below for HASINFO
-      }
-      lds[e * ST + k] = v;
+      const T* src = slab + (int64_t)(s0 + row) * L::SW + L::JCOFF;
+      lds[row * ST + k] = src[k];
     }
     __syncthreads();
-    if (HASINFO) {
-      // weighted rows: wjc[col][row] and wr, computed once per edge.
-      for (int e = t; e < nt; e += 128) {
-        const int64_t ge = s0 + e;
-        const T w00 = info[3 * ge], w01 = info[3 * ge + 1],
-                w11 = info[3 * ge + 2];
-        T* row = lds + e * ST;
-        for (int col = 0; col < 9; ++col) {
-          const T a0 = row[col * 2], a1 = row[col * 2 + 1];
-          row[20 + col * 2] = w00 * a0 + w01 * a1;
-          row[20 + col * 2 + 1] = w01 * a0 + w11 * a1;
-        }
-        const T r0 = row[18], r1 = row[19];
-        row[18] = w00 * r0 + w01 * r1;   // wr replaces r (safe: r only
-        row[19] = w01 * r0 + w11 * r1;   // used via wr below)
-      }
-      __syncthreads();
-    }
     if (t < 90) {
-      const int woff = HASINFO ? 20 : 0;  // weighted rows offset (== raw if no info)
+      const int woff = HASINFO ? 20 : 0;  // weighted rows (== raw if no info)
       for (int e = 0; e < nt; ++e) {
         const T* row = lds + e * ST;
-        if (t < 81) {
-          // Hpp[i][j] += sum_rows jc[row][i] * wjc[row][j]
+        if (t < 81)
           acc += row[ti * 2] * row[woff + tj * 2] +
                  row[ti * 2 + 1] * row[woff + tj * 2 + 1];
-        } else {
+        else
           acc -= row[ti * 2] * row[18] + row[ti * 2 + 1] * row[19];
-        }
       }
     }
     __syncthreads();
@@ -463,6 +435,19 @@ __global__ __launch_bounds__(128) void kAssembleCam(
     atomicAdd(&Hpp[(int64_t)cam * 81 + ti * 9 + tj], acc);
   else if (t < 90)
     atomicAdd(&g[(int64_t)cam * 9 + ti], acc);
+}
+
+// Explicit only: stream the slab's Hpl blocks out as a cam-sorted grad-major
+// copy for the E*w kernel (contiguous reads, coalesced writes).
+template <typename T, bool HASINFO>
+__global__ void kFinalizeCam(int64_t nL, const T* __restrict__ slab,
+                             T* __restrict__ HplCam) {
+  using L = SlabLayout<true, HASINFO>;
+  for (int64_t j = blockIdx.x * (int64_t)kBlk + threadIdx.x; j < nL;
+       j += (int64_t)gridDim.x * kBlk) {
+    const T* row = slab + j * L::SW;
+    for (int k = 0; k < 27; ++k) HplCam[(int64_t)k * nL + j] = row[k];
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -496,10 +481,9 @@ __global__ void kInvert(int nBlk, const T* __restrict__ Hd, T* __restrict__ Hinv
   }
 }
 
-// Rare path: re-run EVERY block, retrying semi-definite ones with a growing
-// relative diagonal jitter (matches the CPU oracle).  Launched only when
-// kInvert reported a failure, so its register/scratch cost is off the hot
-// path.
+// Rare path: retry semi-definite blocks with a growing relative diagonal
+// jitter (matches the CPU oracle); launched only if kInvert reported failure
+// so its register/scratch cost stays off the hot path.
 template <typename T, int D>
 __global__ void kInvertJitter(int nBlk, const T* __restrict__ Hd,
                               T* __restrict__ Hinv, int* fail) {
@@ -529,52 +513,53 @@ __global__ void kInvertJitter(int nBlk, const T* __restrict__ Hd,
 // ---------------------------------------------------------------------------
 // Schur SpMV pieces
 // ---------------------------------------------------------------------------
-// Materialise Hlp = Hpl^T blocks in point-sorted order (grad-major), once
-// per assembly.  The gather by ptPerm is the one unavoidable random-access
-// pass; paying it here makes every PCG iteration's E^T x fully coalesced and
-// atomic-free (the per-edge atomic version measured 674 us/call on Venice-5M,
-// ~2.4x the E*w kernel, bound by 15M fp64 atomics).
-template <typename T>
-__global__ void kTransposeHlp(int64_t nL, const int* __restrict__ ptPerm,
-                              const T* __restrict__ Hpl, T* __restrict__ Hlp) {
-  for (int64_t j = blockIdx.x * (int64_t)kBlk + threadIdx.x; j < nL;
-       j += (int64_t)gridDim.x * kBlk) {
-    const int64_t e = ptPerm[j];
-    for (int a = 0; a < 9; ++a)
-      for (int b = 0; b < 3; ++b)
-        Hlp[((int64_t)(b * 9 + a)) * nL + j] =
-            Hpl[((int64_t)(a * 3 + b)) * nL + e];
-  }
-}
-
-// temp[3*pt] += Hlp_j x[9*cam_j]: one thread per point-sorted edge (fully
-// coalesced Hlp reads; x is L2-resident), wave-level segmented-scan over the
-// point runs so only segment tails touch memory (atomicAdd per tail instead
-// of 3 atomics per edge; a thread-per-point variant measured WORSE -- 27
-// read streams spread across lanes thrash L1).
-template <typename T>
-__global__ void kSpmvEtx(int64_t nL, const int* __restrict__ ptOfPt,
-                         const int* __restrict__ camOfPt,
-                         const T* __restrict__ Hlp,
+// temp[3*pt] = Hpl^T x per local point (fully local, no collective): one
+// thread per primary-order edge (coalesced grad-major Hpl or J reads; the
+// replicated camera vector x is L2-resident), wave segmented-scan over the
+// point runs, atomics only at run tails.  IMP: matrix-free from J
+// (reference C23).
+template <typename T, bool IMP, bool HASINFO>
+__global__ void kSpmvEtx(int64_t nL, const int* __restrict__ camOf,
+                         const int* __restrict__ ptOf,
+                         const T* __restrict__ Hpl, const T* __restrict__ Jc,
+                         const T* __restrict__ Jp, const T* __restrict__ info,
                          const T* __restrict__ x, T* __restrict__ out) {
   const int lane = threadIdx.x & 63;
-  for (int64_t j0 = blockIdx.x * (int64_t)kBlk + threadIdx.x;
-       j0 < ((nL + kBlk - 1) / kBlk) * (int64_t)kBlk;
+  const int64_t nWork = ((nL + kBlk - 1) / kBlk) * (int64_t)kBlk;
+  for (int64_t j0 = blockIdx.x * (int64_t)kBlk + threadIdx.x; j0 < nWork;
        j0 += (int64_t)gridDim.x * kBlk) {
     const bool active = j0 < nL;
     const int64_t j = active ? j0 : nL - 1;
-    const int pt = ptOfPt[j];
+    const int pt = ptOf[j];
     T o0 = 0, o1 = 0, o2 = 0;
     if (active) {
-      const T* xc = x + (int64_t)camOfPt[j] * 9;
-      for (int i = 0; i < 9; ++i) {
-        const T xi = xc[i];
-        o0 += Hlp[((int64_t)(0 * 9 + i)) * nL + j] * xi;
-        o1 += Hlp[((int64_t)(1 * 9 + i)) * nL + j] * xi;
-        o2 += Hlp[((int64_t)(2 * 9 + i)) * nL + j] * xi;
+      const T* xc = x + (int64_t)camOf[j] * 9;
+      if (IMP) {
+        T u0 = T(0), u1 = T(0);
+        for (int i = 0; i < 9; ++i) {
+          const T xi = xc[i];
+          u0 += Jc[((int64_t)(i * 2 + 0)) * nL + j] * xi;
+          u1 += Jc[((int64_t)(i * 2 + 1)) * nL + j] * xi;
+        }
+        if (HASINFO) {
+          const T w00 = info[3 * j], w01 = info[3 * j + 1],
+                  w11 = info[3 * j + 2];
+          const T a = w00 * u0 + w01 * u1;
+          u1 = w01 * u0 + w11 * u1;
+          u0 = a;
+        }
+        o0 = Jp[((int64_t)0) * nL + j] * u0 + Jp[((int64_t)1) * nL + j] * u1;
+        o1 = Jp[((int64_t)2) * nL + j] * u0 + Jp[((int64_t)3) * nL + j] * u1;
+        o2 = Jp[((int64_t)4) * nL + j] * u0 + Jp[((int64_t)5) * nL + j] * u1;
+      } else {
+        for (int i = 0; i < 9; ++i) {
+          const T xi = xc[i];
+          o0 += Hpl[((int64_t)(i * 3 + 0)) * nL + j] * xi;
+          o1 += Hpl[((int64_t)(i * 3 + 1)) * nL + j] * xi;
+          o2 += Hpl[((int64_t)(i * 3 + 2)) * nL + j] * xi;
+        }
       }
     }
-    // inclusive segmented scan over the wave (segments = equal pt id)
     for (int off = 1; off < 64; off <<= 1) {
       const int ppt = __shfl_up(pt, off, 64);
       const T a0 = __shfl_up(o0, off, 64);
@@ -586,8 +571,8 @@ __global__ void kSpmvEtx(int64_t nL, const int* __restrict__ ptOfPt,
         o2 += a2;
       }
     }
-    const int npt_next = __shfl_down(pt, 1, 64);
-    const bool tail = active && (lane == 63 || npt_next != pt || j0 == nL - 1);
+    const int nextPt = __shfl_down(pt, 1, 64);
+    const bool tail = active && (lane == 63 || nextPt != pt || j0 == nL - 1);
     if (tail) {
       atomicAdd(&out[3 * pt], o0);
       atomicAdd(&out[3 * pt + 1], o1);
@@ -596,31 +581,29 @@ __global__ void kSpmvEtx(int64_t nL, const int* __restrict__ ptOfPt,
   }
 }
 
-// out[9*cam] += Hpl_e w[3*pt]: one wave per chunk of one camera's block-row;
-// per-lane partials, wave-wide shuffle reduce, one atomicAdd set per chunk.
+// out[9*cam] += E w partials: one wave per <=256-row chunk of one camera's
+// cam-sorted rows; per-lane partials, wave-wide shuffle reduce, one
+// atomicAdd set per chunk.  Caller allreduces 9*ncam (the ONLY per-PCG-
+// iteration collective).
 template <typename T>
-__global__ __launch_bounds__(64) void kSpmvEx(int nChunks,
-                                              const int* __restrict__ chCam,
-                                              const int* __restrict__ chLo,
-                                              const int* __restrict__ chHi,
-                                              const int* __restrict__ ptOf,
-                                              const T* __restrict__ Hpl,
-                                              int64_t nL,
-                                              const T* __restrict__ w,
-                                              T* __restrict__ out) {
+__global__ __launch_bounds__(64) void kSpmvEx(
+    int nChunks, const int* __restrict__ chCam, const int* __restrict__ chLo,
+    const int* __restrict__ chHi, const int* __restrict__ ptOfCam,
+    const T* __restrict__ HplCam, int64_t nL, const T* __restrict__ w,
+    T* __restrict__ out) {
   const int chunk = blockIdx.x;
   if (chunk >= nChunks) return;
   const int cam = chCam[chunk];
   T acc[9];
   for (int i = 0; i < 9; ++i) acc[i] = T(0);
   const int lo = chLo[chunk], hi = chHi[chunk];
-  for (int e = lo + (int)threadIdx.x; e < hi; e += 64) {
-    const T* wp = w + (int64_t)ptOf[e] * 3;
+  for (int j = lo + (int)threadIdx.x; j < hi; j += 64) {
+    const T* wp = w + (int64_t)ptOfCam[j] * 3;
     const T w0 = wp[0], w1 = wp[1], w2 = wp[2];
     for (int i = 0; i < 9; ++i)
-      acc[i] += Hpl[((int64_t)(i * 3 + 0)) * nL + e] * w0 +
-                Hpl[((int64_t)(i * 3 + 1)) * nL + e] * w1 +
-                Hpl[((int64_t)(i * 3 + 2)) * nL + e] * w2;
+      acc[i] += HplCam[((int64_t)(i * 3 + 0)) * nL + j] * w0 +
+                HplCam[((int64_t)(i * 3 + 1)) * nL + j] * w1 +
+                HplCam[((int64_t)(i * 3 + 2)) * nL + j] * w2;
   }
   for (int off = 32; off > 0; off >>= 1)
     for (int i = 0; i < 9; ++i) acc[i] += __shfl_down(acc[i], off, 64);
@@ -630,52 +613,17 @@ __global__ __launch_bounds__(64) void kSpmvEx(int nChunks,
   }
 }
 
-// Implicit (matrix-free) Schur products from the accepted Jacobians
-// (reference C23, implicit_schur_pcg_solver.cu:21-90): E^T x = Jp^T W (Jc x)
-// per edge with atomic scatter (point degree is low); E w = Jc^T W (Jp w)
-// chunked per camera block-row with a wave reduction.
+// Implicit E w: matrix-free from J in primary order (per-edge atomics into
+// the small replicated camera vector; the memory-saving mode trades the
+// materialised Hpl for this recomputation, reference C23).
 template <typename T, bool HASINFO>
-__global__ void kSpmvEtxImp(int64_t nL, const int* __restrict__ camOf,
-                            const int* __restrict__ ptOf,
-                            const T* __restrict__ Jc, const T* __restrict__ Jp,
-                            const T* __restrict__ info,
-                            const T* __restrict__ x, T* __restrict__ out) {
+__global__ void kSpmvExImp(int64_t nL, const int* __restrict__ camOf,
+                           const int* __restrict__ ptOf,
+                           const T* __restrict__ Jc, const T* __restrict__ Jp,
+                           const T* __restrict__ info, const T* __restrict__ w,
+                           T* __restrict__ out) {
   for (int64_t e = blockIdx.x * (int64_t)kBlk + threadIdx.x; e < nL;
        e += (int64_t)gridDim.x * kBlk) {
-    const T* xc = x + (int64_t)camOf[e] * 9;
-    T u0 = T(0), u1 = T(0);
-    for (int i = 0; i < 9; ++i) {
-      const T xi = xc[i];
-      u0 += Jc[((int64_t)(i * 2 + 0)) * nL + e] * xi;
-      u1 += Jc[((int64_t)(i * 2 + 1)) * nL + e] * xi;
-    }
-    if (HASINFO) {
-      const T w00 = info[3 * e], w01 = info[3 * e + 1], w11 = info[3 * e + 2];
-      const T a = w00 * u0 + w01 * u1;
-      u1 = w01 * u0 + w11 * u1;
-      u0 = a;
-    }
-    T* op = out + (int64_t)ptOf[e] * 3;
-    for (int j = 0; j < 3; ++j)
-      atomicAdd(&op[j], Jp[((int64_t)(j * 2 + 0)) * nL + e] * u0 +
-                            Jp[((int64_t)(j * 2 + 1)) * nL + e] * u1);
-  }
-}
-
-template <typename T, bool HASINFO>
-__global__ __launch_bounds__(64) void kSpmvExImp(
-    int nChunks, const int* __restrict__ chCam, const int* __restrict__ chLo,
-    const int* __restrict__ chHi, const int* __restrict__ ptOf,
-    const T* __restrict__ Jc, const T* __restrict__ Jp,
-    const T* __restrict__ info, int64_t nL, const T* __restrict__ w,
-    T* __restrict__ out) {
-  const int chunk = blockIdx.x;
-  if (chunk >= nChunks) return;
-  const int cam = chCam[chunk];
-  T acc[9];
-  for (int i = 0; i < 9; ++i) acc[i] = T(0);
-  const int lo = chLo[chunk], hi = chHi[chunk];
-  for (int e = lo + (int)threadIdx.x; e < hi; e += 64) {
     const T* wp = w + (int64_t)ptOf[e] * 3;
     T u0 = T(0), u1 = T(0);
     for (int j = 0; j < 3; ++j) {
@@ -688,15 +636,10 @@ __global__ __launch_bounds__(64) void kSpmvExImp(
       u1 = w01 * u0 + w11 * u1;
       u0 = a;
     }
+    T* oc = out + (int64_t)camOf[e] * 9;
     for (int i = 0; i < 9; ++i)
-      acc[i] += Jc[((int64_t)(i * 2 + 0)) * nL + e] * u0 +
-                Jc[((int64_t)(i * 2 + 1)) * nL + e] * u1;
-  }
-  for (int off = 32; off > 0; off >>= 1)
-    for (int i = 0; i < 9; ++i) acc[i] += __shfl_down(acc[i], off, 64);
-  if (threadIdx.x == 0) {
-    T* oc = out + (int64_t)cam * 9;
-    for (int i = 0; i < 9; ++i) atomicAdd(&oc[i], acc[i]);
+      atomicAdd(&oc[i], Jc[((int64_t)(i * 2 + 0)) * nL + e] * u0 +
+                            Jc[((int64_t)(i * 2 + 1)) * nL + e] * u1);
   }
 }
 
@@ -720,9 +663,9 @@ __global__ void kBlockDiagMatVec(int nBlk, const T* __restrict__ A,
 // ---------------------------------------------------------------------------
 // Small vector kernels
 // ---------------------------------------------------------------------------
-// y += sign * (*a) * x, with the scalar produced on-device (removes the
-// p^T q host readback from the PCG loop; rho alone is read back per
-// iteration, needed for the reference's refuse/tol control flow).
+// y += sign * (*a) * x, scalar produced on-device (removes the p^T q host
+// readback from the PCG loop; rho alone is read back per iteration, needed
+// for the reference's refuse/tol control flow).
 template <typename T, int SIGN>
 __global__ void kAxpyS(int64_t n, const double* __restrict__ a,
                        const T* __restrict__ x, T* __restrict__ y) {
@@ -735,12 +678,6 @@ __global__ void kDivScalar(double* out, const double* num, const double* den) {
   *out = *num / *den;
 }
 
-template <typename T>
-__global__ void kAxpy(int64_t n, T a, const T* __restrict__ x, T* __restrict__ y) {
-  for (int64_t i = blockIdx.x * (int64_t)kBlk + threadIdx.x; i < n;
-       i += (int64_t)gridDim.x * kBlk)
-    y[i] += a * x[i];
-}
 template <typename T>
 __global__ void kXpby(int64_t n, const T* __restrict__ x, T b, T* __restrict__ y) {
   for (int64_t i = blockIdx.x * (int64_t)kBlk + threadIdx.x; i < n;
@@ -768,7 +705,13 @@ __global__ void kAddAssign(int64_t n, const T* __restrict__ x, T* __restrict__ y
        i += (int64_t)gridDim.x * kBlk)
     y[i] += x[i];
 }
-// deltaX_p = HllInv * (g_p - temp)
+template <typename T>
+__global__ void kZeroRange(T* p, int64_t n) {
+  for (int64_t i = blockIdx.x * (int64_t)kBlk + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * kBlk)
+    p[i] = T(0);
+}
+// deltaX_p = HllInv * (g_p - temp)  (local point shard)
 template <typename T>
 __global__ void kBackSub(int npt, const T* __restrict__ HllInv,
                          const T* __restrict__ gp, const T* __restrict__ temp,
@@ -776,12 +719,11 @@ __global__ void kBackSub(int npt, const T* __restrict__ HllInv,
   for (int64_t p = blockIdx.x * (int64_t)kBlk + threadIdx.x; p < npt;
        p += (int64_t)gridDim.x * kBlk) {
     const T* inv = HllInv + p * 9;
-    T rhs[3], o[3];
+    T rhs[3];
     for (int i = 0; i < 3; ++i) rhs[i] = gp[3 * p + i] - temp[3 * p + i];
-    for (int i = 0; i < 3; ++i) {
-      o[i] = inv[i * 3] * rhs[0] + inv[i * 3 + 1] * rhs[1] + inv[i * 3 + 2] * rhs[2];
-    }
-    for (int i = 0; i < 3; ++i) dxp[3 * p + i] = o[i];
+    for (int i = 0; i < 3; ++i)
+      dxp[3 * p + i] =
+          inv[i * 3] * rhs[0] + inv[i * 3 + 1] * rhs[1] + inv[i * 3 + 2] * rhs[2];
   }
 }
 
@@ -830,7 +772,10 @@ class GpuEngine final : public Engine<T> {
             const ProblemOption& opt, const std::string& rcclId,
             CustomForward<T> customForward)
       : customFwd_(std::move(customForward)),
-        rank_(opt.rank), world_(opt.worldSize), ncam_(ix.ncam), npt_(ix.npt),
+        rank_(opt.rank),
+        world_(opt.worldSize),
+        ncam_(ix.ncam),
+        npt_(ix.npt),
         analytical_(opt.diff == DiffMode::ANALYTICAL),
         implicit_(opt.schur == SchurMode::IMPLICIT) {
     HIP_CHECK(hipSetDevice(opt.deviceIndex));
@@ -838,6 +783,9 @@ class GpuEngine final : public Engine<T> {
     e0_ = ix.split[rank_];
     e1_ = ix.split[rank_ + 1];
     nL_ = e1_ - e0_;
+    ptLo_ = ix.ptSplit[rank_];
+    ptHi_ = ix.ptSplit[rank_ + 1];
+    npL_ = ptHi_ - ptLo_;
     nc_ = (int64_t)ncam_ * 9;
     np_ = (int64_t)npt_ * 3;
     dim_ = nc_ + np_;
@@ -852,7 +800,7 @@ class GpuEngine final : public Engine<T> {
       hasComm_ = true;
     }
 
-    // Static per-edge data.
+    // Static per-edge data (primary = (pt,cam)-sorted order).
     dCamOf_ = dalloc<int>(nL_);
     dPtOf_ = dalloc<int>(nL_);
     up(dCamOf_, ix.camOf.data() + e0_, nL_);
@@ -868,7 +816,6 @@ class GpuEngine final : public Engine<T> {
     dMeas_ = dalloc<T>(nL_ * 2);
     upCast(dMeas_, ix.measSorted.data() + 2 * e0_, nL_ * 2);
     if (customFwd_) {
-      // row-split measurement copy + leaf slab for the user forward()
       dLeaf_ = dalloc<T>(nL_ * 12);
       dMeasSplit_ = dalloc<T>(nL_ * 2);
       std::vector<T> ms(nL_ * 2);
@@ -883,7 +830,7 @@ class GpuEngine final : public Engine<T> {
       upCast(dInfo_, ix.infoSorted.data() + 3 * e0_, nL_ * 3);
     }
 
-    // Parameters (replicated): [cams | pts].
+    // Parameters (cameras replicated; this rank maintains its point shard).
     dParams_ = dalloc<T>(dim_);
     dParamsBak_ = dalloc<T>(dim_);
     {
@@ -902,10 +849,10 @@ class GpuEngine final : public Engine<T> {
       dJp_[s] = dalloc<T>(nL_ * 6);
     }
 
-    // Linear system.
+    // Linear system (full-size global indexing; only this rank's point
+    // ranges of Hll/g_p/temp are maintained).
     dHpp_ = dalloc<T>((int64_t)ncam_ * 81);
     dHll_ = dalloc<T>((int64_t)npt_ * 9);
-    if (!implicit_) dHpl_ = dalloc<T>(nL_ * 27);
     dG_ = dalloc<T>(dim_);
     dGBak_ = dalloc<T>(dim_);
     dHppD_ = dalloc<T>((int64_t)ncam_ * 81);
@@ -917,7 +864,7 @@ class GpuEngine final : public Engine<T> {
     HIP_CHECK(hipMemsetAsync(dDeltaX_, 0, dim_ * sizeof(T), stream_));
     HIP_CHECK(hipMemsetAsync(dDeltaXBak_, 0, dim_ * sizeof(T), stream_));
 
-    // PCG workspace (camera-sized unless noted).
+    // PCG workspace.
     dP_ = dalloc<T>(nc_);
     dRr_ = dalloc<T>(nc_);
     dZ_ = dalloc<T>(nc_);
@@ -926,60 +873,46 @@ class GpuEngine final : public Engine<T> {
     dXBak_ = dalloc<T>(nc_);
     dW_ = dalloc<T>(np_);
     dTemp_ = dalloc<T>(np_);
+    HIP_CHECK(hipMemsetAsync(dW_, 0, np_ * sizeof(T), stream_));
 
-    dPart_ = dalloc<double>(kRedBlocks + 4);
+    dPart_ = dalloc<double>(kRedBlocks + 8);
     dFail_ = dalloc<int>(2);
 
-    // Chunk table for the E*w block-row kernel.
-    std::vector<int> cCam, cLo, cHi;
-    constexpr int CHUNK = 256;
-    for (int c = ix.camOf.empty() ? 0 : ix.camOf[e0_];
-         c < ncam_ && ix.camRowPtr[c] < e1_; ++c) {
-      const int64_t lo = std::max(ix.camRowPtr[c], e0_);
-      const int64_t hi = std::min(ix.camRowPtr[c + 1], e1_);
-      for (int64_t s = lo; s < hi; s += CHUNK) {
-        cCam.push_back(c);
-        cLo.push_back((int)(s - e0_));
-        cHi.push_back((int)(std::min<int64_t>(s + CHUNK, hi) - e0_));
-      }
-    }
-    nChunks_ = (int)cCam.size();
-    dChCam_ = dalloc<int>(nChunks_);
-    dChLo_ = dalloc<int>(nChunks_);
-    dChHi_ = dalloc<int>(nChunks_);
-    up(dChCam_, cCam.data(), nChunks_);
-    up(dChLo_, cLo.data(), nChunks_);
-    up(dChHi_, cHi.data(), nChunks_);
-
-    // Point-sorted view of the local edges (for the atomic-free E^T x).
+    // Cam-sorted view of the local edges: slab positions + chunk table.
     {
-      std::vector<int> perm(nL_), rowPtr(npt_ + 1, 0), camOfPt(nL_),
-          ptOfPt(nL_);
-      for (int64_t e = 0; e < nL_; ++e) rowPtr[ix.ptOf[e0_ + e] + 1]++;
-      for (int p = 0; p < npt_; ++p) rowPtr[p + 1] += rowPtr[p];
+      std::vector<int> camPos(nL_), rowPtr(ncam_ + 1, 0), ptOfCam(nL_);
+      for (int64_t e = 0; e < nL_; ++e) rowPtr[ix.camOf[e0_ + e] + 1]++;
+      for (int c = 0; c < ncam_; ++c) rowPtr[c + 1] += rowPtr[c];
       std::vector<int> cursor(rowPtr.begin(), rowPtr.end() - 1);
       for (int64_t e = 0; e < nL_; ++e) {
-        const int pos = cursor[ix.ptOf[e0_ + e]]++;
-        perm[pos] = (int)e;
-        camOfPt[pos] = ix.camOf[e0_ + e];
-        ptOfPt[pos] = ix.ptOf[e0_ + e];
+        const int pos = cursor[ix.camOf[e0_ + e]]++;
+        camPos[e] = pos;
+        ptOfCam[pos] = ix.ptOf[e0_ + e];
       }
-      std::vector<int> ptPos(nL_);
-      for (int64_t j = 0; j < nL_; ++j) ptPos[perm[j]] = (int)j;
-      dPtPos_ = dalloc<int>(nL_);
-      up(dPtPos_, ptPos.data(), nL_);
-      dPtPerm_ = dalloc<int>(nL_);
-      dPtRowPtr_ = dalloc<int>(npt_ + 1);
-      dCamOfPt_ = dalloc<int>(nL_);
-      dPtOfPt_ = dalloc<int>(nL_);
-      up(dPtPerm_, perm.data(), nL_);
-      up(dPtRowPtr_, rowPtr.data(), npt_ + 1);
-      up(dCamOfPt_, camOfPt.data(), nL_);
-      up(dPtOfPt_, ptOfPt.data(), nL_);
+      dCamPos_ = dalloc<int>(nL_);
+      dPtOfCam_ = dalloc<int>(nL_);
+      up(dCamPos_, camPos.data(), nL_);
+      up(dPtOfCam_, ptOfCam.data(), nL_);
+      std::vector<int> cCam, cLo, cHi;
+      constexpr int CHUNK = 256;
+      for (int c = 0; c < ncam_; ++c)
+        for (int s = rowPtr[c]; s < rowPtr[c + 1]; s += CHUNK) {
+          cCam.push_back(c);
+          cLo.push_back(s);
+          cHi.push_back(std::min(s + CHUNK, rowPtr[c + 1]));
+        }
+      nChunks_ = (int)cCam.size();
+      dChCam_ = dalloc<int>(nChunks_);
+      dChLo_ = dalloc<int>(nChunks_);
+      dChHi_ = dalloc<int>(nChunks_);
+      up(dChCam_, cCam.data(), nChunks_);
+      up(dChLo_, cLo.data(), nChunks_);
+      up(dChHi_, cHi.data(), nChunks_);
     }
+    dSlab_ = dalloc<T>(nL_ * slabWidth());
     if (!implicit_) {
-      dHlp_ = dalloc<T>(nL_ * 27);
-      dSlab_ = dalloc<T>(nL_ * 40);
+      dHpl_ = dalloc<T>(nL_ * 27);
+      dHplCam_ = dalloc<T>(nL_ * 27);
     }
     sync();
   }
@@ -1003,7 +936,7 @@ class GpuEngine final : public Engine<T> {
                          stream_, nL_, dCamOf_, dPtOf_, dParams_, ncam_,
                          dMeas_, dCamFixed_, dPtFixed_, dR_[cur_], dJc_[cur_],
                          dJp_[cur_], scalarPtr());
-    return globalScalar();
+    return globalScalar(ncclSum);
   }
 
   void acceptForward() override { cur_ ^= 1; }  // accepted set = dX_[cur_^1]
@@ -1033,52 +966,30 @@ class GpuEngine final : public Engine<T> {
                          dJc_[cur_], dJp_[cur_]);
     }
     reduceDetAsync(dR_[cur_], dR_[cur_], nL_ * 2, ROp::SumSq, scalarPtr());
-    return globalScalar();
+    return globalScalar(ncclSum);
   }
 
   void buildLinearSystem() override {
     const int bak = cur_ ^ 1;
     HIP_CHECK(hipMemsetAsync(dHpp_, 0, (int64_t)ncam_ * 81 * sizeof(T), stream_));
-    HIP_CHECK(hipMemsetAsync(dHll_, 0, (int64_t)npt_ * 9 * sizeof(T), stream_));
-    HIP_CHECK(hipMemsetAsync(dG_, 0, dim_ * sizeof(T), stream_));
-    if (hasInfo_) {
-      if (implicit_)
-        hipLaunchKernelGGL((kAssembleEdge<T, true, false>), dim3(gridFor(nL_)),
-                           dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
-                           dR_[bak], dJc_[bak], dJp_[bak], dInfo_, dHll_,
-                           dHpl_, dG_, ncam_, dPtPos_, dSlab_);
+    HIP_CHECK(hipMemsetAsync(dHll_ + (int64_t)ptLo_ * 9, 0,
+                             (int64_t)npL_ * 9 * sizeof(T), stream_));
+    HIP_CHECK(hipMemsetAsync(dG_, 0, nc_ * sizeof(T), stream_));
+    HIP_CHECK(hipMemsetAsync(dG_ + nc_ + (int64_t)ptLo_ * 3, 0,
+                             (int64_t)npL_ * 3 * sizeof(T), stream_));
+    dispatchAssemble(bak);
+    if (!implicit_) {
+      if (hasInfo_)
+        hipLaunchKernelGGL((kFinalizeCam<T, true>), dim3(gridFor(nL_)),
+                           dim3(kBlk), 0, stream_, nL_, dSlab_, dHplCam_);
       else
-        hipLaunchKernelGGL((kAssembleEdge<T, true, true>), dim3(gridFor(nL_)),
-                           dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
-                           dR_[bak], dJc_[bak], dJp_[bak], dInfo_, dHll_,
-                           dHpl_, dG_, ncam_, dPtPos_, dSlab_);
-      if (nChunks_ > 0)
-        hipLaunchKernelGGL((kAssembleCam<T, true>), dim3(nChunks_), dim3(128),
-                           0, stream_, nChunks_, dChCam_, dChLo_, dChHi_, nL_,
-                           dR_[bak], dJc_[bak], dInfo_, dHpp_, dG_);
-    } else {
-      if (implicit_)
-        hipLaunchKernelGGL((kAssembleEdge<T, false, false>),
-                           dim3(gridFor(nL_)), dim3(kBlk), 0, stream_, nL_,
-                           dCamOf_, dPtOf_, dR_[bak], dJc_[bak], dJp_[bak],
-                           nullptr, dHll_, dHpl_, dG_, ncam_, dPtPos_, dSlab_);
-      else
-        hipLaunchKernelGGL((kAssembleEdge<T, false, true>),
-                           dim3(gridFor(nL_)), dim3(kBlk), 0, stream_, nL_,
-                           dCamOf_, dPtOf_, dR_[bak], dJc_[bak], dJp_[bak],
-                           nullptr, dHll_, dHpl_, dG_, ncam_, dPtPos_, dSlab_);
-      if (nChunks_ > 0)
-        hipLaunchKernelGGL((kAssembleCam<T, false>), dim3(nChunks_), dim3(128),
-                           0, stream_, nChunks_, dChCam_, dChLo_, dChHi_, nL_,
-                           dR_[bak], dJc_[bak], nullptr, dHpp_, dG_);
+        hipLaunchKernelGGL((kFinalizeCam<T, false>), dim3(gridFor(nL_)),
+                           dim3(kBlk), 0, stream_, nL_, dSlab_, dHplCam_);
     }
-    if (!implicit_)
-      hipLaunchKernelGGL(kFinalizePt<T>, dim3(gridFor(npt_)), dim3(kBlk), 0,
-                         stream_, npt_, dPtRowPtr_, dSlab_, nL_, dHlp_, dHll_,
-                         dG_ + nc_);
-    allreduce(dHpp_, (int64_t)ncam_ * 81);
-    allreduce(dHll_, (int64_t)npt_ * 9);
-    allreduce(dG_, dim_);
+    // Only the small camera-side quantities cross ranks (the reference
+    // allreduced Hpp, Hll AND g, its site A1).
+    allreduce(dHpp_, (int64_t)ncam_ * 81, ncclSum);
+    allreduce(dG_, nc_, ncclSum);
     sync();
   }
 
@@ -1108,19 +1019,20 @@ class GpuEngine final : public Engine<T> {
     hipLaunchKernelGGL((kDamp<T, 9>), dim3(gridFor((int64_t)ncam_ * 81)),
                        dim3(kBlk), 0, stream_, (int64_t)ncam_ * 81, dHpp_,
                        dHppD_, f, dCamFixed_);
-    hipLaunchKernelGGL((kDamp<T, 3>), dim3(gridFor((int64_t)npt_ * 9)),
-                       dim3(kBlk), 0, stream_, (int64_t)npt_ * 9, dHll_, dHllD_,
-                       f, dPtFixed_);
+    hipLaunchKernelGGL((kDamp<T, 3>), dim3(gridFor((int64_t)npL_ * 9)),
+                       dim3(kBlk), 0, stream_, (int64_t)npL_ * 9,
+                       dHll_ + (int64_t)ptLo_ * 9, dHllD_ + (int64_t)ptLo_ * 9,
+                       f, dPtFixed_ ? dPtFixed_ + ptLo_ : nullptr);
   }
 
   int solveLinear(const SolverOptionPCG& opt) override {
-    // Block inverses (preconditioner + Cinv).  Fast path first; the jitter
-    // retry kernel runs only if some block was numerically semi-definite.
+    // Block inverses (preconditioner replicated; Cinv on the local shard).
     HIP_CHECK(hipMemsetAsync(dFail_, 0, 2 * sizeof(int), stream_));
     hipLaunchKernelGGL((kInvert<T, 9>), dim3(gridFor(ncam_)), dim3(kBlk), 0,
                        stream_, ncam_, dHppD_, dHppInv_, dFail_);
-    hipLaunchKernelGGL((kInvert<T, 3>), dim3(gridFor(npt_)), dim3(kBlk), 0,
-                       stream_, npt_, dHllD_, dHllInv_, dFail_);
+    hipLaunchKernelGGL((kInvert<T, 3>), dim3(gridFor(npL_)), dim3(kBlk), 0,
+                       stream_, npL_, dHllD_ + (int64_t)ptLo_ * 9,
+                       dHllInv_ + (int64_t)ptLo_ * 9, dFail_);
     int fail[2] = {0, 0};
     HIP_CHECK(hipMemcpyAsync(fail, dFail_, 2 * sizeof(int),
                              hipMemcpyDeviceToHost, stream_));
@@ -1129,9 +1041,10 @@ class GpuEngine final : public Engine<T> {
       hipLaunchKernelGGL((kInvertJitter<T, 9>), dim3(gridFor(ncam_)),
                          dim3(kBlk), 0, stream_, ncam_, dHppD_, dHppInv_,
                          dFail_);
-      hipLaunchKernelGGL((kInvertJitter<T, 3>), dim3(gridFor(npt_)),
-                         dim3(kBlk), 0, stream_, npt_, dHllD_, dHllInv_,
-                         dFail_);
+      hipLaunchKernelGGL((kInvertJitter<T, 3>), dim3(gridFor(npL_)),
+                         dim3(kBlk), 0, stream_, npL_,
+                         dHllD_ + (int64_t)ptLo_ * 9,
+                         dHllInv_ + (int64_t)ptLo_ * 9, dFail_);
       HIP_CHECK(hipMemcpyAsync(fail, dFail_, 2 * sizeof(int),
                                hipMemcpyDeviceToHost, stream_));
       sync();
@@ -1140,12 +1053,12 @@ class GpuEngine final : public Engine<T> {
 
     const T* gc = dG_;
     const T* gp = dG_ + nc_;
-    // v = gc/world - E Cinv gp
-    blockMatVec<3, 0>(npt_, dHllInv_, gp, dW_);
+    // v = gc/world - E Cinv gp  (partial, then the 9*ncam allreduce)
+    applyCinv(gp, dW_);
     spmvEx(dW_, dV_);
     hipLaunchKernelGGL(kVMake<T>, dim3(gridFor(nc_)), dim3(kBlk), 0, stream_,
                        nc_, gc, T(1) / T(world_), dV_);
-    allreduce(dV_, nc_);
+    allreduce(dV_, nc_, ncclSum);
     // Warm start: x = deltaX camera part (in place in dDeltaX_).
     T* x = dDeltaX_;
     schurApply(x, dQ_);
@@ -1186,26 +1099,46 @@ class GpuEngine final : public Engine<T> {
       ++n;
       done = std::abs(rho) < opt.tol;
     }
-    // Back-substitution: deltaX_p = Cinv (g_p - E^T x).
+    // Back-substitution: deltaX_p = Cinv (g_p - E^T x), fully local.
     spmvEtx(x, dTemp_);
-    allreduce(dTemp_, np_);
-    hipLaunchKernelGGL(kBackSub<T>, dim3(gridFor(npt_)), dim3(kBlk), 0, stream_,
-                       npt_, dHllInv_, gp, dTemp_, dDeltaX_ + nc_);
+    hipLaunchKernelGGL(kBackSub<T>, dim3(gridFor(npL_)), dim3(kBlk), 0, stream_,
+                       npL_, dHllInv_ + (int64_t)ptLo_ * 9,
+                       gp + (int64_t)ptLo_ * 3, dTemp_ + (int64_t)ptLo_ * 3,
+                       dDeltaX_ + nc_ + (int64_t)ptLo_ * 3);
     sync();
     return n;
   }
 
   double deltaXL2() override {
-    return std::sqrt(reduceDet(dDeltaX_, dDeltaX_, dim_, ROp::SumSq));
+    const double camSS = reduceDet(dDeltaX_, dDeltaX_, nc_, ROp::SumSq);
+    reduceDetAsync(dDeltaX_ + nc_ + (int64_t)ptLo_ * 3,
+                   dDeltaX_ + nc_ + (int64_t)ptLo_ * 3, (int64_t)npL_ * 3,
+                   ROp::SumSq, scalarPtr());
+    return std::sqrt(camSS + globalScalar(ncclSum));
   }
   double xL2() override {
-    return std::sqrt(reduceDet(dParams_, dParams_, dim_, ROp::SumSq));
+    const double camSS = reduceDet(dParams_, dParams_, nc_, ROp::SumSq);
+    reduceDetAsync(dParams_ + nc_ + (int64_t)ptLo_ * 3,
+                   dParams_ + nc_ + (int64_t)ptLo_ * 3, (int64_t)npL_ * 3,
+                   ROp::SumSq, scalarPtr());
+    return std::sqrt(camSS + globalScalar(ncclSum));
   }
-  double gInf() override { return reduceDet(dG_, dG_, dim_, ROp::AbsMax); }
+  double gInf() override {
+    const double camMax = reduceDet(dG_, dG_, nc_, ROp::AbsMax);
+    reduceDetAsync(dG_ + nc_ + (int64_t)ptLo_ * 3,
+                   dG_ + nc_ + (int64_t)ptLo_ * 3, (int64_t)npL_ * 3,
+                   ROp::AbsMax, scalarPtr());
+    const double ptMax = globalScalar(ncclMax);
+    return camMax > ptMax ? camMax : ptMax;
+  }
 
   void updateParams() override {
-    hipLaunchKernelGGL(kAddAssign<T>, dim3(gridFor(dim_)), dim3(kBlk), 0,
-                       stream_, dim_, dDeltaX_, dParams_);
+    hipLaunchKernelGGL(kAddAssign<T>, dim3(gridFor(nc_)), dim3(kBlk), 0,
+                       stream_, nc_, dDeltaX_, dParams_);
+    hipLaunchKernelGGL(kAddAssign<T>, dim3(gridFor((int64_t)npL_ * 3)),
+                       dim3(kBlk), 0, stream_, (int64_t)npL_ * 3,
+                       dDeltaX_ + nc_ + (int64_t)ptLo_ * 3,
+                       dParams_ + nc_ + (int64_t)ptLo_ * 3);
   }
 
   double rhoDenominator(double chi2Backup) override {
@@ -1214,15 +1147,33 @@ class GpuEngine final : public Engine<T> {
     hipLaunchKernelGGL(kRhoDenom<T>, dim3(gridFor(nL_)), dim3(kBlk), 0, stream_,
                        nL_, dCamOf_, dPtOf_, dR_[bak], dJc_[bak], dJp_[bak],
                        dDeltaX_, dDeltaX_ + nc_, scalarPtr());
-    return globalScalar() - chi2Backup;
+    return globalScalar(ncclSum) - chi2Backup;
   }
 
-  void getParams(double* cams, double* pts) const override {
-    std::vector<T> h(dim_);
-    HIP_CHECK(hipMemcpy(h.data(), dParams_, dim_ * sizeof(T),
+  void getParams(double* cams, double* pts) override {
+    std::vector<T> hc(nc_);
+    HIP_CHECK(hipMemcpy(hc.data(), dParams_, nc_ * sizeof(T),
                         hipMemcpyDeviceToHost));
-    for (int64_t i = 0; i < nc_; ++i) cams[i] = (double)h[i];
-    for (int64_t i = 0; i < np_; ++i) pts[i] = (double)h[nc_ + i];
+    for (int64_t i = 0; i < nc_; ++i) cams[i] = (double)hc[i];
+    if (hasComm_) {
+      // point shards: zero the non-local entries, allreduce-sum.
+      T* tmp = dalloc<T>(np_);
+      HIP_CHECK(hipMemsetAsync(tmp, 0, np_ * sizeof(T), stream_));
+      HIP_CHECK(hipMemcpyAsync(tmp + (int64_t)ptLo_ * 3,
+                               dParams_ + nc_ + (int64_t)ptLo_ * 3,
+                               (int64_t)npL_ * 3 * sizeof(T),
+                               hipMemcpyDeviceToDevice, stream_));
+      allreduce(tmp, np_, ncclSum);
+      sync();
+      std::vector<T> h(np_);
+      HIP_CHECK(hipMemcpy(h.data(), tmp, np_ * sizeof(T), hipMemcpyDeviceToHost));
+      for (int64_t i = 0; i < np_; ++i) pts[i] = (double)h[i];
+      return;
+    }
+    std::vector<T> h(np_);
+    HIP_CHECK(hipMemcpy(h.data(), dParams_ + nc_, np_ * sizeof(T),
+                        hipMemcpyDeviceToHost));
+    for (int64_t i = 0; i < np_; ++i) pts[i] = (double)h[i];
   }
 
   DenseDump dump() const override {
@@ -1253,6 +1204,29 @@ class GpuEngine final : public Engine<T> {
   }
 
  private:
+  int slabWidth() const {
+    return (implicit_ ? 0 : 27) + 20 + (hasInfo_ ? 18 : 0);
+  }
+  void dispatchAssemble(int bak) {
+    auto launch = [&](auto hasInfoTag, auto explTag) {
+      constexpr bool HI = decltype(hasInfoTag)::value;
+      constexpr bool EX = decltype(explTag)::value;
+      hipLaunchKernelGGL((kAssembleEdge<T, HI, EX>), dim3(gridFor(nL_)),
+                         dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
+                         dR_[bak], dJc_[bak], dJp_[bak], dInfo_, dHll_, dHpl_,
+                         dG_, ncam_, dCamPos_, dSlab_);
+      if (nChunks_ > 0)
+        hipLaunchKernelGGL((kAssembleCam<T, HI, EX>), dim3(nChunks_), dim3(128),
+                           0, stream_, nChunks_, dChCam_, dChLo_, dChHi_,
+                           dSlab_, dHpp_, dG_);
+    };
+    using TrueT = std::integral_constant<bool, true>;
+    using FalseT = std::integral_constant<bool, false>;
+    if (hasInfo_ && !implicit_) launch(TrueT{}, TrueT{});
+    else if (hasInfo_ && implicit_) launch(TrueT{}, FalseT{});
+    else if (!hasInfo_ && !implicit_) launch(FalseT{}, TrueT{});
+    else launch(FalseT{}, FalseT{});
+  }
   template <typename U>
   U* dalloc(int64_t n) {
     void* p = nullptr;
@@ -1284,7 +1258,6 @@ class GpuEngine final : public Engine<T> {
     return o;
   }
   std::vector<double> transposeJ(const std::vector<double>& v, int w) const {
-    // device: [(col*2+row)*nL + e] -> host: [e][2][w/2... ] w=9: [e*18+row*9+col]
     const int cols = w == 9 ? 9 : 3;
     std::vector<double> o(v.size());
     for (int64_t e = 0; e < nL_; ++e)
@@ -1302,20 +1275,23 @@ class GpuEngine final : public Engine<T> {
     HIP_CHECK(hipMemsetAsync(scalarPtr(), 0, sizeof(double), stream_));
   }
   // Read the device scalar accumulator, allreducing across ranks first.
-  double globalScalar() {
+  double globalScalar(ncclRedOp_t op) {
     if (hasComm_)
-      RCCL_CHECK(ncclAllReduce(scalarPtr(), scalarPtr(), 1, ncclDouble, ncclSum,
+      RCCL_CHECK(ncclAllReduce(scalarPtr(), scalarPtr(), 1, ncclDouble, op,
                                comm_, stream_));
+    return readScalar(scalarPtr());
+  }
+  double readScalar(double* dptr) {
     double h = 0;
-    HIP_CHECK(hipMemcpyAsync(&h, scalarPtr(), sizeof(double),
-                             hipMemcpyDeviceToHost, stream_));
+    HIP_CHECK(hipMemcpyAsync(&h, dptr, sizeof(double), hipMemcpyDeviceToHost,
+                             stream_));
     sync();
     return h;
   }
-  void allreduce(T* buf, int64_t n) {
+  void allreduce(T* buf, int64_t n, ncclRedOp_t op) {
     if (!hasComm_ || n == 0) return;
     RCCL_CHECK(ncclAllReduce(buf, buf, n,
-                             sizeof(T) == 8 ? ncclDouble : ncclFloat, ncclSum,
+                             sizeof(T) == 8 ? ncclDouble : ncclFloat, op,
                              comm_, stream_));
   }
   void reduceDetAsync(const T* a, const T* b, int64_t n, ROp op, double* out) {
@@ -1340,13 +1316,6 @@ class GpuEngine final : public Engine<T> {
         break;
     }
   }
-  double readScalar(double* dptr) {
-    double h = 0;
-    HIP_CHECK(hipMemcpyAsync(&h, dptr, sizeof(double), hipMemcpyDeviceToHost,
-                             stream_));
-    sync();
-    return h;
-  }
   double reduceDet(const T* a, const T* b, int64_t n, ROp op) {
     reduceDetAsync(a, b, n, op, scalarPtr());
     return readScalar(scalarPtr());
@@ -1357,56 +1326,69 @@ class GpuEngine final : public Engine<T> {
                        dim3(gridFor((int64_t)nBlk * D)), dim3(kBlk), 0, stream_,
                        nBlk, A, xv, yv);
   }
+  // w_local = Cinv * in_local  (pointers offset to the local shard)
+  void applyCinv(const T* in, T* out) {
+    blockMatVec<3, 0>(npL_, dHllInv_ + (int64_t)ptLo_ * 9,
+                      in + (int64_t)ptLo_ * 3, out + (int64_t)ptLo_ * 3);
+  }
   void spmvEtx(const T* xv, T* out) {
-    HIP_CHECK(hipMemsetAsync(out, 0, np_ * sizeof(T), stream_));
+    hipLaunchKernelGGL(kZeroRange<T>, dim3(gridFor((int64_t)npL_ * 3)),
+                       dim3(kBlk), 0, stream_, out + (int64_t)ptLo_ * 3,
+                       (int64_t)npL_ * 3);
+    const int bak = cur_ ^ 1;
     if (implicit_) {
-      const int bak = cur_ ^ 1;
       if (hasInfo_)
-        hipLaunchKernelGGL((kSpmvEtxImp<T, true>), dim3(gridFor(nL_)),
+        hipLaunchKernelGGL((kSpmvEtx<T, true, true>), dim3(gridFor(nL_)),
                            dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
-                           dJc_[bak], dJp_[bak], dInfo_, xv, out);
+                           (const T*)nullptr, dJc_[bak], dJp_[bak], dInfo_, xv,
+                           out);
       else
-        hipLaunchKernelGGL((kSpmvEtxImp<T, false>), dim3(gridFor(nL_)),
+        hipLaunchKernelGGL((kSpmvEtx<T, true, false>), dim3(gridFor(nL_)),
                            dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
-                           dJc_[bak], dJp_[bak], nullptr, xv, out);
-      return;
+                           (const T*)nullptr, dJc_[bak], dJp_[bak],
+                           (const T*)nullptr, xv, out);
+    } else {
+      hipLaunchKernelGGL((kSpmvEtx<T, false, false>), dim3(gridFor(nL_)),
+                         dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_, dHpl_,
+                         (const T*)nullptr, (const T*)nullptr,
+                         (const T*)nullptr, xv, out);
     }
-    hipLaunchKernelGGL(kSpmvEtx<T>, dim3(gridFor(nL_)), dim3(kBlk), 0,
-                       stream_, nL_, dPtOfPt_, dCamOfPt_, dHlp_, xv, out);
   }
   void spmvEx(const T* wv, T* out) {
     HIP_CHECK(hipMemsetAsync(out, 0, nc_ * sizeof(T), stream_));
-    if (nChunks_ == 0) return;
+    const int bak = cur_ ^ 1;
     if (implicit_) {
-      const int bak = cur_ ^ 1;
       if (hasInfo_)
-        hipLaunchKernelGGL((kSpmvExImp<T, true>), dim3(nChunks_), dim3(64), 0,
-                           stream_, nChunks_, dChCam_, dChLo_, dChHi_, dPtOf_,
-                           dJc_[bak], dJp_[bak], dInfo_, nL_, wv, out);
+        hipLaunchKernelGGL((kSpmvExImp<T, true>), dim3(gridFor(nL_)),
+                           dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
+                           dJc_[bak], dJp_[bak], dInfo_, wv, out);
       else
-        hipLaunchKernelGGL((kSpmvExImp<T, false>), dim3(nChunks_), dim3(64), 0,
-                           stream_, nChunks_, dChCam_, dChLo_, dChHi_, dPtOf_,
-                           dJc_[bak], dJp_[bak], nullptr, nL_, wv, out);
+        hipLaunchKernelGGL((kSpmvExImp<T, false>), dim3(gridFor(nL_)),
+                           dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
+                           dJc_[bak], dJp_[bak], (const T*)nullptr, wv, out);
       return;
     }
-    hipLaunchKernelGGL(kSpmvEx<T>, dim3(nChunks_), dim3(64), 0, stream_,
-                       nChunks_, dChCam_, dChLo_, dChHi_, dPtOf_, dHpl_, nL_,
-                       wv, out);
+    if (nChunks_ > 0)
+      hipLaunchKernelGGL(kSpmvEx<T>, dim3(nChunks_), dim3(64), 0, stream_,
+                         nChunks_, dChCam_, dChLo_, dChHi_, dPtOfCam_,
+                         dHplCam_, nL_, wv, out);
   }
-  // q = S x = HppD x - E Cinv E^T x   (2 allreduces, reference site A4).
+  // q = S x = HppD x - E Cinv E^T x  (ONE 9*ncam allreduce; the reference's
+  // site A4 needed an additional 3*npt allreduce here).
   void schurApply(const T* xv, T* q) {
     spmvEtx(xv, dTemp_);
-    allreduce(dTemp_, np_);
-    blockMatVec<3, 0>(npt_, dHllInv_, dTemp_, dW_);
+    applyCinv(dTemp_, dW_);
     spmvEx(dW_, q);
-    allreduce(q, nc_);
+    allreduce(q, nc_, ncclSum);
     blockMatVec<9, 1>(ncam_, dHppD_, xv, q);
   }
 
   hipStream_t stream_{};
   ncclComm_t comm_{};
   bool hasComm_ = false;
+  CustomForward<T> customFwd_;
   int rank_, world_, ncam_, npt_;
+  int ptLo_ = 0, ptHi_ = 0, npL_ = 0;
   int64_t e0_ = 0, e1_ = 0, nL_ = 0, nc_ = 0, np_ = 0, dim_ = 0;
   bool hasInfo_ = false;
   bool analytical_ = false;
@@ -1414,13 +1396,12 @@ class GpuEngine final : public Engine<T> {
   int cur_ = 0;
   int nChunks_ = 0;
   int *dCamOf_{}, *dPtOf_{}, *dChCam_{}, *dChLo_{}, *dChHi_{}, *dFail_{};
-  int *dPtPerm_{}, *dPtRowPtr_{}, *dCamOfPt_{}, *dPtOfPt_{}, *dPtPos_{};
+  int *dCamPos_{}, *dPtOfCam_{};
   T *dMeas_{}, *dInfo_{}, *dLeaf_{}, *dMeasSplit_{};
-  CustomForward<T> customFwd_;
   unsigned char *dCamFixed_{}, *dPtFixed_{};
   T *dParams_{}, *dParamsBak_{};
   T *dR_[2]{}, *dJc_[2]{}, *dJp_[2]{};
-  T *dHpp_{}, *dHll_{}, *dHpl_{}, *dHlp_{}, *dSlab_{}, *dG_{}, *dGBak_{};
+  T *dHpp_{}, *dHll_{}, *dHpl_{}, *dHplCam_{}, *dSlab_{}, *dG_{}, *dGBak_{};
   T *dHppD_{}, *dHllD_{}, *dHppInv_{}, *dHllInv_{};
   T *dDeltaX_{}, *dDeltaXBak_{};
   T *dP_{}, *dRr_{}, *dZ_{}, *dQ_{}, *dV_{}, *dW_{}, *dTemp_{}, *dXBak_{};

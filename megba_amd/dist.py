@@ -14,9 +14,10 @@ def gloo_allreduce_callback():
     import torch
     import torch.distributed as dist
 
-    def allreduce(arr: np.ndarray):
+    def allreduce(arr: np.ndarray, op: str = "sum"):
         t = torch.from_numpy(arr)
-        dist.all_reduce(t, op=dist.ReduceOp.SUM)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX if op == "max"
+                        else dist.ReduceOp.SUM)
     return allreduce
 
 

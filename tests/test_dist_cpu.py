@@ -59,6 +59,9 @@ def test_partition_covers_all_edges():
     split = ii["split"]
     assert split[0] == 0 and split[-1] == len(ci)
     assert all(split[i] <= split[i + 1] for i in range(4))
-    # sorted order is (cam, pt)-lexicographic
-    key = ii["cam_of"].astype(np.int64) * (len(pts) + 1) + ii["pt_of"]
+    # sorted order is (pt, cam)-lexicographic; splits are point-aligned
+    key = ii["pt_of"].astype(np.int64) * (len(cams) + 1) + ii["cam_of"]
     assert (np.diff(key) >= 0).all()
+    for s_ in split[1:-1]:
+        if 0 < s_ < len(ci):
+            assert ii["pt_of"][s_] != ii["pt_of"][s_ - 1]
