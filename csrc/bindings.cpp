@@ -372,6 +372,7 @@ PYBIND11_MODULE(_core, m) {
   m.def("jv_angle_axis_to_rotation", [](py::sequence aa) { return jvWrap(jvAngleAxisToRotation(jvList(aa))); });
   m.def("jv_rotation2d", [](const PyJetVec& t) { return jvWrap(jvRotation2D(t.v)); });
   m.def("jv_quaternion_to_rotation", [](py::sequence q) { return jvWrap(jvQuaternionToRotation(jvList(q))); });
+  m.def("jv_rotation_to_quaternion", [](py::sequence R) { return jvWrap(jvRotationToQuaternion(jvList(R))); });
   m.def("jv_normalize_quaternion", [](py::sequence q) { return jvWrap(jvNormalizeQuaternion(jvList(q))); });
   m.def("jv_radial_distortion", [](py::sequence p, py::sequence intr) { return PyJetVec{jvRadialDistortion(jvList(p), jvList(intr))}; });
 

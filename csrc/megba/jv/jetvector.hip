@@ -471,6 +471,161 @@ JetVec<T> jvRadialDistortion(const std::vector<JetVec<T>>& p,
   return intr[0] * d;
 }
 
+// Rotation matrix -> unit quaternion [w,x,y,z] with gradients: per-item
+// Shepperd branch on the largest of {trace, R00, R11, R22} (the reference's
+// RotationToQuaternion kernel, quaternion.cu:102-199, used a static device
+// function-pointer table; here one parameterised branch).  Inputs must be
+// dense JetVectors (geometry-op outputs are).
+namespace {
+template <typename T>
+__global__ void kRot2Quat(int64_t n, int N, const T* const* rv,
+                          const T* const* rg, T* const* qv, T* const* qg) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    T v[9], q[4];
+    for (int k = 0; k < 9; ++k) v[k] = rv[k][i];
+    // local dense copies of grads are too large for registers at runtime N:
+    // recompute via the strided helper form instead.
+    // Build strided accessors: helper expects gr[(k*N+g)*n]; we emulate by
+    // copying pointers — instead call a small inline with direct indexing:
+    const T tr = v[0] + v[4] + v[8];
+    int b = 0;
+    T best = tr;
+    if (v[0] > best) { b = 1; best = v[0]; }
+    if (v[4] > best) { b = 2; best = v[4]; }
+    if (v[8] > best) { b = 3; }
+    const T sg[4][3] = {{1, 1, 1}, {1, -1, -1}, {-1, 1, -1}, {-1, -1, 1}};
+    const int major[4] = {0, 1, 2, 3};
+    const int oth[4][3][4] = {
+        {{1, 7, 5, -1}, {2, 2, 6, -1}, {3, 3, 1, -1}},
+        {{0, 7, 5, -1}, {2, 3, 1, +1}, {3, 2, 6, +1}},
+        {{0, 2, 6, -1}, {1, 3, 1, +1}, {3, 7, 5, +1}},
+        {{0, 3, 1, -1}, {1, 2, 6, +1}, {2, 7, 5, +1}}};
+    const T t = T(1) + sg[b][0] * v[0] + sg[b][1] * v[4] + sg[b][2] * v[8];
+    const T s = ::sqrt(t);
+    const T inv2s = T(0.5) / s;
+    q[major[b]] = s * T(0.5);
+    T comp[3];
+    for (int k = 0; k < 3; ++k) {
+      comp[k] = (v[oth[b][k][1]] + T(oth[b][k][3]) * v[oth[b][k][2]]) * inv2s;
+      q[oth[b][k][0]] = comp[k];
+    }
+    for (int k = 0; k < 4; ++k) qv[k][i] = q[k];
+    for (int g = 0; g < N; ++g) {
+      const T dt = sg[b][0] * rg[0][(int64_t)g * n + i] +
+                   sg[b][1] * rg[4][(int64_t)g * n + i] +
+                   sg[b][2] * rg[8][(int64_t)g * n + i];
+      const T ds = dt * inv2s;
+      qg[major[b]][(int64_t)g * n + i] = ds * T(0.5);
+      for (int k = 0; k < 3; ++k) {
+        const T dn = rg[oth[b][k][1]][(int64_t)g * n + i] +
+                     T(oth[b][k][3]) * rg[oth[b][k][2]][(int64_t)g * n + i];
+        qg[oth[b][k][0]][(int64_t)g * n + i] = dn * inv2s - comp[k] * ds / s;
+      }
+    }
+  }
+}
+}  // namespace
+
+template <typename T>
+std::vector<JetVec<T>> jvRotationToQuaternion(const std::vector<JetVec<T>>& R) {
+  MEGBA_CHECK(R.size() == 9, "rotation needs 9 components");
+  for (const auto& r : R)
+    MEGBA_CHECK(r.kind() == JvKind::DENSE, "rot->quat needs dense JetVectors");
+  const int64_t n = R[0].nItem;
+  const int N = R[0].N;
+  const bool gpu = R[0].onGpu;
+  std::vector<JetVec<T>> q;
+  for (int k = 0; k < 4; ++k) {
+    JetVec<T> o;
+    o.nItem = n;
+    o.N = N;
+    o.onGpu = gpu;
+    o.value = makeBuf<T>(n, gpu);
+    o.grad = makeBuf<T>((int64_t)N * n, gpu);
+    q.push_back(o);
+  }
+  std::vector<const T*> rv(9), rg(9);
+  std::vector<T*> qvp(4), qgp(4);
+  for (int k = 0; k < 9; ++k) {
+    rv[k] = R[k].value->ptr;
+    rg[k] = R[k].grad->ptr;
+  }
+  for (int k = 0; k < 4; ++k) {
+    qvp[k] = q[k].value->ptr;
+    qgp[k] = q[k].grad->ptr;
+  }
+  if (gpu) {
+    // pointer tables on device
+    const T** drv;
+    const T** drg;
+    T** dqv;
+    T** dqg;
+    JV_HIP_CHECK(hipMalloc(&drv, 9 * sizeof(T*)));
+    JV_HIP_CHECK(hipMalloc(&drg, 9 * sizeof(T*)));
+    JV_HIP_CHECK(hipMalloc(&dqv, 4 * sizeof(T*)));
+    JV_HIP_CHECK(hipMalloc(&dqg, 4 * sizeof(T*)));
+    JV_HIP_CHECK(hipMemcpy(drv, rv.data(), 9 * sizeof(T*), hipMemcpyHostToDevice));
+    JV_HIP_CHECK(hipMemcpy(drg, rg.data(), 9 * sizeof(T*), hipMemcpyHostToDevice));
+    JV_HIP_CHECK(hipMemcpy(dqv, qvp.data(), 4 * sizeof(T*), hipMemcpyHostToDevice));
+    JV_HIP_CHECK(hipMemcpy(dqg, qgp.data(), 4 * sizeof(T*), hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(kRot2Quat<T>, dim3(jvGrid(n)), dim3(256), 0, 0, n, N,
+                       drv, drg, dqv, dqg);
+    JV_HIP_CHECK(hipGetLastError());
+    JV_HIP_CHECK(hipDeviceSynchronize());
+    (void)hipFree(drv);
+    (void)hipFree(drg);
+    (void)hipFree(dqv);
+    (void)hipFree(dqg);
+  } else {
+#pragma omp parallel for schedule(static)
+    for (int64_t i = 0; i < n; ++i) {
+      T v[9], qq[4];
+      for (int k = 0; k < 9; ++k) v[k] = rv[k][i];
+      const T tr = v[0] + v[4] + v[8];
+      int b = 0;
+      T best = tr;
+      if (v[0] > best) { b = 1; best = v[0]; }
+      if (v[4] > best) { b = 2; best = v[4]; }
+      if (v[8] > best) { b = 3; }
+      const T sg[4][3] = {{1, 1, 1}, {1, -1, -1}, {-1, 1, -1}, {-1, -1, 1}};
+      const int major[4] = {0, 1, 2, 3};
+      const int oth[4][3][4] = {
+          {{1, 7, 5, -1}, {2, 2, 6, -1}, {3, 3, 1, -1}},
+          {{0, 7, 5, -1}, {2, 3, 1, +1}, {3, 2, 6, +1}},
+          {{0, 2, 6, -1}, {1, 3, 1, +1}, {3, 7, 5, +1}},
+          {{0, 3, 1, -1}, {1, 2, 6, +1}, {2, 7, 5, +1}}};
+      const T t = T(1) + sg[b][0] * v[0] + sg[b][1] * v[4] + sg[b][2] * v[8];
+      const T s = std::sqrt(t);
+      const T inv2s = T(0.5) / s;
+      qq[major[b]] = s * T(0.5);
+      T comp[3];
+      for (int k = 0; k < 3; ++k) {
+        comp[k] = (v[oth[b][k][1]] + T(oth[b][k][3]) * v[oth[b][k][2]]) * inv2s;
+        qq[oth[b][k][0]] = comp[k];
+      }
+      for (int k = 0; k < 4; ++k) qvp[k][i] = qq[k];
+      for (int g = 0; g < N; ++g) {
+        const T dt = sg[b][0] * rg[0][(int64_t)g * n + i] +
+                     sg[b][1] * rg[4][(int64_t)g * n + i] +
+                     sg[b][2] * rg[8][(int64_t)g * n + i];
+        const T ds = dt * inv2s;
+        qgp[major[b]][(int64_t)g * n + i] = ds * T(0.5);
+        for (int k = 0; k < 3; ++k) {
+          const T dn = rg[oth[b][k][1]][(int64_t)g * n + i] +
+                       T(oth[b][k][3]) * rg[oth[b][k][2]][(int64_t)g * n + i];
+          qgp[oth[b][k][0]][(int64_t)g * n + i] = dn * inv2s - comp[k] * ds / s;
+        }
+      }
+    }
+  }
+  return q;
+}
+template std::vector<JetVec<double>> jvRotationToQuaternion<double>(
+    const std::vector<JetVec<double>>&);
+template std::vector<JetVec<float>> jvRotationToQuaternion<float>(
+    const std::vector<JetVec<float>>&);
+
 // Explicit instantiations.
 #define JV_INST(T)                                                            \
   template JetVec<T> jvFromHost<T>(const T*, const T*, int64_t, int, int,     \

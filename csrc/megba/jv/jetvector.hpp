@@ -89,6 +89,9 @@ std::vector<JetVec<T>> jvRotation2D(const JetVec<T>& theta);
 // unit quaternion [w,x,y,z] -> row-major R[9]
 template <typename T>
 std::vector<JetVec<T>> jvQuaternionToRotation(const std::vector<JetVec<T>>& q);
+// row-major R[9] -> unit quaternion [w,x,y,z] (per-item Shepperd branch)
+template <typename T>
+std::vector<JetVec<T>> jvRotationToQuaternion(const std::vector<JetVec<T>>& R);
 // q[4] -> normalised q[4]
 template <typename T>
 std::vector<JetVec<T>> jvNormalizeQuaternion(const std::vector<JetVec<T>>& q);

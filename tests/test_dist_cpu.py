@@ -43,10 +43,18 @@ def test_world2_matches_world1(tmp_path):
     import torch.multiprocessing as mp
     ref = _solve(1)
     out = tmp_path / "chis.json"
-    port = 29511
-    mp.spawn(_worker, args=(2, port, str(out)), nprocs=2, join=True)
+    mp.spawn(_worker, args=(2, 29511, str(out)), nprocs=2, join=True)
     chis = json.loads(out.read_text())
     assert len(chis) == len(ref)
+    np.testing.assert_allclose(chis, ref, rtol=1e-6)
+
+
+def test_world4_matches_world1(tmp_path):
+    import torch.multiprocessing as mp
+    ref = _solve(1)
+    out = tmp_path / "chis4.json"
+    mp.spawn(_worker, args=(4, 29512, str(out)), nprocs=4, join=True)
+    chis = json.loads(out.read_text())
     np.testing.assert_allclose(chis, ref, rtol=1e-6)
 
 

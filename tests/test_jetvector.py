@@ -211,3 +211,59 @@ def test_geo_gpu_matches_cpu():
     for k in range(9):
         np.testing.assert_allclose(out[k][0], ref[k][0], rtol=1e-12)
         np.testing.assert_allclose(out[k][1], ref[k][1], rtol=1e-12)
+
+
+def test_rotation_to_quaternion_roundtrip_cpu():
+    # q -> R -> q' must reproduce q (up to sign), values and gradients.
+    ni = 64
+    q_val = rng.normal(size=(4, ni))
+    q_val /= np.linalg.norm(q_val, axis=0, keepdims=True)
+    # force some items into each Shepperd branch via sign flips
+    q_val[:, ::4] = np.abs(q_val[:, ::4])
+    q = [_core.JetVector(q_val[i], None, N=4, grad_pos=i) for i in range(4)]
+    R = _core.jv_quaternion_to_rotation(q)
+    q2 = _core.jv_rotation_to_quaternion(R)
+    v2 = np.stack([q2[k].to_numpy()[0] for k in range(4)])
+    # fix sign per item (quaternion double cover)
+    sign = np.sign((v2 * q_val).sum(axis=0))
+    np.testing.assert_allclose(v2 * sign, q_val, atol=1e-9)
+    # gradient check vs finite differences through the full chain
+    def chain(qi):
+        w, x, y, z = qi
+        Rm = np.array([
+            [1 - 2 * (y * y + z * z), 2 * (x * y - w * z), 2 * (x * z + w * y)],
+            [2 * (x * y + w * z), 1 - 2 * (x * x + z * z), 2 * (y * z - w * x)],
+            [2 * (x * z - w * y), 2 * (y * z + w * x), 1 - 2 * (x * x + y * y)]])
+        t = np.trace(Rm)
+        b = int(np.argmax([t, Rm[0, 0], Rm[1, 1], Rm[2, 2]]))
+        if b == 0:
+            s = np.sqrt(1 + t)
+            return np.array([s / 2, (Rm[2, 1] - Rm[1, 2]) / (2 * s),
+                             (Rm[0, 2] - Rm[2, 0]) / (2 * s),
+                             (Rm[1, 0] - Rm[0, 1]) / (2 * s)])
+        if b == 1:
+            s = np.sqrt(1 + Rm[0, 0] - Rm[1, 1] - Rm[2, 2])
+            return np.array([(Rm[2, 1] - Rm[1, 2]) / (2 * s), s / 2,
+                             (Rm[1, 0] + Rm[0, 1]) / (2 * s),
+                             (Rm[0, 2] + Rm[2, 0]) / (2 * s)])
+        if b == 2:
+            s = np.sqrt(1 - Rm[0, 0] + Rm[1, 1] - Rm[2, 2])
+            return np.array([(Rm[0, 2] - Rm[2, 0]) / (2 * s),
+                             (Rm[1, 0] + Rm[0, 1]) / (2 * s), s / 2,
+                             (Rm[2, 1] + Rm[1, 2]) / (2 * s)])
+        s = np.sqrt(1 - Rm[0, 0] - Rm[1, 1] + Rm[2, 2])
+        return np.array([(Rm[1, 0] - Rm[0, 1]) / (2 * s),
+                         (Rm[0, 2] + Rm[2, 0]) / (2 * s),
+                         (Rm[2, 1] + Rm[1, 2]) / (2 * s), s / 2])
+
+    grads = [q2[k].to_numpy()[1] for k in range(4)]
+    eps = 1e-7
+    for item in range(0, ni, 9):
+        qi = q_val[:, item]
+        for d in range(4):
+            qp, qm = qi.copy(), qi.copy()
+            qp[d] += eps
+            qm[d] -= eps
+            fd = (chain(qp) - chain(qm)) / (2 * eps)
+            for k in range(4):
+                np.testing.assert_allclose(grads[k][d, item], fd[k], atol=1e-5)
