@@ -643,6 +643,50 @@ __global__ void kSpmvExImp(int64_t nL, const int* __restrict__ camOf,
   }
 }
 
+// Fused preconditioner apply + rho partial: z = Binv r (thread per row) and
+// per-block partials of r.z in one pass (saves two launches per PCG iter).
+template <typename T>
+__global__ void kPrecondRho(int nBlk, const T* __restrict__ Binv,
+                            const T* __restrict__ r, T* __restrict__ z,
+                            double* part) {
+  __shared__ double sm[kBlk];
+  double local = 0.0;
+  for (int64_t idx = blockIdx.x * (int64_t)kBlk + threadIdx.x;
+       idx < (int64_t)nBlk * 9; idx += (int64_t)gridDim.x * kBlk) {
+    const int64_t b = idx / 9;
+    const int rr = (int)(idx % 9);
+    const T* row = Binv + b * 81 + (int64_t)rr * 9;
+    const T* xb = r + b * 9;
+    T sv = T(0);
+    for (int j = 0; j < 9; ++j) sv += row[j] * xb[j];
+    z[idx] = sv;
+    local += (double)sv * (double)r[idx];
+  }
+  sm[threadIdx.x] = local;
+  __syncthreads();
+  for (int st = kBlk / 2; st > 0; st >>= 1) {
+    if (threadIdx.x < st) sm[threadIdx.x] += sm[threadIdx.x + st];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) part[blockIdx.x] = sm[0];
+}
+
+// Fused PCG vector update: xBak = x; x += alpha p; r -= alpha q.
+template <typename T>
+__global__ void kUpdateXR(int64_t n, const double* __restrict__ alpha,
+                          const T* __restrict__ p, const T* __restrict__ q,
+                          T* __restrict__ x, T* __restrict__ xBak,
+                          T* __restrict__ r) {
+  const T a = (T)(*alpha);
+  for (int64_t i = blockIdx.x * (int64_t)kBlk + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * kBlk) {
+    const T xv = x[i];
+    xBak[i] = xv;
+    x[i] = xv + a * p[i];
+    r[i] -= a * q[i];
+  }
+}
+
 // Block-diagonal matvec, one thread per output row.
 // MODE 0: y = A x;  MODE 1: y = A x - y  (the reference's rw=1,dw=-1 gemv).
 template <typename T, int D, int MODE>
@@ -1068,9 +1112,12 @@ class GpuEngine final : public Engine<T> {
     int n = 0;
     double rho = 0.0, rhoPrev = 0.0, rhoMin = INFINITY;
     bool done = false;
+    const int rhoGrid = gridFor(nc_) < kRedBlocks ? gridFor(nc_) : kRedBlocks;
     while (!done && n < opt.maxIter) {
-      blockMatVec<9, 0>(ncam_, dHppInv_, dRr_, dZ_);
-      reduceDetAsync(dRr_, dZ_, nc_, ROp::Dot, slotRho());
+      hipLaunchKernelGGL(kPrecondRho<T>, dim3(rhoGrid), dim3(kBlk), 0, stream_,
+                         ncam_, dHppInv_, dRr_, dZ_, dPart_);
+      hipLaunchKernelGGL((kRedFinal<ROp::Dot>), dim3(1), dim3(kBlk), 0,
+                         stream_, dPart_, rhoGrid, slotRho());
       rho = readScalar(slotRho());
       if (rho > opt.refuseRatio * rhoMin) {
         HIP_CHECK(hipMemcpyAsync(x, dXBak_, nc_ * sizeof(T),
@@ -1089,12 +1136,8 @@ class GpuEngine final : public Engine<T> {
       reduceDetAsync(dP_, dQ_, nc_, ROp::Dot, slotPq());
       hipLaunchKernelGGL(kDivScalar, dim3(1), dim3(1), 0, stream_, slotAlpha(),
                          slotRho(), slotPq());
-      HIP_CHECK(hipMemcpyAsync(dXBak_, x, nc_ * sizeof(T),
-                               hipMemcpyDeviceToDevice, stream_));
-      hipLaunchKernelGGL((kAxpyS<T, 1>), dim3(gridFor(nc_)), dim3(kBlk), 0,
-                         stream_, nc_, slotAlpha(), dP_, x);
-      hipLaunchKernelGGL((kAxpyS<T, -1>), dim3(gridFor(nc_)), dim3(kBlk), 0,
-                         stream_, nc_, slotAlpha(), dQ_, dRr_);
+      hipLaunchKernelGGL(kUpdateXR<T>, dim3(gridFor(nc_)), dim3(kBlk), 0,
+                         stream_, nc_, slotAlpha(), dP_, dQ_, x, dXBak_, dRr_);
       rhoPrev = rho;
       ++n;
       done = std::abs(rho) < opt.tol;
