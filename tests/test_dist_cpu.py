@@ -73,3 +73,23 @@ def test_partition_covers_all_edges():
     for s_ in split[1:-1]:
         if 0 < s_ < len(ci):
             assert ii["pt_of"][s_] != ii["pt_of"][s_ - 1]
+
+
+def test_bench_contract_torchrun_cpu(tmp_path):
+    """Exercises bench.py exactly as the driver launches it (torchrun, one
+    process per 'GPU'), on the CPU engine with 2 ranks over gloo."""
+    import os
+    import subprocess
+    import sys
+    env = dict(os.environ)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29518", "bench.py", "--model", "tiny",
+         "--device", "cpu", "--gpus", "2", "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, timeout=600, env=env)
+    assert r.returncode == 0, r.stderr[-3000:]
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    d = json.loads(line)
+    assert d["n_gpus"] == 2 and d["steps"] == 2
+    assert d["metric"] == "lm_iterations_per_s" and d["value"] > 0
