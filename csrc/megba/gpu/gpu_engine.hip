@@ -385,13 +385,10 @@ __global__ void kAssembleEdge(int64_t nL, const int* __restrict__ camOf,
       }
       T* row = slab + (int64_t)camPos[e] * L::SW;
       if (EXPL) {
-        // edge-major 28-double rows (16B-aligned): E^T x streams each
-        // point's contiguous run with one thread per point, no atomics.
-        T* hrow = Hpl + e * 28;
         for (int a = 0; a < 9; ++a)
           for (int b = 0; b < 3; ++b) {
             const T v = jc[0][a] * wjp[0][b] + jc[1][a] * wjp[1][b];
-            hrow[a * 3 + b] = v;
+            Hpl[((int64_t)(a * 3 + b)) * nL + e] = v;
             row[a * 3 + b] = v;
           }
       }
@@ -635,37 +632,6 @@ __global__ void kSpmvEtx(int64_t nL, const int* __restrict__ camOf,
       atomicAdd(&out[3 * pt + 1], o1);
       atomicAdd(&out[3 * pt + 2], o2);
     }
-  }
-}
-
-// Explicit E^T x: one thread per local point streaming its contiguous
-// edge-major Hpl rows ((pt,cam)-sorted order => runs are contiguous, ~5
-// rows x 224 B each); direct store, no atomics, no zero pass.  The
-// replicated camera vector x is L2-resident.
-template <typename T>
-__global__ void kSpmvEtxPt(int npL, int ptLo,
-                           const int* __restrict__ rowPtrLoc,
-                           const int* __restrict__ camOf,
-                           const T* __restrict__ HplRows,
-                           const T* __restrict__ x, T* __restrict__ out) {
-  for (int64_t i = blockIdx.x * (int64_t)kBlk + threadIdx.x; i < npL;
-       i += (int64_t)gridDim.x * kBlk) {
-    const int lo = rowPtrLoc[i], hi = rowPtrLoc[i + 1];
-    T o0 = 0, o1 = 0, o2 = 0;
-    for (int j = lo; j < hi; ++j) {
-      const T* row = HplRows + (int64_t)j * 28;
-      const T* xc = x + (int64_t)camOf[j] * 9;
-      for (int a = 0; a < 9; ++a) {
-        const T xa = xc[a];
-        o0 += row[a * 3 + 0] * xa;
-        o1 += row[a * 3 + 1] * xa;
-        o2 += row[a * 3 + 2] * xa;
-      }
-    }
-    T* op = out + ((int64_t)ptLo + i) * 3;
-    op[0] = o0;
-    op[1] = o1;
-    op[2] = o2;
   }
 }
 
@@ -1056,15 +1022,8 @@ class GpuEngine final : public Engine<T> {
     }
     dSlab_ = dalloc<T>(nL_ * slabWidth());
     if (!implicit_) {
-      dHpl_ = dalloc<T>(nL_ * 28);  // edge-major 28-double rows
+      dHpl_ = dalloc<T>(nL_ * 27);
       dHplCam_ = dalloc<T>(nL_ * 27);
-    }
-    {
-      std::vector<int> rpl(npL_ + 1);
-      for (int i = 0; i <= npL_; ++i)
-        rpl[i] = (int)(ix.ptRowPtr[ptLo_ + i] - e0_);
-      dPtRowPtrLoc_ = dalloc<int>(npL_ + 1);
-      up(dPtRowPtrLoc_, rpl.data(), npL_ + 1);
     }
     sync();
   }
@@ -1340,11 +1299,11 @@ class GpuEngine final : public Engine<T> {
     d.Hpp = down(dHpp_, (int64_t)ncam_ * 81);
     d.Hll = down(dHll_, (int64_t)npt_ * 9);
     if (!implicit_) {
-      // device Hpl is edge-major [e][28]; dump as [e][9][3]
-      std::vector<double> em = down(dHpl_, nL_ * 28);
-      std::vector<double> o(nL_ * 27);
+      // device Hpl is grad-major [27][nL]; dump as [e][9][3]
+      std::vector<double> gm = down(dHpl_, nL_ * 27);
+      std::vector<double> o(gm.size());
       for (int64_t e = 0; e < nL_; ++e)
-        for (int k = 0; k < 27; ++k) o[e * 27 + k] = em[e * 28 + k];
+        for (int k = 0; k < 27; ++k) o[e * 27 + k] = gm[(int64_t)k * nL_ + e];
       d.Hpl = o;
     }
     d.g = down(dG_, dim_);
@@ -1488,12 +1447,6 @@ class GpuEngine final : public Engine<T> {
                       in + (int64_t)ptLo_ * 3, out + (int64_t)ptLo_ * 3);
   }
   void spmvEtx(const T* xv, T* out) {
-    if (!implicit_) {
-      hipLaunchKernelGGL(kSpmvEtxPt<T>, dim3(gridFor(npL_)), dim3(kBlk), 0,
-                         stream_, npL_, ptLo_, dPtRowPtrLoc_, dCamOf_, dHpl_,
-                         xv, out);
-      return;
-    }
     hipLaunchKernelGGL(kZeroRange<T>, dim3(gridFor((int64_t)npL_ * 3)),
                        dim3(kBlk), 0, stream_, out + (int64_t)ptLo_ * 3,
                        (int64_t)npL_ * 3);
@@ -1509,6 +1462,11 @@ class GpuEngine final : public Engine<T> {
                            dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
                            (const T*)nullptr, dJc_[bak], dJp_[bak],
                            (const T*)nullptr, xv, out);
+    } else {
+      hipLaunchKernelGGL((kSpmvEtx<T, false, false>), dim3(gridFor(nL_)),
+                         dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_, dHpl_,
+                         (const T*)nullptr, (const T*)nullptr,
+                         (const T*)nullptr, xv, out);
     }
   }
   void spmvEx(const T* wv, T* out) {
@@ -1608,7 +1566,7 @@ class GpuEngine final : public Engine<T> {
   int cur_ = 0;
   int nChunks_ = 0;
   int *dCamOf_{}, *dPtOf_{}, *dChCam_{}, *dChLo_{}, *dChHi_{}, *dFail_{};
-  int *dCamPos_{}, *dPtOfCam_{}, *dPtRowPtrLoc_{};
+  int *dCamPos_{}, *dPtOfCam_{};
   T *dMeas_{}, *dInfo_{}, *dLeaf_{}, *dMeasSplit_{};
   unsigned char *dCamFixed_{}, *dPtFixed_{};
   T *dParams_{}, *dParamsBak_{};
