@@ -1516,6 +1516,10 @@ class GpuEngine final : public Engine<T> {
     if (pcgGraphExec_ || pcgGraphTried_) return;
     pcgGraphTried_ = true;
     if (getenv("MEGBA_NO_GRAPH")) return;
+    // Multi-rank: RCCL-in-graph-capture is not exercisable in this round's
+    // single-GPU test environment, so stay eager unless explicitly enabled
+    // (the collective SEQUENCE is identical either way).
+    if (world_ > 1 && !getenv("MEGBA_GRAPH")) return;
     sync();
     hipGraph_t graph = nullptr;
     if (hipStreamBeginCapture(stream_, hipStreamCaptureModeThreadLocal) !=

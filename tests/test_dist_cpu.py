@@ -31,8 +31,19 @@ def _worker(rank, world_size, port, out_path):
         "gloo", init_method=f"tcp://127.0.0.1:{port}",
         rank=rank, world_size=world_size)
     try:
-        chis = _solve(world_size, rank, gloo_allreduce_callback())
+        import megba_amd as mb
+        cams, pts, ci, pi, meas = mb.synthesize_bal(*SHAPE, seed=SEED)
+        p = mb.BAProblem(cams, pts, ci, pi, meas)
+        p.build(device="cpu", rank=rank, world_size=world_size,
+                allreduce=gloo_allreduce_callback())
+        rep = p.solve(max_iter=8, tau=1e4, solver_tol=1e-6,
+                      solver_max_iter=300, solver_refuse_ratio=1e6,
+                      verbose=False)
+        chis = [it["chi2"] for it in rep["iters"]]
         if rank == 0:
+            c2, p2 = p.get_params()   # exercises the point-shard merge
+            np.save(out_path + ".cams.npy", c2)
+            np.save(out_path + ".pts.npy", p2)
             with open(out_path, "w") as f:
                 json.dump(chis, f)
     finally:
@@ -40,13 +51,25 @@ def _worker(rank, world_size, port, out_path):
 
 
 def test_world2_matches_world1(tmp_path):
+    import megba_amd as mb
     import torch.multiprocessing as mp
-    ref = _solve(1)
+    cams, pts, ci, pi, meas = mb.synthesize_bal(*SHAPE, seed=SEED)
+    p1 = mb.BAProblem(cams, pts, ci, pi, meas)
+    p1.build(device="cpu")
+    rep = p1.solve(max_iter=8, tau=1e4, solver_tol=1e-6, solver_max_iter=300,
+                   solver_refuse_ratio=1e6, verbose=False)
+    ref = [it["chi2"] for it in rep["iters"]]
+    c1, q1 = p1.get_params()
     out = tmp_path / "chis.json"
     mp.spawn(_worker, args=(2, 29511, str(out)), nprocs=2, join=True)
     chis = json.loads(out.read_text())
     assert len(chis) == len(ref)
     np.testing.assert_allclose(chis, ref, rtol=1e-6)
+    # final parameters (incl. the merged point shards) must match
+    c2 = np.load(str(out) + ".cams.npy")
+    q2 = np.load(str(out) + ".pts.npy")
+    np.testing.assert_allclose(c2, c1, rtol=1e-6, atol=1e-9)
+    np.testing.assert_allclose(q2, q1, rtol=1e-6, atol=1e-9)
 
 
 def test_world4_matches_world1(tmp_path):
