@@ -66,7 +66,8 @@ class Engine {
   // set (reference computeRhoDenominator, lm_algo.cu:82-126).
   virtual double rhoDenominator(double chi2Backup) = 0;
 
-  // Introspection (tests / write-back).
+  // Introspection (tests / write-back).  COLLECTIVE when worldSize>1: every
+  // rank must call it (point shards are merged with an allreduce).
   virtual void getParams(double* cams, double* pts) = 0;
   virtual DenseDump dump() const = 0;
 };

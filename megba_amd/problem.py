@@ -69,6 +69,9 @@ class BAProblem:
         return getattr(self._core, name)
 
     def params(self):
+        """Solved parameters (cams, pts).  COLLECTIVE when world_size>1:
+        every rank must call it (the point shards are merged with an
+        allreduce)."""
         return self._core.get_params()
 
 

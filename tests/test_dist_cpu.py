@@ -40,8 +40,10 @@ def _worker(rank, world_size, port, out_path):
                       solver_max_iter=300, solver_refuse_ratio=1e6,
                       verbose=False)
         chis = [it["chi2"] for it in rep["iters"]]
+        # get_params is COLLECTIVE at world_size>1 (merges point shards):
+        # every rank must call it.
+        c2, p2 = p.get_params()
         if rank == 0:
-            c2, p2 = p.get_params()   # exercises the point-shard merge
             np.save(out_path + ".cams.npy", c2)
             np.save(out_path + ".pts.npy", p2)
             with open(out_path, "w") as f:
