@@ -34,3 +34,19 @@ def test_cli_runs_cpu(tmp_path):
     assert "Iter" in r.stdout and "Finished" in r.stdout
     c2, *_ = mb.load_bal(out)
     assert not np.allclose(c2, cams)
+
+
+def test_cpp_cli_runs_cpu(tmp_path):
+    import os
+    binpath = "examples/bal_solve_cpp"
+    if not os.path.exists(binpath):
+        import pytest
+        pytest.skip("native example not built")
+    cams, pts, ci, pi, meas = mb.synthesize_bal(8, 50, 400, seed=1)
+    f = tmp_path / "prob.txt"
+    mb.save_bal(f, cams, pts, ci, pi, meas)
+    r = subprocess.run([binpath, "--path", str(f), "--device", "cpu",
+                        "--max_iter", "4", "--tau", "1e4"],
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr
+    assert "final error" in r.stdout
