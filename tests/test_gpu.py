@@ -138,3 +138,21 @@ def test_rccl_world1_via_torchrun(tmp_path):
         capture_output=True, text=True, timeout=600, env=env)
     assert r.returncode == 0, r.stderr[-3000:]
     assert '"lm_iterations_per_s"' in r.stdout
+
+
+@pytest.mark.gpu
+def test_cpp_cli_gpu(tmp_path):
+    import os
+    import subprocess
+    binpath = "examples/bal_solve_cpp"
+    if not os.path.exists(binpath):
+        pytest.skip("native example not built")
+    cams, pts, ci, pi, meas = mb.synthesize_bal(10, 80, 700, seed=2)
+    f = tmp_path / "prob.txt"
+    from megba_amd import save_bal
+    save_bal(f, cams, pts, ci, pi, meas)
+    r = subprocess.run([binpath, "--path", str(f), "--device", "gpu",
+                        "--max_iter", "4", "--tau", "1e4"],
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr
+    assert "final error" in r.stdout
