@@ -69,3 +69,16 @@ def test_implicit_fp32_analytical_gpu():
     p.build(device="gpu", dtype="float32", diff="analytical", schur="implicit")
     rep = p.solve(max_iter=5, verbose=False)
     assert rep["final_chi2"] < rep["iters"][0]["chi2"]
+
+
+@pytest.mark.gpu
+def test_implicit_full_solve_gpu():
+    # Multiple accept/reject cycles: catches stale-buffer bugs in the
+    # implicit path (the PCG operator must always read the freshly accepted
+    # Jacobians, even across the double-buffer swaps).
+    ex, im = _pair("gpu")
+    kw = dict(max_iter=8, solver_tol=1e-6, solver_max_iter=300,
+              solver_refuse_ratio=1e6, verbose=False)
+    r1, r2 = ex.solve(**kw), im.solve(**kw)
+    np.testing.assert_allclose([i["chi2"] for i in r2["iters"]],
+                               [i["chi2"] for i in r1["iters"]], rtol=1e-5)
