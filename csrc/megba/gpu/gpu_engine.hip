@@ -1315,7 +1315,9 @@ class GpuEngine final : public Engine<T> {
 
  private:
   int slabWidth() const {
-    return (implicit_ ? 0 : 27) + 20 + (hasInfo_ ? 18 : 0);
+    // must match SlabLayout<EXPL,HASINFO>::SW
+    return (implicit_ ? 0 : 27) + 20 + (hasInfo_ ? 18 : 0) +
+           (implicit_ ? 6 : 0);
   }
   void dispatchAssemble(int bak) {
     auto launch = [&](auto hasInfoTag, auto explTag) {
