@@ -584,22 +584,15 @@ __global__ void kSpmvEtx(int64_t nL, const int* __restrict__ camOf,
                          const T* __restrict__ Hpl, const T* __restrict__ Jc,
                          const T* __restrict__ Jp, const T* __restrict__ info,
                          const T* __restrict__ x, T* __restrict__ out) {
-  // Two consecutive edges per lane: pairs sharing a point combine locally,
-  // halving the wave-scan work per edge; segment pieces that end inside a
-  // lane flush with the scanned prefix of earlier lanes.
   const int lane = threadIdx.x & 63;
-  const int64_t nPair = (nL + 1) / 2;
-  const int64_t nWork = ((nPair + kBlk - 1) / kBlk) * (int64_t)kBlk;
-  for (int64_t t = blockIdx.x * (int64_t)kBlk + threadIdx.x; t < nWork;
-       t += (int64_t)gridDim.x * kBlk) {
-    const int64_t eA = 2 * t;
-    const int64_t eB = eA + 1;
-    const bool actA = eA < nL;
-    const bool actB = eB < nL;
-    const int ptA = ptOf[actA ? eA : nL - 1];
-    const int ptB = actB ? ptOf[eB] : ptA;
-    T oA[3] = {0, 0, 0}, oB[3] = {0, 0, 0};
-    auto compute = [&](int64_t j, T o[3]) {
+  const int64_t nWork = ((nL + kBlk - 1) / kBlk) * (int64_t)kBlk;
+  for (int64_t j0 = blockIdx.x * (int64_t)kBlk + threadIdx.x; j0 < nWork;
+       j0 += (int64_t)gridDim.x * kBlk) {
+    const bool active = j0 < nL;
+    const int64_t j = active ? j0 : nL - 1;
+    const int pt = ptOf[j];
+    T o0 = 0, o1 = 0, o2 = 0;
+    if (active) {
       const T* xc = x + (int64_t)camOf[j] * 9;
       if (IMP) {
         T u0 = T(0), u1 = T(0);
@@ -615,66 +608,35 @@ __global__ void kSpmvEtx(int64_t nL, const int* __restrict__ camOf,
           u1 = w01 * u0 + w11 * u1;
           u0 = a;
         }
-        o[0] = Jp[((int64_t)0) * nL + j] * u0 + Jp[((int64_t)1) * nL + j] * u1;
-        o[1] = Jp[((int64_t)2) * nL + j] * u0 + Jp[((int64_t)3) * nL + j] * u1;
-        o[2] = Jp[((int64_t)4) * nL + j] * u0 + Jp[((int64_t)5) * nL + j] * u1;
+        o0 = Jp[((int64_t)0) * nL + j] * u0 + Jp[((int64_t)1) * nL + j] * u1;
+        o1 = Jp[((int64_t)2) * nL + j] * u0 + Jp[((int64_t)3) * nL + j] * u1;
+        o2 = Jp[((int64_t)4) * nL + j] * u0 + Jp[((int64_t)5) * nL + j] * u1;
       } else {
         for (int i = 0; i < 9; ++i) {
           const T xi = xc[i];
-          o[0] += Hpl[((int64_t)(i * 3 + 0)) * nL + j] * xi;
-          o[1] += Hpl[((int64_t)(i * 3 + 1)) * nL + j] * xi;
-          o[2] += Hpl[((int64_t)(i * 3 + 2)) * nL + j] * xi;
+          o0 += Hpl[((int64_t)(i * 3 + 0)) * nL + j] * xi;
+          o1 += Hpl[((int64_t)(i * 3 + 1)) * nL + j] * xi;
+          o2 += Hpl[((int64_t)(i * 3 + 2)) * nL + j] * xi;
         }
       }
-    };
-    if (actA) compute(eA, oA);
-    if (actB) compute(eB, oB);
-    // lane-local combine; scan runs on the LAST half's (ptB) values
-    T s0, s1, s2;
-    const bool samePt = ptA == ptB;
-    if (samePt) {
-      s0 = oA[0] + oB[0];
-      s1 = oA[1] + oB[1];
-      s2 = oA[2] + oB[2];
-    } else {
-      s0 = oB[0];
-      s1 = oB[1];
-      s2 = oB[2];
     }
     for (int off = 1; off < 64; off <<= 1) {
-      const int ppt = __shfl_up(ptB, off, 64);
-      const T a0 = __shfl_up(s0, off, 64);
-      const T a1 = __shfl_up(s1, off, 64);
-      const T a2 = __shfl_up(s2, off, 64);
-      if (lane >= off && ppt == ptB) {
-        s0 += a0;
-        s1 += a1;
-        s2 += a2;
+      const int ppt = __shfl_up(pt, off, 64);
+      const T a0 = __shfl_up(o0, off, 64);
+      const T a1 = __shfl_up(o1, off, 64);
+      const T a2 = __shfl_up(o2, off, 64);
+      if (lane >= off && ppt == pt) {
+        o0 += a0;
+        o1 += a1;
+        o2 += a2;
       }
     }
-    // flush a segment that ends at this lane's first half
-    const int prevKey = __shfl_up(ptB, 1, 64);
-    const T p0 = __shfl_up(s0, 1, 64);
-    const T p1 = __shfl_up(s1, 1, 64);
-    const T p2 = __shfl_up(s2, 1, 64);
-    if (actA && !samePt) {
-      T f0 = oA[0], f1 = oA[1], f2 = oA[2];
-      if (lane > 0 && prevKey == ptA) {
-        f0 += p0;
-        f1 += p1;
-        f2 += p2;
-      }
-      atomicAdd(&out[3 * ptA], f0);
-      atomicAdd(&out[3 * ptA + 1], f1);
-      atomicAdd(&out[3 * ptA + 2], f2);
-    }
-    const int nextA = __shfl_down(ptA, 1, 64);
-    const bool tail =
-        (actA || actB) && (lane == 63 || nextA != ptB || eB >= nL - 1);
+    const int nextPt = __shfl_down(pt, 1, 64);
+    const bool tail = active && (lane == 63 || nextPt != pt || j0 == nL - 1);
     if (tail) {
-      atomicAdd(&out[3 * ptB], s0);
-      atomicAdd(&out[3 * ptB + 1], s1);
-      atomicAdd(&out[3 * ptB + 2], s2);
+      atomicAdd(&out[3 * pt], o0);
+      atomicAdd(&out[3 * pt + 1], o1);
+      atomicAdd(&out[3 * pt + 2], o2);
     }
   }
 }
@@ -1490,20 +1452,20 @@ class GpuEngine final : public Engine<T> {
     const int bak = cur_ ^ 1;
     if (implicit_) {
       if (hasInfo_)
-        hipLaunchKernelGGL((kSpmvEtx<T, true, true>),
-                           dim3(gridFor((nL_ + 1) / 2)), dim3(kBlk), 0,
-                           stream_, nL_, dCamOf_, dPtOf_, (const T*)nullptr,
-                           dJc_[bak], dJp_[bak], dInfo_, xv, out);
+        hipLaunchKernelGGL((kSpmvEtx<T, true, true>), dim3(gridFor(nL_)),
+                           dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
+                           (const T*)nullptr, dJc_[bak], dJp_[bak], dInfo_, xv,
+                           out);
       else
-        hipLaunchKernelGGL((kSpmvEtx<T, true, false>),
-                           dim3(gridFor((nL_ + 1) / 2)), dim3(kBlk), 0,
-                           stream_, nL_, dCamOf_, dPtOf_, (const T*)nullptr,
-                           dJc_[bak], dJp_[bak], (const T*)nullptr, xv, out);
+        hipLaunchKernelGGL((kSpmvEtx<T, true, false>), dim3(gridFor(nL_)),
+                           dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
+                           (const T*)nullptr, dJc_[bak], dJp_[bak],
+                           (const T*)nullptr, xv, out);
     } else {
-      hipLaunchKernelGGL((kSpmvEtx<T, false, false>),
-                         dim3(gridFor((nL_ + 1) / 2)), dim3(kBlk), 0, stream_,
-                         nL_, dCamOf_, dPtOf_, dHpl_, (const T*)nullptr,
-                         (const T*)nullptr, (const T*)nullptr, xv, out);
+      hipLaunchKernelGGL((kSpmvEtx<T, false, false>), dim3(gridFor(nL_)),
+                         dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_, dHpl_,
+                         (const T*)nullptr, (const T*)nullptr,
+                         (const T*)nullptr, xv, out);
     }
   }
   void spmvEx(const T* wv, T* out) {
