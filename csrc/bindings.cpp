@@ -359,7 +359,7 @@ PYBIND11_MODULE(_core, m) {
       .def_property_readonly("N", [](const PyJetVec& a) { return a.v.N; })
       .def("to_numpy", &jvDownload);
   m.def("jv_scalar", [](double s, int N) { return PyJetVec{jvScalar<double>(s, N)}; });
-  auto binop = [](const char* name, JvOp op, py::module_& m) {};
+  m.def("jv_pool_trim", []() { jvPoolTrim(); });
   m.def("jv_add", [](const PyJetVec& a, const PyJetVec& b) { return PyJetVec{jvBinary(JvOp::Add, a.v, b.v)}; });
   m.def("jv_sub", [](const PyJetVec& a, const PyJetVec& b) { return PyJetVec{jvBinary(JvOp::Sub, a.v, b.v)}; });
   m.def("jv_mul", [](const PyJetVec& a, const PyJetVec& b) { return PyJetVec{jvBinary(JvOp::Mul, a.v, b.v)}; });

@@ -25,11 +25,11 @@ class JV:
     def __init__(self, raw):
         self.raw = raw
 
-    def _coerce(self, o, n_grad=12):
+    def _coerce(self, o):
         if isinstance(o, JV):
             return o.raw
         if isinstance(o, (int, float)):
-            return _core.jv_scalar(float(o), n_grad)
+            return _core.jv_scalar(float(o), self.raw.N)
         return o
 
     def __add__(self, o):

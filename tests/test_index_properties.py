@@ -1,0 +1,63 @@
+"""Property tests for the index builder: partition invariants across random
+problem shapes and world sizes."""
+import numpy as np
+from hypothesis import given, settings, strategies as st
+
+import megba_amd as mb
+
+
+@st.composite
+def problems(draw):
+    ncam = draw(st.integers(2, 12))
+    npt = draw(st.integers(4, 60))
+    nobs = draw(st.integers(2 * npt + 2 * ncam, 400))
+    seed = draw(st.integers(0, 10_000))
+    return ncam, npt, nobs, seed
+
+
+@given(problems(), st.integers(1, 6))
+@settings(max_examples=25, deadline=None)
+def test_partition_invariants(shape, world):
+    ncam, npt, nobs, seed = shape
+    if world > npt:
+        return
+    cams, pts, ci, pi, meas = mb.synthesize_bal(ncam, npt, nobs, seed=seed)
+    p = mb.BAProblem(cams, pts, ci, pi, meas)
+    p.build(device="cpu", rank=0, world_size=world)
+    ii = p.index_info()
+    split, pt_split = ii["split"], ii["pt_split"]
+    pt_of, cam_of, perm = ii["pt_of"], ii["cam_of"], ii["perm"]
+    # (pt, cam)-lexicographic order
+    key = pt_of.astype(np.int64) * (ncam + 1) + cam_of
+    assert (np.diff(key) >= 0).all()
+    # permutation is a bijection preserving the data
+    assert sorted(perm) == list(range(nobs))
+    np.testing.assert_array_equal(pt_of, pi[perm])
+    np.testing.assert_array_equal(cam_of, ci[perm])
+    # splits cover everything, are point-aligned, and match pt_split
+    assert split[0] == 0 and split[-1] == nobs
+    assert pt_split[0] == 0 and pt_split[-1] == npt
+    rowptr = ii["pt_rowptr"]
+    for r in range(world + 1):
+        assert split[r] == rowptr[pt_split[r]]
+    for r in range(world):
+        # every edge in rank r's range has a point in rank r's point range
+        lo, hi = split[r], split[r + 1]
+        if lo < hi:
+            assert pt_of[lo:hi].min() >= pt_split[r]
+            assert pt_of[lo:hi].max() < pt_split[r + 1]
+
+
+def test_solve_fp32_tracks_fp64():
+    cams, pts, ci, pi, meas = mb.synthesize_bal(12, 120, 1100, seed=3)
+    p64 = mb.BAProblem(cams, pts, ci, pi, meas)
+    p64.build(device="cpu", dtype="float64")
+    p32 = mb.BAProblem(cams, pts, ci, pi, meas)
+    p32.build(device="cpu", dtype="float32")
+    kw = dict(max_iter=5, solver_tol=1e-3, solver_max_iter=100,
+              solver_refuse_ratio=1e6, verbose=False)
+    r64, r32 = p64.solve(**kw), p32.solve(**kw)
+    # fp32 follows the fp64 trajectory loosely
+    np.testing.assert_allclose(r32["iters"][0]["chi2"],
+                               r64["iters"][0]["chi2"], rtol=1e-4)
+    assert r32["final_chi2"] < 1.5 * r64["final_chi2"] + 1e-3
