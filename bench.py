@@ -77,9 +77,14 @@ def main():
             from megba_amd.dist import gloo_allreduce_callback
             allreduce = gloo_allreduce_callback()
 
+    device_index = local_rank
+    if args.device == "gpu":
+        from megba_amd import _core
+        ngpu = max(1, _core.hip_device_count())
+        device_index = local_rank % ngpu
     p = mb.BAProblem(cams, pts, ci, pi, meas)
     p.build(device=args.device, dtype=args.dtype, rank=rank,
-            world_size=world, device_index=local_rank, diff=args.diff,
+            world_size=world, device_index=device_index, diff=args.diff,
             schur=args.schur, allreduce=allreduce, rccl_id=rccl_id)
 
     # Fixed-work steps: every LM iteration runs the full solver_max_iter=100
