@@ -1001,7 +1001,8 @@ class GpuEngine final : public Engine<T> {
       up(dCamPos_, camPos.data(), nL_);
       up(dPtOfCam_, ptOfCam.data(), nL_);
       std::vector<int> cCam, cLo, cHi;
-      constexpr int CHUNK = 256;
+      int CHUNK = 256;  // cam-chunk rows (tunable: MEGBA_CHUNK)
+      if (const char* c = getenv("MEGBA_CHUNK")) CHUNK = std::atoi(c);
       for (int c = 0; c < ncam_; ++c)
         for (int s = rowPtr[c]; s < rowPtr[c + 1]; s += CHUNK) {
           cCam.push_back(c);
