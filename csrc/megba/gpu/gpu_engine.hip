@@ -419,13 +419,16 @@ __global__ void kAssembleEdge(int64_t nL, const int* __restrict__ camOf,
       for (int k = 0; k < 3; ++k)
         gp3[k] = -(jp[0][k] * wr[0] + jp[1][k] * wr[1]);
     }
-    // segmented scan over the wave's point runs
+    // segmented scan over the wave's point runs (early exit once no lane
+    // continues a segment at this distance — monotone in off)
     for (int off = 1; off < 64; off <<= 1) {
       const int ppt = __shfl_up(pt, off, 64);
+      const bool join = lane >= off && ppt == pt;
+      if (__ballot(join) == 0ull) break;
       T a6[6], a3[3];
       for (int k = 0; k < 6; ++k) a6[k] = __shfl_up(hll6[k], off, 64);
       for (int k = 0; k < 3; ++k) a3[k] = __shfl_up(gp3[k], off, 64);
-      if (lane >= off && ppt == pt) {
+      if (join) {
         for (int k = 0; k < 6; ++k) hll6[k] += a6[k];
         for (int k = 0; k < 3; ++k) gp3[k] += a3[k];
       }
@@ -622,10 +625,16 @@ __global__ void kSpmvEtx(int64_t nL, const int* __restrict__ camOf,
     }
     for (int off = 1; off < 64; off <<= 1) {
       const int ppt = __shfl_up(pt, off, 64);
+      const bool join = lane >= off && ppt == pt;
+      // Point runs are short (degree ~5): once no lane continues a segment
+      // at distance `off`, no lane can at any larger distance either — skip
+      // the remaining dependent shuffle rounds (the scan chain is this
+      // kernel's issue-stall bound, 70% SQ_WAIT_INST_ANY).
+      if (__ballot(join) == 0ull) break;
       const T a0 = __shfl_up(o0, off, 64);
       const T a1 = __shfl_up(o1, off, 64);
       const T a2 = __shfl_up(o2, off, 64);
-      if (lane >= off && ppt == pt) {
+      if (join) {
         o0 += a0;
         o1 += a1;
         o2 += a2;
