@@ -114,9 +114,11 @@ struct PyProblem {
       std::string id;
       if (!rcclId.is_none()) id = py::cast<std::string>(rcclId);
       if (isDouble)
-        engD = makeGpuEngine<double>(prob, ix, opt, id, cf);
+        engD = makeGpuEngine<double>(prob, ix, opt, id, cf,
+                                     wrapAllreduce<double>(allreduce));
       else
-        engF = makeGpuEngine<float>(prob, ix, opt, id);
+        engF = makeGpuEngine<float>(prob, ix, opt, id, nullptr,
+                                    wrapAllreduce<float>(allreduce));
 #else
       MEGBA_CHECK(false, "built without GPU support");
 #endif

@@ -883,8 +883,9 @@ class GpuEngine final : public Engine<T> {
  public:
   GpuEngine(const BAProblemHost& prob, const ProblemIndex& ix,
             const ProblemOption& opt, const std::string& rcclId,
-            CustomForward<T> customForward)
+            CustomForward<T> customForward, HostAllreduce<T> hostAllreduce)
       : customFwd_(std::move(customForward)),
+        hostAr_(std::move(hostAllreduce)),
         rank_(opt.rank),
         world_(opt.worldSize),
         ncam_(ix.ncam),
@@ -905,12 +906,17 @@ class GpuEngine final : public Engine<T> {
     hasInfo_ = !ix.infoSorted.empty();
 
     if (world_ > 1) {
-      MEGBA_CHECK(rcclId.size() == sizeof(ncclUniqueId),
-                  "worldSize>1 requires the RCCL unique id");
-      ncclUniqueId id;
-      std::memcpy(&id, rcclId.data(), sizeof(id));
-      RCCL_CHECK(ncclCommInitRank(&comm_, world_, id, rank_));
-      hasComm_ = true;
+      if (rcclId.empty() && hostAr_) {
+        // gloo-backed testing fallback (several ranks may share one GPU)
+      } else {
+        MEGBA_CHECK(rcclId.size() == sizeof(ncclUniqueId),
+                    "worldSize>1 requires the RCCL unique id (or a host "
+                    "allreduce fallback)");
+        ncclUniqueId id;
+        std::memcpy(&id, rcclId.data(), sizeof(id));
+        RCCL_CHECK(ncclCommInitRank(&comm_, world_, id, rank_));
+        hasComm_ = true;
+      }
     }
 
     // Static per-edge data (primary = (pt,cam)-sorted order).
@@ -1400,9 +1406,17 @@ class GpuEngine final : public Engine<T> {
   }
   // Read the device scalar accumulator, allreducing across ranks first.
   double globalScalar(ncclRedOp_t op) {
-    if (hasComm_)
+    if (hasComm_) {
       RCCL_CHECK(ncclAllReduce(scalarPtr(), scalarPtr(), 1, ncclDouble, op,
                                comm_, stream_));
+      return readScalar(scalarPtr());
+    }
+    if (hostAr_ && world_ > 1) {
+      double h = readScalar(scalarPtr());
+      T v = (T)h;
+      hostAr_(&v, 1, op == ncclMax ? 'm' : 's');
+      return (double)v;
+    }
     return readScalar(scalarPtr());
   }
   double readScalar(double* dptr) {
@@ -1413,10 +1427,20 @@ class GpuEngine final : public Engine<T> {
     return h;
   }
   void allreduce(T* buf, int64_t n, ncclRedOp_t op) {
-    if (!hasComm_ || n == 0) return;
-    RCCL_CHECK(ncclAllReduce(buf, buf, n,
-                             sizeof(T) == 8 ? ncclDouble : ncclFloat, op,
-                             comm_, stream_));
+    if (n == 0 || world_ == 1) return;
+    if (hasComm_) {
+      RCCL_CHECK(ncclAllReduce(buf, buf, n,
+                               sizeof(T) == 8 ? ncclDouble : ncclFloat, op,
+                               comm_, stream_));
+      return;
+    }
+    if (!hostAr_) return;
+    // gloo-backed testing fallback: bounce through the host.
+    sync();
+    std::vector<T> h(n);
+    HIP_CHECK(hipMemcpy(h.data(), buf, n * sizeof(T), hipMemcpyDeviceToHost));
+    hostAr_(h.data(), n, op == ncclMax ? 'm' : 's');
+    HIP_CHECK(hipMemcpy(buf, h.data(), n * sizeof(T), hipMemcpyHostToDevice));
   }
   void reduceDetAsync(const T* a, const T* b, int64_t n, ROp op, double* out) {
     switch (op) {
@@ -1573,6 +1597,7 @@ class GpuEngine final : public Engine<T> {
   ncclComm_t comm_{};
   bool hasComm_ = false;
   CustomForward<T> customFwd_;
+  HostAllreduce<T> hostAr_;
   int rank_, world_, ncam_, npt_;
   int ptLo_ = 0, ptHi_ = 0, npL_ = 0;
   int64_t e0_ = 0, e1_ = 0, nL_ = 0, nc_ = 0, np_ = 0, dim_ = 0;
@@ -1604,17 +1629,19 @@ std::unique_ptr<Engine<T>> makeGpuEngine(const BAProblemHost& prob,
                                          const ProblemIndex& ix,
                                          const ProblemOption& opt,
                                          const std::string& rcclId,
-                                         CustomForward<T> customForward) {
+                                         CustomForward<T> customForward,
+                                         HostAllreduce<T> hostAllreduce) {
   return std::make_unique<GpuEngine<T>>(prob, ix, opt, rcclId,
-                                        std::move(customForward));
+                                        std::move(customForward),
+                                        std::move(hostAllreduce));
 }
 
 template std::unique_ptr<Engine<double>> makeGpuEngine<double>(
     const BAProblemHost&, const ProblemIndex&, const ProblemOption&,
-    const std::string&, CustomForward<double>);
+    const std::string&, CustomForward<double>, HostAllreduce<double>);
 template std::unique_ptr<Engine<float>> makeGpuEngine<float>(
     const BAProblemHost&, const ProblemIndex&, const ProblemOption&,
-    const std::string&, CustomForward<float>);
+    const std::string&, CustomForward<float>, HostAllreduce<float>);
 
 std::string rcclUniqueIdString() {
   ncclUniqueId id;

@@ -6,6 +6,7 @@
 #include <string>
 
 #include "../common.hpp"
+#include "../cpu_engine.hpp"  // HostAllreduce
 #include "../custom.hpp"
 #include "../engine.hpp"
 #include "../problem.hpp"
@@ -16,12 +17,17 @@ namespace megba {
 // by rank 0 (rcclUniqueIdString) and broadcast out-of-band (bench.py uses a
 // torch.distributed gloo store for the exchange; RCCL itself then runs
 // natively over xGMI with no Python in the loop).
+// `hostAllreduce`: testing fallback — when worldSize>1 and rcclId is empty,
+// collectives bounce through the host callback (e.g. torch.distributed gloo),
+// so the sharded GPU code paths can be exercised by several ranks sharing
+// one GPU.  Production multi-GPU uses RCCL (rcclId non-empty).
 template <typename T>
 std::unique_ptr<Engine<T>> makeGpuEngine(const BAProblemHost& prob,
                                          const ProblemIndex& ix,
                                          const ProblemOption& opt,
                                          const std::string& rcclId,
-                                         CustomForward<T> customForward = nullptr);
+                                         CustomForward<T> customForward = nullptr,
+                                         HostAllreduce<T> hostAllreduce = nullptr);
 
 std::string rcclUniqueIdString();
 int hipDeviceCountSafe();
