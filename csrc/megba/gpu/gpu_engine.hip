@@ -1280,7 +1280,7 @@ class GpuEngine final : public Engine<T> {
     HIP_CHECK(hipMemcpy(hc.data(), dParams_, nc_ * sizeof(T),
                         hipMemcpyDeviceToHost));
     for (int64_t i = 0; i < nc_; ++i) cams[i] = (double)hc[i];
-    if (hasComm_) {
+    if (world_ > 1 && (hasComm_ || hostAr_)) {
       // point shards: zero the non-local entries, allreduce-sum.
       if (!dPtMerge_) dPtMerge_ = dalloc<T>(np_);
       T* tmp = dPtMerge_;
