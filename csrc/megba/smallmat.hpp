@@ -83,7 +83,7 @@ MEGBA_HD inline bool spdInvertPacked(const T* a, T* inv) {
   for (int i = 0, ii = 0; i < D; ii += ++i) {
     const T dinv = T(1) / L[ii + i];
     L[ii + i] = dinv;
-    for (int j = 0, jj = 0; j < i; jj += ++j) {
+    for (int j = 0; j < i; ++j) {
       T s = T(0);
       // s = -sum_{k=j..i-1} L[i][k] * Linv[k][j]; Linv rows < i already done,
       // and row i's entries at k < j are the only overwritten ones (not read).

@@ -81,7 +81,11 @@ def load_bal(path):
     with open(path) as f:
         header = f.readline().split()
         ncam, npt, nobs = int(header[0]), int(header[1]), int(header[2])
-        body = np.fromstring(f.read(), sep=" ")
+        text = f.read()
+    try:
+        body = np.fromstring(text, sep=" ")
+    except Exception:  # numpy versions without text-mode fromstring
+        body = np.array(text.split(), dtype=np.float64)
     obs = body[:nobs * 4].reshape(nobs, 4)
     cam_idx = obs[:, 0].astype(np.int32)
     pt_idx = obs[:, 1].astype(np.int32)
