@@ -586,23 +586,14 @@ __global__ void kInvertJitter(int nBlk, const T* __restrict__ Hd,
 // 1.08 GB block read keeps full memory parallelism (the fused scan variant
 // was latency-bound at 44% of HBM peak, 70% SQ_WAIT_INST_ANY); (B) a tiny
 // per-point reduction over the 3-stream contribution array (~0.14 GB).
-// XLDS: stage the whole replicated camera vector (9*ncam, 128 KB on
-// Venice) in LDS so the per-edge 9-load gather hits LDS instead of L2
-// (the gather is this kernel's gap to the E*w kernel's bandwidth).
-template <typename T, bool IMP, bool HASINFO, bool XLDS>
-__global__ void kSpmvEtxA(int64_t nL, int nc, const int* __restrict__ camOf,
+template <typename T, bool IMP, bool HASINFO>
+__global__ void kSpmvEtxA(int64_t nL, const int* __restrict__ camOf,
                           const T* __restrict__ Hpl, const T* __restrict__ Jc,
                           const T* __restrict__ Jp, const T* __restrict__ info,
                           const T* __restrict__ x, T* __restrict__ contrib) {
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  T* xs = (T*)smem;
-  if (XLDS) {
-    for (int i = threadIdx.x; i < nc; i += kBlk) xs[i] = x[i];
-    __syncthreads();
-  }
   for (int64_t j = blockIdx.x * (int64_t)kBlk + threadIdx.x; j < nL;
        j += (int64_t)gridDim.x * kBlk) {
-    const T* xc = (XLDS ? xs : x) + (int64_t)camOf[j] * 9;
+    const T* xc = x + (int64_t)camOf[j] * 9;
     T o0 = 0, o1 = 0, o2 = 0;
     if (IMP) {
       T u0 = T(0), u1 = T(0);
@@ -1491,36 +1482,24 @@ class GpuEngine final : public Engine<T> {
     blockMatVec<3, 0>(npL_, dHllInv_ + (int64_t)ptLo_ * 9,
                       in + (int64_t)ptLo_ * 3, out + (int64_t)ptLo_ * 3);
   }
-  template <bool IMP, bool HASINFO>
-  void launchEtxA(const T* hpl, const T* jc, const T* jp, const T* info,
-                  const T* xv) {
-    const size_t xBytes = (size_t)nc_ * sizeof(T);
-    if (xBytes <= 120 * 1024) {
-      static bool attrSet = false;
-      if (!attrSet) {
-        (void)hipFuncSetAttribute(
-            reinterpret_cast<const void*>(&kSpmvEtxA<T, IMP, HASINFO, true>),
-            hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
-        attrSet = true;
-      }
-      hipLaunchKernelGGL((kSpmvEtxA<T, IMP, HASINFO, true>),
-                         dim3(gridFor(nL_)), dim3(kBlk), xBytes, stream_, nL_,
-                         (int)nc_, dCamOf_, hpl, jc, jp, info, xv, dContrib_);
-    } else {
-      hipLaunchKernelGGL((kSpmvEtxA<T, IMP, HASINFO, false>),
-                         dim3(gridFor(nL_)), dim3(kBlk), 0, stream_, nL_,
-                         (int)nc_, dCamOf_, hpl, jc, jp, info, xv, dContrib_);
-    }
-  }
   void spmvEtx(const T* xv, T* out) {
     const int bak = cur_ ^ 1;
     if (implicit_) {
       if (hasInfo_)
-        launchEtxA<true, true>(nullptr, dJc_[bak], dJp_[bak], dInfo_, xv);
+        hipLaunchKernelGGL((kSpmvEtxA<T, true, true>), dim3(gridFor(nL_)),
+                           dim3(kBlk), 0, stream_, nL_, dCamOf_,
+                           (const T*)nullptr, dJc_[bak], dJp_[bak], dInfo_, xv,
+                           dContrib_);
       else
-        launchEtxA<true, false>(nullptr, dJc_[bak], dJp_[bak], nullptr, xv);
+        hipLaunchKernelGGL((kSpmvEtxA<T, true, false>), dim3(gridFor(nL_)),
+                           dim3(kBlk), 0, stream_, nL_, dCamOf_,
+                           (const T*)nullptr, dJc_[bak], dJp_[bak],
+                           (const T*)nullptr, xv, dContrib_);
     } else {
-      launchEtxA<false, false>(dHpl_, nullptr, nullptr, nullptr, xv);
+      hipLaunchKernelGGL((kSpmvEtxA<T, false, false>), dim3(gridFor(nL_)),
+                         dim3(kBlk), 0, stream_, nL_, dCamOf_, dHpl_,
+                         (const T*)nullptr, (const T*)nullptr,
+                         (const T*)nullptr, xv, dContrib_);
     }
     hipLaunchKernelGGL(kSpmvEtxB<T>, dim3(gridFor(npL_)), dim3(kBlk), 0,
                        stream_, npL_, ptLo_, dPtRowPtrLoc_, dContrib_, nL_,
