@@ -581,67 +581,72 @@ __global__ void kInvertJitter(int nBlk, const T* __restrict__ Hd,
 // replicated camera vector x is L2-resident), wave segmented-scan over the
 // point runs, atomics only at run tails.  IMP: matrix-free from J
 // (reference C23).
-// E^T x in two phases: (A) per-edge products into a point-sorted
-// contribution array [3][nL] — pure streaming, no cross-lane work, so the
-// 1.08 GB block read keeps full memory parallelism (the fused scan variant
-// was latency-bound at 44% of HBM peak, 70% SQ_WAIT_INST_ANY); (B) a tiny
-// per-point reduction over the 3-stream contribution array (~0.14 GB).
 template <typename T, bool IMP, bool HASINFO>
-__global__ void kSpmvEtxA(int64_t nL, const int* __restrict__ camOf,
-                          const T* __restrict__ Hpl, const T* __restrict__ Jc,
-                          const T* __restrict__ Jp, const T* __restrict__ info,
-                          const T* __restrict__ x, T* __restrict__ contrib) {
-  for (int64_t j = blockIdx.x * (int64_t)kBlk + threadIdx.x; j < nL;
-       j += (int64_t)gridDim.x * kBlk) {
-    const T* xc = x + (int64_t)camOf[j] * 9;
+__global__ void kSpmvEtx(int64_t nL, const int* __restrict__ camOf,
+                         const int* __restrict__ ptOf,
+                         const T* __restrict__ Hpl, const T* __restrict__ Jc,
+                         const T* __restrict__ Jp, const T* __restrict__ info,
+                         const T* __restrict__ x, T* __restrict__ out) {
+  const int lane = threadIdx.x & 63;
+  const int64_t nWork = ((nL + kBlk - 1) / kBlk) * (int64_t)kBlk;
+  for (int64_t j0 = blockIdx.x * (int64_t)kBlk + threadIdx.x; j0 < nWork;
+       j0 += (int64_t)gridDim.x * kBlk) {
+    const bool active = j0 < nL;
+    const int64_t j = active ? j0 : nL - 1;
+    const int pt = ptOf[j];
     T o0 = 0, o1 = 0, o2 = 0;
-    if (IMP) {
-      T u0 = T(0), u1 = T(0);
-      for (int i = 0; i < 9; ++i) {
-        const T xi = xc[i];
-        u0 += Jc[((int64_t)(i * 2 + 0)) * nL + j] * xi;
-        u1 += Jc[((int64_t)(i * 2 + 1)) * nL + j] * xi;
-      }
-      if (HASINFO) {
-        const T w00 = info[3 * j], w01 = info[3 * j + 1], w11 = info[3 * j + 2];
-        const T a = w00 * u0 + w01 * u1;
-        u1 = w01 * u0 + w11 * u1;
-        u0 = a;
-      }
-      o0 = Jp[((int64_t)0) * nL + j] * u0 + Jp[((int64_t)1) * nL + j] * u1;
-      o1 = Jp[((int64_t)2) * nL + j] * u0 + Jp[((int64_t)3) * nL + j] * u1;
-      o2 = Jp[((int64_t)4) * nL + j] * u0 + Jp[((int64_t)5) * nL + j] * u1;
-    } else {
-      for (int i = 0; i < 9; ++i) {
-        const T xi = xc[i];
-        o0 += Hpl[((int64_t)(i * 3 + 0)) * nL + j] * xi;
-        o1 += Hpl[((int64_t)(i * 3 + 1)) * nL + j] * xi;
-        o2 += Hpl[((int64_t)(i * 3 + 2)) * nL + j] * xi;
+    if (active) {
+      const T* xc = x + (int64_t)camOf[j] * 9;
+      if (IMP) {
+        T u0 = T(0), u1 = T(0);
+        for (int i = 0; i < 9; ++i) {
+          const T xi = xc[i];
+          u0 += Jc[((int64_t)(i * 2 + 0)) * nL + j] * xi;
+          u1 += Jc[((int64_t)(i * 2 + 1)) * nL + j] * xi;
+        }
+        if (HASINFO) {
+          const T w00 = info[3 * j], w01 = info[3 * j + 1],
+                  w11 = info[3 * j + 2];
+          const T a = w00 * u0 + w01 * u1;
+          u1 = w01 * u0 + w11 * u1;
+          u0 = a;
+        }
+        o0 = Jp[((int64_t)0) * nL + j] * u0 + Jp[((int64_t)1) * nL + j] * u1;
+        o1 = Jp[((int64_t)2) * nL + j] * u0 + Jp[((int64_t)3) * nL + j] * u1;
+        o2 = Jp[((int64_t)4) * nL + j] * u0 + Jp[((int64_t)5) * nL + j] * u1;
+      } else {
+        for (int i = 0; i < 9; ++i) {
+          const T xi = xc[i];
+          o0 += Hpl[((int64_t)(i * 3 + 0)) * nL + j] * xi;
+          o1 += Hpl[((int64_t)(i * 3 + 1)) * nL + j] * xi;
+          o2 += Hpl[((int64_t)(i * 3 + 2)) * nL + j] * xi;
+        }
       }
     }
-    contrib[j] = o0;
-    contrib[nL + j] = o1;
-    contrib[2 * nL + j] = o2;
-  }
-}
-
-template <typename T>
-__global__ void kSpmvEtxB(int npL, int ptLo, const int* __restrict__ rowPtrLoc,
-                          const T* __restrict__ contrib, int64_t nL,
-                          T* __restrict__ out) {
-  for (int64_t i = blockIdx.x * (int64_t)kBlk + threadIdx.x; i < npL;
-       i += (int64_t)gridDim.x * kBlk) {
-    const int lo = rowPtrLoc[i], hi = rowPtrLoc[i + 1];
-    T o0 = 0, o1 = 0, o2 = 0;
-    for (int j = lo; j < hi; ++j) {
-      o0 += contrib[j];
-      o1 += contrib[nL + j];
-      o2 += contrib[2 * nL + j];
+    for (int off = 1; off < 64; off <<= 1) {
+      const int ppt = __shfl_up(pt, off, 64);
+      const bool join = lane >= off && ppt == pt;
+      // Point runs are short (degree ~5): once no lane continues a segment
+      // at distance `off`, no lane can at any larger distance either — skip
+      // the remaining dependent shuffle rounds (the scan chain is this
+      // kernel's issue-stall bound, 70% SQ_WAIT_INST_ANY).
+      if (__ballot(join) == 0ull) break;
+      const T a0 = __shfl_up(o0, off, 64);
+      const T a1 = __shfl_up(o1, off, 64);
+      const T a2 = __shfl_up(o2, off, 64);
+      if (join) {
+        o0 += a0;
+        o1 += a1;
+        o2 += a2;
+      }
     }
-    T* op = out + ((int64_t)ptLo + i) * 3;
-    op[0] = o0;
-    op[1] = o1;
-    op[2] = o2;
+    const int nextPt = __shfl_down(pt, 1, 64);
+    const bool tail = active && (lane == 63 || nextPt != pt || j0 == nL - 1);
+    if (tail) {
+      atomicAdd(&out[3 * pt], o0);
+      atomicAdd(&out[3 * pt + 1], o1);
+      atomicAdd(&out[3 * pt + 2], o2);
+    }
   }
 }
 
@@ -1026,14 +1031,6 @@ class GpuEngine final : public Engine<T> {
       up(dChCam_, cCam.data(), nChunks_);
       up(dChLo_, cLo.data(), nChunks_);
       up(dChHi_, cHi.data(), nChunks_);
-    }
-    dContrib_ = dalloc<T>(nL_ * 3);
-    {
-      std::vector<int> rpl(npL_ + 1);
-      for (int i = 0; i <= npL_; ++i)
-        rpl[i] = (int)(ix.ptRowPtr[ptLo_ + i] - e0_);
-      dPtRowPtrLoc_ = dalloc<int>(npL_ + 1);
-      up(dPtRowPtrLoc_, rpl.data(), npL_ + 1);
     }
     dSlab_ = dalloc<T>(nL_ * slabWidth());
     if (!implicit_) {
@@ -1483,27 +1480,27 @@ class GpuEngine final : public Engine<T> {
                       in + (int64_t)ptLo_ * 3, out + (int64_t)ptLo_ * 3);
   }
   void spmvEtx(const T* xv, T* out) {
+    hipLaunchKernelGGL(kZeroRange<T>, dim3(gridFor((int64_t)npL_ * 3)),
+                       dim3(kBlk), 0, stream_, out + (int64_t)ptLo_ * 3,
+                       (int64_t)npL_ * 3);
     const int bak = cur_ ^ 1;
     if (implicit_) {
       if (hasInfo_)
-        hipLaunchKernelGGL((kSpmvEtxA<T, true, true>), dim3(gridFor(nL_)),
-                           dim3(kBlk), 0, stream_, nL_, dCamOf_,
+        hipLaunchKernelGGL((kSpmvEtx<T, true, true>), dim3(gridFor(nL_)),
+                           dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
                            (const T*)nullptr, dJc_[bak], dJp_[bak], dInfo_, xv,
-                           dContrib_);
+                           out);
       else
-        hipLaunchKernelGGL((kSpmvEtxA<T, true, false>), dim3(gridFor(nL_)),
-                           dim3(kBlk), 0, stream_, nL_, dCamOf_,
+        hipLaunchKernelGGL((kSpmvEtx<T, true, false>), dim3(gridFor(nL_)),
+                           dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
                            (const T*)nullptr, dJc_[bak], dJp_[bak],
-                           (const T*)nullptr, xv, dContrib_);
+                           (const T*)nullptr, xv, out);
     } else {
-      hipLaunchKernelGGL((kSpmvEtxA<T, false, false>), dim3(gridFor(nL_)),
-                         dim3(kBlk), 0, stream_, nL_, dCamOf_, dHpl_,
+      hipLaunchKernelGGL((kSpmvEtx<T, false, false>), dim3(gridFor(nL_)),
+                         dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_, dHpl_,
                          (const T*)nullptr, (const T*)nullptr,
-                         (const T*)nullptr, xv, dContrib_);
+                         (const T*)nullptr, xv, out);
     }
-    hipLaunchKernelGGL(kSpmvEtxB<T>, dim3(gridFor(npL_)), dim3(kBlk), 0,
-                       stream_, npL_, ptLo_, dPtRowPtrLoc_, dContrib_, nL_,
-                       out);
   }
   void spmvEx(const T* wv, T* out) {
     HIP_CHECK(hipMemsetAsync(out, 0, nc_ * sizeof(T), stream_));
@@ -1611,7 +1608,7 @@ class GpuEngine final : public Engine<T> {
   int cur_ = 0;
   int nChunks_ = 0;
   int *dCamOf_{}, *dPtOf_{}, *dChCam_{}, *dChLo_{}, *dChHi_{}, *dFail_{};
-  int *dCamPos_{}, *dPtOfCam_{}, *dPtRowPtrLoc_{};
+  int *dCamPos_{}, *dPtOfCam_{};
   T *dMeas_{}, *dInfo_{}, *dLeaf_{}, *dMeasSplit_{};
   unsigned char *dCamFixed_{}, *dPtFixed_{};
   T *dParams_{}, *dParamsBak_{};
@@ -1620,7 +1617,7 @@ class GpuEngine final : public Engine<T> {
   T *dHppD_{}, *dHllD_{}, *dHppInv_{}, *dHllInv_{};
   T *dDeltaX_{}, *dDeltaXBak_{};
   T *dP_{}, *dRr_{}, *dZ_{}, *dQ_{}, *dV_{}, *dW_{}, *dTemp_{}, *dXBak_{},
-      *dXBakPrev_{}, *dPtMerge_{}, *dContrib_{};
+      *dXBakPrev_{}, *dPtMerge_{};
   hipGraphExec_t pcgGraphExec_{};
   bool pcgGraphTried_ = false;
   double* dPart_{};
