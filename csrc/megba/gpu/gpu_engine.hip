@@ -999,6 +999,7 @@ class GpuEngine final : public Engine<T> {
 
     dPart_ = dalloc<double>(kRedBlocks + 8);
     dFail_ = dalloc<int>(2);
+    HIP_CHECK(hipHostMalloc((void**)&hScalar_, sizeof(double)));
 
     // Cam-sorted view of the local edges: slab positions + chunk table.
     {
@@ -1041,6 +1042,7 @@ class GpuEngine final : public Engine<T> {
   }
 
   ~GpuEngine() override {
+    if (hScalar_) (void)hipHostFree(hScalar_);
     if (pcgGraphExec_) (void)hipGraphExecDestroy(pcgGraphExec_);
     for (void* p : allocs_) (void)hipFree(p);
     if (hasComm_) (void)ncclCommDestroy(comm_);
@@ -1420,11 +1422,11 @@ class GpuEngine final : public Engine<T> {
     return readScalar(scalarPtr());
   }
   double readScalar(double* dptr) {
-    double h = 0;
-    HIP_CHECK(hipMemcpyAsync(&h, dptr, sizeof(double), hipMemcpyDeviceToHost,
-                             stream_));
+    // pinned-host destination: the PCG loop does one of these per iteration
+    HIP_CHECK(hipMemcpyAsync(hScalar_, dptr, sizeof(double),
+                             hipMemcpyDeviceToHost, stream_));
     sync();
-    return h;
+    return *hScalar_;
   }
   void allreduce(T* buf, int64_t n, ncclRedOp_t op) {
     if (n == 0 || world_ == 1) return;
@@ -1621,6 +1623,7 @@ class GpuEngine final : public Engine<T> {
   hipGraphExec_t pcgGraphExec_{};
   bool pcgGraphTried_ = false;
   double* dPart_{};
+  double* hScalar_{};
   std::vector<void*> allocs_;
 };
 
