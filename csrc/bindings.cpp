@@ -425,9 +425,8 @@ PYBIND11_MODULE(_core, m) {
   m.def("rccl_unique_id", []() { return py::bytes(rcclUniqueIdString()); });
   m.def("hip_device_count", []() { return hipDeviceCountSafe(); });
   m.def("hip_mem_info", []() {
-    size_t freeB = 0, totalB = 0;
-    if (hipMemGetInfo(&freeB, &totalB) != hipSuccess) return py::make_tuple(0, 0);
-    return py::make_tuple((long long)freeB, (long long)totalB);
+    auto p = hipMemInfoSafe();
+    return py::make_tuple(p.first, p.second);
   });
 #else
   m.attr("has_gpu_support") = false;

@@ -33,6 +33,7 @@
 #include <cstdlib>
 #include <cstring>
 #include <type_traits>
+#include <utility>
 #include <vector>
 
 #include "../analytical.hpp"
@@ -1656,6 +1657,12 @@ int hipDeviceCountSafe() {
   int n = 0;
   if (hipGetDeviceCount(&n) != hipSuccess) return 0;
   return n;
+}
+
+std::pair<long long, long long> hipMemInfoSafe() {
+  size_t freeB = 0, totalB = 0;
+  if (hipMemGetInfo(&freeB, &totalB) != hipSuccess) return {0, 0};
+  return {(long long)freeB, (long long)totalB};
 }
 
 }  // namespace megba

@@ -4,6 +4,7 @@
 
 #include <memory>
 #include <string>
+#include <utility>
 
 #include "../common.hpp"
 #include "../cpu_engine.hpp"  // HostAllreduce
@@ -31,5 +32,6 @@ std::unique_ptr<Engine<T>> makeGpuEngine(const BAProblemHost& prob,
 
 std::string rcclUniqueIdString();
 int hipDeviceCountSafe();
+std::pair<long long, long long> hipMemInfoSafe();  // (free, total) bytes
 
 }  // namespace megba
