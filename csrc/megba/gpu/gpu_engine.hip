@@ -789,6 +789,40 @@ __global__ void kDivScalar(double* out, const double* num, const double* den) {
 }
 __global__ void kSetScalar(double* out, double v) { *out = v; }
 __global__ void kCopyScalar(double* dst, const double* src) { *dst = *src; }
+// Fused reduction finals with the PCG scalar epilogues (fewer launches per
+// iteration; the scalars never leave the device except the rho readback).
+__global__ void kRedFinalRhoBeta(const double* part, int nb, double* rho,
+                                 const double* rhoPrev, double* beta) {
+  __shared__ double sm[kBlk];
+  double v = 0.0;
+  for (int i = threadIdx.x; i < nb; i += kBlk) v += part[i];
+  sm[threadIdx.x] = v;
+  __syncthreads();
+  for (int s = kBlk / 2; s > 0; s >>= 1) {
+    if (threadIdx.x < s) sm[threadIdx.x] += sm[threadIdx.x + s];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    *rho = sm[0];
+    *beta = sm[0] / *rhoPrev;
+  }
+}
+__global__ void kRedFinalAlpha(const double* part, int nb, const double* rho,
+                               double* alpha, double* rhoPrev) {
+  __shared__ double sm[kBlk];
+  double v = 0.0;
+  for (int i = threadIdx.x; i < nb; i += kBlk) v += part[i];
+  sm[threadIdx.x] = v;
+  __syncthreads();
+  for (int s = kBlk / 2; s > 0; s >>= 1) {
+    if (threadIdx.x < s) sm[threadIdx.x] += sm[threadIdx.x + s];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    *alpha = *rho / sm[0];
+    *rhoPrev = *rho;
+  }
+}
 template <typename T>
 __global__ void kXpbyS(int64_t n, const T* __restrict__ x,
                        const double* __restrict__ b, T* __restrict__ y) {
@@ -1531,21 +1565,19 @@ class GpuEngine final : public Engine<T> {
                              hipMemcpyDeviceToDevice, stream_));
     hipLaunchKernelGGL(kPrecondRho<T>, dim3(rhoGrid), dim3(kBlk), 0, stream_,
                        ncam_, dHppInv_, dRr_, dZ_, dPart_);
-    hipLaunchKernelGGL((kRedFinal<ROp::Dot>), dim3(1), dim3(kBlk), 0, stream_,
-                       dPart_, rhoGrid, slotRho());
-    hipLaunchKernelGGL(kDivScalar, dim3(1), dim3(1), 0, stream_, slotBeta(),
-                       slotRho(), slotRhoPrev());
+    hipLaunchKernelGGL(kRedFinalRhoBeta, dim3(1), dim3(kBlk), 0, stream_,
+                       dPart_, rhoGrid, slotRho(), slotRhoPrev(), slotBeta());
     hipLaunchKernelGGL(kXpbyS<T>, dim3(gridFor(nc_)), dim3(kBlk), 0, stream_,
                        nc_, dZ_, slotBeta(), dP_);
     schurApply(dP_, dQ_);
-    reduceDetAsync(dP_, dQ_, nc_, ROp::Dot, slotPq());
-    hipLaunchKernelGGL(kDivScalar, dim3(1), dim3(1), 0, stream_, slotAlpha(),
-                       slotRho(), slotPq());
+    hipLaunchKernelGGL((kRedPartial<T, ROp::Dot>), dim3(kRedBlocks),
+                       dim3(kBlk), 0, stream_, dP_, dQ_, nc_, dPart_);
+    hipLaunchKernelGGL(kRedFinalAlpha, dim3(1), dim3(kBlk), 0, stream_,
+                       dPart_, kRedBlocks, slotRho(), slotAlpha(),
+                       slotRhoPrev());
     hipLaunchKernelGGL(kUpdateXR<T>, dim3(gridFor(nc_)), dim3(kBlk), 0,
                        stream_, nc_, slotAlpha(), dP_, dQ_, dDeltaX_, dXBak_,
                        dRr_);
-    hipLaunchKernelGGL(kCopyScalar, dim3(1), dim3(1), 0, stream_,
-                       slotRhoPrev(), slotRho());
   }
 
   void ensurePcgGraph() {
