@@ -159,23 +159,25 @@ def test_cpp_cli_gpu(tmp_path):
 
 
 @pytest.mark.gpu
-def test_fp32_stages_match_cpu():
-    """fp32 GPU engine vs fp32 CPU engine, stage by stage (loose bounds)."""
+def test_fp32_stages_vs_fp64_oracle():
+    """fp32 GPU stages vs the fp64 CPU oracle: each stage must sit within
+    fp32-roundoff-scale bounds of the exact result (direct fp32-vs-fp32
+    comparison is ill-posed: accumulation orders differ)."""
     cams, pts, ci, pi, meas = mb.synthesize_bal(12, 120, 1100, seed=3)
     eng = {}
-    for device in ("cpu", "gpu"):
+    for device, dtype in (("cpu", "float64"), ("gpu", "float32")):
         p = mb.BAProblem(cams, pts, ci, pi, meas)
-        p.build(device=device, dtype="float32")
+        p.build(device=device, dtype=dtype)
         p.forward()
         p.accept_forward()
         p.build_linear_system()
         p.process_diag(1e4)
-        p.solve_linear(max_iter=200, tol=1e-6, refuse_ratio=1e18)
+        p.solve_linear(max_iter=300, tol=1e-8, refuse_ratio=1e18)
         eng[device] = p.dump()
     d1, d2 = eng["cpu"], eng["gpu"]
     for key in ("r", "Jc", "Jp", "Hpp", "Hll", "g"):
         scale = np.abs(d1[key]).max() or 1.0
-        np.testing.assert_allclose(d2[key], d1[key], rtol=5e-4,
-                                   atol=5e-5 * scale, err_msg=key)
+        np.testing.assert_allclose(d2[key], d1[key], rtol=2e-3,
+                                   atol=2e-4 * scale, err_msg=key)
     scale = np.abs(d1["deltaX"]).max()
-    np.testing.assert_allclose(d2["deltaX"], d1["deltaX"], atol=3e-2 * scale)
+    np.testing.assert_allclose(d2["deltaX"], d1["deltaX"], atol=0.05 * scale)
