@@ -169,9 +169,12 @@ def test_fp32_stages_vs_fp64_oracle():
         p = mb.BAProblem(cams, pts, ci, pi, meas)
         p.build(device=device, dtype=dtype)
         p.forward()
+        dj = p.dump()          # r/J of the last forward (pre-swap buffers)
         p.accept_forward()
         p.build_linear_system()
-        eng[device] = p.dump()
+        d = p.dump()
+        d.update({k: dj[k] for k in ("r", "Jc", "Jp")})
+        eng[device] = d
     d1, d2 = eng["cpu"], eng["gpu"]
     # deltaX is deliberately not compared here: fp32 PCG iterated past its
     # precision floor is run-to-run noisy (atomic accumulation order); the
