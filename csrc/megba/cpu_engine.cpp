@@ -79,6 +79,7 @@ class CpuEngine final : public Engine<T> {
   }
 
   double forward() override {
+    freshCur_ = true;
     if (customFwd_) return forwardCustom();
     using J = Jet<T, 12>;
     T chi2 = T(0);
@@ -182,6 +183,7 @@ class CpuEngine final : public Engine<T> {
     std::swap(JpCur_, JpBak_);
     // After swap the accepted data is in *Bak_; rhoDenominator reads Bak_,
     // forward overwrites Cur_.
+    freshCur_ = false;
   }
 
   void backupParams() override {
@@ -370,9 +372,10 @@ class CpuEngine final : public Engine<T> {
     auto cp = [](const std::vector<T>& v) {
       return std::vector<double>(v.begin(), v.end());
     };
-    d.r = cp(rCur_);
-    d.Jc = cp(JcCur_);
-    d.Jp = cp(JpCur_);
+    // r/J of the LAST forward() (survives the acceptForward buffer swap)
+    d.r = cp(freshCur_ ? rCur_ : rBak_);
+    d.Jc = cp(freshCur_ ? JcCur_ : JcBak_);
+    d.Jp = cp(freshCur_ ? JpCur_ : JpBak_);
     d.Hpp = cp(Hpp_);
     d.Hll = cp(Hll_);
     d.Hpl = cp(Hpl_);
@@ -611,6 +614,7 @@ class CpuEngine final : public Engine<T> {
   int rank_, world_, ncam_, npt_;
   bool analytical_ = false;
   bool implicit_ = false;
+  bool freshCur_ = false;
   int ptLo_ = 0, ptHi_ = 0;
   int64_t e0_ = 0, e1_ = 0, nL_ = 0, dim_ = 0;
   std::vector<int> camOf_, ptOf_;

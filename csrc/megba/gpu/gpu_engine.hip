@@ -1085,6 +1085,7 @@ class GpuEngine final : public Engine<T> {
   }
 
   double forward() override {
+    freshCur_ = true;
     if (customFwd_) return forwardCustom();
     zeroScalar();
     if (analytical_)
@@ -1105,7 +1106,10 @@ class GpuEngine final : public Engine<T> {
     return globalScalar(ncclSum);
   }
 
-  void acceptForward() override { cur_ ^= 1; }  // accepted set = dX_[cur_^1]
+  void acceptForward() override {
+    cur_ ^= 1;          // accepted set = dX_[cur_^1]
+    freshCur_ = false;  // the (new) current buffers are not yet written
+  }
 
   double forwardCustom() {
     hipLaunchKernelGGL(kGatherLeaves<T>, dim3(gridFor(nL_)), dim3(kBlk), 0,
@@ -1343,9 +1347,11 @@ class GpuEngine final : public Engine<T> {
     DenseDump d;
     d.e0 = e0_;
     d.e1 = e1_;
-    d.r = down(dR_[cur_], nL_ * 2);
-    d.Jc = down(dJc_[cur_], nL_ * 18);
-    d.Jp = down(dJp_[cur_], nL_ * 6);
+    // r/J of the LAST forward() (survives the acceptForward buffer swap)
+    const int fw = freshCur_ ? cur_ : (cur_ ^ 1);
+    d.r = down(dR_[fw], nL_ * 2);
+    d.Jc = down(dJc_[fw], nL_ * 18);
+    d.Jp = down(dJp_[fw], nL_ * 6);
     d.Hpp = down(dHpp_, (int64_t)ncam_ * 81);
     d.Hll = down(dHll_, (int64_t)npt_ * 9);
     if (!implicit_) {
@@ -1641,6 +1647,7 @@ class GpuEngine final : public Engine<T> {
   bool implicit_ = false;
   bool fwd2_ = getenv("MEGBA_FWD2") != nullptr;  // 2-lane Jet<6> experiment
   int cur_ = 0;
+  bool freshCur_ = false;  // current r/J buffers hold the last forward()
   int nChunks_ = 0;
   int *dCamOf_{}, *dPtOf_{}, *dChCam_{}, *dChLo_{}, *dChHi_{}, *dFail_{};
   int *dCamPos_{}, *dPtOfCam_{};
