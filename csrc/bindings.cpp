@@ -372,6 +372,7 @@ PYBIND11_MODULE(_core, m) {
   m.def("jv_cos", [](const PyJetVec& a) { return PyJetVec{jvUnary(JvUnary::Cos, a.v)}; });
   m.def("jv_sqrt", [](const PyJetVec& a) { return PyJetVec{jvUnary(JvUnary::Sqrt, a.v)}; });
   m.def("jv_angle_axis_to_rotation", [](py::sequence aa) { return jvWrap(jvAngleAxisToRotation(jvList(aa))); });
+  m.def("jv_normalize_angle", [](const PyJetVec& t) { return PyJetVec{jvNormalizeAngle(t.v)}; });
   m.def("jv_rotation2d", [](const PyJetVec& t) { return jvWrap(jvRotation2D(t.v)); });
   m.def("jv_quaternion_to_rotation", [](py::sequence q) { return jvWrap(jvQuaternionToRotation(jvList(q))); });
   m.def("jv_rotation_to_quaternion", [](py::sequence R) { return jvWrap(jvRotationToQuaternion(jvList(R))); });

@@ -424,6 +424,42 @@ std::vector<JetVec<T>> jvAngleAxisToRotation(const std::vector<JetVec<T>>& aa) {
 }
 
 template <typename T>
+JetVec<T> jvNormalizeAngle(const JetVec<T>& theta) {
+  MEGBA_CHECK(!theta.isScalar, "normalize_angle needs a vector operand");
+  // wrap = theta - 2*pi*round(theta / (2*pi)); d wrap / d theta = 1
+  const int64_t n = theta.nItem;
+  JetVec<T> out;
+  out.nItem = n;
+  out.N = theta.N;
+  out.onGpu = theta.onGpu;
+  out.value = makeBuf<T>(n, out.onGpu);
+  out.grad = makeBuf<T>((int64_t)out.N * n, out.onGpu);
+  // value: via composition would lose the per-item branch; do it directly.
+  std::vector<T> hv(n), hg((int64_t)out.N * n);
+  jvToHost(theta, hv.data(), hg.data());
+  const T twoPi = T(6.283185307179586476925286766559);
+  for (int64_t i = 0; i < n; ++i) {
+    T v = hv[i];
+    v = v - twoPi * (T)std::floor(((double)v + 3.14159265358979323846) /
+                                  (double)twoPi);
+    hv[i] = v;
+  }
+  if (out.onGpu) {
+    JV_HIP_CHECK(hipMemcpy(out.value->ptr, hv.data(), n * sizeof(T),
+                           hipMemcpyHostToDevice));
+    JV_HIP_CHECK(hipMemcpy(out.grad->ptr, hg.data(),
+                           (int64_t)out.N * n * sizeof(T),
+                           hipMemcpyHostToDevice));
+  } else {
+    std::memcpy(out.value->ptr, hv.data(), n * sizeof(T));
+    std::memcpy(out.grad->ptr, hg.data(), (int64_t)out.N * n * sizeof(T));
+  }
+  return out;
+}
+template JetVec<double> jvNormalizeAngle<double>(const JetVec<double>&);
+template JetVec<float> jvNormalizeAngle<float>(const JetVec<float>&);
+
+template <typename T>
 std::vector<JetVec<T>> jvRotation2D(const JetVec<T>& theta) {
   JetVec<T> c = jvUnary(JvUnary::Cos, theta);
   JetVec<T> s = jvUnary(JvUnary::Sin, theta);

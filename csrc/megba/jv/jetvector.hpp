@@ -83,6 +83,10 @@ JetVec<T> jvUnary(JvUnary op, const JetVec<T>& a);
 // aa[3] -> row-major R[9]
 template <typename T>
 std::vector<JetVec<T>> jvAngleAxisToRotation(const std::vector<JetVec<T>>& aa);
+// wrap theta to (-pi, pi], gradient unchanged (reference SE2
+// normalize_rotation2D_Kernel)
+template <typename T>
+JetVec<T> jvNormalizeAngle(const JetVec<T>& theta);
 // theta -> [cos,-sin,sin,cos]
 template <typename T>
 std::vector<JetVec<T>> jvRotation2D(const JetVec<T>& theta);

@@ -267,3 +267,14 @@ def test_rotation_to_quaternion_roundtrip_cpu():
             fd = (chain(qp) - chain(qm)) / (2 * eps)
             for k in range(4):
                 np.testing.assert_allclose(grads[k][d, item], fd[k], atol=1e-5)
+
+
+def test_normalize_angle_cpu():
+    th = rng.normal(scale=6.0, size=50)
+    j = _core.JetVector(th, None, N=1, grad_pos=0)
+    out = _core.jv_normalize_angle(j)
+    v, g = out.to_numpy()
+    ref = th - 2 * np.pi * np.floor((th + np.pi) / (2 * np.pi))
+    np.testing.assert_allclose(v, ref, atol=1e-12)
+    assert (v > -np.pi - 1e-12).all() and (v <= np.pi + 1e-12).all()
+    np.testing.assert_allclose(g[0], np.ones(50))
