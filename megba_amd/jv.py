@@ -83,6 +83,10 @@ def abs(a):  # noqa: A001
     return JV(_core.jv_abs(a.raw))
 
 
+def normalize_angle(a):
+    return JV(_core.jv_normalize_angle(a.raw))
+
+
 def angle_axis_to_rotation(aa):
     return [JV(r) for r in _core.jv_angle_axis_to_rotation([a.raw for a in aa])]
 
