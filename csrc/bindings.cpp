@@ -90,7 +90,8 @@ struct PyProblem {
 
   void build(const std::string& device, const std::string& dtype, int rank,
              int worldSize, int deviceIndex, const std::string& diff,
-             const std::string& schur, py::object allreduce, py::object rcclId,
+             const std::string& schur, const std::string& loss,
+             double lossDelta, py::object allreduce, py::object rcclId,
              py::object customForward) {
     opt.rank = rank;
     opt.worldSize = worldSize;
@@ -98,6 +99,12 @@ struct PyProblem {
     opt.device = device == "gpu" ? Device::GPU : Device::CPU;
     opt.diff = diff == "analytical" ? DiffMode::ANALYTICAL : DiffMode::AUTO;
     opt.schur = schur == "implicit" ? SchurMode::IMPLICIT : SchurMode::EXPLICIT;
+    opt.loss = loss == "huber" ? LossKind::HUBER
+               : loss == "cauchy" ? LossKind::CAUCHY
+                                  : LossKind::NONE;
+    MEGBA_CHECK(loss == "none" || loss == "huber" || loss == "cauchy",
+                "loss must be none|huber|cauchy");
+    opt.lossDelta = lossDelta;
     isDouble = dtype != "float32";
     MEGBA_CHECK(customForward.is_none() || isDouble,
                 "custom_forward requires float64");
@@ -393,6 +400,7 @@ PYBIND11_MODULE(_core, m) {
            py::arg("dtype") = "float64", py::arg("rank") = 0,
            py::arg("world_size") = 1, py::arg("device_index") = 0,
            py::arg("diff") = "auto", py::arg("schur") = "explicit",
+           py::arg("loss") = "none", py::arg("loss_delta") = 1.0,
            py::arg("allreduce") = py::none(), py::arg("rccl_id") = py::none(),
            py::arg("custom_forward") = py::none())
       .def("solve", &PyProblem::solve, py::arg("max_iter") = 20,

@@ -18,6 +18,9 @@ namespace megba {
 enum class Device { CPU, GPU };
 enum class DiffMode { AUTO, ANALYTICAL };
 enum class SchurMode { EXPLICIT, IMPLICIT };
+// Robust loss (beyond the reference, which only has the 2x2 information
+// matrix): IRLS reweighting with rho-consistent cost, s = r^T r.
+enum class LossKind { NONE, HUBER, CAUCHY };
 
 struct SolverOptionPCG {
   int maxIter = 100;
@@ -39,6 +42,8 @@ struct ProblemOption {
   Device device = Device::CPU;
   DiffMode diff = DiffMode::AUTO;
   SchurMode schur = SchurMode::EXPLICIT;
+  LossKind loss = LossKind::NONE;
+  double lossDelta = 1.0;
   int rank = 0;        // this process' rank (one process per GPU)
   int worldSize = 1;
   int deviceIndex = 0; // HIP device ordinal for this rank

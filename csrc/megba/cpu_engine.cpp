@@ -23,7 +23,9 @@ class CpuEngine final : public Engine<T> {
         ncam_(ix.ncam),
         npt_(ix.npt),
         analytical_(opt.diff == DiffMode::ANALYTICAL),
-        implicit_(opt.schur == SchurMode::IMPLICIT) {
+        implicit_(opt.schur == SchurMode::IMPLICIT),
+        lossKind_((int)opt.loss),
+        lossD2_((T)(opt.lossDelta * opt.lossDelta)) {
     e0_ = ix.split[rank_];
     e1_ = ix.split[rank_ + 1];
     nL_ = e1_ - e0_;
@@ -90,9 +92,9 @@ class CpuEngine final : public Engine<T> {
       if (analytical_) {
         T res[2], jc[2][9], jp[2][3];
         balAnalytical<T>(cp, pp, &meas_[2 * e], res, jc, jp);
+        chi2 += lossRho(lossKind_, lossD2_, res[0] * res[0] + res[1] * res[1]);
         for (int row = 0; row < 2; ++row) {
           rCur_[2 * e + row] = res[row];
-          chi2 += res[row] * res[row];
           for (int i = 0; i < 9; ++i) JcCur_[18 * e + 9 * row + i] = jc[row][i];
           for (int i = 0; i < 3; ++i) JpCur_[6 * e + 3 * row + i] = jp[row][i];
         }
@@ -103,9 +105,10 @@ class CpuEngine final : public Engine<T> {
       for (int i = 0; i < 9; ++i) cam[i] = J::leaf(cp[i], i);
       for (int i = 0; i < 3; ++i) pt[i] = J::leaf(pp[i], 9 + i);
       balReprojectionError<T, J>(cam, pt, &meas_[2 * e], res);
+      chi2 += lossRho(lossKind_, lossD2_,
+                      res[0].v * res[0].v + res[1].v * res[1].v);
       for (int row = 0; row < 2; ++row) {
         rCur_[2 * e + row] = res[row].v;
-        chi2 += res[row].v * res[row].v;
         for (int i = 0; i < 9; ++i) JcCur_[18 * e + 9 * row + i] = res[row].d[i];
         for (int i = 0; i < 3; ++i) JpCur_[6 * e + 3 * row + i] = res[row].d[9 + i];
       }
@@ -339,12 +342,15 @@ class CpuEngine final : public Engine<T> {
       const T* Jp = &JpBak_[6 * e];
       const T* dc = &dxc[(size_t)camOf_[e] * 9];
       const T* dp = &dxp[(size_t)ptOf_[e] * 3];
+      T acc2[2];
       for (int row = 0; row < 2; ++row) {
         T acc = rBak_[2 * e + row];
         for (int i = 0; i < 9; ++i) acc += Jc[9 * row + i] * dc[i];
         for (int i = 0; i < 3; ++i) acc += Jp[3 * row + i] * dp[i];
-        s += acc * acc;
+        acc2[row] = acc;
       }
+      s += lossRho(lossKind_, lossD2_,
+                   acc2[0] * acc2[0] + acc2[1] * acc2[1]);
     }
     T buf = s;
     if (ar_) ar_(&buf, 1, 's');
@@ -415,10 +421,12 @@ class CpuEngine final : public Engine<T> {
     T chi2 = T(0);
 #pragma omp parallel for schedule(static) reduction(+ : chi2)
     for (int64_t e = 0; e < nL_; ++e) {
+      const T v0 = res[0].value->ptr[e];
+      const T v1 = res[1].value->ptr[e];
+      chi2 += lossRho(lossKind_, lossD2_, v0 * v0 + v1 * v1);
       for (int row = 0; row < 2; ++row) {
         const T v = res[row].value->ptr[e];
         rCur_[2 * e + row] = v;
-        chi2 += v * v;
         const T* g = res[row].grad->ptr;
         for (int k = 0; k < 9; ++k)
           JcCur_[18 * e + 9 * row + k] = g[(size_t)k * nL_ + e];
@@ -444,6 +452,8 @@ class CpuEngine final : public Engine<T> {
   }
 
   // Weighted rows of the ACCEPTED (post-acceptForward) jacobian set.
+  // Includes the robust-loss IRLS weight (applied to the weighted side only,
+  // so H = sum w J^T W J, g = -sum w J^T W r).
   inline void weightedRows(int64_t e, T wJc[2][9], T wJp[2][3], T wr[2]) {
     const T* Jc = &JcBak_[18 * e];
     const T* Jp = &JpBak_[6 * e];
@@ -471,6 +481,20 @@ class CpuEngine final : public Engine<T> {
       }
       wr[0] = r[0];
       wr[1] = r[1];
+    }
+    if (lossKind_) {
+      const T w =
+          lossWeight(lossKind_, lossD2_, r[0] * r[0] + r[1] * r[1]);
+      for (int i = 0; i < 9; ++i) {
+        wJc[0][i] *= w;
+        wJc[1][i] *= w;
+      }
+      for (int i = 0; i < 3; ++i) {
+        wJp[0][i] *= w;
+        wJp[1][i] *= w;
+      }
+      wr[0] *= w;
+      wr[1] *= w;
     }
   }
 
@@ -540,6 +564,12 @@ class CpuEngine final : public Engine<T> {
       const T a = w00 * u0 + w01 * u1;
       u1 = w01 * u0 + w11 * u1;
       u0 = a;
+    }
+    if (lossKind_) {
+      const T r0 = rBak_[2 * e], r1 = rBak_[2 * e + 1];
+      const T w = lossWeight(lossKind_, lossD2_, r0 * r0 + r1 * r1);
+      u0 *= w;
+      u1 *= w;
     }
   }
 
@@ -615,6 +645,8 @@ class CpuEngine final : public Engine<T> {
   bool analytical_ = false;
   bool implicit_ = false;
   bool freshCur_ = false;
+  int lossKind_ = 0;
+  T lossD2_ = T(1);
   int ptLo_ = 0, ptHi_ = 0;
   int64_t e0_ = 0, e1_ = 0, nL_ = 0, dim_ = 0;
   std::vector<int> camOf_, ptOf_;

@@ -134,7 +134,8 @@ __global__ __launch_bounds__(256, 2) void kForward(
     const T* __restrict__ params, int ncam, const T* __restrict__ meas,
     const unsigned char* __restrict__ camFixed,
     const unsigned char* __restrict__ ptFixed, T* __restrict__ rOut,
-    T* __restrict__ Jc, T* __restrict__ Jp, double* chi2Acc) {
+    T* __restrict__ Jc, T* __restrict__ Jp, double* chi2Acc, int lossKind,
+    T lossD2) {
   using J3 = Jet<T, 3>;
   __shared__ double sm[kBlk];
   double chi2 = 0.0;
@@ -163,11 +164,11 @@ __global__ __launch_bounds__(256, 2) void kForward(
           Jp[((int64_t)((col - 9) * 2 + row)) * nL + e] =
               pfix ? T(0) : res[row].d[j];
       }
-      if (sub == 0) {
-        rOut[(int64_t)row * nL + e] = res[row].v;
-        chi2 += (double)res[row].v * (double)res[row].v;
-      }
+      if (sub == 0) rOut[(int64_t)row * nL + e] = res[row].v;
     }
+    if (sub == 0)
+      chi2 += (double)lossRho(lossKind, lossD2,
+                              res[0].v * res[0].v + res[1].v * res[1].v);
   }
   sm[threadIdx.x] = chi2;
   __syncthreads();
@@ -187,7 +188,8 @@ __global__ __launch_bounds__(256, 1) void kForward2(
     const T* __restrict__ params, int ncam, const T* __restrict__ meas,
     const unsigned char* __restrict__ camFixed,
     const unsigned char* __restrict__ ptFixed, T* __restrict__ rOut,
-    T* __restrict__ Jc, T* __restrict__ Jp, double* chi2Acc) {
+    T* __restrict__ Jc, T* __restrict__ Jp, double* chi2Acc, int lossKind,
+    T lossD2) {
   using J6 = Jet<T, 6>;
   __shared__ double sm[kBlk];
   double chi2 = 0.0;
@@ -216,11 +218,11 @@ __global__ __launch_bounds__(256, 1) void kForward2(
           Jp[((int64_t)((col - 9) * 2 + row)) * nL + e] =
               pfix ? T(0) : res[row].d[j];
       }
-      if (sub == 0) {
-        rOut[(int64_t)row * nL + e] = res[row].v;
-        chi2 += (double)res[row].v * (double)res[row].v;
-      }
+      if (sub == 0) rOut[(int64_t)row * nL + e] = res[row].v;
     }
+    if (sub == 0)
+      chi2 += (double)lossRho(lossKind, lossD2,
+                              res[0].v * res[0].v + res[1].v * res[1].v);
   }
   sm[threadIdx.x] = chi2;
   __syncthreads();
@@ -241,7 +243,8 @@ __global__ void kForwardAnalytical(int64_t nL, const int* __restrict__ camOf,
                                    const unsigned char* __restrict__ camFixed,
                                    const unsigned char* __restrict__ ptFixed,
                                    T* __restrict__ rOut, T* __restrict__ Jc,
-                                   T* __restrict__ Jp, double* chi2Acc) {
+                                   T* __restrict__ Jp, double* chi2Acc,
+                                   int lossKind, T lossD2) {
   __shared__ double sm[kBlk];
   double chi2 = 0.0;
   const T* ptsBase = params + (int64_t)ncam * 9;
@@ -252,11 +255,12 @@ __global__ void kForwardAnalytical(int64_t nL, const int* __restrict__ camOf,
     const T m[2] = {meas[2 * e], meas[2 * e + 1]};
     T res[2], jc[2][9], jp[2][3];
     balAnalytical<T>(cp, pp, m, res, jc, jp);
+    chi2 += (double)lossRho(lossKind, lossD2,
+                            res[0] * res[0] + res[1] * res[1]);
     const bool cfix = camFixed && camFixed[camOf[e]];
     const bool pfix = ptFixed && ptFixed[ptOf[e]];
     for (int row = 0; row < 2; ++row) {
       rOut[(int64_t)row * nL + e] = res[row];
-      chi2 += (double)res[row] * (double)res[row];
       for (int col = 0; col < 9; ++col)
         Jc[((int64_t)(col * 2 + row)) * nL + e] = cfix ? T(0) : jc[row][col];
       for (int col = 0; col < 3; ++col)
@@ -270,6 +274,26 @@ __global__ void kForwardAnalytical(int64_t nL, const int* __restrict__ camOf,
     __syncthreads();
   }
   if (threadIdx.x == 0) atomicAdd(chi2Acc, sm[0]);
+}
+
+// chi2 with a robust loss, from the residual buffers (custom-edge path).
+template <typename T>
+__global__ void kChi2Loss(int64_t nL, const T* __restrict__ r, int lossKind,
+                          T lossD2, double* acc) {
+  __shared__ double sm[kBlk];
+  double local = 0.0;
+  for (int64_t e = blockIdx.x * (int64_t)kBlk + threadIdx.x; e < nL;
+       e += (int64_t)gridDim.x * kBlk) {
+    const T r0 = r[e], r1 = r[nL + e];
+    local += (double)lossRho(lossKind, lossD2, r0 * r0 + r1 * r1);
+  }
+  sm[threadIdx.x] = local;
+  __syncthreads();
+  for (int s = kBlk / 2; s > 0; s >>= 1) {
+    if (threadIdx.x < s) sm[threadIdx.x] += sm[threadIdx.x + s];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) atomicAdd(acc, sm[0]);
 }
 
 // Custom-edge support: gather parameter leaves, repack residual JetVectors.
@@ -339,7 +363,7 @@ __global__ void kAssembleEdge(int64_t nL, const int* __restrict__ camOf,
                               const T* __restrict__ info, T* __restrict__ Hll,
                               T* __restrict__ Hpl, T* __restrict__ g, int ncam,
                               const int* __restrict__ camPos,
-                              T* __restrict__ slab) {
+                              T* __restrict__ slab, int lossKind, T lossD2) {
   using L = SlabLayout<EXPL, HASINFO>;
   T* gp = g + (int64_t)ncam * 9;
   const int lane = threadIdx.x & 63;
@@ -366,7 +390,22 @@ __global__ void kAssembleEdge(int64_t nL, const int* __restrict__ camOf,
       T wjp[2][3], wr[2];
       T wjc0[9], wjc1[9];
       if (HASINFO) {
-        const T w00 = info[3 * e], w01 = info[3 * e + 1], w11 = info[3 * e + 2];
+        // HASINFO also covers robust-loss runs without an information
+        // matrix (info == nullptr -> identity), since both need the
+        // weighted Jc rows stored separately in the slab.
+        T w00 = T(1), w01 = T(0), w11 = T(1);
+        if (info) {
+          w00 = info[3 * e];
+          w01 = info[3 * e + 1];
+          w11 = info[3 * e + 2];
+        }
+        if (lossKind) {
+          const T w = lossWeight(lossKind, lossD2,
+                                 rr[0] * rr[0] + rr[1] * rr[1]);
+          w00 *= w;
+          w01 *= w;
+          w11 *= w;
+        }
         for (int k = 0; k < 9; ++k) {
           wjc0[k] = w00 * jc[0][k] + w01 * jc[1][k];
           wjc1[k] = w01 * jc[0][k] + w11 * jc[1][k];
@@ -587,6 +626,7 @@ __global__ void kSpmvEtx(int64_t nL, const int* __restrict__ camOf,
                          const int* __restrict__ ptOf,
                          const T* __restrict__ Hpl, const T* __restrict__ Jc,
                          const T* __restrict__ Jp, const T* __restrict__ info,
+                         const T* __restrict__ rBak, int lossKind, T lossD2,
                          const T* __restrict__ x, T* __restrict__ out) {
   const int lane = threadIdx.x & 63;
   const int64_t nWork = ((nL + kBlk - 1) / kBlk) * (int64_t)kBlk;
@@ -611,6 +651,12 @@ __global__ void kSpmvEtx(int64_t nL, const int* __restrict__ camOf,
           const T a = w00 * u0 + w01 * u1;
           u1 = w01 * u0 + w11 * u1;
           u0 = a;
+        }
+        if (lossKind) {
+          const T r0 = rBak[j], r1 = rBak[nL + j];
+          const T w = lossWeight(lossKind, lossD2, r0 * r0 + r1 * r1);
+          u0 *= w;
+          u1 *= w;
         }
         o0 = Jp[((int64_t)0) * nL + j] * u0 + Jp[((int64_t)1) * nL + j] * u1;
         o1 = Jp[((int64_t)2) * nL + j] * u0 + Jp[((int64_t)3) * nL + j] * u1;
@@ -883,21 +929,24 @@ __global__ void kRhoDenom(int64_t nL, const int* __restrict__ camOf,
                           const int* __restrict__ ptOf, const T* __restrict__ r,
                           const T* __restrict__ Jc, const T* __restrict__ Jp,
                           const T* __restrict__ dxc, const T* __restrict__ dxp,
-                          double* acc) {
+                          double* acc, int lossKind, T lossD2) {
   __shared__ double sm[kBlk];
   double local = 0.0;
   for (int64_t e = blockIdx.x * (int64_t)kBlk + threadIdx.x; e < nL;
        e += (int64_t)gridDim.x * kBlk) {
     const T* dc = dxc + (int64_t)camOf[e] * 9;
     const T* dp = dxp + (int64_t)ptOf[e] * 3;
+    T acc2[2];
     for (int row = 0; row < 2; ++row) {
       T s = r[(int64_t)row * nL + e];
       for (int k = 0; k < 9; ++k)
         s += Jc[((int64_t)(k * 2 + row)) * nL + e] * dc[k];
       for (int k = 0; k < 3; ++k)
         s += Jp[((int64_t)(k * 2 + row)) * nL + e] * dp[k];
-      local += (double)s * (double)s;
+      acc2[row] = s;
     }
+    local += (double)lossRho(lossKind, lossD2,
+                             acc2[0] * acc2[0] + acc2[1] * acc2[1]);
   }
   sm[threadIdx.x] = local;
   __syncthreads();
@@ -926,7 +975,9 @@ class GpuEngine final : public Engine<T> {
         ncam_(ix.ncam),
         npt_(ix.npt),
         analytical_(opt.diff == DiffMode::ANALYTICAL),
-        implicit_(opt.schur == SchurMode::IMPLICIT) {
+        implicit_(opt.schur == SchurMode::IMPLICIT),
+        lossKind_((int)opt.loss),
+        lossD2_((T)(opt.lossDelta * opt.lossDelta)) {
     HIP_CHECK(hipSetDevice(opt.deviceIndex));
     HIP_CHECK(hipStreamCreate(&stream_));
     e0_ = ix.split[rank_];
@@ -1092,17 +1143,17 @@ class GpuEngine final : public Engine<T> {
       hipLaunchKernelGGL(kForwardAnalytical<T>, dim3(gridFor(nL_)), dim3(kBlk),
                          0, stream_, nL_, dCamOf_, dPtOf_, dParams_, ncam_,
                          dMeas_, dCamFixed_, dPtFixed_, dR_[cur_], dJc_[cur_],
-                         dJp_[cur_], scalarPtr());
-    else if (fwd2_)
+                         dJp_[cur_], scalarPtr(), lossKind_, lossD2_);
+    else if (fwd2_ && lossKind_ == 0)
       hipLaunchKernelGGL(kForward2<T>, dim3(gridFor(nL_ * 2)), dim3(kBlk), 0,
                          stream_, nL_, dCamOf_, dPtOf_, dParams_, ncam_,
                          dMeas_, dCamFixed_, dPtFixed_, dR_[cur_], dJc_[cur_],
-                         dJp_[cur_], scalarPtr());
+                         dJp_[cur_], scalarPtr(), lossKind_, lossD2_);
     else
       hipLaunchKernelGGL(kForward<T>, dim3(gridFor(nL_ * 4)), dim3(kBlk), 0,
                          stream_, nL_, dCamOf_, dPtOf_, dParams_, ncam_,
                          dMeas_, dCamFixed_, dPtFixed_, dR_[cur_], dJc_[cur_],
-                         dJp_[cur_], scalarPtr());
+                         dJp_[cur_], scalarPtr(), lossKind_, lossD2_);
     return globalScalar(ncclSum);
   }
 
@@ -1135,7 +1186,14 @@ class GpuEngine final : public Engine<T> {
                          dCamOf_, dPtOf_, dCamFixed_, dPtFixed_, dR_[cur_],
                          dJc_[cur_], dJp_[cur_]);
     }
-    reduceDetAsync(dR_[cur_], dR_[cur_], nL_ * 2, ROp::SumSq, scalarPtr());
+    if (lossKind_) {
+      zeroScalar();
+      hipLaunchKernelGGL(kChi2Loss<T>, dim3(gridFor(nL_)), dim3(kBlk), 0,
+                         stream_, nL_, dR_[cur_], lossKind_, lossD2_,
+                         scalarPtr());
+    } else {
+      reduceDetAsync(dR_[cur_], dR_[cur_], nL_ * 2, ROp::SumSq, scalarPtr());
+    }
     return globalScalar(ncclSum);
   }
 
@@ -1149,7 +1207,7 @@ class GpuEngine final : public Engine<T> {
                              (int64_t)npL_ * 3 * sizeof(T), stream_));
     dispatchAssemble(bak);
     if (!implicit_) {
-      if (hasInfo_)
+      if (weighted())
         hipLaunchKernelGGL((kFinalizeCam<T, true>), dim3(gridFor(nL_)),
                            dim3(kBlk), 0, stream_, nL_, dSlab_, dHplCam_);
       else
@@ -1312,7 +1370,8 @@ class GpuEngine final : public Engine<T> {
     zeroScalar();
     hipLaunchKernelGGL(kRhoDenom<T>, dim3(gridFor(nL_)), dim3(kBlk), 0, stream_,
                        nL_, dCamOf_, dPtOf_, dR_[bak], dJc_[bak], dJp_[bak],
-                       dDeltaX_, dDeltaX_ + nc_, scalarPtr());
+                       dDeltaX_, dDeltaX_ + nc_, scalarPtr(), lossKind_,
+                       lossD2_);
     return globalScalar(ncclSum) - chi2Backup;
   }
 
@@ -1373,19 +1432,20 @@ class GpuEngine final : public Engine<T> {
   }
 
  private:
+  bool weighted() const { return hasInfo_ || lossKind_ != 0; }
   int slabWidth() const {
-    // must match SlabLayout<EXPL,HASINFO>::SW
-    return (implicit_ ? 0 : 27) + 20 + (hasInfo_ ? 18 : 0) +
+    // must match SlabLayout<EXPL, weighted()>::SW
+    return (implicit_ ? 0 : 27) + 20 + (weighted() ? 18 : 0) +
            (implicit_ ? 6 : 0);
   }
   void dispatchAssemble(int bak) {
-    auto launch = [&](auto hasInfoTag, auto explTag) {
-      constexpr bool HI = decltype(hasInfoTag)::value;
+    auto launch = [&](auto weightedTag, auto explTag) {
+      constexpr bool HI = decltype(weightedTag)::value;
       constexpr bool EX = decltype(explTag)::value;
       hipLaunchKernelGGL((kAssembleEdge<T, HI, EX>), dim3(gridFor(nL_)),
                          dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
                          dR_[bak], dJc_[bak], dJp_[bak], dInfo_, dHll_, dHpl_,
-                         dG_, ncam_, dCamPos_, dSlab_);
+                         dG_, ncam_, dCamPos_, dSlab_, lossKind_, lossD2_);
       if (nChunks_ > 0)
         hipLaunchKernelGGL((kAssembleCam<T, HI, EX>), dim3(nChunks_), dim3(128),
                            0, stream_, nChunks_, dChCam_, dChLo_, dChHi_,
@@ -1393,9 +1453,9 @@ class GpuEngine final : public Engine<T> {
     };
     using TrueT = std::integral_constant<bool, true>;
     using FalseT = std::integral_constant<bool, false>;
-    if (hasInfo_ && !implicit_) launch(TrueT{}, TrueT{});
-    else if (hasInfo_ && implicit_) launch(TrueT{}, FalseT{});
-    else if (!hasInfo_ && !implicit_) launch(FalseT{}, TrueT{});
+    if (weighted() && !implicit_) launch(TrueT{}, TrueT{});
+    else if (weighted() && implicit_) launch(TrueT{}, FalseT{});
+    else if (!weighted() && !implicit_) launch(FalseT{}, TrueT{});
     else launch(FalseT{}, FalseT{});
   }
   template <typename U>
@@ -1531,25 +1591,27 @@ class GpuEngine final : public Engine<T> {
       if (hasInfo_)
         hipLaunchKernelGGL((kSpmvEtx<T, true, true>), dim3(gridFor(nL_)),
                            dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
-                           (const T*)nullptr, dJc_[bak], dJp_[bak], dInfo_, xv,
-                           out);
+                           (const T*)nullptr, dJc_[bak], dJp_[bak], dInfo_,
+                           dR_[bak], lossKind_, lossD2_, xv, out);
       else
         hipLaunchKernelGGL((kSpmvEtx<T, true, false>), dim3(gridFor(nL_)),
                            dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
                            (const T*)nullptr, dJc_[bak], dJp_[bak],
-                           (const T*)nullptr, xv, out);
+                           (const T*)nullptr, dR_[bak], lossKind_, lossD2_,
+                           xv, out);
     } else {
       hipLaunchKernelGGL((kSpmvEtx<T, false, false>), dim3(gridFor(nL_)),
                          dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_, dHpl_,
                          (const T*)nullptr, (const T*)nullptr,
-                         (const T*)nullptr, xv, out);
+                         (const T*)nullptr, (const T*)nullptr, 0, T(0), xv,
+                         out);
     }
   }
   void spmvEx(const T* wv, T* out) {
     HIP_CHECK(hipMemsetAsync(out, 0, nc_ * sizeof(T), stream_));
     if (implicit_) {
       if (nChunks_ == 0) return;
-      if (hasInfo_)
+      if (weighted())
         hipLaunchKernelGGL((kSpmvExImp<T, true>), dim3(nChunks_), dim3(64), 0,
                            stream_, nChunks_, dChCam_, dChLo_, dChHi_,
                            dPtOfCam_, dSlab_, wv, out);
@@ -1645,6 +1707,8 @@ class GpuEngine final : public Engine<T> {
   bool hasInfo_ = false;
   bool analytical_ = false;
   bool implicit_ = false;
+  int lossKind_ = 0;
+  T lossD2_ = T(1);
   bool fwd2_ = getenv("MEGBA_FWD2") != nullptr;  // 2-lane Jet<6> experiment
   int cur_ = 0;
   bool freshCur_ = false;  // current r/J buffers hold the last forward()

@@ -183,4 +183,38 @@ MEGBA_HD inline Jet<T, N> abs(const Jet<T, N>& a) {
   return a.v < T(0) ? -a : a;
 }
 
+// Robust-loss transforms (host + device).  s = squared residual norm.
+// rho: the robust cost; w = d rho / d s: the IRLS weight applied to the
+// weighted J rows and residual in assembly (H ~ sum w J^T W J).
+template <typename T>
+MEGBA_HD inline T lossRho(int kind, T d2, T s) {
+  if (kind == 1) {  // Huber (delta^2 = d2)
+#ifdef __HIP_DEVICE_COMPILE__
+    return s <= d2 ? s : T(2) * ::sqrt(d2 * s) - d2;
+#else
+    return s <= d2 ? s : T(2) * std::sqrt(d2 * s) - d2;
+#endif
+  }
+  if (kind == 2) {  // Cauchy
+#ifdef __HIP_DEVICE_COMPILE__
+    return d2 * ::log(T(1) + s / d2);
+#else
+    return d2 * std::log(T(1) + s / d2);
+#endif
+  }
+  return s;
+}
+template <typename T>
+MEGBA_HD inline T lossWeight(int kind, T d2, T s) {
+  if (kind == 1) {
+#ifdef __HIP_DEVICE_COMPILE__
+    return s <= d2 ? T(1) : ::sqrt(d2 / s);
+#else
+    return s <= d2 ? T(1) : std::sqrt(d2 / s);
+#endif
+  }
+  if (kind == 2) return T(1) / (T(1) + s / d2);
+  return T(1);
+}
+
 }  // namespace megba

@@ -38,14 +38,19 @@ class BAProblem:
 
     # -- build -------------------------------------------------------------
     def build(self, device="cpu", dtype="float64", rank=0, world_size=1,
-              device_index=0, diff="auto", schur="explicit", allreduce=None,
-              rccl_id=None, custom_forward=None):
-        """custom_forward: optional callable (cam_jvs[9], pt_jvs[3],
+              device_index=0, diff="auto", schur="explicit", loss="none",
+              loss_delta=1.0, allreduce=None, rccl_id=None,
+              custom_forward=None):
+        """loss: robust loss ("none" | "huber" | "cauchy") with scale
+        loss_delta -- IRLS reweighting, rho-consistent cost (beyond the
+        reference, which has only the 2x2 information matrix).
+        custom_forward: optional callable (cam_jvs[9], pt_jvs[3],
         meas_jvs[2]) -> (res0, res1) of JetVector, evaluated per forward pass
         (runtime user-defined edges; see megba_amd.jv helpers)."""
         self._core.build(device=device, dtype=dtype, rank=rank,
                          world_size=world_size, device_index=device_index,
-                         diff=diff, schur=schur, allreduce=allreduce,
+                         diff=diff, schur=schur, loss=loss,
+                         loss_delta=loss_delta, allreduce=allreduce,
                          rccl_id=rccl_id, custom_forward=custom_forward)
         self._built = True
         return self

@@ -1,0 +1,128 @@
+"""Robust loss (Huber / Cauchy IRLS) — a capability beyond the reference
+(MegBA only supports the 2x2 information matrix,
+/root/reference/include/edge/base_edge.h information member).
+
+Checks: (a) rho-consistent chi2 (loss=none == plain sum r^2; huber chi2
+matches a numpy oracle); (b) huber recovers a corrupted problem better
+than plain L2; (c) cauchy runs and decreases cost; (d, gpu) GPU and CPU
+trajectories agree with loss enabled in every pipeline mode."""
+import numpy as np
+import pytest
+
+import megba_amd as mb
+from conftest import bal_residual_np
+
+
+def _chi2_np(cams, pts, ci, pi, meas, loss="none", delta=1.0):
+    s = 0.0
+    d2 = delta * delta
+    for e in range(len(ci)):
+        r = bal_residual_np(cams[ci[e]], pts[pi[e]], meas[e])
+        q = float(r @ r)
+        if loss == "huber":
+            s += q if q <= d2 else 2 * np.sqrt(d2 * q) - d2
+        elif loss == "cauchy":
+            s += d2 * np.log1p(q / d2)
+        else:
+            s += q
+    return s
+
+
+def _corrupted_problem(seed=11, frac=0.15, mag=60.0):
+    cams, pts, ci, pi, meas = mb.synthesize_bal(8, 60, 480, seed=seed,
+                                                pixel_noise=0.5)
+    rng = np.random.default_rng(seed + 1)
+    bad = rng.random(len(ci)) < frac
+    meas = meas.copy()
+    meas[bad] += rng.normal(scale=mag, size=(bad.sum(), 2))
+    return cams, pts, ci, pi, meas, bad
+
+
+def _solve(cams, pts, ci, pi, meas, loss, delta=2.0, **build_kw):
+    p = mb.BAProblem(cams, pts, ci, pi, meas)
+    p.build(device=build_kw.pop("device", "cpu"), loss=loss,
+            loss_delta=delta, **build_kw)
+    p.solve(max_iter=25, tau=1e4, epsilon1=1e-10, epsilon2=1e-14,
+            solver_tol=1e-8, solver_max_iter=300, solver_refuse_ratio=1e9,
+            verbose=False)
+    return p.get_params()
+
+
+def test_chi2_matches_rho_oracle():
+    cams, pts, ci, pi, meas = mb.synthesize_bal(5, 30, 220, seed=3)
+    for loss, delta in [("none", 1.0), ("huber", 2.0), ("cauchy", 3.0)]:
+        p = mb.BAProblem(cams, pts, ci, pi, meas)
+        p.build(device="cpu", loss=loss, loss_delta=delta)
+        rep = p.solve(max_iter=0, verbose=False)
+        want = _chi2_np(cams, pts, ci, pi, meas, loss, delta)
+        assert rep["iters"][0]["chi2"] == pytest.approx(want, rel=1e-10)
+
+
+def test_huber_beats_l2_on_outliers():
+    cams, pts, ci, pi, meas, bad = _corrupted_problem()
+    c_l2, p_l2 = _solve(cams, pts, ci, pi, meas, "none")
+    c_hu, p_hu = _solve(cams, pts, ci, pi, meas, "huber", delta=2.0)
+    # Judge on the CLEAN observations only: the huber solution must fit the
+    # inlier structure substantially better than the L2 one.
+    good = ~bad
+    gi = np.where(good)[0]
+    e_l2 = _chi2_np(c_l2, p_l2, ci[gi], pi[gi], meas[gi])
+    e_hu = _chi2_np(c_hu, p_hu, ci[gi], pi[gi], meas[gi])
+    assert e_hu < 0.5 * e_l2, (e_hu, e_l2)
+
+
+def test_cauchy_decreases_cost():
+    cams, pts, ci, pi, meas, _ = _corrupted_problem(seed=21)
+    p = mb.BAProblem(cams, pts, ci, pi, meas)
+    p.build(device="cpu", loss="cauchy", loss_delta=2.0)
+    rep = p.solve(max_iter=15, verbose=False)
+    # Cauchy saturates outliers, so the initial robust cost is already
+    # small; require a solid but not extreme reduction.
+    assert rep["final_chi2"] < 0.7 * rep["iters"][0]["chi2"]
+
+
+def test_loss_rejects_bad_name():
+    cams, pts, ci, pi, meas = mb.synthesize_bal(3, 12, 60, seed=1)
+    p = mb.BAProblem(cams, pts, ci, pi, meas)
+    with pytest.raises(RuntimeError):
+        p.build(device="cpu", loss="tukey")
+
+
+def test_huber_with_information_matrix():
+    # Loss weight composes with the per-edge 2x2 information matrix.
+    cams, pts, ci, pi, meas, _ = _corrupted_problem(seed=31)
+    rng = np.random.default_rng(5)
+    info = np.zeros((len(ci), 3))
+    info[:, 0] = rng.uniform(0.5, 2.0, len(ci))   # i00
+    info[:, 1] = 0.0                              # i01
+    info[:, 2] = rng.uniform(0.5, 2.0, len(ci))   # i11
+    p = mb.BAProblem(cams, pts, ci, pi, meas, info=info)
+    p.build(device="cpu", loss="huber", loss_delta=2.0)
+    rep = p.solve(max_iter=8, verbose=False)
+    assert rep["final_chi2"] < rep["iters"][0]["chi2"]
+    assert np.isfinite(rep["final_chi2"])
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("loss", ["huber", "cauchy"])
+@pytest.mark.parametrize("mode", [
+    dict(diff="auto", schur="explicit"),
+    dict(diff="analytical", schur="explicit"),
+    dict(diff="auto", schur="implicit"),
+])
+def test_gpu_matches_cpu_trajectory(loss, mode):
+    cams, pts, ci, pi, meas, _ = _corrupted_problem(seed=41)
+
+    def run(device):
+        p = mb.BAProblem(cams, pts, ci, pi, meas)
+        p.build(device=device, loss=loss, loss_delta=2.0, **mode)
+        rep = p.solve(max_iter=8, tau=1e4, solver_tol=1e-8,
+                      solver_max_iter=200, solver_refuse_ratio=1e9,
+                      verbose=False)
+        return [it["chi2"] for it in rep["iters"]]
+
+    c = run("cpu")
+    g = run("gpu")
+    assert len(c) == len(g)
+    for a, b in zip(c, g):
+        assert a == pytest.approx(b, rel=1e-6, abs=1e-9)
