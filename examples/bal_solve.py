@@ -48,6 +48,8 @@ def main():
     ap.add_argument("--dtype", default="float64", choices=["float64", "float32"])
     ap.add_argument("--diff", default="auto", choices=["auto", "analytical"])
     ap.add_argument("--schur", default="explicit", choices=["explicit", "implicit"])
+    ap.add_argument("--loss", default="none", choices=["none", "huber", "cauchy"])
+    ap.add_argument("--loss_delta", type=float, default=1.0)
     ap.add_argument("--out", default="", help="write solved BAL file here")
     args = ap.parse_args()
 
@@ -84,7 +86,8 @@ def main():
     p = mb.BAProblem(cams, pts, ci, pi, meas)
     p.build(device=args.device, dtype=args.dtype, rank=rank, world_size=world,
             device_index=int(os.environ.get("LOCAL_RANK", rank)),
-            diff=args.diff, schur=args.schur, allreduce=allreduce,
+            diff=args.diff, schur=args.schur, loss=args.loss,
+            loss_delta=args.loss_delta, allreduce=allreduce,
             rccl_id=rccl_id)
     p.solve(max_iter=args.max_iter, tau=args.tau, epsilon1=args.epsilon1,
             epsilon2=args.epsilon2, solver_max_iter=args.solver_max_iter,

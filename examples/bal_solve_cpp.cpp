@@ -40,6 +40,8 @@ static BAProblemHost loadBal(const std::string& path) {
 
 int main(int argc, char** argv) {
   std::string path, device = "gpu", diff = "auto", schur = "explicit";
+  std::string loss = "none";
+  double lossDelta = 1.0;
   AlgoOptionLM algo;
   SolverOptionPCG sopt;
   sopt.maxIter = 50;
@@ -60,6 +62,8 @@ int main(int argc, char** argv) {
     else if (arg("--tau")) algo.initialRegion = std::atof(argv[++i]);
     else if (arg("--epsilon1")) algo.epsilon1 = std::atof(argv[++i]);
     else if (arg("--epsilon2")) algo.epsilon2 = std::atof(argv[++i]);
+    else if (arg("--loss")) loss = argv[++i];
+    else if (arg("--loss_delta")) lossDelta = std::atof(argv[++i]);
     else {
       std::cerr << "unknown/incomplete flag " << argv[i] << "\n";
       return 2;
@@ -78,6 +82,10 @@ int main(int argc, char** argv) {
   opt.device = device == "cpu" ? Device::CPU : Device::GPU;
   opt.diff = diff == "analytical" ? DiffMode::ANALYTICAL : DiffMode::AUTO;
   opt.schur = schur == "implicit" ? SchurMode::IMPLICIT : SchurMode::EXPLICIT;
+  opt.loss = loss == "huber" ? LossKind::HUBER
+             : loss == "cauchy" ? LossKind::CAUCHY
+                                : LossKind::NONE;
+  opt.lossDelta = lossDelta;
   ProblemIndex ix = buildIndex(prob, 1);
   std::unique_ptr<Engine<double>> eng;
   if (opt.device == Device::CPU)
