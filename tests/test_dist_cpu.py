@@ -118,3 +118,43 @@ def test_bench_contract_torchrun_cpu(tmp_path):
     d = json.loads(line)
     assert d["n_gpus"] == 2 and d["steps"] == 2
     assert d["metric"] == "lm_iterations_per_s" and d["value"] > 0
+
+def _worker_loss(rank, world_size, port, out_path):
+    import torch.distributed as dist
+    from megba_amd.dist import gloo_allreduce_callback
+    dist.init_process_group(
+        "gloo", init_method=f"tcp://127.0.0.1:{port}",
+        rank=rank, world_size=world_size)
+    try:
+        import megba_amd as mb
+        cams, pts, ci, pi, meas = mb.synthesize_bal(*SHAPE, seed=SEED)
+        p = mb.BAProblem(cams, pts, ci, pi, meas)
+        p.build(device="cpu", rank=rank, world_size=world_size,
+                loss="huber", loss_delta=2.0,
+                allreduce=gloo_allreduce_callback())
+        rep = p.solve(max_iter=6, tau=1e4, solver_tol=1e-6,
+                      solver_max_iter=300, solver_refuse_ratio=1e6,
+                      verbose=False)
+        p.get_params()
+        if rank == 0:
+            with open(out_path, "w") as f:
+                json.dump([it["chi2"] for it in rep["iters"]], f)
+    finally:
+        dist.destroy_process_group()
+
+
+def test_world2_robust_loss_matches_world1(tmp_path):
+    """Robust loss weighting is per-edge and must commute with the edge
+    partition: sharded IRLS == single-process IRLS."""
+    import megba_amd as mb
+    import torch.multiprocessing as mp
+    cams, pts, ci, pi, meas = mb.synthesize_bal(*SHAPE, seed=SEED)
+    p1 = mb.BAProblem(cams, pts, ci, pi, meas)
+    p1.build(device="cpu", loss="huber", loss_delta=2.0)
+    rep = p1.solve(max_iter=6, tau=1e4, solver_tol=1e-6, solver_max_iter=300,
+                   solver_refuse_ratio=1e6, verbose=False)
+    ref = [it["chi2"] for it in rep["iters"]]
+    out = tmp_path / "chis_loss.json"
+    mp.spawn(_worker_loss, args=(2, 29513, str(out)), nprocs=2, join=True)
+    chis = json.loads(out.read_text())
+    np.testing.assert_allclose(chis, ref, rtol=1e-6)
