@@ -104,7 +104,9 @@ def main():
         try:
             import torch
             if torch.cuda.is_available():
-                torch.cuda.synchronize()
+                # explicit device: this rank's engine device, not torch's
+                # current device (0), matters on an 8-GPU node.
+                torch.cuda.synchronize(device_index)
         except Exception:
             pass
 
