@@ -158,3 +158,48 @@ def test_world2_robust_loss_matches_world1(tmp_path):
     mp.spawn(_worker_loss, args=(2, 29513, str(out)), nprocs=2, join=True)
     chis = json.loads(out.read_text())
     np.testing.assert_allclose(chis, ref, rtol=1e-6)
+
+
+def _worker_tiny(rank, world_size, port, out_path):
+    import torch.distributed as dist
+    from megba_amd.dist import gloo_allreduce_callback
+    dist.init_process_group(
+        "gloo", init_method=f"tcp://127.0.0.1:{port}",
+        rank=rank, world_size=world_size)
+    try:
+        import megba_amd as mb
+        cams, pts, ci, pi, meas = mb.synthesize_bal(3, 8, 20, seed=2)
+        p = mb.BAProblem(cams, pts, ci, pi, meas)
+        p.build(device="cpu", rank=rank, world_size=world_size,
+                allreduce=gloo_allreduce_callback())
+        rep = p.solve(max_iter=4, tau=1e4, solver_tol=1e-8,
+                      solver_max_iter=100, solver_refuse_ratio=1e9,
+                      verbose=False)
+        p.get_params()
+        if rank == 0:
+            with open(out_path, "w") as f:
+                json.dump([it["chi2"] for it in rep["iters"]], f)
+    finally:
+        dist.destroy_process_group()
+
+
+def test_more_ranks_than_points_worth_of_edges(tmp_path):
+    """Degenerate partition: 8 points over 4 ranks -> thin or empty shards.
+    The distributed path must stay correct (and not deadlock) even when a
+    rank owns few or zero edges."""
+    import megba_amd as mb
+    import torch.multiprocessing as mp
+    cams, pts, ci, pi, meas = mb.synthesize_bal(3, 8, 20, seed=2)
+    p1 = mb.BAProblem(cams, pts, ci, pi, meas)
+    p1.build(device="cpu")
+    rep = p1.solve(max_iter=4, tau=1e4, solver_tol=1e-8,
+                   solver_max_iter=100, solver_refuse_ratio=1e9,
+                   verbose=False)
+    ref = [it["chi2"] for it in rep["iters"]]
+    out = tmp_path / "chis_tiny.json"
+    mp.spawn(_worker_tiny, args=(4, 29514, str(out)), nprocs=4, join=True)
+    chis = json.loads(out.read_text())
+    # Summation order differs between world sizes, and this deliberately
+    # tiny/ill-conditioned problem amplifies it through the trajectory;
+    # the check here is no-deadlock + same optimization outcome.
+    np.testing.assert_allclose(chis, ref, rtol=5e-2)
