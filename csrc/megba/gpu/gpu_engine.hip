@@ -179,60 +179,6 @@ __global__ __launch_bounds__(256, 2) void kForward(
   if (threadIdx.x == 0) atomicAdd(chi2Acc, sm[0]);
 }
 
-// Experimental: 2-lane split with Jet<T,6> (half the value redundancy of the
-// 4-lane version; higher register pressure).  Selected at runtime if it
-// proves faster; kept behind the same layout contract.
-template <typename T>
-__global__ __launch_bounds__(256, 1) void kForward2(
-    int64_t nL, const int* __restrict__ camOf, const int* __restrict__ ptOf,
-    const T* __restrict__ params, int ncam, const T* __restrict__ meas,
-    const unsigned char* __restrict__ camFixed,
-    const unsigned char* __restrict__ ptFixed, T* __restrict__ rOut,
-    T* __restrict__ Jc, T* __restrict__ Jp, double* chi2Acc, int lossKind,
-    T lossD2) {
-  using J6 = Jet<T, 6>;
-  __shared__ double sm[kBlk];
-  double chi2 = 0.0;
-  const T* ptsBase = params + (int64_t)ncam * 9;
-  const int64_t nWork = nL * 2;
-  for (int64_t i = blockIdx.x * (int64_t)kBlk + threadIdx.x; i < nWork;
-       i += (int64_t)gridDim.x * kBlk) {
-    const int64_t e = i >> 1;
-    const int sub = (int)(i & 1);
-    const int base = 6 * sub;           // gradient columns [base,base+6)
-    const T* cp = params + (int64_t)camOf[e] * 9;
-    const T* pp = ptsBase + (int64_t)ptOf[e] * 3;
-    J6 cam[9], pt[3], res[2];
-    for (int k = 0; k < 9; ++k) cam[k] = J6::leaf(cp[k], k - base);
-    for (int k = 0; k < 3; ++k) pt[k] = J6::leaf(pp[k], 9 + k - base);
-    const T m[2] = {meas[2 * e], meas[2 * e + 1]};
-    balReprojectionError<T, J6>(cam, pt, m, res);
-    const bool cfix = camFixed && camFixed[camOf[e]];
-    const bool pfix = ptFixed && ptFixed[ptOf[e]];
-    for (int row = 0; row < 2; ++row) {
-      for (int j = 0; j < 6; ++j) {
-        const int col = base + j;
-        if (col < 9)
-          Jc[((int64_t)(col * 2 + row)) * nL + e] = cfix ? T(0) : res[row].d[j];
-        else
-          Jp[((int64_t)((col - 9) * 2 + row)) * nL + e] =
-              pfix ? T(0) : res[row].d[j];
-      }
-      if (sub == 0) rOut[(int64_t)row * nL + e] = res[row].v;
-    }
-    if (sub == 0)
-      chi2 += (double)lossRho(lossKind, lossD2,
-                              res[0].v * res[0].v + res[1].v * res[1].v);
-  }
-  sm[threadIdx.x] = chi2;
-  __syncthreads();
-  for (int s = kBlk / 2; s > 0; s >>= 1) {
-    if (threadIdx.x < s) sm[threadIdx.x] += sm[threadIdx.x + s];
-    __syncthreads();
-  }
-  if (threadIdx.x == 0) atomicAdd(chi2Acc, sm[0]);
-}
-
 // Analytical-derivative forward: one thread per edge, closed-form residual +
 // 2x12 Jacobian (reference C11, src/geo/analytical_derivatives.cu).
 template <typename T>
@@ -1144,11 +1090,6 @@ class GpuEngine final : public Engine<T> {
                          0, stream_, nL_, dCamOf_, dPtOf_, dParams_, ncam_,
                          dMeas_, dCamFixed_, dPtFixed_, dR_[cur_], dJc_[cur_],
                          dJp_[cur_], scalarPtr(), lossKind_, lossD2_);
-    else if (fwd2_ && lossKind_ == 0)
-      hipLaunchKernelGGL(kForward2<T>, dim3(gridFor(nL_ * 2)), dim3(kBlk), 0,
-                         stream_, nL_, dCamOf_, dPtOf_, dParams_, ncam_,
-                         dMeas_, dCamFixed_, dPtFixed_, dR_[cur_], dJc_[cur_],
-                         dJp_[cur_], scalarPtr(), lossKind_, lossD2_);
     else
       hipLaunchKernelGGL(kForward<T>, dim3(gridFor(nL_ * 4)), dim3(kBlk), 0,
                          stream_, nL_, dCamOf_, dPtOf_, dParams_, ncam_,
@@ -1709,7 +1650,6 @@ class GpuEngine final : public Engine<T> {
   bool implicit_ = false;
   int lossKind_ = 0;
   T lossD2_ = T(1);
-  bool fwd2_ = getenv("MEGBA_FWD2") != nullptr;  // 2-lane Jet<6> experiment
   int cur_ = 0;
   bool freshCur_ = false;  // current r/J buffers hold the last forward()
   int nChunks_ = 0;
