@@ -110,11 +110,18 @@ def test_huber_with_information_matrix():
     dict(diff="analytical", schur="explicit"),
     dict(diff="auto", schur="implicit"),
 ])
-def test_gpu_matches_cpu_trajectory(loss, mode):
+@pytest.mark.parametrize("with_info", [False, True])
+def test_gpu_matches_cpu_trajectory(loss, mode, with_info):
     cams, pts, ci, pi, meas, _ = _corrupted_problem(seed=41)
+    info = None
+    if with_info:
+        rng = np.random.default_rng(9)
+        info = np.zeros((len(ci), 3))
+        info[:, 0] = rng.uniform(0.5, 2.0, len(ci))
+        info[:, 2] = rng.uniform(0.5, 2.0, len(ci))
 
     def run(device):
-        p = mb.BAProblem(cams, pts, ci, pi, meas)
+        p = mb.BAProblem(cams, pts, ci, pi, meas, info=info)
         p.build(device=device, loss=loss, loss_delta=2.0, **mode)
         rep = p.solve(max_iter=8, tau=1e4, solver_tol=1e-8,
                       solver_max_iter=200, solver_refuse_ratio=1e9,
