@@ -13,9 +13,7 @@
 #include <string>
 
 #include "megba/common.hpp"
-#include "megba/cpu_engine.hpp"
-#include "megba/gpu/gpu_engine.hpp"
-#include "megba/lm.hpp"
+#include "megba/graph_api.hpp"
 #include "megba/problem.hpp"
 
 using namespace megba;
@@ -75,9 +73,30 @@ int main(int argc, char** argv) {
                  "[--max_iter N] [--solver_* ...] [--tau T]\n";
     return 2;
   }
-  BAProblemHost prob = loadBal(path);
-  std::cout << "solving " << path << " (" << prob.ncam << " cams, " << prob.npt
-            << " pts, " << prob.nobs << " obs), device=" << device << "\n";
+  BAProblemHost raw = loadBal(path);
+  std::cout << "solving " << path << " (" << raw.ncam << " cams, " << raw.npt
+            << " pts, " << raw.nobs << " obs), device=" << device << "\n";
+
+  // Build through the g2o-style graph API — the same construction flow as
+  // the reference demo (examples/BAL_Double.cpp:60-164: one vertex per
+  // camera/point, one edge per observation, then solve + writeBack).
+  std::vector<BaseVertex> camVs, ptVs;
+  camVs.reserve(raw.ncam);
+  ptVs.reserve(raw.npt);
+  for (int i = 0; i < raw.ncam; ++i)
+    camVs.emplace_back(VertexKind::CAMERA, &raw.cams[9 * i]);
+  for (int i = 0; i < raw.npt; ++i)
+    ptVs.emplace_back(VertexKind::POINT, &raw.pts[3 * i]);
+  GraphProblem graph;
+  for (auto& v : camVs) graph.appendVertex(&v);
+  for (auto& v : ptVs) graph.appendVertex(&v);
+  for (int64_t k = 0; k < raw.nobs; ++k) {
+    ReprojectionEdge e;
+    e.appendVertex(&camVs[raw.camIdx[k]]).appendVertex(&ptVs[raw.ptIdx[k]])
+        .setMeasurement(raw.meas[2 * k], raw.meas[2 * k + 1]);
+    graph.appendEdge(e);
+  }
+
   ProblemOption opt;
   opt.device = device == "cpu" ? Device::CPU : Device::GPU;
   opt.diff = diff == "analytical" ? DiffMode::ANALYTICAL : DiffMode::AUTO;
@@ -86,13 +105,7 @@ int main(int argc, char** argv) {
              : loss == "cauchy" ? LossKind::CAUCHY
                                 : LossKind::NONE;
   opt.lossDelta = lossDelta;
-  ProblemIndex ix = buildIndex(prob, 1);
-  std::unique_ptr<Engine<double>> eng;
-  if (opt.device == Device::CPU)
-    eng = makeCpuEngine<double>(prob, ix, opt, nullptr);
-  else
-    eng = makeGpuEngine<double>(prob, ix, opt, std::string());
-  LMReport rep = runLM<double>(*eng, algo, sopt);
+  LMReport rep = graph.solve(opt, algo, sopt);
   std::cout << "final error: " << rep.finalChi2 / 2 << " after "
             << rep.acceptedSteps << " accepted / " << rep.rejectedSteps
             << " rejected steps, " << rep.totalMs << " ms\n";

@@ -9,6 +9,8 @@ over xGMI (one process per GPU).
 """
 from .problem import BAProblem, load_bal, save_bal  # noqa: F401
 from .synth import synthesize_bal  # noqa: F401
+from .graph import (  # noqa: F401
+    GraphProblem, CameraVertex, PointVertex, ReprojectionEdge)
 
 __version__ = "0.1.0"
 
