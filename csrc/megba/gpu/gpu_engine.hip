@@ -499,6 +499,62 @@ __global__ void kFinalizeCam(int64_t nL, const T* __restrict__ slab,
   }
 }
 
+// Implicit only: stream the slab's weighted-Jc and Jp back out as cam-sorted
+// GRAD-MAJOR copies.  The first E w implementation read the 26-value
+// edge-major slab rows directly each PCG iteration; with 64 lanes striding
+// SW*sizeof(T) apart every one of its ~26 loads is address-divergent
+// (~64 TA requests/instruction), which measured 2.2x the byte floor at 29M
+// edges (profiles/r01_final13682_implicit_fp32.md).  Paying one extra
+// coalesced-write pass per ACCEPTED LM step makes the per-iteration reads
+// fully coalesced instead.
+template <typename T, bool HASINFO>
+__global__ void kFinalizeCamImp(int64_t nL, const T* __restrict__ slab,
+                                T* __restrict__ JcCam,
+                                T* __restrict__ JpCam) {
+  using L = SlabLayout<false, HASINFO>;
+  constexpr int woff = HASINFO ? L::WJCOFF : L::JCOFF;
+  for (int64_t j = blockIdx.x * (int64_t)kBlk + threadIdx.x; j < nL;
+       j += (int64_t)gridDim.x * kBlk) {
+    const T* row = slab + j * L::SW;
+    for (int k = 0; k < 18; ++k) JcCam[(int64_t)k * nL + j] = row[woff + k];
+    for (int k = 0; k < 6; ++k) JpCam[(int64_t)k * nL + j] = row[L::JPOFF + k];
+  }
+}
+
+// Implicit E w over the grad-major cam-sorted copies (weights already folded
+// into JcCam): u = Jp w, acc += wJc^T u; chunk/wave/atomic structure as
+// kSpmvEx.
+template <typename T>
+__global__ __launch_bounds__(64) void kSpmvExImpG(
+    int nChunks, const int* __restrict__ chCam, const int* __restrict__ chLo,
+    const int* __restrict__ chHi, const int* __restrict__ ptOfCam,
+    const T* __restrict__ JcCam, const T* __restrict__ JpCam, int64_t nL,
+    const T* __restrict__ w, T* __restrict__ out) {
+  const int chunk = blockIdx.x;
+  if (chunk >= nChunks) return;
+  const int cam = chCam[chunk];
+  T acc[9];
+  for (int i = 0; i < 9; ++i) acc[i] = T(0);
+  const int lo = chLo[chunk], hi = chHi[chunk];
+  for (int j = lo + (int)threadIdx.x; j < hi; j += 64) {
+    const T* wp = w + (int64_t)ptOfCam[j] * 3;
+    T u0 = T(0), u1 = T(0);
+    for (int c = 0; c < 3; ++c) {
+      u0 += JpCam[((int64_t)(c * 2 + 0)) * nL + j] * wp[c];
+      u1 += JpCam[((int64_t)(c * 2 + 1)) * nL + j] * wp[c];
+    }
+    for (int i = 0; i < 9; ++i)
+      acc[i] += JcCam[((int64_t)(i * 2 + 0)) * nL + j] * u0 +
+                JcCam[((int64_t)(i * 2 + 1)) * nL + j] * u1;
+  }
+  for (int off = 32; off > 0; off >>= 1)
+    for (int i = 0; i < 9; ++i) acc[i] += __shfl_down(acc[i], off, 64);
+  if (threadIdx.x == 0) {
+    T* oc = out + (int64_t)cam * 9;
+    for (int i = 0; i < 9; ++i) atomicAdd(&oc[i], acc[i]);
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Damping, block inverse
 // ---------------------------------------------------------------------------
@@ -1066,6 +1122,10 @@ class GpuEngine final : public Engine<T> {
       up(dChHi_, cHi.data(), nChunks_);
     }
     dSlab_ = dalloc<T>(nL_ * slabWidth());
+    if (implicit_) {
+      dJcCam_ = dalloc<T>(nL_ * 18);
+      dJpCam_ = dalloc<T>(nL_ * 6);
+    }
     if (!implicit_) {
       dHpl_ = dalloc<T>(nL_ * 27);
       dHplCam_ = dalloc<T>(nL_ * 27);
@@ -1154,6 +1214,15 @@ class GpuEngine final : public Engine<T> {
       else
         hipLaunchKernelGGL((kFinalizeCam<T, false>), dim3(gridFor(nL_)),
                            dim3(kBlk), 0, stream_, nL_, dSlab_, dHplCam_);
+    } else {
+      if (weighted())
+        hipLaunchKernelGGL((kFinalizeCamImp<T, true>), dim3(gridFor(nL_)),
+                           dim3(kBlk), 0, stream_, nL_, dSlab_, dJcCam_,
+                           dJpCam_);
+      else
+        hipLaunchKernelGGL((kFinalizeCamImp<T, false>), dim3(gridFor(nL_)),
+                           dim3(kBlk), 0, stream_, nL_, dSlab_, dJcCam_,
+                           dJpCam_);
     }
     // Only the small camera-side quantities cross ranks (the reference
     // allreduced Hpp, Hll AND g, its site A1).
@@ -1552,14 +1621,9 @@ class GpuEngine final : public Engine<T> {
     HIP_CHECK(hipMemsetAsync(out, 0, nc_ * sizeof(T), stream_));
     if (implicit_) {
       if (nChunks_ == 0) return;
-      if (weighted())
-        hipLaunchKernelGGL((kSpmvExImp<T, true>), dim3(nChunks_), dim3(64), 0,
-                           stream_, nChunks_, dChCam_, dChLo_, dChHi_,
-                           dPtOfCam_, dSlab_, wv, out);
-      else
-        hipLaunchKernelGGL((kSpmvExImp<T, false>), dim3(nChunks_), dim3(64), 0,
-                           stream_, nChunks_, dChCam_, dChLo_, dChHi_,
-                           dPtOfCam_, dSlab_, wv, out);
+      hipLaunchKernelGGL(kSpmvExImpG<T>, dim3(nChunks_), dim3(64), 0,
+                         stream_, nChunks_, dChCam_, dChLo_, dChHi_,
+                         dPtOfCam_, dJcCam_, dJpCam_, nL_, wv, out);
       return;
     }
     if (nChunks_ > 0)
@@ -1660,6 +1724,7 @@ class GpuEngine final : public Engine<T> {
   T *dParams_{}, *dParamsBak_{};
   T *dR_[2]{}, *dJc_[2]{}, *dJp_[2]{};
   T *dHpp_{}, *dHll_{}, *dHpl_{}, *dHplCam_{}, *dSlab_{}, *dG_{}, *dGBak_{};
+  T *dJcCam_{}, *dJpCam_{};  // implicit: cam-sorted grad-major wJc / Jp
   T *dHppD_{}, *dHllD_{}, *dHppInv_{}, *dHllInv_{};
   T *dDeltaX_{}, *dDeltaXBak_{};
   T *dP_{}, *dRr_{}, *dZ_{}, *dQ_{}, *dV_{}, *dW_{}, *dTemp_{}, *dXBak_{},
