@@ -731,43 +731,6 @@ __global__ __launch_bounds__(64) void kSpmvEx(
   }
 }
 
-// Implicit E w: matrix-free, chunked over the cam-sorted slab rows (which
-// carry Jc/wJc/Jp): u = Jp w (raw), acc += wJc^T u  (Jc^T W Jp w with W
-// symmetric).  Wave reduce, one atomicAdd set per chunk — a per-edge atomic
-// variant measured 2.43 ms/call on Venice-5M vs ~0.6 ms for this one.
-template <typename T, bool HASINFO>
-__global__ __launch_bounds__(64) void kSpmvExImp(
-    int nChunks, const int* __restrict__ chCam, const int* __restrict__ chLo,
-    const int* __restrict__ chHi, const int* __restrict__ ptOfCam,
-    const T* __restrict__ slab, const T* __restrict__ w,
-    T* __restrict__ out) {
-  using L = SlabLayout<false, HASINFO>;
-  const int chunk = blockIdx.x;
-  if (chunk >= nChunks) return;
-  const int cam = chCam[chunk];
-  T acc[9];
-  for (int i = 0; i < 9; ++i) acc[i] = T(0);
-  const int lo = chLo[chunk], hi = chHi[chunk];
-  const int woff = HASINFO ? L::WJCOFF : L::JCOFF;
-  for (int j = lo + (int)threadIdx.x; j < hi; j += 64) {
-    const T* row = slab + (int64_t)j * L::SW;
-    const T* wp = w + (int64_t)ptOfCam[j] * 3;
-    T u0 = T(0), u1 = T(0);
-    for (int c = 0; c < 3; ++c) {
-      u0 += row[L::JPOFF + c * 2] * wp[c];
-      u1 += row[L::JPOFF + c * 2 + 1] * wp[c];
-    }
-    for (int i = 0; i < 9; ++i)
-      acc[i] += row[woff + i * 2] * u0 + row[woff + i * 2 + 1] * u1;
-  }
-  for (int off = 32; off > 0; off >>= 1)
-    for (int i = 0; i < 9; ++i) acc[i] += __shfl_down(acc[i], off, 64);
-  if (threadIdx.x == 0) {
-    T* oc = out + (int64_t)cam * 9;
-    for (int i = 0; i < 9; ++i) atomicAdd(&oc[i], acc[i]);
-  }
-}
-
 // Fused preconditioner apply + rho partial: z = Binv r (thread per row) and
 // per-block partials of r.z in one pass (saves two launches per PCG iter).
 template <typename T>
