@@ -11,9 +11,11 @@ run trafalgar257 float64 analytical explicit 10
 run venice1778 float64 auto explicit 10
 run venice1778 float64 analytical explicit 10
 run venice1778 float64 auto implicit 10
+run venice1778 float32 auto implicit 10
 run venice1778 float32 analytical explicit 10
 run final13682 float32 analytical implicit 4
 run final13682 float32 analytical explicit 4
+run synth20k float64 auto implicit 4
 run synth20k float64 auto explicit 4
 } > gpurun_out/bench_table.jsonl
 python - <<'PYEOF'
