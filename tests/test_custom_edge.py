@@ -65,3 +65,39 @@ def test_custom_solve_gpu():
     r1, r2 = ref.solve(**kw), cust.solve(**kw)
     np.testing.assert_allclose([i["chi2"] for i in r2["iters"]],
                                [i["chi2"] for i in r1["iters"]], rtol=1e-6)
+
+
+def _pair_loss(device):
+    # custom forward composed with robust loss: the engine's chi2 must apply
+    # rho to the user-residual (kChi2Loss on GPU), and assembly must weight
+    # the repacked J rows identically to the built-in path.
+    cams, pts, ci, pi, meas = mb.synthesize_bal(12, 120, 1100, seed=13)
+    rng = np.random.default_rng(1)
+    bad = rng.random(len(ci)) < 0.1
+    meas = meas.copy()
+    meas[bad] += rng.normal(scale=40.0, size=(int(bad.sum()), 2))
+    ref = mb.BAProblem(cams, pts, ci, pi, meas)
+    ref.build(device=device, loss="huber", loss_delta=2.0)
+    cust = mb.BAProblem(cams, pts, ci, pi, meas)
+    cust.build(device=device, loss="huber", loss_delta=2.0,
+               custom_forward=bal_forward)
+    return ref, cust
+
+
+def test_custom_forward_with_loss_cpu():
+    ref, cust = _pair_loss("cpu")
+    kw = dict(max_iter=5, solver_tol=1e-6, solver_max_iter=200,
+              solver_refuse_ratio=1e6, verbose=False)
+    r1, r2 = ref.solve(**kw), cust.solve(**kw)
+    np.testing.assert_allclose([i["chi2"] for i in r2["iters"]],
+                               [i["chi2"] for i in r1["iters"]], rtol=1e-6)
+
+
+@pytest.mark.gpu
+def test_custom_forward_with_loss_gpu():
+    ref, cust = _pair_loss("gpu")
+    kw = dict(max_iter=5, solver_tol=1e-6, solver_max_iter=200,
+              solver_refuse_ratio=1e6, verbose=False)
+    r1, r2 = ref.solve(**kw), cust.solve(**kw)
+    np.testing.assert_allclose([i["chi2"] for i in r2["iters"]],
+                               [i["chi2"] for i in r1["iters"]], rtol=1e-6)
