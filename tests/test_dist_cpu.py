@@ -203,3 +203,45 @@ def test_more_ranks_than_points_worth_of_edges(tmp_path):
     # tiny/ill-conditioned problem amplifies it through the trajectory;
     # the check here is no-deadlock + same optimization outcome.
     np.testing.assert_allclose(chis, ref, rtol=5e-2)
+
+
+def _worker8(rank, world_size, port, out_path):
+    import torch.distributed as dist
+    from megba_amd.dist import gloo_allreduce_callback
+    dist.init_process_group(
+        "gloo", init_method=f"tcp://127.0.0.1:{port}",
+        rank=rank, world_size=world_size)
+    try:
+        import megba_amd as mb
+        cams, pts, ci, pi, meas = mb.synthesize_bal(*SHAPE, seed=SEED)
+        p = mb.BAProblem(cams, pts, ci, pi, meas)
+        p.build(device="cpu", rank=rank, world_size=world_size,
+                schur="implicit", allreduce=gloo_allreduce_callback())
+        rep = p.solve(max_iter=5, tau=1e4, solver_tol=1e-6,
+                      solver_max_iter=200, solver_refuse_ratio=1e6,
+                      verbose=False)
+        p.get_params()
+        if rank == 0:
+            with open(out_path, "w") as f:
+                json.dump([it["chi2"] for it in rep["iters"]], f)
+    finally:
+        dist.destroy_process_group()
+
+
+def test_world8_implicit_matches_world1():
+    """World size 8 (the driver's max scaling config), implicit mode (the
+    flagship bench mode): the sharded trajectory must track world-1."""
+    import tempfile
+    import megba_amd as mb
+    import torch.multiprocessing as mp
+    cams, pts, ci, pi, meas = mb.synthesize_bal(*SHAPE, seed=SEED)
+    p1 = mb.BAProblem(cams, pts, ci, pi, meas)
+    p1.build(device="cpu", schur="implicit")
+    rep = p1.solve(max_iter=5, tau=1e4, solver_tol=1e-6, solver_max_iter=200,
+                   solver_refuse_ratio=1e6, verbose=False)
+    ref = [it["chi2"] for it in rep["iters"]]
+    with tempfile.TemporaryDirectory() as d:
+        out = os.path.join(d, "chis8.json")
+        mp.spawn(_worker8, args=(8, 29515, out), nprocs=8, join=True)
+        chis = json.loads(open(out).read())
+    np.testing.assert_allclose(chis, ref, rtol=1e-5)
