@@ -96,6 +96,13 @@ struct PyProblem {
     opt.rank = rank;
     opt.worldSize = worldSize;
     opt.deviceIndex = deviceIndex;
+    MEGBA_CHECK(device == "cpu" || device == "gpu", "device must be cpu|gpu");
+    MEGBA_CHECK(dtype == "float64" || dtype == "float32",
+                "dtype must be float64|float32");
+    MEGBA_CHECK(diff == "auto" || diff == "analytical",
+                "diff must be auto|analytical");
+    MEGBA_CHECK(schur == "explicit" || schur == "implicit",
+                "schur must be explicit|implicit");
     opt.device = device == "gpu" ? Device::GPU : Device::CPU;
     opt.diff = diff == "analytical" ? DiffMode::ANALYTICAL : DiffMode::AUTO;
     opt.schur = schur == "implicit" ? SchurMode::IMPLICIT : SchurMode::EXPLICIT;
