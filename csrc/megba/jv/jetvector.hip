@@ -341,6 +341,19 @@ static JetVec<T> denseLike(const JetVec<T>& a, const JetVec<T>& b) {
 
 template <typename T>
 JetVec<T> jvBinary(JvOp op, const JetVec<T>& a, const JetVec<T>& b) {
+  if (a.isScalar && b.isScalar) {
+    // Pure-scalar op: host arithmetic, scalar result (the reference's
+    // PURE_SCALAR_OP dispatch, src/operator/jet_vector.cpp).
+    MEGBA_CHECK(a.N == b.N, "JetVector gradient-width mismatch");
+    T r = T(0);
+    switch (op) {
+      case JvOp::Add: r = a.scalarVal + b.scalarVal; break;
+      case JvOp::Sub: r = a.scalarVal - b.scalarVal; break;
+      case JvOp::Mul: r = a.scalarVal * b.scalarVal; break;
+      case JvOp::Div: r = a.scalarVal / b.scalarVal; break;
+    }
+    return jvScalar<T>(r, a.N);
+  }
   JetVec<T> out = denseLike(a, b);
   switch (op) {
     case JvOp::Add: dispatchBinary<T, 0>(a, b, out); break;

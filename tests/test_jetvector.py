@@ -79,6 +79,18 @@ def run_binary_suite(gpu):
         rs = Ref(np.full(NI, 1.75))
         check(op(a, s), ref_op(ra, rs))
         check(op(s, b), ref_op(rs, rb))
+        # scalar x JPV and scalar x scalar (completes the reference's
+        # 39-variant kind matrix, jet_vector_math_impl.cu dispatchers)
+        check(op(c, s), ref_op(rc, rs))
+        check(op(s, d), ref_op(rs, rd))
+        # pure-scalar op yields a scalar JV (reference PURE_SCALAR_OP);
+        # materialize it by adding a zero dense vector.
+        ss = op(s, _core.jv_scalar(0.5, N))
+        z, rz = dense(gpu)
+        zero = _core.jv_sub(z, z)
+        check(_core.jv_add(ss, zero),
+              ref_op(rs, Ref(np.full(NI, 0.5))) + Ref(np.zeros(NI),
+                                                      np.zeros((N, NI))))
 
 
 def run_unary_suite(gpu):
