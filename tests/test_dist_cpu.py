@@ -118,6 +118,13 @@ def test_bench_contract_torchrun_cpu(tmp_path):
     d = json.loads(line)
     assert d["n_gpus"] == 2 and d["steps"] == 2
     assert d["metric"] == "lm_iterations_per_s" and d["value"] > 0
+    # the driver's contract fields must all be present
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"):
+        assert key in d, key
+    assert d["higher_is_better"] is True and d["data"] == "synthetic"
+    assert d["config"]["parallelism"] == "edge-dp2"
 
 def _worker_loss(rank, world_size, port, out_path):
     import torch.distributed as dist
