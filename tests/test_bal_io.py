@@ -50,3 +50,18 @@ def test_cpp_cli_runs_cpu(tmp_path):
                        capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, r.stderr
     assert "final error" in r.stdout
+
+
+def test_cli_out_roundtrip(tmp_path):
+    """bal_solve.py --out writes a loadable BAL file of the solved state."""
+    import subprocess
+    import sys
+    out = str(tmp_path / "solved.txt")
+    r = subprocess.run([sys.executable, "examples/bal_solve.py",
+                        "--synthetic", "ladybug", "--device", "cpu",
+                        "--max_iter", "2", "--out", out],
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-500:]
+    cams, pts, ci, pi, meas = mb.load_bal(out)
+    assert cams.shape == (49, 9) and pts.shape == (7776, 3)
+    assert len(ci) == len(pi) == len(meas)
