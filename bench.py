@@ -48,6 +48,10 @@ def main():
                     choices=["explicit", "implicit"])
     ap.add_argument("--verbose", action="store_true")
     args = ap.parse_args()
+    if args.steps < 1:
+        ap.error("--steps must be >= 1")
+    if args.warmup < 0:
+        ap.error("--warmup must be >= 0")
 
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))

@@ -118,21 +118,31 @@ struct PyProblem {
     CustomForward<double> cf = wrapCustomForward(customForward);
     ix = buildIndex(prob, worldSize);
     if (opt.device == Device::CPU) {
+      // A CPU world>1 run without an allreduce hook would silently skip
+      // every reduction (each collective is guarded by `if (ar_)`) and
+      // produce wrong Hpp/g/chi2 per rank; refuse it like the GPU path
+      // refuses a missing rccl_id.
+      MEGBA_CHECK(worldSize == 1 || !allreduce.is_none(),
+                  "CPU world_size>1 requires an allreduce callback");
       if (isDouble)
         engD = makeCpuEngine<double>(prob, ix, opt,
-                                     wrapAllreduce<double>(allreduce), cf);
+                                     wrapAllreduce<double>(allreduce), cf,
+                                     wrapAllreduce<double>(allreduce));
       else
-        engF = makeCpuEngine<float>(prob, ix, opt, wrapAllreduce<float>(allreduce));
+        engF = makeCpuEngine<float>(prob, ix, opt, wrapAllreduce<float>(allreduce),
+                                    nullptr, wrapAllreduce<double>(allreduce));
     } else {
 #ifdef MEGBA_WITH_GPU
       std::string id;
       if (!rcclId.is_none()) id = py::cast<std::string>(rcclId);
       if (isDouble)
         engD = makeGpuEngine<double>(prob, ix, opt, id, cf,
+                                     wrapAllreduce<double>(allreduce),
                                      wrapAllreduce<double>(allreduce));
       else
         engF = makeGpuEngine<float>(prob, ix, opt, id, nullptr,
-                                    wrapAllreduce<float>(allreduce));
+                                    wrapAllreduce<float>(allreduce),
+                                    wrapAllreduce<double>(allreduce));
 #else
       MEGBA_CHECK(false, "built without GPU support");
 #endif

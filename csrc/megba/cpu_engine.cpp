@@ -20,8 +20,10 @@ class CpuEngine final : public Engine<T> {
  public:
   CpuEngine(const BAProblemHost& prob, const ProblemIndex& ix,
             const ProblemOption& opt, HostAllreduce<T> allreduce,
-            CustomForward<T> customForward)
+            CustomForward<T> customForward,
+            HostAllreduce<double> allreduceScalar)
       : ar_(std::move(allreduce)),
+        arD_(std::move(allreduceScalar)),
         customFwd_(std::move(customForward)),
         rank_(opt.rank),
         world_(opt.worldSize),
@@ -132,9 +134,7 @@ class CpuEngine final : public Engine<T> {
       }
       zeroFixed(e);
     }
-    T buf = chi2;
-    if (ar_) ar_(&buf, 1, 's');
-    return (double)buf;
+    return scalarAr(chi2, 's');
   }
 
   void buildLinearSystem() override {
@@ -336,31 +336,28 @@ class CpuEngine final : public Engine<T> {
     double s = 0;
     const int64_t nc = (int64_t)ncam_ * 9;
     for (int64_t i = 0; i < nc; ++i) s += (double)deltaX_[i] * deltaX_[i];
-    T sp = T(0);
+    double sp = 0;
     for (int64_t i = nc + (int64_t)ptLo_ * 3; i < nc + (int64_t)ptHi_ * 3; ++i)
-      sp += deltaX_[i] * deltaX_[i];
-    if (ar_) ar_(&sp, 1, 's');
-    return std::sqrt(s + (double)sp);
+      sp += (double)deltaX_[i] * deltaX_[i];
+    return std::sqrt(s + scalarAr(sp, 's'));
   }
   double xL2() override {
     double s = 0;
     for (const T v : cams_) s += (double)v * v;
-    T sp = T(0);
+    double sp = 0;
     for (int64_t i = (int64_t)ptLo_ * 3; i < (int64_t)ptHi_ * 3; ++i)
-      sp += pts_[i] * pts_[i];
-    if (ar_) ar_(&sp, 1, 's');
-    return std::sqrt(s + (double)sp);
+      sp += (double)pts_[i] * pts_[i];
+    return std::sqrt(s + scalarAr(sp, 's'));
   }
   double gInf() override {
     double m = 0;
     const int64_t nc = (int64_t)ncam_ * 9;
     for (int64_t i = 0; i < nc; ++i)
       m = std::max(m, std::abs((double)g_[i]));
-    T mp = T(0);
+    double mp = 0;
     for (int64_t i = nc + (int64_t)ptLo_ * 3; i < nc + (int64_t)ptHi_ * 3; ++i)
-      mp = std::max(mp, (T)std::abs((double)g_[i]));
-    if (ar_) ar_(&mp, 1, 'm');
-    return std::max(m, (double)mp);
+      mp = std::max(mp, std::abs((double)g_[i]));
+    return std::max(m, scalarAr(mp, 'm'));
   }
 
   void updateParams() override {
@@ -392,9 +389,7 @@ class CpuEngine final : public Engine<T> {
       s += lossRho(lossKind_, lossD2_,
                    acc2[0] * acc2[0] + acc2[1] * acc2[1]);
     }
-    T buf = s;
-    if (ar_) ar_(&buf, 1, 's');
-    return (double)buf - chi2Backup;
+    return scalarAr(s, 's') - chi2Backup;
   }
 
   // ---- debug access -------------------------------------------------------
@@ -475,9 +470,7 @@ class CpuEngine final : public Engine<T> {
       }
       zeroFixed(e);
     }
-    T buf = chi2;
-    if (ar_) ar_(&buf, 1, 's');
-    return (double)buf;
+    return scalarAr(chi2, 's');
   }
 
   // Fixed vertices (g2o parity, reference base_vertex.h `fixed`): their J
@@ -697,7 +690,23 @@ class CpuEngine final : public Engine<T> {
     return s;
   }
 
+  // Reduce a control-flow scalar across ranks in full double precision
+  // (falls back to the T-typed callback only if no double variant exists).
+  double scalarAr(double v, char op) {
+    if (arD_) {
+      arD_(&v, 1, op);
+      return v;
+    }
+    if (ar_) {
+      T t = (T)v;
+      ar_(&t, 1, op);
+      return (double)t;
+    }
+    return v;
+  }
+
   HostAllreduce<T> ar_;
+  HostAllreduce<double> arD_;
   CustomForward<T> customFwd_;
   int rank_, world_, ncam_, npt_;
   bool analytical_ = false;
@@ -725,17 +734,19 @@ std::unique_ptr<Engine<T>> makeCpuEngine(const BAProblemHost& prob,
                                          const ProblemIndex& ix,
                                          const ProblemOption& opt,
                                          HostAllreduce<T> allreduce,
-                                         CustomForward<T> customForward) {
+                                         CustomForward<T> customForward,
+                                         HostAllreduce<double> allreduceScalar) {
   return std::make_unique<CpuEngine<T>>(prob, ix, opt, std::move(allreduce),
-                                        std::move(customForward));
+                                        std::move(customForward),
+                                        std::move(allreduceScalar));
 }
 
 template std::unique_ptr<Engine<double>> makeCpuEngine<double>(
     const BAProblemHost&, const ProblemIndex&, const ProblemOption&,
-    HostAllreduce<double>, CustomForward<double>);
+    HostAllreduce<double>, CustomForward<double>, HostAllreduce<double>);
 template std::unique_ptr<Engine<float>> makeCpuEngine<float>(
     const BAProblemHost&, const ProblemIndex&, const ProblemOption&,
-    HostAllreduce<float>, CustomForward<float>);
+    HostAllreduce<float>, CustomForward<float>, HostAllreduce<double>);
 
 // Instantiate the LM driver here as well.
 template LMReport runLM<double>(Engine<double>&, const AlgoOptionLM&,

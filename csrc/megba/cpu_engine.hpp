@@ -21,11 +21,13 @@ namespace megba {
 template <typename T>
 using HostAllreduce = std::function<void(T*, std::size_t, char)>;
 
+// `allreduceScalar`: double-typed variant of the same callback used for the
+// control-flow scalars (chi2, norms, rho denominator) so fp32 multi-rank
+// runs keep full double precision in the reductions that steer LM.
 template <typename T>
-std::unique_ptr<Engine<T>> makeCpuEngine(const BAProblemHost& prob,
-                                         const ProblemIndex& ix,
-                                         const ProblemOption& opt,
-                                         HostAllreduce<T> allreduce,
-                                         CustomForward<T> customForward = nullptr);
+std::unique_ptr<Engine<T>> makeCpuEngine(
+    const BAProblemHost& prob, const ProblemIndex& ix, const ProblemOption& opt,
+    HostAllreduce<T> allreduce, CustomForward<T> customForward = nullptr,
+    HostAllreduce<double> allreduceScalar = nullptr);
 
 }  // namespace megba

@@ -815,7 +815,9 @@ __global__ void kRedFinalRhoBeta(const double* part, int nb, double* rho,
   }
   if (threadIdx.x == 0) {
     *rho = sm[0];
-    *beta = sm[0] / *rhoPrev;
+    // rhoPrev==0 only after an exactly-converged iteration under fixed-work
+    // mode; beta=0 (restart direction) instead of 0/0 = NaN.
+    *beta = *rhoPrev != 0.0 ? sm[0] / *rhoPrev : 0.0;
   }
 }
 __global__ void kRedFinalAlpha(const double* part, int nb, const double* rho,
@@ -830,7 +832,10 @@ __global__ void kRedFinalAlpha(const double* part, int nb, const double* rho,
     __syncthreads();
   }
   if (threadIdx.x == 0) {
-    *alpha = *rho / sm[0];
+    // dot(p,q) can reach exactly 0 when PCG has converged but fixed-work
+    // mode (tol=0) forces further iterations; alpha=0 then leaves x/r
+    // unchanged instead of propagating NaN into deltaX.
+    *alpha = sm[0] != 0.0 ? *rho / sm[0] : 0.0;
     *rhoPrev = *rho;
   }
 }
@@ -932,9 +937,11 @@ class GpuEngine final : public Engine<T> {
  public:
   GpuEngine(const BAProblemHost& prob, const ProblemIndex& ix,
             const ProblemOption& opt, const std::string& rcclId,
-            CustomForward<T> customForward, HostAllreduce<T> hostAllreduce)
+            CustomForward<T> customForward, HostAllreduce<T> hostAllreduce,
+            HostAllreduce<double> hostAllreduceScalar)
       : customFwd_(std::move(customForward)),
         hostAr_(std::move(hostAllreduce)),
+        hostArD_(std::move(hostAllreduceScalar)),
         rank_(opt.rank),
         world_(opt.worldSize),
         ncam_(ix.ncam),
@@ -1487,8 +1494,15 @@ class GpuEngine final : public Engine<T> {
                                comm_, stream_));
       return readScalar(scalarPtr());
     }
-    if (hostAr_ && world_ > 1) {
+    if ((hostArD_ || hostAr_) && world_ > 1) {
+      // Host-callback fallback: reduce in full double (the device
+      // accumulator already is double) so fp32 multi-rank runs don't lose
+      // precision in chi2/deltaXL2/gInf.
       double h = readScalar(scalarPtr());
+      if (hostArD_) {
+        hostArD_(&h, 1, op == ncclMax ? 'm' : 's');
+        return h;
+      }
       T v = (T)h;
       hostAr_(&v, 1, op == ncclMax ? 'm' : 's');
       return (double)v;
@@ -1669,6 +1683,7 @@ class GpuEngine final : public Engine<T> {
   bool hasComm_ = false;
   CustomForward<T> customFwd_;
   HostAllreduce<T> hostAr_;
+  HostAllreduce<double> hostArD_;
   int rank_, world_, ncam_, npt_;
   int ptLo_ = 0, ptHi_ = 0, npL_ = 0;
   int64_t e0_ = 0, e1_ = 0, nL_ = 0, nc_ = 0, np_ = 0, dim_ = 0;
@@ -1705,18 +1720,22 @@ std::unique_ptr<Engine<T>> makeGpuEngine(const BAProblemHost& prob,
                                          const ProblemOption& opt,
                                          const std::string& rcclId,
                                          CustomForward<T> customForward,
-                                         HostAllreduce<T> hostAllreduce) {
+                                         HostAllreduce<T> hostAllreduce,
+                                         HostAllreduce<double> hostAllreduceScalar) {
   return std::make_unique<GpuEngine<T>>(prob, ix, opt, rcclId,
                                         std::move(customForward),
-                                        std::move(hostAllreduce));
+                                        std::move(hostAllreduce),
+                                        std::move(hostAllreduceScalar));
 }
 
 template std::unique_ptr<Engine<double>> makeGpuEngine<double>(
     const BAProblemHost&, const ProblemIndex&, const ProblemOption&,
-    const std::string&, CustomForward<double>, HostAllreduce<double>);
+    const std::string&, CustomForward<double>, HostAllreduce<double>,
+    HostAllreduce<double>);
 template std::unique_ptr<Engine<float>> makeGpuEngine<float>(
     const BAProblemHost&, const ProblemIndex&, const ProblemOption&,
-    const std::string&, CustomForward<float>, HostAllreduce<float>);
+    const std::string&, CustomForward<float>, HostAllreduce<float>,
+    HostAllreduce<double>);
 
 std::string rcclUniqueIdString() {
   ncclUniqueId id;

@@ -22,13 +22,14 @@ namespace megba {
 // collectives bounce through the host callback (e.g. torch.distributed gloo),
 // so the sharded GPU code paths can be exercised by several ranks sharing
 // one GPU.  Production multi-GPU uses RCCL (rcclId non-empty).
+// `hostAllreduceScalar`: double-typed variant of the host callback used for
+// control-flow scalars so fp32 gloo-fallback runs reduce in full double.
 template <typename T>
-std::unique_ptr<Engine<T>> makeGpuEngine(const BAProblemHost& prob,
-                                         const ProblemIndex& ix,
-                                         const ProblemOption& opt,
-                                         const std::string& rcclId,
-                                         CustomForward<T> customForward = nullptr,
-                                         HostAllreduce<T> hostAllreduce = nullptr);
+std::unique_ptr<Engine<T>> makeGpuEngine(
+    const BAProblemHost& prob, const ProblemIndex& ix, const ProblemOption& opt,
+    const std::string& rcclId, CustomForward<T> customForward = nullptr,
+    HostAllreduce<T> hostAllreduce = nullptr,
+    HostAllreduce<double> hostAllreduceScalar = nullptr);
 
 std::string rcclUniqueIdString();
 int hipDeviceCountSafe();
