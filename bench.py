@@ -80,7 +80,25 @@ def main():
         # gloo for bootstrap only; the GPU engine runs RCCL natively.
         dist.init_process_group("gloo", rank=rank, world_size=world)
         if args.device == "gpu":
+            from megba_amd import _core
             from megba_amd.dist import broadcast_rccl_id
+            ngpu = max(1, _core.hip_device_count())
+            dev = local_rank % ngpu
+            if world > 1 and not os.environ.get("MEGBA_NO_PREFLIGHT"):
+                # Pre-flight: throwaway comm + 1-element allreduce under a
+                # watchdog, so a wedged RCCL bootstrap aborts the whole job
+                # with a clear message instead of hanging the scale run.
+                pf_id = broadcast_rccl_id(rank)
+                try:
+                    t_pf = _core.rccl_preflight(pf_id, rank, world, dev,
+                                                120.0)
+                except Exception as e:
+                    print(f"RCCL BOOTSTRAP FAILED (rank {rank}/{world}, "
+                          f"device {dev}): {e}", file=sys.stderr, flush=True)
+                    sys.exit(1)
+                if rank == 0:
+                    print(f"# rccl preflight ok ({t_pf:.2f}s, world={world})",
+                          file=sys.stderr)
             rccl_id = broadcast_rccl_id(rank)
         else:
             from megba_amd.dist import gloo_allreduce_callback

@@ -449,6 +449,15 @@ PYBIND11_MODULE(_core, m) {
 #ifdef MEGBA_WITH_GPU
   m.attr("has_gpu_support") = true;
   m.def("rccl_unique_id", []() { return py::bytes(rcclUniqueIdString()); });
+  m.def("rccl_preflight",
+        [](py::bytes id, int rank, int world, int deviceIndex,
+           double timeoutSec) {
+          std::string s = id;
+          py::gil_scoped_release rel;
+          return rcclPreflight(s, rank, world, deviceIndex, timeoutSec);
+        },
+        py::arg("id"), py::arg("rank"), py::arg("world"),
+        py::arg("device_index") = 0, py::arg("timeout_sec") = 120.0);
   m.def("hip_device_count", []() { return hipDeviceCountSafe(); });
   m.def("hip_mem_info", []() {
     auto p = hipMemInfoSafe();

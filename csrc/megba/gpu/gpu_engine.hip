@@ -29,9 +29,14 @@
 #include <hip/hip_runtime.h>
 #include <rccl/rccl.h>
 
+#include <chrono>
 #include <cmath>
+#include <condition_variable>
 #include <cstdlib>
 #include <cstring>
+#include <memory>
+#include <mutex>
+#include <thread>
 #include <type_traits>
 #include <utility>
 #include <vector>
@@ -759,20 +764,54 @@ __global__ void kPrecondRho(int nBlk, const T* __restrict__ Binv,
   if (threadIdx.x == 0) part[blockIdx.x] = sm[0];
 }
 
-// Fused PCG vector update: xBak = x; x += alpha p; r -= alpha q.
+// Fused PCG vector update incl. the two-deep backup rotation:
+// xBakPrev = xBak; xBak = x; x += alpha p; r -= alpha q.
+// (The rotation is a data move, not a pointer swap, so the kernel's
+// pointers stay stable across hipGraph replays.)
 template <typename T>
 __global__ void kUpdateXR(int64_t n, const double* __restrict__ alpha,
                           const T* __restrict__ p, const T* __restrict__ q,
                           T* __restrict__ x, T* __restrict__ xBak,
-                          T* __restrict__ r) {
+                          T* __restrict__ xBakPrev, T* __restrict__ r) {
   const T a = (T)(*alpha);
   for (int64_t i = blockIdx.x * (int64_t)kBlk + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * kBlk) {
+    xBakPrev[i] = xBak[i];
     const T xv = x[i];
     xBak[i] = xv;
     x[i] = xv + a * p[i];
     r[i] -= a * q[i];
   }
+}
+
+// B-apply (y = A x - y, the Schur S-apply tail) fused with the per-block
+// p^T q dot partials: saves the separate full pass over p and q per PCG
+// iteration (the reference used a standalone cublasDot, its :368-385).
+template <typename T>
+__global__ void kBApplyDot(int nBlk, const T* __restrict__ A,
+                           const T* __restrict__ x, T* __restrict__ y,
+                           double* part) {
+  __shared__ double sm[kBlk];
+  double local = 0.0;
+  for (int64_t idx = blockIdx.x * (int64_t)kBlk + threadIdx.x;
+       idx < (int64_t)nBlk * 9; idx += (int64_t)gridDim.x * kBlk) {
+    const int64_t b = idx / 9;
+    const int rrow = (int)(idx % 9);
+    const T* row = A + b * 81 + (int64_t)rrow * 9;
+    const T* xb = x + b * 9;
+    T s = T(0);
+    for (int j = 0; j < 9; ++j) s += row[j] * xb[j];
+    const T q = s - y[idx];
+    y[idx] = q;
+    local += (double)x[idx] * (double)q;
+  }
+  sm[threadIdx.x] = local;
+  __syncthreads();
+  for (int st = kBlk / 2; st > 0; st >>= 1) {
+    if (threadIdx.x < st) sm[threadIdx.x] += sm[threadIdx.x + st];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) part[blockIdx.x] = sm[0];
 }
 
 // Block-diagonal matvec, one thread per output row.
@@ -975,6 +1014,18 @@ class GpuEngine final : public Engine<T> {
         RCCL_CHECK(ncclCommInitRank(&comm_, world_, id, rank_));
         hasComm_ = true;
       }
+    } else if (getenv("MEGBA_FORCE_RCCL")) {
+      // 1-GPU hardening mode: run a real world-1 RCCL communicator so
+      // ncclCommInitRank, every ncclAllReduce call site and RCCL-under-
+      // hipGraph-capture are exercised on a single GPU (the multi-rank
+      // collective SEQUENCE is identical; only the ring is trivial).
+      ncclUniqueId id;
+      if (rcclId.size() == sizeof(id))
+        std::memcpy(&id, rcclId.data(), sizeof(id));
+      else
+        RCCL_CHECK(ncclGetUniqueId(&id));
+      RCCL_CHECK(ncclCommInitRank(&comm_, 1, id, 0));
+      hasComm_ = true;
     }
 
     // Static per-edge data (primary = (pt,cam)-sorted order).
@@ -1055,7 +1106,10 @@ class GpuEngine final : public Engine<T> {
     dTemp_ = dalloc<T>(np_);
     HIP_CHECK(hipMemsetAsync(dW_, 0, np_ * sizeof(T), stream_));
 
-    dPart_ = dalloc<double>(kRedBlocks + 8);
+    // Partial-sum scratch: big enough for both the fixed-shape two-pass
+    // reductions (kRedBlocks) and the fused B-apply+dot grid over nc_.
+    partCap_ = kRedBlocks > gridFor(nc_) ? kRedBlocks : gridFor(nc_);
+    dPart_ = dalloc<double>(partCap_ + 8);
     dFail_ = dalloc<int>(2);
     HIP_CHECK(hipHostMalloc((void**)&hScalar_, sizeof(double)));
 
@@ -1287,21 +1341,35 @@ class GpuEngine final : public Engine<T> {
     double rho = 0.0, rhoMin = INFINITY;
     bool done = false;
     ensurePcgGraph();
-    while (!done && n < opt.maxIter) {
-      if (pcgGraphExec_) {
-        HIP_CHECK(hipGraphLaunch(pcgGraphExec_, stream_));
-      } else {
-        pcgBody();
+    // Fixed-work mode (tol<=0 with refuse disabled, the bench contract):
+    // rho steers nothing, so skip the per-iteration host readback and
+    // enqueue all maxIter bodies back-to-back — ONE host sync per solve
+    // instead of one per iteration (the N=8 constant-term killer).
+    const bool fixedWork = opt.tol <= 0.0 && opt.refuseRatio >= 1e29;
+    if (fixedWork) {
+      for (; n < opt.maxIter; ++n) {
+        if (pcgGraphExec_)
+          HIP_CHECK(hipGraphLaunch(pcgGraphExec_, stream_));
+        else
+          pcgBody();
       }
-      rho = readScalar(slotRho());
-      if (rho > opt.refuseRatio * rhoMin) {
-        HIP_CHECK(hipMemcpyAsync(x, dXBakPrev_, nc_ * sizeof(T),
-                                 hipMemcpyDeviceToDevice, stream_));
-        break;
+    } else {
+      while (!done && n < opt.maxIter) {
+        if (pcgGraphExec_) {
+          HIP_CHECK(hipGraphLaunch(pcgGraphExec_, stream_));
+        } else {
+          pcgBody();
+        }
+        rho = readScalar(slotRho());
+        if (rho > opt.refuseRatio * rhoMin) {
+          HIP_CHECK(hipMemcpyAsync(x, dXBakPrev_, nc_ * sizeof(T),
+                                   hipMemcpyDeviceToDevice, stream_));
+          break;
+        }
+        rhoMin = rhoMin < rho ? rhoMin : rho;
+        ++n;
+        done = std::abs(rho) < opt.tol;
       }
-      rhoMin = rhoMin < rho ? rhoMin : rho;
-      ++n;
-      done = std::abs(rho) < opt.tol;
     }
     // Back-substitution: deltaX_p = Cinv (g_p - E^T x), fully local.
     spmvEtx(x, dTemp_);
@@ -1478,12 +1546,12 @@ class GpuEngine final : public Engine<T> {
     return o;
   }
   void sync() { HIP_CHECK(hipStreamSynchronize(stream_)); }
-  double* scalarPtr() { return dPart_ + kRedBlocks; }
-  double* slotRho() { return dPart_ + kRedBlocks + 1; }
-  double* slotPq() { return dPart_ + kRedBlocks + 2; }
-  double* slotAlpha() { return dPart_ + kRedBlocks + 3; }
-  double* slotRhoPrev() { return dPart_ + kRedBlocks + 4; }
-  double* slotBeta() { return dPart_ + kRedBlocks + 5; }
+  double* scalarPtr() { return dPart_ + partCap_; }
+  double* slotRho() { return dPart_ + partCap_ + 1; }
+  double* slotPq() { return dPart_ + partCap_ + 2; }
+  double* slotAlpha() { return dPart_ + partCap_ + 3; }
+  double* slotRhoPrev() { return dPart_ + partCap_ + 4; }
+  double* slotBeta() { return dPart_ + partCap_ + 5; }
   void zeroScalar() {
     HIP_CHECK(hipMemsetAsync(scalarPtr(), 0, sizeof(double), stream_));
   }
@@ -1517,7 +1585,7 @@ class GpuEngine final : public Engine<T> {
     return *hScalar_;
   }
   void allreduce(T* buf, int64_t n, ncclRedOp_t op) {
-    if (n == 0 || world_ == 1) return;
+    if (n == 0 || (world_ == 1 && !hasComm_)) return;
     if (hasComm_) {
       RCCL_CHECK(ncclAllReduce(buf, buf, n,
                                sizeof(T) == 8 ? ncclDouble : ncclFloat, op,
@@ -1609,44 +1677,50 @@ class GpuEngine final : public Engine<T> {
                          dHplCam_, nL_, wv, out);
   }
   // One full PCG iteration (captured as a hipGraph when possible).
+  // vs round 1: the p^T q dot partial is fused into the B-apply, the
+  // two-deep x backup rotation is fused into kUpdateXR, and the standalone
+  // kRedPartial pass + dXBakPrev memcpy are gone (~2 launches + one full
+  // read pass over p,q per iteration — part of the N=8 constant term,
+  // PLAN_r02 item 3).
   void pcgBody() {
     const int rhoGrid = gridFor(nc_) < kRedBlocks ? gridFor(nc_) : kRedBlocks;
-    HIP_CHECK(hipMemcpyAsync(dXBakPrev_, dXBak_, nc_ * sizeof(T),
-                             hipMemcpyDeviceToDevice, stream_));
     hipLaunchKernelGGL(kPrecondRho<T>, dim3(rhoGrid), dim3(kBlk), 0, stream_,
                        ncam_, dHppInv_, dRr_, dZ_, dPart_);
     hipLaunchKernelGGL(kRedFinalRhoBeta, dim3(1), dim3(kBlk), 0, stream_,
                        dPart_, rhoGrid, slotRho(), slotRhoPrev(), slotBeta());
     hipLaunchKernelGGL(kXpbyS<T>, dim3(gridFor(nc_)), dim3(kBlk), 0, stream_,
                        nc_, dZ_, slotBeta(), dP_);
-    schurApply(dP_, dQ_);
-    hipLaunchKernelGGL((kRedPartial<T, ROp::Dot>), dim3(kRedBlocks),
-                       dim3(kBlk), 0, stream_, dP_, dQ_, nc_, dPart_);
+    schurApply(dP_, dQ_, /*withDot=*/true);
     hipLaunchKernelGGL(kRedFinalAlpha, dim3(1), dim3(kBlk), 0, stream_,
-                       dPart_, kRedBlocks, slotRho(), slotAlpha(),
+                       dPart_, gridFor(nc_), slotRho(), slotAlpha(),
                        slotRhoPrev());
     hipLaunchKernelGGL(kUpdateXR<T>, dim3(gridFor(nc_)), dim3(kBlk), 0,
                        stream_, nc_, slotAlpha(), dP_, dQ_, dDeltaX_, dXBak_,
-                       dRr_);
+                       dXBakPrev_, dRr_);
   }
 
+  // (Re-)capture the PCG body as a hipGraph.  Implicit mode reads the
+  // accepted J set through double-buffered pointers that swap on accept, so
+  // its graph is re-captured lazily whenever the accepted buffer set
+  // changed since the last capture; the topology is identical, so the
+  // existing executable graph is patched in place with hipGraphExecUpdate
+  // (falling back to re-instantiation, then to eager execution).  Multi-
+  // rank RCCL collectives are captured too (allreduce is stream-ordered on
+  // stream_); any capture failure falls back to the eager path permanently.
   void ensurePcgGraph() {
-    if (pcgGraphExec_ || pcgGraphTried_) return;
-    pcgGraphTried_ = true;
-    if (getenv("MEGBA_NO_GRAPH")) return;
-    // Implicit mode reads the accepted J set through double-buffered
-    // pointers that swap on every accept; a captured graph would freeze the
-    // capture-time pointers and read stale Jacobians.  Stay eager.
-    if (implicit_) return;
-    // Multi-rank: RCCL-in-graph-capture is not exercisable in this round's
-    // single-GPU test environment, so stay eager unless explicitly enabled
-    // (the collective SEQUENCE is identical either way).
-    if (world_ > 1 && !getenv("MEGBA_GRAPH")) return;
+    if (pcgGraphFailed_ || getenv("MEGBA_NO_GRAPH")) return;
+    // host-callback collectives (gloo testing fallback) cannot be captured
+    if (world_ > 1 && !hasComm_) return;
+    const int bak = cur_ ^ 1;
+    if (pcgGraphExec_ && (!implicit_ || graphBak_ == bak)) return;
     sync();
     hipGraph_t graph = nullptr;
     if (hipStreamBeginCapture(stream_, hipStreamCaptureModeThreadLocal) !=
-        hipSuccess)
+        hipSuccess) {
+      pcgGraphFailed_ = true;
+      (void)hipGetLastError();
       return;
+    }
     bool ok = true;
     try {
       pcgBody();
@@ -1656,26 +1730,47 @@ class GpuEngine final : public Engine<T> {
     if (hipStreamEndCapture(stream_, &graph) != hipSuccess || !ok || !graph) {
       if (graph) (void)hipGraphDestroy(graph);
       (void)hipGetLastError();
+      pcgGraphFailed_ = true;
       return;
+    }
+    if (pcgGraphExec_) {
+      hipGraphNode_t errNode = nullptr;
+      hipGraphExecUpdateResult res{};
+      if (hipGraphExecUpdate(pcgGraphExec_, graph, &errNode, &res) ==
+          hipSuccess) {
+        (void)hipGraphDestroy(graph);
+        graphBak_ = bak;
+        return;
+      }
+      (void)hipGetLastError();
+      (void)hipGraphExecDestroy(pcgGraphExec_);
+      pcgGraphExec_ = nullptr;
     }
     hipGraphExec_t exec = nullptr;
     if (hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0) != hipSuccess) {
       (void)hipGraphDestroy(graph);
       (void)hipGetLastError();
+      pcgGraphFailed_ = true;
       return;
     }
     (void)hipGraphDestroy(graph);
     pcgGraphExec_ = exec;
+    graphBak_ = bak;
   }
 
   // q = S x = HppD x - E Cinv E^T x  (ONE 9*ncam allreduce; the reference's
-  // site A4 needed an additional 3*npt allreduce here).
-  void schurApply(const T* xv, T* q) {
+  // site A4 needed an additional 3*npt allreduce here).  withDot fuses the
+  // x^T q dot partials into the B-apply pass (used by the PCG body).
+  void schurApply(const T* xv, T* q, bool withDot = false) {
     spmvEtx(xv, dTemp_);
     applyCinv(dTemp_, dW_);
     spmvEx(dW_, q);
     allreduce(q, nc_, ncclSum);
-    blockMatVec<9, 1>(ncam_, dHppD_, xv, q);
+    if (withDot)
+      hipLaunchKernelGGL(kBApplyDot<T>, dim3(gridFor(nc_)), dim3(kBlk), 0,
+                         stream_, ncam_, dHppD_, xv, q, dPart_);
+    else
+      blockMatVec<9, 1>(ncam_, dHppD_, xv, q);
   }
 
   hipStream_t stream_{};
@@ -1708,7 +1803,9 @@ class GpuEngine final : public Engine<T> {
   T *dP_{}, *dRr_{}, *dZ_{}, *dQ_{}, *dV_{}, *dW_{}, *dTemp_{}, *dXBak_{},
       *dXBakPrev_{}, *dPtMerge_{};
   hipGraphExec_t pcgGraphExec_{};
-  bool pcgGraphTried_ = false;
+  bool pcgGraphFailed_ = false;
+  int graphBak_ = -1;  // the cur_^1 value the current graph was captured for
+  int partCap_ = kRedBlocks;
   double* dPart_{};
   double* hScalar_{};
   std::vector<void*> allocs_;
@@ -1736,6 +1833,71 @@ template std::unique_ptr<Engine<float>> makeGpuEngine<float>(
     const BAProblemHost&, const ProblemIndex&, const ProblemOption&,
     const std::string&, CustomForward<float>, HostAllreduce<float>,
     HostAllreduce<double>);
+
+// Pre-flight bootstrap self-test for the driver's multi-GPU scale run:
+// init a throwaway communicator from a dedicated unique id, run a
+// 1-element allreduce, verify the sum, tear down — all under a watchdog so
+// a wedged rendezvous aborts with a clear message instead of hanging the
+// whole job (the unique id is single-use; the engine gets its own).
+// Returns elapsed seconds.  Reference anchor for what is being guarded:
+// /root/reference/src/resource/handle_manager.cpp:17-21 (comm init) + the
+// collective sites of SURVEY.md section 2b.
+double rcclPreflight(const std::string& idBytes, int rank, int world,
+                     int deviceIndex, double timeoutSec) {
+  MEGBA_CHECK(idBytes.size() == sizeof(ncclUniqueId),
+              "rccl_preflight: bad unique id size");
+  struct State {
+    std::mutex m;
+    std::condition_variable cv;
+    bool done = false;
+    std::string err;
+  };
+  auto st = std::make_shared<State>();
+  const auto t0 = std::chrono::steady_clock::now();
+  ncclUniqueId id;
+  std::memcpy(&id, idBytes.data(), sizeof(id));
+  std::thread([st, id, rank, world, deviceIndex]() {
+    std::string err;
+    try {
+      HIP_CHECK(hipSetDevice(deviceIndex));
+      ncclComm_t comm{};
+      RCCL_CHECK(ncclCommInitRank(&comm, world, id, rank));
+      double* buf = nullptr;
+      HIP_CHECK(hipMalloc((void**)&buf, sizeof(double)));
+      const double one = 1.0;
+      HIP_CHECK(hipMemcpy(buf, &one, sizeof(double), hipMemcpyHostToDevice));
+      RCCL_CHECK(ncclAllReduce(buf, buf, 1, ncclDouble, ncclSum, comm,
+                               /*stream=*/0));
+      HIP_CHECK(hipStreamSynchronize(0));
+      double out = 0.0;
+      HIP_CHECK(hipMemcpy(&out, buf, sizeof(double), hipMemcpyDeviceToHost));
+      (void)hipFree(buf);
+      (void)ncclCommDestroy(comm);
+      MEGBA_CHECK(std::abs(out - (double)world) < 1e-9,
+                  "rccl_preflight: allreduce sum mismatch");
+    } catch (const std::exception& e) {
+      err = e.what();
+    }
+    {
+      std::lock_guard<std::mutex> lk(st->m);
+      st->done = true;
+      st->err = err;
+    }
+    st->cv.notify_all();
+  }).detach();
+  std::unique_lock<std::mutex> lk(st->m);
+  if (!st->cv.wait_for(lk, std::chrono::duration<double>(timeoutSec),
+                       [&] { return st->done; }))
+    MEGBA_CHECK(false,
+                "rccl_preflight: timed out after " +
+                    std::to_string(timeoutSec) +
+                    "s waiting for comm init + allreduce (rank " +
+                    std::to_string(rank) + "/" + std::to_string(world) +
+                    ") — a peer rank is missing or the rendezvous is wedged");
+  MEGBA_CHECK(st->err.empty(), "rccl_preflight: " + st->err);
+  return std::chrono::duration<double>(std::chrono::steady_clock::now() - t0)
+      .count();
+}
 
 std::string rcclUniqueIdString() {
   ncclUniqueId id;

@@ -32,6 +32,10 @@ std::unique_ptr<Engine<T>> makeGpuEngine(
     HostAllreduce<double> hostAllreduceScalar = nullptr);
 
 std::string rcclUniqueIdString();
+// Bootstrap self-test: comm init + 1-element allreduce under a watchdog;
+// throws (instead of hanging) when a peer is missing.  Returns seconds.
+double rcclPreflight(const std::string& idBytes, int rank, int world,
+                     int deviceIndex, double timeoutSec);
 int hipDeviceCountSafe();
 std::pair<long long, long long> hipMemInfoSafe();  // (free, total) bytes
 
