@@ -631,10 +631,15 @@ __global__ void kInvertJitter(int nBlk, const T* __restrict__ Hd,
 template <typename T, bool IMP, bool HASINFO>
 __global__ void kSpmvEtx(int64_t nL, const int* __restrict__ camOf,
                          const int* __restrict__ ptOf,
-                         const T* __restrict__ Hpl, const T* __restrict__ Jc,
-                         const T* __restrict__ Jp, const T* __restrict__ info,
-                         const T* __restrict__ rBak, int lossKind, T lossD2,
+                         const T* __restrict__ Hpl,
+                         const T* const* __restrict__ jSlots,
+                         const T* __restrict__ info, int lossKind, T lossD2,
                          const T* __restrict__ x, T* __restrict__ out) {
+  // IMP reads the accepted J set through the device pointer slots so a
+  // captured graph stays valid across accept-time buffer flips.
+  const T* Jc = IMP ? jSlots[0] : nullptr;
+  const T* Jp = IMP ? jSlots[1] : nullptr;
+  const T* rBak = IMP ? jSlots[2] : nullptr;
   const int lane = threadIdx.x & 63;
   const int64_t nWork = ((nL + kBlk - 1) / kBlk) * (int64_t)kBlk;
   for (int64_t j0 = blockIdx.x * (int64_t)kBlk + threadIdx.x; j0 < nWork;
@@ -834,6 +839,19 @@ __global__ void kBlockDiagMatVec(int nBlk, const T* __restrict__ A,
 // ---------------------------------------------------------------------------
 // Small vector kernels
 // ---------------------------------------------------------------------------
+// Device-side pointer slots for the implicit J buffers: the double-buffered
+// accepted set flips on every LM accept, but the captured PCG graph freezes
+// kernel arguments — so the kernels dereference this slot array instead and
+// acceptForward rewrites it (kernel args carry the values; no host-buffer
+// lifetime to manage).  Slots: [0]=Jc, [1]=Jp, [2]=r (all accepted/bak).
+template <typename T>
+__global__ void kSetPtrSlots(const T** slots, const T* a, const T* b,
+                             const T* c) {
+  slots[0] = a;
+  slots[1] = b;
+  slots[2] = c;
+}
+
 __global__ void kDivScalar(double* out, const double* num, const double* den) {
   *out = *num / *den;
 }
@@ -1146,6 +1164,8 @@ class GpuEngine final : public Engine<T> {
       up(dChHi_, cHi.data(), nChunks_);
     }
     dSlab_ = dalloc<T>(nL_ * slabWidth());
+    dJSlots_ = dalloc<const T*>(3);
+    updateJSlots();
     if (implicit_) {
       dJcCam_ = dalloc<T>(nL_ * 18);
       dJpCam_ = dalloc<T>(nL_ * 6);
@@ -1185,6 +1205,14 @@ class GpuEngine final : public Engine<T> {
   void acceptForward() override {
     cur_ ^= 1;          // accepted set = dX_[cur_^1]
     freshCur_ = false;  // the (new) current buffers are not yet written
+    updateJSlots();
+  }
+
+  // Point the device slots at the accepted (bak) buffer set.
+  void updateJSlots() {
+    const int bak = cur_ ^ 1;
+    hipLaunchKernelGGL(kSetPtrSlots<T>, dim3(1), dim3(1), 0, stream_,
+                       dJSlots_, dJc_[bak], dJp_[bak], dR_[bak]);
   }
 
   double forwardCustom() {
@@ -1641,25 +1669,22 @@ class GpuEngine final : public Engine<T> {
     hipLaunchKernelGGL(kZeroRange<T>, dim3(gridFor((int64_t)npL_ * 3)),
                        dim3(kBlk), 0, stream_, out + (int64_t)ptLo_ * 3,
                        (int64_t)npL_ * 3);
-    const int bak = cur_ ^ 1;
     if (implicit_) {
       if (hasInfo_)
         hipLaunchKernelGGL((kSpmvEtx<T, true, true>), dim3(gridFor(nL_)),
                            dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
-                           (const T*)nullptr, dJc_[bak], dJp_[bak], dInfo_,
-                           dR_[bak], lossKind_, lossD2_, xv, out);
+                           (const T*)nullptr, (const T* const*)dJSlots_,
+                           dInfo_, lossKind_, lossD2_, xv, out);
       else
         hipLaunchKernelGGL((kSpmvEtx<T, true, false>), dim3(gridFor(nL_)),
                            dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
-                           (const T*)nullptr, dJc_[bak], dJp_[bak],
-                           (const T*)nullptr, dR_[bak], lossKind_, lossD2_,
-                           xv, out);
+                           (const T*)nullptr, (const T* const*)dJSlots_,
+                           (const T*)nullptr, lossKind_, lossD2_, xv, out);
     } else {
       hipLaunchKernelGGL((kSpmvEtx<T, false, false>), dim3(gridFor(nL_)),
                          dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_, dHpl_,
-                         (const T*)nullptr, (const T*)nullptr,
-                         (const T*)nullptr, (const T*)nullptr, 0, T(0), xv,
-                         out);
+                         (const T* const*)nullptr, (const T*)nullptr, 0,
+                         T(0), xv, out);
     }
   }
   void spmvEx(const T* wv, T* out) {
@@ -1699,20 +1724,15 @@ class GpuEngine final : public Engine<T> {
                        dXBakPrev_, dRr_);
   }
 
-  // (Re-)capture the PCG body as a hipGraph.  Implicit mode reads the
-  // accepted J set through double-buffered pointers that swap on accept, so
-  // its graph is re-captured lazily whenever the accepted buffer set
-  // changed since the last capture; the topology is identical, so the
-  // existing executable graph is patched in place with hipGraphExecUpdate
-  // (falling back to re-instantiation, then to eager execution).  Multi-
-  // rank RCCL collectives are captured too (allreduce is stream-ordered on
+  // Capture the PCG body as a hipGraph, once.  The implicit path reads the
+  // double-buffered accepted J set through device pointer slots (dJSlots_),
+  // so accept-time buffer flips never invalidate the graph.  Multi-rank
+  // RCCL collectives are captured too (ncclAllReduce is stream-ordered on
   // stream_); any capture failure falls back to the eager path permanently.
   void ensurePcgGraph() {
-    if (pcgGraphFailed_ || getenv("MEGBA_NO_GRAPH")) return;
+    if (pcgGraphExec_ || pcgGraphFailed_ || getenv("MEGBA_NO_GRAPH")) return;
     // host-callback collectives (gloo testing fallback) cannot be captured
     if (world_ > 1 && !hasComm_) return;
-    const int bak = cur_ ^ 1;
-    if (pcgGraphExec_ && (!implicit_ || graphBak_ == bak)) return;
     sync();
     hipGraph_t graph = nullptr;
     if (hipStreamBeginCapture(stream_, hipStreamCaptureModeThreadLocal) !=
@@ -1733,19 +1753,6 @@ class GpuEngine final : public Engine<T> {
       pcgGraphFailed_ = true;
       return;
     }
-    if (pcgGraphExec_) {
-      hipGraphNode_t errNode = nullptr;
-      hipGraphExecUpdateResult res{};
-      if (hipGraphExecUpdate(pcgGraphExec_, graph, &errNode, &res) ==
-          hipSuccess) {
-        (void)hipGraphDestroy(graph);
-        graphBak_ = bak;
-        return;
-      }
-      (void)hipGetLastError();
-      (void)hipGraphExecDestroy(pcgGraphExec_);
-      pcgGraphExec_ = nullptr;
-    }
     hipGraphExec_t exec = nullptr;
     if (hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0) != hipSuccess) {
       (void)hipGraphDestroy(graph);
@@ -1755,7 +1762,6 @@ class GpuEngine final : public Engine<T> {
     }
     (void)hipGraphDestroy(graph);
     pcgGraphExec_ = exec;
-    graphBak_ = bak;
   }
 
   // q = S x = HppD x - E Cinv E^T x  (ONE 9*ncam allreduce; the reference's
@@ -1804,7 +1810,7 @@ class GpuEngine final : public Engine<T> {
       *dXBakPrev_{}, *dPtMerge_{};
   hipGraphExec_t pcgGraphExec_{};
   bool pcgGraphFailed_ = false;
-  int graphBak_ = -1;  // the cur_^1 value the current graph was captured for
+  const T** dJSlots_{};  // device slots: accepted {Jc, Jp, r} pointers
   int partCap_ = kRedBlocks;
   double* dPart_{};
   double* hScalar_{};
