@@ -9,6 +9,7 @@
 
 #include <memory>
 
+#include "megba/bal_functor.hpp"
 #include "megba/common.hpp"
 #include "megba/custom.hpp"
 #include "megba/jv/jetvector.hpp"
@@ -58,9 +59,18 @@ struct PyProblem {
             py::array_t<int, py::array::c_style | py::array::forcecast> ptIdx,
             py::array_t<double, py::array::c_style | py::array::forcecast> meas,
             py::object info, py::object camFixed, py::object ptFixed) {
-    MEGBA_CHECK(cams.ndim() == 2 && cams.shape(1) == 9, "cams must be (ncam,9)");
+    // Block dims are inferred from the array shapes (reference analogue:
+    // ProblemOption.N, include/common.h:27-46).  Compiled set:
+    // camDim in {9,6,4}, ptDim = 3, resDim in {2,3}.
+    MEGBA_CHECK(cams.ndim() == 2 && (cams.shape(1) == 9 || cams.shape(1) == 6 ||
+                                     cams.shape(1) == 4),
+                "cams must be (ncam, 9|6|4)");
     MEGBA_CHECK(pts.ndim() == 2 && pts.shape(1) == 3, "pts must be (npt,3)");
-    MEGBA_CHECK(meas.ndim() == 2 && meas.shape(1) == 2, "meas must be (nobs,2)");
+    MEGBA_CHECK(meas.ndim() == 2 && (meas.shape(1) == 2 || meas.shape(1) == 3),
+                "meas must be (nobs, 2|3)");
+    prob.camDim = (int)cams.shape(1);
+    prob.ptDim = (int)pts.shape(1);
+    prob.resDim = (int)meas.shape(1);
     prob.ncam = (int)cams.shape(0);
     prob.npt = (int)pts.shape(0);
     prob.nobs = (int64_t)meas.shape(0);
@@ -70,11 +80,13 @@ struct PyProblem {
     prob.ptIdx.assign(ptIdx.data(), ptIdx.data() + ptIdx.size());
     prob.meas.assign(meas.data(), meas.data() + meas.size());
     if (!info.is_none()) {
+      const int rw = prob.resDim * (prob.resDim + 1) / 2;
       auto infoArr = py::cast<
           py::array_t<double, py::array::c_style | py::array::forcecast>>(info);
-      MEGBA_CHECK(infoArr.ndim() == 2 && infoArr.shape(1) == 3 &&
+      MEGBA_CHECK(infoArr.ndim() == 2 && infoArr.shape(1) == rw &&
                       infoArr.shape(0) == prob.nobs,
-                  "info must be (nobs,3): w00,w01,w11");
+                  "info must be (nobs, resDim*(resDim+1)/2) packed-upper "
+                  "symmetric (RD=2: w00,w01,w11)");
       prob.info.assign(infoArr.data(), infoArr.data() + infoArr.size());
     }
     auto readMask = [](py::object o, int n, std::vector<uint8_t>& dst) {
@@ -92,10 +104,32 @@ struct PyProblem {
              int worldSize, int deviceIndex, const std::string& diff,
              const std::string& schur, const std::string& loss,
              double lossDelta, py::object allreduce, py::object rcclId,
-             py::object customForward) {
+             py::object customForward, py::object intrinsics) {
     opt.rank = rank;
     opt.worldSize = worldSize;
     opt.deviceIndex = deviceIndex;
+    opt.camDim = prob.camDim;
+    opt.ptDim = prob.ptDim;
+    opt.resDim = prob.resDim;
+    if (!intrinsics.is_none()) {
+      auto arr = py::cast<
+          py::array_t<double, py::array::c_style | py::array::forcecast>>(
+          intrinsics);
+      MEGBA_CHECK(arr.size() == 3, "intrinsics must be [f, k1, k2]");
+      for (int i = 0; i < 3; ++i) opt.intr[i] = arr.data()[i];
+    }
+    MEGBA_CHECK(!customForward.is_none() ||
+                    hasBuiltinResidual(prob.camDim, prob.ptDim, prob.resDim),
+                "no built-in residual for (camDim,ptDim,resDim) = (" +
+                    std::to_string(prob.camDim) + "," +
+                    std::to_string(prob.ptDim) + "," +
+                    std::to_string(prob.resDim) +
+                    "): provide custom_forward (built-ins: (9,3,2) BAL, "
+                    "(6,3,2) BAL fixed-intrinsics, (6,3,3) SE3 point)");
+    MEGBA_CHECK(diff != "analytical" ||
+                    (prob.camDim == 9 && prob.resDim == 2),
+                "analytical diff is only available for the BAL (9,3,2) "
+                "model");
     MEGBA_CHECK(device == "cpu" || device == "gpu", "device must be cpu|gpu");
     MEGBA_CHECK(dtype == "float64" || dtype == "float32",
                 "dtype must be float64|float32");
@@ -274,8 +308,8 @@ struct PyProblem {
   double gInf() { return withEngine([&](auto& e) { return e.gInf(); }); }
 
   py::tuple getParams() {
-    py::array_t<double> cams({prob.ncam, 9});
-    py::array_t<double> pts({prob.npt, 3});
+    py::array_t<double> cams({prob.ncam, prob.camDim});
+    py::array_t<double> pts({prob.npt, prob.ptDim});
     withEngine([&](auto& e) {
       e.getParams(cams.mutable_data(), pts.mutable_data());
       return 0;
@@ -419,7 +453,8 @@ PYBIND11_MODULE(_core, m) {
            py::arg("diff") = "auto", py::arg("schur") = "explicit",
            py::arg("loss") = "none", py::arg("loss_delta") = 1.0,
            py::arg("allreduce") = py::none(), py::arg("rccl_id") = py::none(),
-           py::arg("custom_forward") = py::none())
+           py::arg("custom_forward") = py::none(),
+           py::arg("intrinsics") = py::none())
       .def("solve", &PyProblem::solve, py::arg("max_iter") = 20,
            py::arg("tau") = 1e4, py::arg("epsilon1") = 1.0,
            py::arg("epsilon2") = 1e-10, py::arg("solver_max_iter") = 100,

@@ -60,4 +60,68 @@ MEGBA_HD inline void balReprojectionError(const JT cam[9], const JT pt[3],
   res[1] = scaled * yp - meas[1];
 }
 
+// (6,3,2) built-in: BAL with per-problem FIXED intrinsics (calibrated
+// camera rig) — camera = [angle-axis(3), t(3)], intr = {f, k1, k2} plain
+// constants.  Same projection model as BAL, intrinsics out of the state.
+template <typename T, typename JT>
+MEGBA_HD inline void balFixedIntrError(const JT cam[6], const JT pt[3],
+                                       const T meas[2], const T intr[3],
+                                       JT res[2]) {
+  JT P[3];
+  angleAxisRotatePoint<T, JT>(cam, pt, P);
+  P[0] += cam[3];
+  P[1] += cam[4];
+  P[2] += cam[5];
+  const JT invNegZ = T(-1) / P[2];
+  const JT xp = P[0] * invNegZ;
+  const JT yp = P[1] * invNegZ;
+  const JT r2 = xp * xp + yp * yp;
+  const JT scaled = intr[0] * (T(1) + r2 * (intr[1] + intr[2] * r2));
+  res[0] = scaled * xp - meas[0];
+  res[1] = scaled * yp - meas[1];
+}
+
+// (6,3,3) built-in: SE3 point-alignment residual r = R(aa) p + t - meas
+// (3D registration / pose-graph-style landmark edge).
+template <typename T, typename JT>
+MEGBA_HD inline void se3PointError(const JT cam[6], const JT pt[3],
+                                   const T meas[3], JT res[3]) {
+  JT P[3];
+  angleAxisRotatePoint<T, JT>(cam, pt, P);
+  res[0] = P[0] + cam[3] - meas[0];
+  res[1] = P[1] + cam[4] - meas[1];
+  res[2] = P[2] + cam[5] - meas[2];
+}
+
+// Does a built-in residual exist for these dims?  (Other combinations are
+// supported via the custom-forward path only.)
+MEGBA_HD inline bool hasBuiltinResidual(int cd, int pd, int rd) {
+  return (cd == 9 && pd == 3 && rd == 2) || (cd == 6 && pd == 3 && rd == 2) ||
+         (cd == 6 && pd == 3 && rd == 3);
+}
+
+// Dims-dispatched built-in residual (compile-time selection; `intr` only
+// read by the fixed-intrinsics model).
+template <typename T, typename JT, int CD, int PD, int RD>
+MEGBA_HD inline void builtinResidual(const JT* cam, const JT* pt,
+                                     const T* meas, const T* intr, JT* res) {
+  if constexpr (CD == 9 && PD == 3 && RD == 2) {
+    (void)intr;
+    balReprojectionError<T, JT>(cam, pt, meas, res);
+  } else if constexpr (CD == 6 && PD == 3 && RD == 2) {
+    balFixedIntrError<T, JT>(cam, pt, meas, intr, res);
+  } else if constexpr (CD == 6 && PD == 3 && RD == 3) {
+    (void)intr;
+    se3PointError<T, JT>(cam, pt, meas, res);
+  } else {
+    // custom-forward-only dims: engines refuse AUTO diff without a custom
+    // forward at build time, so this is never reached.
+    for (int i = 0; i < RD; ++i) res[i] = JT(T(0));
+    (void)cam;
+    (void)pt;
+    (void)meas;
+    (void)intr;
+  }
+}
+
 }  // namespace megba

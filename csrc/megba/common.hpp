@@ -47,6 +47,16 @@ struct ProblemOption {
   int rank = 0;        // this process' rank (one process per GPU)
   int worldSize = 1;
   int deviceIndex = 0; // HIP device ordinal for this rank
+  // Block dimensions (camera/point/residual).  The engines are compiled for
+  // a fixed practical set ({9,6,4} x {3} x {2,3}); the reference took these
+  // as runtime values everywhere (build_linear_system.cu:48-146,
+  // common.h:27-46 ProblemOption.N).  Dims select the built-in residual:
+  // (9,3,2) BAL, (6,3,2) BAL with fixed intrinsics, (6,3,3) SE3 point
+  // alignment; other combinations require a custom forward.
+  int camDim = 9;
+  int ptDim = 3;
+  int resDim = 2;
+  double intr[3] = {1.0, 0.0, 0.0};  // f,k1,k2 for the (6,3,2) built-in
 };
 
 struct IterLog {

@@ -22,16 +22,24 @@
 namespace megba {
 
 // Raw problem as handed over from Python (original observation order).
+// Block dims are carried here (default BAL 9/3/2); the reference's
+// runtime-dim analogue is ProblemOption.N + per-kernel cameraDim/pointDim/
+// resDim arguments (include/common.h:27-46, build_linear_system.cu:48-146).
 struct BAProblemHost {
   int ncam = 0;
   int npt = 0;
   int64_t nobs = 0;
-  std::vector<double> cams;  // ncam*9
-  std::vector<double> pts;   // npt*3
+  int camDim = 9;
+  int ptDim = 3;
+  int resDim = 2;
+  std::vector<double> cams;  // ncam*camDim
+  std::vector<double> pts;   // npt*ptDim
   std::vector<int> camIdx;   // nobs
   std::vector<int> ptIdx;    // nobs
-  std::vector<double> meas;  // nobs*2 ([obs][2])
-  std::vector<double> info;  // optional nobs*3 (2x2 sym weights w00,w01,w11); empty = identity
+  std::vector<double> meas;  // nobs*resDim ([obs][resDim])
+  // optional nobs * resDim*(resDim+1)/2 packed-upper symmetric per-edge
+  // information matrix (RD=2: w00,w01,w11); empty = identity
+  std::vector<double> info;
   std::vector<uint8_t> camFixed;  // optional ncam (g2o-style fixed vertices)
   std::vector<uint8_t> ptFixed;   // optional npt
 };
@@ -39,10 +47,11 @@ struct BAProblemHost {
 struct ProblemIndex {
   int ncam = 0, npt = 0;
   int64_t nobs = 0;
+  int resDim = 2;
   // Arrays in (point, camera)-sorted order:
   std::vector<int> camOf, ptOf;     // nobs
-  std::vector<double> measSorted;   // nobs*2
-  std::vector<double> infoSorted;   // nobs*3 or empty
+  std::vector<double> measSorted;   // nobs*resDim
+  std::vector<double> infoSorted;   // nobs*resDim*(resDim+1)/2 or empty
   std::vector<int64_t> ptRowPtr;    // npt+1: edge range of each point
   std::vector<int64_t> split;       // worldSize+1 edge partition (point-aligned)
   std::vector<int> ptSplit;         // worldSize+1 point-id partition
@@ -57,6 +66,9 @@ inline ProblemIndex buildIndex(const BAProblemHost& p, int worldSize) {
   ix.ncam = p.ncam;
   ix.npt = p.npt;
   ix.nobs = p.nobs;
+  ix.resDim = p.resDim;
+  const int rd = p.resDim;
+  const int rw = rd * (rd + 1) / 2;
   const int64_t n = p.nobs;
 
   // Stable counting sort: by camera, then by point -> (pt, cam) order.
@@ -89,20 +101,17 @@ inline ProblemIndex buildIndex(const BAProblemHost& p, int worldSize) {
 
   ix.camOf.resize(n);
   ix.ptOf.resize(n);
-  ix.measSorted.resize(n * 2);
+  ix.measSorted.resize(n * rd);
   const bool hasInfo = !p.info.empty();
-  if (hasInfo) ix.infoSorted.resize(n * 3);
+  if (hasInfo) ix.infoSorted.resize(n * rw);
   for (int64_t k = 0; k < n; ++k) {
     const int64_t e = ix.perm[k];
     ix.camOf[k] = p.camIdx[e];
     ix.ptOf[k] = p.ptIdx[e];
-    ix.measSorted[2 * k] = p.meas[2 * e];
-    ix.measSorted[2 * k + 1] = p.meas[2 * e + 1];
-    if (hasInfo) {
-      ix.infoSorted[3 * k] = p.info[3 * e];
-      ix.infoSorted[3 * k + 1] = p.info[3 * e + 1];
-      ix.infoSorted[3 * k + 2] = p.info[3 * e + 2];
-    }
+    for (int d = 0; d < rd; ++d) ix.measSorted[rd * k + d] = p.meas[rd * e + d];
+    if (hasInfo)
+      for (int d = 0; d < rw; ++d)
+        ix.infoSorted[rw * k + d] = p.info[rw * e + d];
   }
 
   // Every vertex must be observed (else its Hessian block is singular).
@@ -136,6 +145,9 @@ inline ProblemIndex buildIndex(const BAProblemHost& p, int worldSize) {
         hi = mid;
     }
     lo = std::max(lo, prevPt + 1);  // at least one point per rank
+    // ...and leave at least one point for every later rank (a skewed
+    // distribution can otherwise push the split past npt).
+    lo = std::min(lo, p.npt - (worldSize - r));
     ix.ptSplit[r] = lo;
     ix.split[r] = ix.ptRowPtr[lo];
     prevPt = lo;
