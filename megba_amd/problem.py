@@ -11,11 +11,19 @@ import numpy as np
 class BAProblem:
     """A BAL-family bundle adjustment problem.
 
-    cams: (ncam, 9) [angle-axis(3), t(3), f, k1, k2]
+    Block dims are inferred from the array shapes; the compiled set is
+    camDim in {9, 6, 4}, ptDim = 3, resDim in {2, 3} (the reference takes
+    these as runtime kernel arguments, build_linear_system.cu:48-146).
+
+    cams: (ncam, 9) BAL [angle-axis(3), t(3), f, k1, k2]  -- or (ncam, 6)
+          calibrated [angle-axis(3), t(3)] (pass intrinsics= to build()),
+          or (ncam, 4) with a custom_forward.
     pts:  (npt, 3)
     cam_idx/pt_idx: (nobs,) int
-    meas: (nobs, 2)
-    info: optional (nobs, 3) upper-tri 2x2 information [w00, w01, w11]
+    meas: (nobs, 2) image observations, or (nobs, 3) for 3D residuals
+          (e.g. the built-in (6,3,3) SE3 point-alignment model)
+    info: optional (nobs, resDim*(resDim+1)/2) packed-upper symmetric
+          information (resDim=2: [w00, w01, w11])
     """
 
     def __init__(self, cams, pts, cam_idx, pt_idx, meas, info=None,
@@ -40,18 +48,20 @@ class BAProblem:
     def build(self, device="cpu", dtype="float64", rank=0, world_size=1,
               device_index=0, diff="auto", schur="explicit", loss="none",
               loss_delta=1.0, allreduce=None, rccl_id=None,
-              custom_forward=None):
+              custom_forward=None, intrinsics=None):
         """loss: robust loss ("none" | "huber" | "cauchy") with scale
         loss_delta -- IRLS reweighting, rho-consistent cost (beyond the
         reference, which has only the 2x2 information matrix).
-        custom_forward: optional callable (cam_jvs[9], pt_jvs[3],
-        meas_jvs[2]) -> (res0, res1) of JetVector, evaluated per forward pass
-        (runtime user-defined edges; see megba_amd.jv helpers)."""
+        custom_forward: optional callable (cam_jvs[camDim], pt_jvs[3],
+        meas_jvs[resDim]) -> resDim JetVectors, evaluated per forward pass
+        (runtime user-defined edges; see megba_amd.jv helpers).
+        intrinsics: [f, k1, k2] for the (6,3,2) fixed-intrinsics built-in."""
         self._core.build(device=device, dtype=dtype, rank=rank,
                          world_size=world_size, device_index=device_index,
                          diff=diff, schur=schur, loss=loss,
                          loss_delta=loss_delta, allreduce=allreduce,
-                         rccl_id=rccl_id, custom_forward=custom_forward)
+                         rccl_id=rccl_id, custom_forward=custom_forward,
+                         intrinsics=intrinsics)
         self._built = True
         return self
 
