@@ -17,12 +17,12 @@ import pytest
 import megba_amd as mb
 from megba_amd import jv
 
-RNG = np.random.default_rng(17)
 INTR = [420.0, -1e-7, 3e-13]
 
 
 def _synth_calibrated(ncam=10, npt=90, nobs=800, seed=4):
     """Calibrated-camera problem: cams (ncam,6), shared intrinsics INTR."""
+    RNG = np.random.default_rng(seed + 100)
     cams9, pts, ci, pi, meas = mb.synthesize_bal(ncam, npt, nobs, seed=seed)
     cams9[:, 6] = INTR[0]
     cams9[:, 7] = INTR[1]
@@ -70,6 +70,7 @@ def _bal_fixed_intr_res(cam6, pt, meas2):
 
 
 def test_632_jacobian_vs_fd():
+    RNG = np.random.default_rng(1)
     cams9, pts, ci, pi, meas = _synth_calibrated()
     cams6 = cams9[:, :6].copy()
     p = mb.BAProblem(cams6, pts, ci, pi, meas)
@@ -116,6 +117,7 @@ def test_632_matches_932_with_same_params():
 
 
 def test_632_lm_decreases():
+    RNG = np.random.default_rng(2)
     cams9, pts, ci, pi, meas = _synth_calibrated(14, 130, 1200, seed=8)
     cams6 = cams9[:, :6] + RNG.normal(0, 0.01, (14, 6))  # start off-optimum
     p = mb.BAProblem(cams6, pts, ci, pi, meas)
@@ -129,6 +131,7 @@ def test_632_lm_decreases():
 
 
 def _synth_se3(ncam=8, npt=70, nobs=500, seed=6):
+    RNG = np.random.default_rng(seed + 200)
     cams = np.zeros((ncam, 6))
     cams[:, :3] = RNG.normal(0, 0.3, (ncam, 3))
     cams[:, 3:] = RNG.normal(0, 1.0, (ncam, 3))
@@ -155,6 +158,7 @@ def _synth_se3(ncam=8, npt=70, nobs=500, seed=6):
 
 
 def test_633_se3_fd_and_solve():
+    RNG = np.random.default_rng(3)
     cams, pts, ci, pi, meas, se3 = _synth_se3()
     # perturb initial state
     cams0 = cams + RNG.normal(0, 0.02, cams.shape)
@@ -200,6 +204,7 @@ def _wp_np(cam4, pt, meas2):
 
 
 def test_432_custom_forward_fd_and_solve():
+    RNG = np.random.default_rng(4)
     ncam, npt, nobs = 6, 50, 360
     cams = RNG.normal(0, 0.1, (ncam, 4))
     pts = RNG.normal(0, 1.0, (npt, 3))
