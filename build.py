@@ -32,6 +32,13 @@ SOURCES = [
     # (path, is_device_code)
     ("megba/cpu_engine.cpp", False),
     ("megba/gpu/gpu_engine.hip", True),
+    # one TU per block-dimension set so hipcc compiles them in parallel
+    ("megba/gpu/gpu_dims_932.hip", True),
+    ("megba/gpu/gpu_dims_632.hip", True),
+    ("megba/gpu/gpu_dims_432.hip", True),
+    ("megba/gpu/gpu_dims_933.hip", True),
+    ("megba/gpu/gpu_dims_633.hip", True),
+    ("megba/gpu/gpu_dims_433.hip", True),
     ("megba/jv/jetvector.hip", True),
     ("bindings.cpp", False),
 ]

@@ -1,50 +1,21 @@
-// MI355X (gfx950) engine: HIP kernels + host orchestration + RCCL collectives.
-//
-// Kernel / distribution design (MI355X-first redesign, not a port —
-// behavioural anchors cite /root/reference):
-//  * Observations are (point, camera)-sorted and partitioned on POINT
-//    boundaries: each rank owns its points outright, so the whole point side
-//    (Hll, g_p, Cinv, E^T x, back-substitution) is communication-free and
-//    the only per-PCG-iteration collective is an allreduce of the replicated
-//    camera vector (9*ncam = 128 KB on Venice).  The reference replicates
-//    the point side instead and allreduces 3*npt words (24 MB on Venice)
-//    per iteration (its sites A1/A3-A6) — a ~190x traffic reduction sized
-//    for xGMI's per-link ring bandwidth.
-//  * kForward fuses the ENTIRE vectorised-autodiff forward pass into one
-//    kernel: each observation is handled by 4 consecutive lanes, each lane
-//    carrying a Jet<T,3> slice of the 12-wide dual part, so every
-//    intermediate of the reprojection expression lives in VGPRs.  The
-//    reference launches ~1 CUDA kernel per elementary op and streams
-//    (N+1)*nItem doubles through HBM each time
-//    (src/operator/jet_vector_math_impl.cu).
-//  * Assembly transports the camera-side per-edge data through a cam-sorted
-//    edge-major slab (contiguous ~400 B per edge, ~1 line-fetch per 64 B)
-//    instead of per-element transpose gathers (8x amplification) or
-//    per-element atomics (the all-atomic version measured 40.8 ms on
-//    Venice-5M, profiles/r01_venice_single_gpu.md).  Point-side Hll/g_p
-//    accumulate with a wave-level segmented scan over the point-sorted runs
-//    — atomics only at segment tails.
-//  * Control-flow scalars (rho, p^T q, norms) use fixed-shape two-pass
-//    deterministic reductions so every rank takes identical PCG branches.
+// GPU engine: dims dispatch + RCCL/HIP utilities.
+// The templated implementation (GpuEngine<T, CD, PD, RD> + all kernels)
+// lives in gpu_engine_impl.hpp and is explicitly instantiated by the
+// gpu_dims_*.hip TUs (parallel compilation); this TU routes runtime dims
+// to the right instantiation (the reference passed cameraDim/pointDim/
+// resDim as runtime kernel arguments, build_linear_system.cu:48-146).
 #include <hip/hip_runtime.h>
 #include <rccl/rccl.h>
 
 #include <chrono>
 #include <cmath>
 #include <condition_variable>
-#include <cstdlib>
 #include <cstring>
 #include <memory>
 #include <mutex>
+#include <string>
 #include <thread>
-#include <type_traits>
-#include <utility>
-#include <vector>
 
-#include "../analytical.hpp"
-#include "../bal_functor.hpp"
-#include "../jv/jetvector.hpp"
-#include "../smallmat.hpp"
 #include "gpu_engine.hpp"
 
 namespace megba {
@@ -63,1759 +34,15 @@ namespace megba {
                 std::string("RCCL error: ") + ncclGetErrorString(_e) + " @ " #expr); \
   } while (0)
 
-namespace {
-
-constexpr int kBlk = 256;
-constexpr int kRedBlocks = 256;  // fixed -> deterministic reductions
-
-inline int gridFor(int64_t n) {
-  int64_t g = (n + kBlk - 1) / kBlk;
-  return (int)(g < 1 ? 1 : (g > 8192 ? 8192 : g));
-}
-
-// ---------------------------------------------------------------------------
-// Reductions
-// ---------------------------------------------------------------------------
-enum class ROp { Dot, SumSq, AbsMax };
-
-template <typename T, ROp OP>
-__global__ void kRedPartial(const T* a, const T* b, int64_t n, double* part) {
-  __shared__ double sm[kBlk];
-  double v = 0.0;
-  for (int64_t i = blockIdx.x * (int64_t)kBlk + threadIdx.x; i < n;
-       i += (int64_t)gridDim.x * kBlk) {
-    const double x = (double)a[i];
-    if (OP == ROp::Dot)
-      v += x * (double)b[i];
-    else if (OP == ROp::SumSq)
-      v += x * x;
-    else
-      v = fmax(v, fabs(x));
-  }
-  sm[threadIdx.x] = v;
-  __syncthreads();
-  for (int s = kBlk / 2; s > 0; s >>= 1) {
-    if (threadIdx.x < s) {
-      if (OP == ROp::AbsMax)
-        sm[threadIdx.x] = fmax(sm[threadIdx.x], sm[threadIdx.x + s]);
-      else
-        sm[threadIdx.x] += sm[threadIdx.x + s];
-    }
-    __syncthreads();
-  }
-  if (threadIdx.x == 0) part[blockIdx.x] = sm[0];
-}
-
-template <ROp OP>
-__global__ void kRedFinal(const double* part, int nb, double* out) {
-  __shared__ double sm[kBlk];
-  double v = 0.0;
-  for (int i = threadIdx.x; i < nb; i += kBlk) {
-    if (OP == ROp::AbsMax)
-      v = fmax(v, part[i]);
-    else
-      v += part[i];
-  }
-  sm[threadIdx.x] = v;
-  __syncthreads();
-  for (int s = kBlk / 2; s > 0; s >>= 1) {
-    if (threadIdx.x < s) {
-      if (OP == ROp::AbsMax)
-        sm[threadIdx.x] = fmax(sm[threadIdx.x], sm[threadIdx.x + s]);
-      else
-        sm[threadIdx.x] += sm[threadIdx.x + s];
-    }
-    __syncthreads();
-  }
-  if (threadIdx.x == 0) *out = sm[0];
-}
-
-// ---------------------------------------------------------------------------
-// Forward (fused register autodiff)
-// ---------------------------------------------------------------------------
-template <typename T>
-__global__ __launch_bounds__(256, 2) void kForward(
-    int64_t nL, const int* __restrict__ camOf, const int* __restrict__ ptOf,
-    const T* __restrict__ params, int ncam, const T* __restrict__ meas,
-    const unsigned char* __restrict__ camFixed,
-    const unsigned char* __restrict__ ptFixed, T* __restrict__ rOut,
-    T* __restrict__ Jc, T* __restrict__ Jp, double* chi2Acc, int lossKind,
-    T lossD2) {
-  using J3 = Jet<T, 3>;
-  __shared__ double sm[kBlk];
-  double chi2 = 0.0;
-  const T* ptsBase = params + (int64_t)ncam * 9;
-  const int64_t nWork = nL * 4;
-  for (int64_t i = blockIdx.x * (int64_t)kBlk + threadIdx.x; i < nWork;
-       i += (int64_t)gridDim.x * kBlk) {
-    const int64_t e = i >> 2;
-    const int sub = (int)(i & 3);       // this lane's 3-wide gradient slice
-    const int base = 3 * sub;           // gradient columns [base,base+3)
-    const T* cp = params + (int64_t)camOf[e] * 9;
-    const T* pp = ptsBase + (int64_t)ptOf[e] * 3;
-    J3 cam[9], pt[3], res[2];
-    for (int k = 0; k < 9; ++k) cam[k] = J3::leaf(cp[k], k - base);
-    for (int k = 0; k < 3; ++k) pt[k] = J3::leaf(pp[k], 9 + k - base);
-    const T m[2] = {meas[2 * e], meas[2 * e + 1]};
-    balReprojectionError<T, J3>(cam, pt, m, res);
-    const bool cfix = camFixed && camFixed[camOf[e]];
-    const bool pfix = ptFixed && ptFixed[ptOf[e]];
-    for (int row = 0; row < 2; ++row) {
-      for (int j = 0; j < 3; ++j) {
-        const int col = base + j;
-        if (col < 9)
-          Jc[((int64_t)(col * 2 + row)) * nL + e] = cfix ? T(0) : res[row].d[j];
-        else
-          Jp[((int64_t)((col - 9) * 2 + row)) * nL + e] =
-              pfix ? T(0) : res[row].d[j];
-      }
-      if (sub == 0) rOut[(int64_t)row * nL + e] = res[row].v;
-    }
-    if (sub == 0)
-      chi2 += (double)lossRho(lossKind, lossD2,
-                              res[0].v * res[0].v + res[1].v * res[1].v);
-  }
-  sm[threadIdx.x] = chi2;
-  __syncthreads();
-  for (int s = kBlk / 2; s > 0; s >>= 1) {
-    if (threadIdx.x < s) sm[threadIdx.x] += sm[threadIdx.x + s];
-    __syncthreads();
-  }
-  if (threadIdx.x == 0) atomicAdd(chi2Acc, sm[0]);
-}
-
-// Analytical-derivative forward: one thread per edge, closed-form residual +
-// 2x12 Jacobian (reference C11, src/geo/analytical_derivatives.cu).
-template <typename T>
-__global__ void kForwardAnalytical(int64_t nL, const int* __restrict__ camOf,
-                                   const int* __restrict__ ptOf,
-                                   const T* __restrict__ params, int ncam,
-                                   const T* __restrict__ meas,
-                                   const unsigned char* __restrict__ camFixed,
-                                   const unsigned char* __restrict__ ptFixed,
-                                   T* __restrict__ rOut, T* __restrict__ Jc,
-                                   T* __restrict__ Jp, double* chi2Acc,
-                                   int lossKind, T lossD2) {
-  __shared__ double sm[kBlk];
-  double chi2 = 0.0;
-  const T* ptsBase = params + (int64_t)ncam * 9;
-  for (int64_t e = blockIdx.x * (int64_t)kBlk + threadIdx.x; e < nL;
-       e += (int64_t)gridDim.x * kBlk) {
-    const T* cp = params + (int64_t)camOf[e] * 9;
-    const T* pp = ptsBase + (int64_t)ptOf[e] * 3;
-    const T m[2] = {meas[2 * e], meas[2 * e + 1]};
-    T res[2], jc[2][9], jp[2][3];
-    balAnalytical<T>(cp, pp, m, res, jc, jp);
-    chi2 += (double)lossRho(lossKind, lossD2,
-                            res[0] * res[0] + res[1] * res[1]);
-    const bool cfix = camFixed && camFixed[camOf[e]];
-    const bool pfix = ptFixed && ptFixed[ptOf[e]];
-    for (int row = 0; row < 2; ++row) {
-      rOut[(int64_t)row * nL + e] = res[row];
-      for (int col = 0; col < 9; ++col)
-        Jc[((int64_t)(col * 2 + row)) * nL + e] = cfix ? T(0) : jc[row][col];
-      for (int col = 0; col < 3; ++col)
-        Jp[((int64_t)(col * 2 + row)) * nL + e] = pfix ? T(0) : jp[row][col];
-    }
-  }
-  sm[threadIdx.x] = chi2;
-  __syncthreads();
-  for (int s = kBlk / 2; s > 0; s >>= 1) {
-    if (threadIdx.x < s) sm[threadIdx.x] += sm[threadIdx.x + s];
-    __syncthreads();
-  }
-  if (threadIdx.x == 0) atomicAdd(chi2Acc, sm[0]);
-}
-
-// chi2 with a robust loss, from the residual buffers (custom-edge path).
-template <typename T>
-__global__ void kChi2Loss(int64_t nL, const T* __restrict__ r, int lossKind,
-                          T lossD2, double* acc) {
-  __shared__ double sm[kBlk];
-  double local = 0.0;
-  for (int64_t e = blockIdx.x * (int64_t)kBlk + threadIdx.x; e < nL;
-       e += (int64_t)gridDim.x * kBlk) {
-    const T r0 = r[e], r1 = r[nL + e];
-    local += (double)lossRho(lossKind, lossD2, r0 * r0 + r1 * r1);
-  }
-  sm[threadIdx.x] = local;
-  __syncthreads();
-  for (int s = kBlk / 2; s > 0; s >>= 1) {
-    if (threadIdx.x < s) sm[threadIdx.x] += sm[threadIdx.x + s];
-    __syncthreads();
-  }
-  if (threadIdx.x == 0) atomicAdd(acc, sm[0]);
-}
-
-// Custom-edge support: gather parameter leaves, repack residual JetVectors.
-template <typename T>
-__global__ void kGatherLeaves(int64_t nL, const int* __restrict__ camOf,
-                              const int* __restrict__ ptOf,
-                              const T* __restrict__ params, int ncam,
-                              T* __restrict__ leaf /*[12][nL]*/) {
-  const T* ptsBase = params + (int64_t)ncam * 9;
-  for (int64_t e = blockIdx.x * (int64_t)kBlk + threadIdx.x; e < nL;
-       e += (int64_t)gridDim.x * kBlk) {
-    const T* cp = params + (int64_t)camOf[e] * 9;
-    const T* pp = ptsBase + (int64_t)ptOf[e] * 3;
-    for (int k = 0; k < 9; ++k) leaf[(int64_t)k * nL + e] = cp[k];
-    for (int k = 0; k < 3; ++k) leaf[(int64_t)(9 + k) * nL + e] = pp[k];
-  }
-}
-
-template <typename T>
-__global__ void kRepackRes(int64_t nL, int row, const T* __restrict__ rv,
-                           const T* __restrict__ rg /*[12][nL]*/,
-                           const int* __restrict__ camOf,
-                           const int* __restrict__ ptOf,
-                           const unsigned char* __restrict__ camFixed,
-                           const unsigned char* __restrict__ ptFixed,
-                           T* __restrict__ rOut, T* __restrict__ Jc,
-                           T* __restrict__ Jp) {
-  for (int64_t e = blockIdx.x * (int64_t)kBlk + threadIdx.x; e < nL;
-       e += (int64_t)gridDim.x * kBlk) {
-    rOut[(int64_t)row * nL + e] = rv[e];
-    const bool cfix = camFixed && camFixed[camOf[e]];
-    const bool pfix = ptFixed && ptFixed[ptOf[e]];
-    for (int k = 0; k < 9; ++k)
-      Jc[((int64_t)(k * 2 + row)) * nL + e] =
-          cfix ? T(0) : rg[(int64_t)k * nL + e];
-    for (int k = 0; k < 3; ++k)
-      Jp[((int64_t)(k * 2 + row)) * nL + e] =
-          pfix ? T(0) : rg[(int64_t)(9 + k) * nL + e];
-  }
-}
-
-// ---------------------------------------------------------------------------
-// Assembly
-// ---------------------------------------------------------------------------
-// Slab row layout (doubles), written at the edge's CAM-sorted position:
-//   [0..26]                 Hpl block (EXPL only)
-//   [JCOFF..JCOFF+17]       Jc rows (col-major pairs: jc[col][row])
-//   [JCOFF+18, JCOFF+19]    weighted residual rows
-//   [JCOFF+20..JCOFF+37]    weighted Jc rows (HASINFO only)
-template <bool EXPL, bool HASINFO>
-struct SlabLayout {
-  static constexpr int JCOFF = EXPL ? 27 : 0;
-  static constexpr int WROFF = JCOFF + 18;
-  static constexpr int WJCOFF = WROFF + 2;
-  static constexpr int JPOFF = WJCOFF + (HASINFO ? 18 : 0);  // implicit only
-  static constexpr int SW = JPOFF + (EXPL ? 0 : 6);
-};
-
-// Per-edge pass, primary ((pt,cam)-sorted) order: Hpl grad-major (for E^T x),
-// the cam-sorted slab row, and the point-side Hll/g_p via a wave-level
-// segmented scan (atomics only at point-run tails).
-template <typename T, bool HASINFO, bool EXPL>
-__global__ void kAssembleEdge(int64_t nL, const int* __restrict__ camOf,
-                              const int* __restrict__ ptOf,
-                              const T* __restrict__ r, const T* __restrict__ Jc,
-                              const T* __restrict__ Jp,
-                              const T* __restrict__ info, T* __restrict__ Hll,
-                              T* __restrict__ Hpl, T* __restrict__ g, int ncam,
-                              const int* __restrict__ camPos,
-                              T* __restrict__ slab, int lossKind, T lossD2) {
-  using L = SlabLayout<EXPL, HASINFO>;
-  T* gp = g + (int64_t)ncam * 9;
-  const int lane = threadIdx.x & 63;
-  const int64_t nWork = ((nL + kBlk - 1) / kBlk) * (int64_t)kBlk;
-  for (int64_t e0i = blockIdx.x * (int64_t)kBlk + threadIdx.x; e0i < nWork;
-       e0i += (int64_t)gridDim.x * kBlk) {
-    const bool active = e0i < nL;
-    const int64_t e = active ? e0i : nL - 1;
-    const int pt = ptOf[e];
-    T hll6[6] = {0, 0, 0, 0, 0, 0};
-    T gp3[3] = {0, 0, 0};
-    if (active) {
-      T jc[2][9], jp[2][3], rr[2];
-      for (int k = 0; k < 9; ++k) {
-        jc[0][k] = Jc[((int64_t)(k * 2 + 0)) * nL + e];
-        jc[1][k] = Jc[((int64_t)(k * 2 + 1)) * nL + e];
-      }
-      for (int k = 0; k < 3; ++k) {
-        jp[0][k] = Jp[((int64_t)(k * 2 + 0)) * nL + e];
-        jp[1][k] = Jp[((int64_t)(k * 2 + 1)) * nL + e];
-      }
-      rr[0] = r[e];
-      rr[1] = r[nL + e];
-      T wjp[2][3], wr[2];
-      T wjc0[9], wjc1[9];
-      if (HASINFO) {
-        // HASINFO also covers robust-loss runs without an information
-        // matrix (info == nullptr -> identity), since both need the
-        // weighted Jc rows stored separately in the slab.
-        T w00 = T(1), w01 = T(0), w11 = T(1);
-        if (info) {
-          w00 = info[3 * e];
-          w01 = info[3 * e + 1];
-          w11 = info[3 * e + 2];
-        }
-        if (lossKind) {
-          const T w = lossWeight(lossKind, lossD2,
-                                 rr[0] * rr[0] + rr[1] * rr[1]);
-          w00 *= w;
-          w01 *= w;
-          w11 *= w;
-        }
-        for (int k = 0; k < 9; ++k) {
-          wjc0[k] = w00 * jc[0][k] + w01 * jc[1][k];
-          wjc1[k] = w01 * jc[0][k] + w11 * jc[1][k];
-        }
-        for (int k = 0; k < 3; ++k) {
-          wjp[0][k] = w00 * jp[0][k] + w01 * jp[1][k];
-          wjp[1][k] = w01 * jp[0][k] + w11 * jp[1][k];
-        }
-        wr[0] = w00 * rr[0] + w01 * rr[1];
-        wr[1] = w01 * rr[0] + w11 * rr[1];
-      } else {
-        for (int k = 0; k < 3; ++k) {
-          wjp[0][k] = jp[0][k];
-          wjp[1][k] = jp[1][k];
-        }
-        wr[0] = rr[0];
-        wr[1] = rr[1];
-      }
-      T* row = slab + (int64_t)camPos[e] * L::SW;
-      if (EXPL) {
-        for (int a = 0; a < 9; ++a)
-          for (int b = 0; b < 3; ++b) {
-            const T v = jc[0][a] * wjp[0][b] + jc[1][a] * wjp[1][b];
-            Hpl[((int64_t)(a * 3 + b)) * nL + e] = v;
-            row[a * 3 + b] = v;
-          }
-      }
-      for (int k = 0; k < 9; ++k) {
-        row[L::JCOFF + k * 2] = jc[0][k];
-        row[L::JCOFF + k * 2 + 1] = jc[1][k];
-      }
-      row[L::WROFF] = wr[0];
-      row[L::WROFF + 1] = wr[1];
-      if (HASINFO)
-        for (int k = 0; k < 9; ++k) {
-          row[L::WJCOFF + k * 2] = wjc0[k];
-          row[L::WJCOFF + k * 2 + 1] = wjc1[k];
-        }
-      if (!EXPL)
-        for (int k = 0; k < 3; ++k) {
-          row[L::JPOFF + k * 2] = jp[0][k];
-          row[L::JPOFF + k * 2 + 1] = jp[1][k];
-        }
-      // point-side contributions
-      hll6[0] = jp[0][0] * wjp[0][0] + jp[1][0] * wjp[1][0];
-      hll6[1] = jp[0][0] * wjp[0][1] + jp[1][0] * wjp[1][1];
-      hll6[2] = jp[0][0] * wjp[0][2] + jp[1][0] * wjp[1][2];
-      hll6[3] = jp[0][1] * wjp[0][1] + jp[1][1] * wjp[1][1];
-      hll6[4] = jp[0][1] * wjp[0][2] + jp[1][1] * wjp[1][2];
-      hll6[5] = jp[0][2] * wjp[0][2] + jp[1][2] * wjp[1][2];
-      for (int k = 0; k < 3; ++k)
-        gp3[k] = -(jp[0][k] * wr[0] + jp[1][k] * wr[1]);
-    }
-    // segmented scan over the wave's point runs (early exit once no lane
-    // continues a segment at this distance — monotone in off)
-    for (int off = 1; off < 64; off <<= 1) {
-      const int ppt = __shfl_up(pt, off, 64);
-      const bool join = lane >= off && ppt == pt;
-      if (__ballot(join) == 0ull) break;
-      T a6[6], a3[3];
-      for (int k = 0; k < 6; ++k) a6[k] = __shfl_up(hll6[k], off, 64);
-      for (int k = 0; k < 3; ++k) a3[k] = __shfl_up(gp3[k], off, 64);
-      if (join) {
-        for (int k = 0; k < 6; ++k) hll6[k] += a6[k];
-        for (int k = 0; k < 3; ++k) gp3[k] += a3[k];
-      }
-    }
-    const int nextPt = __shfl_down(pt, 1, 64);
-    const bool tail = active && (lane == 63 || nextPt != pt || e0i == nL - 1);
-    if (tail) {
-      T* H = Hll + (int64_t)pt * 9;
-      atomicAdd(&H[0], hll6[0]);
-      atomicAdd(&H[1], hll6[1]);
-      atomicAdd(&H[2], hll6[2]);
-      atomicAdd(&H[3], hll6[1]);
-      atomicAdd(&H[4], hll6[3]);
-      atomicAdd(&H[5], hll6[4]);
-      atomicAdd(&H[6], hll6[2]);
-      atomicAdd(&H[7], hll6[4]);
-      atomicAdd(&H[8], hll6[5]);
-      for (int k = 0; k < 3; ++k) atomicAdd(&gp[(int64_t)pt * 3 + k], gp3[k]);
-    }
-  }
-}
-
-// Camera blocks: one 128-thread block per <=256-row chunk of one camera's
-// cam-sorted slab rows.  Rows are contiguous, staged through LDS in 128-row
-// tiles; threads 0..80 each own one Hpp element, 81..89 one g_c element;
-// one atomicAdd per output per chunk.
-template <typename T, bool HASINFO, bool EXPL>
-__global__ __launch_bounds__(128) void kAssembleCam(
-    int nChunks, const int* __restrict__ chCam, const int* __restrict__ chLo,
-    const int* __restrict__ chHi, const T* __restrict__ slab,
-    T* __restrict__ Hpp, T* __restrict__ g) {
-  using L = SlabLayout<EXPL, HASINFO>;
-  constexpr int TE = 128;
-  constexpr int ST = HASINFO ? 38 : 20;  // jc 18 + wr 2 (+ wjc 18)
-  __shared__ T lds[TE * ST];
-  const int chunk = blockIdx.x;
-  if (chunk >= nChunks) return;
-  const int cam = chCam[chunk];
-  const int lo = chLo[chunk], hi = chHi[chunk];
-  const int t = threadIdx.x;
-  T acc = T(0);
-  const int ti = t < 81 ? t / 9 : t - 81;
-  const int tj = t < 81 ? t % 9 : 0;
-  for (int s0 = lo; s0 < hi; s0 += TE) {
-    const int nt = min(TE, hi - s0);
-    for (int idx = t; idx < nt * ST; idx += 128) {
-      const int row = idx / ST;
-      const int k = idx % ST;
-      const T* src = slab + (int64_t)(s0 + row) * L::SW + L::JCOFF;
-      lds[row * ST + k] = src[k];
-    }
-    __syncthreads();
-    if (t < 90) {
-      const int woff = HASINFO ? 20 : 0;  // weighted rows (== raw if no info)
-      for (int e = 0; e < nt; ++e) {
-        const T* row = lds + e * ST;
-        if (t < 81)
-          acc += row[ti * 2] * row[woff + tj * 2] +
-                 row[ti * 2 + 1] * row[woff + tj * 2 + 1];
-        else
-          acc -= row[ti * 2] * row[18] + row[ti * 2 + 1] * row[19];
-      }
-    }
-    __syncthreads();
-  }
-  if (t < 81)
-    atomicAdd(&Hpp[(int64_t)cam * 81 + ti * 9 + tj], acc);
-  else if (t < 90)
-    atomicAdd(&g[(int64_t)cam * 9 + ti], acc);
-}
-
-// Explicit only: stream the slab's Hpl blocks out as a cam-sorted grad-major
-// copy for the E*w kernel (contiguous reads, coalesced writes).
-template <typename T, bool HASINFO>
-__global__ void kFinalizeCam(int64_t nL, const T* __restrict__ slab,
-                             T* __restrict__ HplCam) {
-  using L = SlabLayout<true, HASINFO>;
-  for (int64_t j = blockIdx.x * (int64_t)kBlk + threadIdx.x; j < nL;
-       j += (int64_t)gridDim.x * kBlk) {
-    const T* row = slab + j * L::SW;
-    for (int k = 0; k < 27; ++k) HplCam[(int64_t)k * nL + j] = row[k];
-  }
-}
-
-// Implicit only: stream the slab's weighted-Jc and Jp back out as cam-sorted
-// GRAD-MAJOR copies.  The first E w implementation read the 26-value
-// edge-major slab rows directly each PCG iteration; with 64 lanes striding
-// SW*sizeof(T) apart every one of its ~26 loads is address-divergent
-// (~64 TA requests/instruction), which measured 2.2x the byte floor at 29M
-// edges (profiles/r01_final13682_implicit_fp32.md).  Paying one extra
-// coalesced-write pass per ACCEPTED LM step makes the per-iteration reads
-// fully coalesced instead.
-template <typename T, bool HASINFO>
-__global__ void kFinalizeCamImp(int64_t nL, const T* __restrict__ slab,
-                                T* __restrict__ JcCam,
-                                T* __restrict__ JpCam) {
-  using L = SlabLayout<false, HASINFO>;
-  constexpr int woff = HASINFO ? L::WJCOFF : L::JCOFF;
-  for (int64_t j = blockIdx.x * (int64_t)kBlk + threadIdx.x; j < nL;
-       j += (int64_t)gridDim.x * kBlk) {
-    const T* row = slab + j * L::SW;
-    for (int k = 0; k < 18; ++k) JcCam[(int64_t)k * nL + j] = row[woff + k];
-    for (int k = 0; k < 6; ++k) JpCam[(int64_t)k * nL + j] = row[L::JPOFF + k];
-  }
-}
-
-// Implicit E w over the grad-major cam-sorted copies (weights already folded
-// into JcCam): u = Jp w, acc += wJc^T u; chunk/wave/atomic structure as
-// kSpmvEx.
-template <typename T>
-__global__ __launch_bounds__(64) void kSpmvExImpG(
-    int nChunks, const int* __restrict__ chCam, const int* __restrict__ chLo,
-    const int* __restrict__ chHi, const int* __restrict__ ptOfCam,
-    const T* __restrict__ JcCam, const T* __restrict__ JpCam, int64_t nL,
-    const T* __restrict__ w, T* __restrict__ out) {
-  const int chunk = blockIdx.x;
-  if (chunk >= nChunks) return;
-  const int cam = chCam[chunk];
-  T acc[9];
-  for (int i = 0; i < 9; ++i) acc[i] = T(0);
-  const int lo = chLo[chunk], hi = chHi[chunk];
-  for (int j = lo + (int)threadIdx.x; j < hi; j += 64) {
-    const T* wp = w + (int64_t)ptOfCam[j] * 3;
-    T u0 = T(0), u1 = T(0);
-    for (int c = 0; c < 3; ++c) {
-      u0 += JpCam[((int64_t)(c * 2 + 0)) * nL + j] * wp[c];
-      u1 += JpCam[((int64_t)(c * 2 + 1)) * nL + j] * wp[c];
-    }
-    for (int i = 0; i < 9; ++i)
-      acc[i] += JcCam[((int64_t)(i * 2 + 0)) * nL + j] * u0 +
-                JcCam[((int64_t)(i * 2 + 1)) * nL + j] * u1;
-  }
-  for (int off = 32; off > 0; off >>= 1)
-    for (int i = 0; i < 9; ++i) acc[i] += __shfl_down(acc[i], off, 64);
-  if (threadIdx.x == 0) {
-    T* oc = out + (int64_t)cam * 9;
-    for (int i = 0; i < 9; ++i) atomicAdd(&oc[i], acc[i]);
-  }
-}
-
-// ---------------------------------------------------------------------------
-// Damping, block inverse
-// ---------------------------------------------------------------------------
-template <typename T, int D>
-__global__ void kDamp(int64_t nElem, const T* __restrict__ H, T* __restrict__ Hd,
-                      T f, const unsigned char* __restrict__ fixed) {
-  for (int64_t i = blockIdx.x * (int64_t)kBlk + threadIdx.x; i < nElem;
-       i += (int64_t)gridDim.x * kBlk) {
-    const int within = (int)(i % (D * D));
-    const bool diag = within / D == within % D;
-    if (fixed && fixed[i / (D * D)]) {
-      // fixed vertex: identity block => deltaX = 0 for it (J columns are
-      // already zeroed by the forward kernels)
-      Hd[i] = diag ? T(1) : T(0);
-      continue;
-    }
-    const T v = H[i];
-    Hd[i] = diag ? v * f : v;
-  }
-}
-
-template <typename T, int D>
-__global__ void kInvert(int nBlk, const T* __restrict__ Hd, T* __restrict__ Hinv,
-                        int* fail) {
-  for (int64_t b = blockIdx.x * (int64_t)kBlk + threadIdx.x; b < nBlk;
-       b += (int64_t)gridDim.x * kBlk) {
-    if (!spdInvertPacked<T, D>(Hd + b * D * D, Hinv + b * D * D))
-      atomicOr(fail, 1);  // rare: retried by kInvertJitter
-  }
-}
-
-// Rare path: retry semi-definite blocks with a growing relative diagonal
-// jitter (matches the CPU oracle); launched only if kInvert reported failure
-// so its register/scratch cost stays off the hot path.
-template <typename T, int D>
-__global__ void kInvertJitter(int nBlk, const T* __restrict__ Hd,
-                              T* __restrict__ Hinv, int* fail) {
-  for (int64_t b = blockIdx.x * (int64_t)kBlk + threadIdx.x; b < nBlk;
-       b += (int64_t)gridDim.x * kBlk) {
-    const T* a = Hd + b * D * D;
-    T* out = Hinv + b * D * D;
-    if (spdInvertPacked<T, D>(a, out)) continue;
-    T buf[D * D];
-    T mx = T(0);
-    for (int i = 0; i < D; ++i) {
-      const T ad = (T)fabs((double)a[i * D + i]);
-      mx = ad > mx ? ad : mx;
-    }
-    const T eps = (mx > T(0) ? mx : T(1)) * T(1e-10);
-    bool ok = false;
-    T jit = eps;
-    for (int k = 0; k < 40 && !ok; ++k, jit *= T(10)) {
-      for (int i = 0; i < D * D; ++i) buf[i] = a[i];
-      for (int i = 0; i < D; ++i) buf[i * D + i] += jit;
-      ok = spdInvertPacked<T, D>(buf, out);
-    }
-    if (!ok) atomicOr(fail + 1, 1);
-  }
-}
-
-// ---------------------------------------------------------------------------
-// Schur SpMV pieces
-// ---------------------------------------------------------------------------
-// temp[3*pt] = Hpl^T x per local point (fully local, no collective): one
-// thread per primary-order edge (coalesced grad-major Hpl or J reads; the
-// replicated camera vector x is L2-resident), wave segmented-scan over the
-// point runs, atomics only at run tails.  IMP: matrix-free from J
-// (reference C23).
-template <typename T, bool IMP, bool HASINFO>
-__global__ void kSpmvEtx(int64_t nL, const int* __restrict__ camOf,
-                         const int* __restrict__ ptOf,
-                         const T* __restrict__ Hpl,
-                         const T* const* __restrict__ jSlots,
-                         const T* __restrict__ info, int lossKind, T lossD2,
-                         const T* __restrict__ x, T* __restrict__ out) {
-  // IMP reads the accepted J set through the device pointer slots so a
-  // captured graph stays valid across accept-time buffer flips.
-  const T* Jc = IMP ? jSlots[0] : nullptr;
-  const T* Jp = IMP ? jSlots[1] : nullptr;
-  const T* rBak = IMP ? jSlots[2] : nullptr;
-  const int lane = threadIdx.x & 63;
-  const int64_t nWork = ((nL + kBlk - 1) / kBlk) * (int64_t)kBlk;
-  for (int64_t j0 = blockIdx.x * (int64_t)kBlk + threadIdx.x; j0 < nWork;
-       j0 += (int64_t)gridDim.x * kBlk) {
-    const bool active = j0 < nL;
-    const int64_t j = active ? j0 : nL - 1;
-    const int pt = ptOf[j];
-    T o0 = 0, o1 = 0, o2 = 0;
-    if (active) {
-      const T* xc = x + (int64_t)camOf[j] * 9;
-      if (IMP) {
-        T u0 = T(0), u1 = T(0);
-        for (int i = 0; i < 9; ++i) {
-          const T xi = xc[i];
-          u0 += Jc[((int64_t)(i * 2 + 0)) * nL + j] * xi;
-          u1 += Jc[((int64_t)(i * 2 + 1)) * nL + j] * xi;
-        }
-        if (HASINFO) {
-          const T w00 = info[3 * j], w01 = info[3 * j + 1],
-                  w11 = info[3 * j + 2];
-          const T a = w00 * u0 + w01 * u1;
-          u1 = w01 * u0 + w11 * u1;
-          u0 = a;
-        }
-        if (lossKind) {
-          const T r0 = rBak[j], r1 = rBak[nL + j];
-          const T w = lossWeight(lossKind, lossD2, r0 * r0 + r1 * r1);
-          u0 *= w;
-          u1 *= w;
-        }
-        o0 = Jp[((int64_t)0) * nL + j] * u0 + Jp[((int64_t)1) * nL + j] * u1;
-        o1 = Jp[((int64_t)2) * nL + j] * u0 + Jp[((int64_t)3) * nL + j] * u1;
-        o2 = Jp[((int64_t)4) * nL + j] * u0 + Jp[((int64_t)5) * nL + j] * u1;
-      } else {
-        for (int i = 0; i < 9; ++i) {
-          const T xi = xc[i];
-          o0 += Hpl[((int64_t)(i * 3 + 0)) * nL + j] * xi;
-          o1 += Hpl[((int64_t)(i * 3 + 1)) * nL + j] * xi;
-          o2 += Hpl[((int64_t)(i * 3 + 2)) * nL + j] * xi;
-        }
-      }
-    }
-    for (int off = 1; off < 64; off <<= 1) {
-      const int ppt = __shfl_up(pt, off, 64);
-      const bool join = lane >= off && ppt == pt;
-      // Point runs are short (degree ~5): once no lane continues a segment
-      // at distance `off`, no lane can at any larger distance either — skip
-      // the remaining dependent shuffle rounds (the scan chain is this
-      // kernel's issue-stall bound, 70% SQ_WAIT_INST_ANY).
-      if (__ballot(join) == 0ull) break;
-      const T a0 = __shfl_up(o0, off, 64);
-      const T a1 = __shfl_up(o1, off, 64);
-      const T a2 = __shfl_up(o2, off, 64);
-      if (join) {
-        o0 += a0;
-        o1 += a1;
-        o2 += a2;
-      }
-    }
-    const int nextPt = __shfl_down(pt, 1, 64);
-    const bool tail = active && (lane == 63 || nextPt != pt || j0 == nL - 1);
-    if (tail) {
-      atomicAdd(&out[3 * pt], o0);
-      atomicAdd(&out[3 * pt + 1], o1);
-      atomicAdd(&out[3 * pt + 2], o2);
-    }
-  }
-}
-
-// out[9*cam] += E w partials: one wave per <=256-row chunk of one camera's
-// cam-sorted rows; per-lane partials, wave-wide shuffle reduce, one
-// atomicAdd set per chunk.  Caller allreduces 9*ncam (the ONLY per-PCG-
-// iteration collective).
-template <typename T>
-__global__ __launch_bounds__(64) void kSpmvEx(
-    int nChunks, const int* __restrict__ chCam, const int* __restrict__ chLo,
-    const int* __restrict__ chHi, const int* __restrict__ ptOfCam,
-    const T* __restrict__ HplCam, int64_t nL, const T* __restrict__ w,
-    T* __restrict__ out) {
-  const int chunk = blockIdx.x;
-  if (chunk >= nChunks) return;
-  const int cam = chCam[chunk];
-  T acc[9];
-  for (int i = 0; i < 9; ++i) acc[i] = T(0);
-  const int lo = chLo[chunk], hi = chHi[chunk];
-  for (int j = lo + (int)threadIdx.x; j < hi; j += 64) {
-    const T* wp = w + (int64_t)ptOfCam[j] * 3;
-    const T w0 = wp[0], w1 = wp[1], w2 = wp[2];
-    for (int i = 0; i < 9; ++i)
-      acc[i] += HplCam[((int64_t)(i * 3 + 0)) * nL + j] * w0 +
-                HplCam[((int64_t)(i * 3 + 1)) * nL + j] * w1 +
-                HplCam[((int64_t)(i * 3 + 2)) * nL + j] * w2;
-  }
-  for (int off = 32; off > 0; off >>= 1)
-    for (int i = 0; i < 9; ++i) acc[i] += __shfl_down(acc[i], off, 64);
-  if (threadIdx.x == 0) {
-    T* oc = out + (int64_t)cam * 9;
-    for (int i = 0; i < 9; ++i) atomicAdd(&oc[i], acc[i]);
-  }
-}
-
-// Fused preconditioner apply + rho partial: z = Binv r (thread per row) and
-// per-block partials of r.z in one pass (saves two launches per PCG iter).
-template <typename T>
-__global__ void kPrecondRho(int nBlk, const T* __restrict__ Binv,
-                            const T* __restrict__ r, T* __restrict__ z,
-                            double* part) {
-  __shared__ double sm[kBlk];
-  double local = 0.0;
-  for (int64_t idx = blockIdx.x * (int64_t)kBlk + threadIdx.x;
-       idx < (int64_t)nBlk * 9; idx += (int64_t)gridDim.x * kBlk) {
-    const int64_t b = idx / 9;
-    const int rr = (int)(idx % 9);
-    const T* row = Binv + b * 81 + (int64_t)rr * 9;
-    const T* xb = r + b * 9;
-    T sv = T(0);
-    for (int j = 0; j < 9; ++j) sv += row[j] * xb[j];
-    z[idx] = sv;
-    local += (double)sv * (double)r[idx];
-  }
-  sm[threadIdx.x] = local;
-  __syncthreads();
-  for (int st = kBlk / 2; st > 0; st >>= 1) {
-    if (threadIdx.x < st) sm[threadIdx.x] += sm[threadIdx.x + st];
-    __syncthreads();
-  }
-  if (threadIdx.x == 0) part[blockIdx.x] = sm[0];
-}
-
-// Fused PCG vector update incl. the two-deep backup rotation:
-// xBakPrev = xBak; xBak = x; x += alpha p; r -= alpha q.
-// (The rotation is a data move, not a pointer swap, so the kernel's
-// pointers stay stable across hipGraph replays.)
-template <typename T>
-__global__ void kUpdateXR(int64_t n, const double* __restrict__ alpha,
-                          const T* __restrict__ p, const T* __restrict__ q,
-                          T* __restrict__ x, T* __restrict__ xBak,
-                          T* __restrict__ xBakPrev, T* __restrict__ r) {
-  const T a = (T)(*alpha);
-  for (int64_t i = blockIdx.x * (int64_t)kBlk + threadIdx.x; i < n;
-       i += (int64_t)gridDim.x * kBlk) {
-    xBakPrev[i] = xBak[i];
-    const T xv = x[i];
-    xBak[i] = xv;
-    x[i] = xv + a * p[i];
-    r[i] -= a * q[i];
-  }
-}
-
-// B-apply (y = A x - y, the Schur S-apply tail) fused with the per-block
-// p^T q dot partials: saves the separate full pass over p and q per PCG
-// iteration (the reference used a standalone cublasDot, its :368-385).
-template <typename T>
-__global__ void kBApplyDot(int nBlk, const T* __restrict__ A,
-                           const T* __restrict__ x, T* __restrict__ y,
-                           double* part) {
-  __shared__ double sm[kBlk];
-  double local = 0.0;
-  for (int64_t idx = blockIdx.x * (int64_t)kBlk + threadIdx.x;
-       idx < (int64_t)nBlk * 9; idx += (int64_t)gridDim.x * kBlk) {
-    const int64_t b = idx / 9;
-    const int rrow = (int)(idx % 9);
-    const T* row = A + b * 81 + (int64_t)rrow * 9;
-    const T* xb = x + b * 9;
-    T s = T(0);
-    for (int j = 0; j < 9; ++j) s += row[j] * xb[j];
-    const T q = s - y[idx];
-    y[idx] = q;
-    local += (double)x[idx] * (double)q;
-  }
-  sm[threadIdx.x] = local;
-  __syncthreads();
-  for (int st = kBlk / 2; st > 0; st >>= 1) {
-    if (threadIdx.x < st) sm[threadIdx.x] += sm[threadIdx.x + st];
-    __syncthreads();
-  }
-  if (threadIdx.x == 0) part[blockIdx.x] = sm[0];
-}
-
-// Block-diagonal matvec, one thread per output row.
-// MODE 0: y = A x;  MODE 1: y = A x - y  (the reference's rw=1,dw=-1 gemv).
-template <typename T, int D, int MODE>
-__global__ void kBlockDiagMatVec(int nBlk, const T* __restrict__ A,
-                                 const T* __restrict__ x, T* __restrict__ y) {
-  for (int64_t idx = blockIdx.x * (int64_t)kBlk + threadIdx.x;
-       idx < (int64_t)nBlk * D; idx += (int64_t)gridDim.x * kBlk) {
-    const int64_t b = idx / D;
-    const int rrow = (int)(idx % D);
-    const T* row = A + b * D * D + (int64_t)rrow * D;
-    const T* xb = x + b * D;
-    T s = T(0);
-    for (int j = 0; j < D; ++j) s += row[j] * xb[j];
-    y[idx] = (MODE == 0) ? s : s - y[idx];
-  }
-}
-
-// ---------------------------------------------------------------------------
-// Small vector kernels
-// ---------------------------------------------------------------------------
-// Device-side pointer slots for the implicit J buffers: the double-buffered
-// accepted set flips on every LM accept, but the captured PCG graph freezes
-// kernel arguments — so the kernels dereference this slot array instead and
-// acceptForward rewrites it (kernel args carry the values; no host-buffer
-// lifetime to manage).  Slots: [0]=Jc, [1]=Jp, [2]=r (all accepted/bak).
-template <typename T>
-__global__ void kSetPtrSlots(const T** slots, const T* a, const T* b,
-                             const T* c) {
-  slots[0] = a;
-  slots[1] = b;
-  slots[2] = c;
-}
-
-__global__ void kDivScalar(double* out, const double* num, const double* den) {
-  *out = *num / *den;
-}
-__global__ void kSetScalar(double* out, double v) { *out = v; }
-__global__ void kCopyScalar(double* dst, const double* src) { *dst = *src; }
-// Fused reduction finals with the PCG scalar epilogues (fewer launches per
-// iteration; the scalars never leave the device except the rho readback).
-__global__ void kRedFinalRhoBeta(const double* part, int nb, double* rho,
-                                 const double* rhoPrev, double* beta) {
-  __shared__ double sm[kBlk];
-  double v = 0.0;
-  for (int i = threadIdx.x; i < nb; i += kBlk) v += part[i];
-  sm[threadIdx.x] = v;
-  __syncthreads();
-  for (int s = kBlk / 2; s > 0; s >>= 1) {
-    if (threadIdx.x < s) sm[threadIdx.x] += sm[threadIdx.x + s];
-    __syncthreads();
-  }
-  if (threadIdx.x == 0) {
-    *rho = sm[0];
-    // rhoPrev==0 only after an exactly-converged iteration under fixed-work
-    // mode; beta=0 (restart direction) instead of 0/0 = NaN.
-    *beta = *rhoPrev != 0.0 ? sm[0] / *rhoPrev : 0.0;
-  }
-}
-__global__ void kRedFinalAlpha(const double* part, int nb, const double* rho,
-                               double* alpha, double* rhoPrev) {
-  __shared__ double sm[kBlk];
-  double v = 0.0;
-  for (int i = threadIdx.x; i < nb; i += kBlk) v += part[i];
-  sm[threadIdx.x] = v;
-  __syncthreads();
-  for (int s = kBlk / 2; s > 0; s >>= 1) {
-    if (threadIdx.x < s) sm[threadIdx.x] += sm[threadIdx.x + s];
-    __syncthreads();
-  }
-  if (threadIdx.x == 0) {
-    // dot(p,q) can reach exactly 0 when PCG has converged but fixed-work
-    // mode (tol=0) forces further iterations; alpha=0 then leaves x/r
-    // unchanged instead of propagating NaN into deltaX.
-    *alpha = sm[0] != 0.0 ? *rho / sm[0] : 0.0;
-    *rhoPrev = *rho;
-  }
-}
-template <typename T>
-__global__ void kXpbyS(int64_t n, const T* __restrict__ x,
-                       const double* __restrict__ b, T* __restrict__ y) {
-  const T bv = (T)(*b);
-  for (int64_t i = blockIdx.x * (int64_t)kBlk + threadIdx.x; i < n;
-       i += (int64_t)gridDim.x * kBlk)
-    y[i] = x[i] + bv * y[i];
-}
-
-// r = v - q
-template <typename T>
-__global__ void kSub(int64_t n, const T* __restrict__ v, const T* __restrict__ q,
-                     T* __restrict__ r) {
-  for (int64_t i = blockIdx.x * (int64_t)kBlk + threadIdx.x; i < n;
-       i += (int64_t)gridDim.x * kBlk)
-    r[i] = v[i] - q[i];
-}
-// v = gc * s - v   (s = 1/worldSize pre-compensation, reference :478)
-template <typename T>
-__global__ void kVMake(int64_t n, const T* __restrict__ gc, T s, T* __restrict__ v) {
-  for (int64_t i = blockIdx.x * (int64_t)kBlk + threadIdx.x; i < n;
-       i += (int64_t)gridDim.x * kBlk)
-    v[i] = gc[i] * s - v[i];
-}
-template <typename T>
-__global__ void kAddAssign(int64_t n, const T* __restrict__ x, T* __restrict__ y) {
-  for (int64_t i = blockIdx.x * (int64_t)kBlk + threadIdx.x; i < n;
-       i += (int64_t)gridDim.x * kBlk)
-    y[i] += x[i];
-}
-template <typename T>
-__global__ void kZeroRange(T* p, int64_t n) {
-  for (int64_t i = blockIdx.x * (int64_t)kBlk + threadIdx.x; i < n;
-       i += (int64_t)gridDim.x * kBlk)
-    p[i] = T(0);
-}
-// deltaX_p = HllInv * (g_p - temp)  (local point shard)
-template <typename T>
-__global__ void kBackSub(int npt, const T* __restrict__ HllInv,
-                         const T* __restrict__ gp, const T* __restrict__ temp,
-                         T* __restrict__ dxp) {
-  for (int64_t p = blockIdx.x * (int64_t)kBlk + threadIdx.x; p < npt;
-       p += (int64_t)gridDim.x * kBlk) {
-    const T* inv = HllInv + p * 9;
-    T rhs[3];
-    for (int i = 0; i < 3; ++i) rhs[i] = gp[3 * p + i] - temp[3 * p + i];
-    for (int i = 0; i < 3; ++i)
-      dxp[3 * p + i] =
-          inv[i * 3] * rhs[0] + inv[i * 3 + 1] * rhs[1] + inv[i * 3 + 2] * rhs[2];
-  }
-}
-
-// ---------------------------------------------------------------------------
-// rho denominator:  sum over edges of (J dx + r)^2   (backup J/r)
-// ---------------------------------------------------------------------------
-template <typename T>
-__global__ void kRhoDenom(int64_t nL, const int* __restrict__ camOf,
-                          const int* __restrict__ ptOf, const T* __restrict__ r,
-                          const T* __restrict__ Jc, const T* __restrict__ Jp,
-                          const T* __restrict__ dxc, const T* __restrict__ dxp,
-                          double* acc, int lossKind, T lossD2) {
-  __shared__ double sm[kBlk];
-  double local = 0.0;
-  for (int64_t e = blockIdx.x * (int64_t)kBlk + threadIdx.x; e < nL;
-       e += (int64_t)gridDim.x * kBlk) {
-    const T* dc = dxc + (int64_t)camOf[e] * 9;
-    const T* dp = dxp + (int64_t)ptOf[e] * 3;
-    T acc2[2];
-    for (int row = 0; row < 2; ++row) {
-      T s = r[(int64_t)row * nL + e];
-      for (int k = 0; k < 9; ++k)
-        s += Jc[((int64_t)(k * 2 + row)) * nL + e] * dc[k];
-      for (int k = 0; k < 3; ++k)
-        s += Jp[((int64_t)(k * 2 + row)) * nL + e] * dp[k];
-      acc2[row] = s;
-    }
-    local += (double)lossRho(lossKind, lossD2,
-                             acc2[0] * acc2[0] + acc2[1] * acc2[1]);
-  }
-  sm[threadIdx.x] = local;
-  __syncthreads();
-  for (int s = kBlk / 2; s > 0; s >>= 1) {
-    if (threadIdx.x < s) sm[threadIdx.x] += sm[threadIdx.x + s];
-    __syncthreads();
-  }
-  if (threadIdx.x == 0) atomicAdd(acc, sm[0]);
-}
-
-}  // namespace
-
-// ===========================================================================
-// Host engine
-// ===========================================================================
-template <typename T>
-class GpuEngine final : public Engine<T> {
- public:
-  GpuEngine(const BAProblemHost& prob, const ProblemIndex& ix,
-            const ProblemOption& opt, const std::string& rcclId,
-            CustomForward<T> customForward, HostAllreduce<T> hostAllreduce,
-            HostAllreduce<double> hostAllreduceScalar)
-      : customFwd_(std::move(customForward)),
-        hostAr_(std::move(hostAllreduce)),
-        hostArD_(std::move(hostAllreduceScalar)),
-        rank_(opt.rank),
-        world_(opt.worldSize),
-        ncam_(ix.ncam),
-        npt_(ix.npt),
-        analytical_(opt.diff == DiffMode::ANALYTICAL),
-        implicit_(opt.schur == SchurMode::IMPLICIT),
-        lossKind_((int)opt.loss),
-        lossD2_((T)(opt.lossDelta * opt.lossDelta)) {
-    HIP_CHECK(hipSetDevice(opt.deviceIndex));
-    HIP_CHECK(hipStreamCreate(&stream_));
-    e0_ = ix.split[rank_];
-    e1_ = ix.split[rank_ + 1];
-    nL_ = e1_ - e0_;
-    ptLo_ = ix.ptSplit[rank_];
-    ptHi_ = ix.ptSplit[rank_ + 1];
-    npL_ = ptHi_ - ptLo_;
-    nc_ = (int64_t)ncam_ * 9;
-    np_ = (int64_t)npt_ * 3;
-    dim_ = nc_ + np_;
-    hasInfo_ = !ix.infoSorted.empty();
-
-    if (world_ > 1) {
-      if (rcclId.empty() && hostAr_) {
-        // gloo-backed testing fallback (several ranks may share one GPU)
-      } else {
-        MEGBA_CHECK(rcclId.size() == sizeof(ncclUniqueId),
-                    "worldSize>1 requires the RCCL unique id (or a host "
-                    "allreduce fallback)");
-        ncclUniqueId id;
-        std::memcpy(&id, rcclId.data(), sizeof(id));
-        RCCL_CHECK(ncclCommInitRank(&comm_, world_, id, rank_));
-        hasComm_ = true;
-      }
-    } else if (getenv("MEGBA_FORCE_RCCL")) {
-      // 1-GPU hardening mode: run a real world-1 RCCL communicator so
-      // ncclCommInitRank, every ncclAllReduce call site and RCCL-under-
-      // hipGraph-capture are exercised on a single GPU (the multi-rank
-      // collective SEQUENCE is identical; only the ring is trivial).
-      ncclUniqueId id;
-      if (rcclId.size() == sizeof(id))
-        std::memcpy(&id, rcclId.data(), sizeof(id));
-      else
-        RCCL_CHECK(ncclGetUniqueId(&id));
-      RCCL_CHECK(ncclCommInitRank(&comm_, 1, id, 0));
-      hasComm_ = true;
-    }
-
-    // Static per-edge data (primary = (pt,cam)-sorted order).
-    dCamOf_ = dalloc<int>(nL_);
-    dPtOf_ = dalloc<int>(nL_);
-    up(dCamOf_, ix.camOf.data() + e0_, nL_);
-    up(dPtOf_, ix.ptOf.data() + e0_, nL_);
-    if (!prob.camFixed.empty()) {
-      dCamFixed_ = dalloc<unsigned char>(ncam_);
-      up(dCamFixed_, prob.camFixed.data(), ncam_);
-    }
-    if (!prob.ptFixed.empty()) {
-      dPtFixed_ = dalloc<unsigned char>(npt_);
-      up(dPtFixed_, prob.ptFixed.data(), npt_);
-    }
-    dMeas_ = dalloc<T>(nL_ * 2);
-    upCast(dMeas_, ix.measSorted.data() + 2 * e0_, nL_ * 2);
-    if (customFwd_) {
-      dLeaf_ = dalloc<T>(nL_ * 12);
-      dMeasSplit_ = dalloc<T>(nL_ * 2);
-      std::vector<T> ms(nL_ * 2);
-      for (int64_t e = 0; e < nL_; ++e) {
-        ms[e] = (T)ix.measSorted[2 * (e0_ + e)];
-        ms[nL_ + e] = (T)ix.measSorted[2 * (e0_ + e) + 1];
-      }
-      up(dMeasSplit_, ms.data(), nL_ * 2);
-    }
-    if (hasInfo_) {
-      dInfo_ = dalloc<T>(nL_ * 3);
-      upCast(dInfo_, ix.infoSorted.data() + 3 * e0_, nL_ * 3);
-    }
-
-    // Parameters (cameras replicated; this rank maintains its point shard).
-    dParams_ = dalloc<T>(dim_);
-    dParamsBak_ = dalloc<T>(dim_);
-    {
-      std::vector<T> h(dim_);
-      for (int64_t i = 0; i < nc_; ++i) h[i] = (T)prob.cams[i];
-      for (int64_t i = 0; i < np_; ++i) h[nc_ + i] = (T)prob.pts[i];
-      up(dParams_, h.data(), dim_);
-      HIP_CHECK(hipMemcpyAsync(dParamsBak_, dParams_, dim_ * sizeof(T),
-                               hipMemcpyDeviceToDevice, stream_));
-    }
-
-    // Residual / Jacobian double-buffer (current + accepted).
-    for (int s = 0; s < 2; ++s) {
-      dR_[s] = dalloc<T>(nL_ * 2);
-      dJc_[s] = dalloc<T>(nL_ * 18);
-      dJp_[s] = dalloc<T>(nL_ * 6);
-    }
-
-    // Linear system (full-size global indexing; only this rank's point
-    // ranges of Hll/g_p/temp are maintained).
-    dHpp_ = dalloc<T>((int64_t)ncam_ * 81);
-    dHll_ = dalloc<T>((int64_t)npt_ * 9);
-    dG_ = dalloc<T>(dim_);
-    dGBak_ = dalloc<T>(dim_);
-    dHppD_ = dalloc<T>((int64_t)ncam_ * 81);
-    dHllD_ = dalloc<T>((int64_t)npt_ * 9);
-    dHppInv_ = dalloc<T>((int64_t)ncam_ * 81);
-    dHllInv_ = dalloc<T>((int64_t)npt_ * 9);
-    dDeltaX_ = dalloc<T>(dim_);
-    dDeltaXBak_ = dalloc<T>(dim_);
-    HIP_CHECK(hipMemsetAsync(dDeltaX_, 0, dim_ * sizeof(T), stream_));
-    HIP_CHECK(hipMemsetAsync(dDeltaXBak_, 0, dim_ * sizeof(T), stream_));
-
-    // PCG workspace.
-    dP_ = dalloc<T>(nc_);
-    dRr_ = dalloc<T>(nc_);
-    dZ_ = dalloc<T>(nc_);
-    dQ_ = dalloc<T>(nc_);
-    dV_ = dalloc<T>(nc_);
-    dXBak_ = dalloc<T>(nc_);
-    dXBakPrev_ = dalloc<T>(nc_);
-    HIP_CHECK(hipMemsetAsync(dXBak_, 0, nc_ * sizeof(T), stream_));
-    HIP_CHECK(hipMemsetAsync(dXBakPrev_, 0, nc_ * sizeof(T), stream_));
-    dW_ = dalloc<T>(np_);
-    dTemp_ = dalloc<T>(np_);
-    HIP_CHECK(hipMemsetAsync(dW_, 0, np_ * sizeof(T), stream_));
-
-    // Partial-sum scratch: big enough for both the fixed-shape two-pass
-    // reductions (kRedBlocks) and the fused B-apply+dot grid over nc_.
-    partCap_ = kRedBlocks > gridFor(nc_) ? kRedBlocks : gridFor(nc_);
-    dPart_ = dalloc<double>(partCap_ + 8);
-    dFail_ = dalloc<int>(2);
-    HIP_CHECK(hipHostMalloc((void**)&hScalar_, sizeof(double)));
-
-    // Cam-sorted view of the local edges: slab positions + chunk table.
-    {
-      std::vector<int> camPos(nL_), rowPtr(ncam_ + 1, 0), ptOfCam(nL_);
-      for (int64_t e = 0; e < nL_; ++e) rowPtr[ix.camOf[e0_ + e] + 1]++;
-      for (int c = 0; c < ncam_; ++c) rowPtr[c + 1] += rowPtr[c];
-      std::vector<int> cursor(rowPtr.begin(), rowPtr.end() - 1);
-      for (int64_t e = 0; e < nL_; ++e) {
-        const int pos = cursor[ix.camOf[e0_ + e]]++;
-        camPos[e] = pos;
-        ptOfCam[pos] = ix.ptOf[e0_ + e];
-      }
-      dCamPos_ = dalloc<int>(nL_);
-      dPtOfCam_ = dalloc<int>(nL_);
-      up(dCamPos_, camPos.data(), nL_);
-      up(dPtOfCam_, ptOfCam.data(), nL_);
-      std::vector<int> cCam, cLo, cHi;
-      int CHUNK = 256;  // cam-chunk rows (tunable: MEGBA_CHUNK)
-      if (const char* c = getenv("MEGBA_CHUNK")) CHUNK = std::atoi(c);
-      for (int c = 0; c < ncam_; ++c)
-        for (int s = rowPtr[c]; s < rowPtr[c + 1]; s += CHUNK) {
-          cCam.push_back(c);
-          cLo.push_back(s);
-          cHi.push_back(std::min(s + CHUNK, rowPtr[c + 1]));
-        }
-      nChunks_ = (int)cCam.size();
-      dChCam_ = dalloc<int>(nChunks_);
-      dChLo_ = dalloc<int>(nChunks_);
-      dChHi_ = dalloc<int>(nChunks_);
-      up(dChCam_, cCam.data(), nChunks_);
-      up(dChLo_, cLo.data(), nChunks_);
-      up(dChHi_, cHi.data(), nChunks_);
-    }
-    dSlab_ = dalloc<T>(nL_ * slabWidth());
-    dJSlots_ = dalloc<const T*>(3);
-    updateJSlots();
-    if (implicit_) {
-      dJcCam_ = dalloc<T>(nL_ * 18);
-      dJpCam_ = dalloc<T>(nL_ * 6);
-    }
-    if (!implicit_) {
-      dHpl_ = dalloc<T>(nL_ * 27);
-      dHplCam_ = dalloc<T>(nL_ * 27);
-    }
-    sync();
-  }
-
-  ~GpuEngine() override {
-    if (hScalar_) (void)hipHostFree(hScalar_);
-    if (pcgGraphExec_) (void)hipGraphExecDestroy(pcgGraphExec_);
-    for (void* p : allocs_) (void)hipFree(p);
-    if (hasComm_) (void)ncclCommDestroy(comm_);
-    (void)hipStreamDestroy(stream_);
-  }
-
-  double forward() override {
-    freshCur_ = true;
-    if (customFwd_) return forwardCustom();
-    zeroScalar();
-    if (analytical_)
-      hipLaunchKernelGGL(kForwardAnalytical<T>, dim3(gridFor(nL_)), dim3(kBlk),
-                         0, stream_, nL_, dCamOf_, dPtOf_, dParams_, ncam_,
-                         dMeas_, dCamFixed_, dPtFixed_, dR_[cur_], dJc_[cur_],
-                         dJp_[cur_], scalarPtr(), lossKind_, lossD2_);
-    else
-      hipLaunchKernelGGL(kForward<T>, dim3(gridFor(nL_ * 4)), dim3(kBlk), 0,
-                         stream_, nL_, dCamOf_, dPtOf_, dParams_, ncam_,
-                         dMeas_, dCamFixed_, dPtFixed_, dR_[cur_], dJc_[cur_],
-                         dJp_[cur_], scalarPtr(), lossKind_, lossD2_);
-    return globalScalar(ncclSum);
-  }
-
-  void acceptForward() override {
-    cur_ ^= 1;          // accepted set = dX_[cur_^1]
-    freshCur_ = false;  // the (new) current buffers are not yet written
-    updateJSlots();
-  }
-
-  // Point the device slots at the accepted (bak) buffer set.
-  void updateJSlots() {
-    const int bak = cur_ ^ 1;
-    hipLaunchKernelGGL(kSetPtrSlots<T>, dim3(1), dim3(1), 0, stream_,
-                       dJSlots_, dJc_[bak], dJp_[bak], dR_[bak]);
-  }
-
-  double forwardCustom() {
-    hipLaunchKernelGGL(kGatherLeaves<T>, dim3(gridFor(nL_)), dim3(kBlk), 0,
-                       stream_, nL_, dCamOf_, dPtOf_, dParams_, ncam_, dLeaf_);
-    sync();  // JetVector ops run on the null stream
-    std::vector<JetVec<T>> camL, ptL, ms, res;
-    for (int k = 0; k < 9; ++k)
-      camL.push_back(jvView<T>(dLeaf_ + (int64_t)k * nL_, nL_, 12, k, true));
-    for (int k = 0; k < 3; ++k)
-      ptL.push_back(
-          jvView<T>(dLeaf_ + (int64_t)(9 + k) * nL_, nL_, 12, 9 + k, true));
-    for (int r = 0; r < 2; ++r)
-      ms.push_back(jvView<T>(dMeasSplit_ + (int64_t)r * nL_, nL_, 12, -1, true));
-    customFwd_(camL, ptL, ms, res);
-    MEGBA_CHECK(res.size() == 2, "custom forward must return 2 residuals");
-    HIP_CHECK(hipDeviceSynchronize());
-    for (int r = 0; r < 2; ++r) {
-      MEGBA_CHECK(res[r].kind() == JvKind::DENSE && res[r].nItem == nL_ &&
-                      res[r].N == 12 && res[r].onGpu,
-                  "custom residual must be a dense GPU JetVector (N=12)");
-      hipLaunchKernelGGL(kRepackRes<T>, dim3(gridFor(nL_)), dim3(kBlk), 0,
-                         stream_, nL_, r, res[r].value->ptr, res[r].grad->ptr,
-                         dCamOf_, dPtOf_, dCamFixed_, dPtFixed_, dR_[cur_],
-                         dJc_[cur_], dJp_[cur_]);
-    }
-    if (lossKind_) {
-      zeroScalar();
-      hipLaunchKernelGGL(kChi2Loss<T>, dim3(gridFor(nL_)), dim3(kBlk), 0,
-                         stream_, nL_, dR_[cur_], lossKind_, lossD2_,
-                         scalarPtr());
-    } else {
-      reduceDetAsync(dR_[cur_], dR_[cur_], nL_ * 2, ROp::SumSq, scalarPtr());
-    }
-    return globalScalar(ncclSum);
-  }
-
-  void buildLinearSystem() override {
-    const int bak = cur_ ^ 1;
-    HIP_CHECK(hipMemsetAsync(dHpp_, 0, (int64_t)ncam_ * 81 * sizeof(T), stream_));
-    HIP_CHECK(hipMemsetAsync(dHll_ + (int64_t)ptLo_ * 9, 0,
-                             (int64_t)npL_ * 9 * sizeof(T), stream_));
-    HIP_CHECK(hipMemsetAsync(dG_, 0, nc_ * sizeof(T), stream_));
-    HIP_CHECK(hipMemsetAsync(dG_ + nc_ + (int64_t)ptLo_ * 3, 0,
-                             (int64_t)npL_ * 3 * sizeof(T), stream_));
-    dispatchAssemble(bak);
-    if (!implicit_) {
-      if (weighted())
-        hipLaunchKernelGGL((kFinalizeCam<T, true>), dim3(gridFor(nL_)),
-                           dim3(kBlk), 0, stream_, nL_, dSlab_, dHplCam_);
-      else
-        hipLaunchKernelGGL((kFinalizeCam<T, false>), dim3(gridFor(nL_)),
-                           dim3(kBlk), 0, stream_, nL_, dSlab_, dHplCam_);
-    } else {
-      if (weighted())
-        hipLaunchKernelGGL((kFinalizeCamImp<T, true>), dim3(gridFor(nL_)),
-                           dim3(kBlk), 0, stream_, nL_, dSlab_, dJcCam_,
-                           dJpCam_);
-      else
-        hipLaunchKernelGGL((kFinalizeCamImp<T, false>), dim3(gridFor(nL_)),
-                           dim3(kBlk), 0, stream_, nL_, dSlab_, dJcCam_,
-                           dJpCam_);
-    }
-    // Only the small camera-side quantities cross ranks (the reference
-    // allreduced Hpp, Hll AND g, its site A1).
-    allreduce(dHpp_, (int64_t)ncam_ * 81, ncclSum);
-    allreduce(dG_, nc_, ncclSum);
-    sync();
-  }
-
-  void backupParams() override {
-    HIP_CHECK(hipMemcpyAsync(dParamsBak_, dParams_, dim_ * sizeof(T),
-                             hipMemcpyDeviceToDevice, stream_));
-  }
-  void rollbackParams() override {
-    HIP_CHECK(hipMemcpyAsync(dParams_, dParamsBak_, dim_ * sizeof(T),
-                             hipMemcpyDeviceToDevice, stream_));
-  }
-  void backupGDx() override {
-    HIP_CHECK(hipMemcpyAsync(dDeltaXBak_, dDeltaX_, dim_ * sizeof(T),
-                             hipMemcpyDeviceToDevice, stream_));
-    HIP_CHECK(hipMemcpyAsync(dGBak_, dG_, dim_ * sizeof(T),
-                             hipMemcpyDeviceToDevice, stream_));
-  }
-  void rollbackGDx() override {
-    HIP_CHECK(hipMemcpyAsync(dDeltaX_, dDeltaXBak_, dim_ * sizeof(T),
-                             hipMemcpyDeviceToDevice, stream_));
-    HIP_CHECK(hipMemcpyAsync(dG_, dGBak_, dim_ * sizeof(T),
-                             hipMemcpyDeviceToDevice, stream_));
-  }
-
-  void processDiag(double region) override {
-    const T f = T(1) + T(1) / (T)region;
-    hipLaunchKernelGGL((kDamp<T, 9>), dim3(gridFor((int64_t)ncam_ * 81)),
-                       dim3(kBlk), 0, stream_, (int64_t)ncam_ * 81, dHpp_,
-                       dHppD_, f, dCamFixed_);
-    hipLaunchKernelGGL((kDamp<T, 3>), dim3(gridFor((int64_t)npL_ * 9)),
-                       dim3(kBlk), 0, stream_, (int64_t)npL_ * 9,
-                       dHll_ + (int64_t)ptLo_ * 9, dHllD_ + (int64_t)ptLo_ * 9,
-                       f, dPtFixed_ ? dPtFixed_ + ptLo_ : nullptr);
-  }
-
-  int solveLinear(const SolverOptionPCG& opt) override {
-    // Block inverses (preconditioner replicated; Cinv on the local shard).
-    HIP_CHECK(hipMemsetAsync(dFail_, 0, 2 * sizeof(int), stream_));
-    hipLaunchKernelGGL((kInvert<T, 9>), dim3(gridFor(ncam_)), dim3(kBlk), 0,
-                       stream_, ncam_, dHppD_, dHppInv_, dFail_);
-    hipLaunchKernelGGL((kInvert<T, 3>), dim3(gridFor(npL_)), dim3(kBlk), 0,
-                       stream_, npL_, dHllD_ + (int64_t)ptLo_ * 9,
-                       dHllInv_ + (int64_t)ptLo_ * 9, dFail_);
-    int fail[2] = {0, 0};
-    HIP_CHECK(hipMemcpyAsync(fail, dFail_, 2 * sizeof(int),
-                             hipMemcpyDeviceToHost, stream_));
-    sync();
-    if (fail[0]) {
-      hipLaunchKernelGGL((kInvertJitter<T, 9>), dim3(gridFor(ncam_)),
-                         dim3(kBlk), 0, stream_, ncam_, dHppD_, dHppInv_,
-                         dFail_);
-      hipLaunchKernelGGL((kInvertJitter<T, 3>), dim3(gridFor(npL_)),
-                         dim3(kBlk), 0, stream_, npL_,
-                         dHllD_ + (int64_t)ptLo_ * 9,
-                         dHllInv_ + (int64_t)ptLo_ * 9, dFail_);
-      HIP_CHECK(hipMemcpyAsync(fail, dFail_, 2 * sizeof(int),
-                               hipMemcpyDeviceToHost, stream_));
-      sync();
-      MEGBA_CHECK(!fail[1], "singular Hessian block");
-    }
-
-    const T* gc = dG_;
-    const T* gp = dG_ + nc_;
-    // v = gc/world - E Cinv gp  (partial, then the 9*ncam allreduce)
-    applyCinv(gp, dW_);
-    spmvEx(dW_, dV_);
-    hipLaunchKernelGGL(kVMake<T>, dim3(gridFor(nc_)), dim3(kBlk), 0, stream_,
-                       nc_, gc, T(1) / T(world_), dV_);
-    allreduce(dV_, nc_, ncclSum);
-    // Warm start: x = deltaX camera part (in place in dDeltaX_).
-    T* x = dDeltaX_;
-    schurApply(x, dQ_);
-    hipLaunchKernelGGL(kSub<T>, dim3(gridFor(nc_)), dim3(kBlk), 0, stream_, nc_,
-                       dV_, dQ_, dRr_);
-
-    // The loop body is value-uniform (beta/alpha live on-device; the first
-    // iteration gets beta = rho/INF = 0 against a zeroed p), so it is
-    // captured once as a hipGraph and replayed with ONE host interaction per
-    // iteration: the rho readback that drives the reference's refuse/tol
-    // control flow.  Refuse semantics are preserved with a two-deep x backup
-    // (the body runs speculatively; on refuse x is restored to the value two
-    // updates back, exactly what the reference's pre-update check restores).
-    HIP_CHECK(hipMemsetAsync(dP_, 0, nc_ * sizeof(T), stream_));
-    hipLaunchKernelGGL(kSetScalar, dim3(1), dim3(1), 0, stream_, slotRhoPrev(),
-                       INFINITY);
-    int n = 0;
-    double rho = 0.0, rhoMin = INFINITY;
-    bool done = false;
-    ensurePcgGraph();
-    // Fixed-work mode (tol<=0 with refuse disabled, the bench contract):
-    // rho steers nothing, so skip the per-iteration host readback and
-    // enqueue all maxIter bodies back-to-back — ONE host sync per solve
-    // instead of one per iteration (the N=8 constant-term killer).
-    const bool fixedWork = opt.tol <= 0.0 && opt.refuseRatio >= 1e29;
-    if (fixedWork) {
-      for (; n < opt.maxIter; ++n) {
-        if (pcgGraphExec_)
-          HIP_CHECK(hipGraphLaunch(pcgGraphExec_, stream_));
-        else
-          pcgBody();
-      }
-    } else {
-      while (!done && n < opt.maxIter) {
-        if (pcgGraphExec_) {
-          HIP_CHECK(hipGraphLaunch(pcgGraphExec_, stream_));
-        } else {
-          pcgBody();
-        }
-        rho = readScalar(slotRho());
-        if (rho > opt.refuseRatio * rhoMin) {
-          HIP_CHECK(hipMemcpyAsync(x, dXBakPrev_, nc_ * sizeof(T),
-                                   hipMemcpyDeviceToDevice, stream_));
-          break;
-        }
-        rhoMin = rhoMin < rho ? rhoMin : rho;
-        ++n;
-        done = std::abs(rho) < opt.tol;
-      }
-    }
-    // Back-substitution: deltaX_p = Cinv (g_p - E^T x), fully local.
-    spmvEtx(x, dTemp_);
-    hipLaunchKernelGGL(kBackSub<T>, dim3(gridFor(npL_)), dim3(kBlk), 0, stream_,
-                       npL_, dHllInv_ + (int64_t)ptLo_ * 9,
-                       gp + (int64_t)ptLo_ * 3, dTemp_ + (int64_t)ptLo_ * 3,
-                       dDeltaX_ + nc_ + (int64_t)ptLo_ * 3);
-    sync();
-    return n;
-  }
-
-  double deltaXL2() override {
-    const double camSS = reduceDet(dDeltaX_, dDeltaX_, nc_, ROp::SumSq);
-    reduceDetAsync(dDeltaX_ + nc_ + (int64_t)ptLo_ * 3,
-                   dDeltaX_ + nc_ + (int64_t)ptLo_ * 3, (int64_t)npL_ * 3,
-                   ROp::SumSq, scalarPtr());
-    return std::sqrt(camSS + globalScalar(ncclSum));
-  }
-  double xL2() override {
-    const double camSS = reduceDet(dParams_, dParams_, nc_, ROp::SumSq);
-    reduceDetAsync(dParams_ + nc_ + (int64_t)ptLo_ * 3,
-                   dParams_ + nc_ + (int64_t)ptLo_ * 3, (int64_t)npL_ * 3,
-                   ROp::SumSq, scalarPtr());
-    return std::sqrt(camSS + globalScalar(ncclSum));
-  }
-  double gInf() override {
-    const double camMax = reduceDet(dG_, dG_, nc_, ROp::AbsMax);
-    reduceDetAsync(dG_ + nc_ + (int64_t)ptLo_ * 3,
-                   dG_ + nc_ + (int64_t)ptLo_ * 3, (int64_t)npL_ * 3,
-                   ROp::AbsMax, scalarPtr());
-    const double ptMax = globalScalar(ncclMax);
-    return camMax > ptMax ? camMax : ptMax;
-  }
-
-  void updateParams() override {
-    hipLaunchKernelGGL(kAddAssign<T>, dim3(gridFor(nc_)), dim3(kBlk), 0,
-                       stream_, nc_, dDeltaX_, dParams_);
-    hipLaunchKernelGGL(kAddAssign<T>, dim3(gridFor((int64_t)npL_ * 3)),
-                       dim3(kBlk), 0, stream_, (int64_t)npL_ * 3,
-                       dDeltaX_ + nc_ + (int64_t)ptLo_ * 3,
-                       dParams_ + nc_ + (int64_t)ptLo_ * 3);
-  }
-
-  double rhoDenominator(double chi2Backup) override {
-    const int bak = cur_ ^ 1;
-    zeroScalar();
-    hipLaunchKernelGGL(kRhoDenom<T>, dim3(gridFor(nL_)), dim3(kBlk), 0, stream_,
-                       nL_, dCamOf_, dPtOf_, dR_[bak], dJc_[bak], dJp_[bak],
-                       dDeltaX_, dDeltaX_ + nc_, scalarPtr(), lossKind_,
-                       lossD2_);
-    return globalScalar(ncclSum) - chi2Backup;
-  }
-
-  void getParams(double* cams, double* pts) override {
-    std::vector<T> hc(nc_);
-    HIP_CHECK(hipMemcpy(hc.data(), dParams_, nc_ * sizeof(T),
-                        hipMemcpyDeviceToHost));
-    for (int64_t i = 0; i < nc_; ++i) cams[i] = (double)hc[i];
-    if (world_ > 1 && (hasComm_ || hostAr_)) {
-      // point shards: zero the non-local entries, allreduce-sum.
-      if (!dPtMerge_) dPtMerge_ = dalloc<T>(np_);
-      T* tmp = dPtMerge_;
-      HIP_CHECK(hipMemsetAsync(tmp, 0, np_ * sizeof(T), stream_));
-      HIP_CHECK(hipMemcpyAsync(tmp + (int64_t)ptLo_ * 3,
-                               dParams_ + nc_ + (int64_t)ptLo_ * 3,
-                               (int64_t)npL_ * 3 * sizeof(T),
-                               hipMemcpyDeviceToDevice, stream_));
-      allreduce(tmp, np_, ncclSum);
-      sync();
-      std::vector<T> h(np_);
-      HIP_CHECK(hipMemcpy(h.data(), tmp, np_ * sizeof(T), hipMemcpyDeviceToHost));
-      for (int64_t i = 0; i < np_; ++i) pts[i] = (double)h[i];
-      return;
-    }
-    std::vector<T> h(np_);
-    HIP_CHECK(hipMemcpy(h.data(), dParams_ + nc_, np_ * sizeof(T),
-                        hipMemcpyDeviceToHost));
-    for (int64_t i = 0; i < np_; ++i) pts[i] = (double)h[i];
-  }
-
-  DenseDump dump() const override {
-    DenseDump d;
-    d.e0 = e0_;
-    d.e1 = e1_;
-    // r/J of the LAST forward() (survives the acceptForward buffer swap)
-    const int fw = freshCur_ ? cur_ : (cur_ ^ 1);
-    d.r = down(dR_[fw], nL_ * 2);
-    d.Jc = down(dJc_[fw], nL_ * 18);
-    d.Jp = down(dJp_[fw], nL_ * 6);
-    d.Hpp = down(dHpp_, (int64_t)ncam_ * 81);
-    d.Hll = down(dHll_, (int64_t)npt_ * 9);
-    if (!implicit_) {
-      // device Hpl is grad-major [27][nL]; dump as [e][9][3]
-      std::vector<double> gm = down(dHpl_, nL_ * 27);
-      std::vector<double> o(gm.size());
-      for (int64_t e = 0; e < nL_; ++e)
-        for (int k = 0; k < 27; ++k) o[e * 27 + k] = gm[(int64_t)k * nL_ + e];
-      d.Hpl = o;
-    }
-    d.g = down(dG_, dim_);
-    d.deltaX = down(dDeltaX_, dim_);
-    // Dump layouts match the CPU engine: r [e][2], Jc [e][2][9], Jp [e][2][3]
-    // (the device layout is gradient-major; transpose here).
-    d.r = transposeR(d.r);
-    d.Jc = transposeJ(d.Jc, 9);
-    d.Jp = transposeJ(d.Jp, 6);
-    return d;
-  }
-
- private:
-  bool weighted() const { return hasInfo_ || lossKind_ != 0; }
-  int slabWidth() const {
-    // must match SlabLayout<EXPL, weighted()>::SW
-    return (implicit_ ? 0 : 27) + 20 + (weighted() ? 18 : 0) +
-           (implicit_ ? 6 : 0);
-  }
-  void dispatchAssemble(int bak) {
-    auto launch = [&](auto weightedTag, auto explTag) {
-      constexpr bool HI = decltype(weightedTag)::value;
-      constexpr bool EX = decltype(explTag)::value;
-      hipLaunchKernelGGL((kAssembleEdge<T, HI, EX>), dim3(gridFor(nL_)),
-                         dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
-                         dR_[bak], dJc_[bak], dJp_[bak], dInfo_, dHll_, dHpl_,
-                         dG_, ncam_, dCamPos_, dSlab_, lossKind_, lossD2_);
-      if (nChunks_ > 0)
-        hipLaunchKernelGGL((kAssembleCam<T, HI, EX>), dim3(nChunks_), dim3(128),
-                           0, stream_, nChunks_, dChCam_, dChLo_, dChHi_,
-                           dSlab_, dHpp_, dG_);
-    };
-    using TrueT = std::integral_constant<bool, true>;
-    using FalseT = std::integral_constant<bool, false>;
-    if (weighted() && !implicit_) launch(TrueT{}, TrueT{});
-    else if (weighted() && implicit_) launch(TrueT{}, FalseT{});
-    else if (!weighted() && !implicit_) launch(FalseT{}, TrueT{});
-    else launch(FalseT{}, FalseT{});
-  }
-  template <typename U>
-  U* dalloc(int64_t n) {
-    void* p = nullptr;
-    HIP_CHECK(hipMalloc(&p, (n > 0 ? n : 1) * sizeof(U)));
-    allocs_.push_back(p);
-    return (U*)p;
-  }
-  template <typename U>
-  void up(U* dst, const U* src, int64_t n) {
-    if (n > 0)
-      HIP_CHECK(hipMemcpyAsync(dst, src, n * sizeof(U), hipMemcpyHostToDevice,
-                               stream_));
-  }
-  void upCast(T* dst, const double* src, int64_t n) {
-    std::vector<T> h(n);
-    for (int64_t i = 0; i < n; ++i) h[i] = (T)src[i];
-    if (n > 0)
-      HIP_CHECK(hipMemcpy(dst, h.data(), n * sizeof(T), hipMemcpyHostToDevice));
-  }
-  std::vector<double> down(const T* src, int64_t n) const {
-    std::vector<T> h(n);
-    HIP_CHECK(hipMemcpy(h.data(), (void*)src, n * sizeof(T), hipMemcpyDeviceToHost));
-    return std::vector<double>(h.begin(), h.end());
-  }
-  std::vector<double> transposeR(const std::vector<double>& v) const {
-    std::vector<double> o(v.size());
-    for (int64_t e = 0; e < nL_; ++e)
-      for (int row = 0; row < 2; ++row) o[2 * e + row] = v[row * nL_ + e];
-    return o;
-  }
-  std::vector<double> transposeJ(const std::vector<double>& v, int w) const {
-    const int cols = w == 9 ? 9 : 3;
-    std::vector<double> o(v.size());
-    for (int64_t e = 0; e < nL_; ++e)
-      for (int col = 0; col < cols; ++col)
-        for (int row = 0; row < 2; ++row)
-          o[e * cols * 2 + row * cols + col] = v[((int64_t)(col * 2 + row)) * nL_ + e];
-    return o;
-  }
-  void sync() { HIP_CHECK(hipStreamSynchronize(stream_)); }
-  double* scalarPtr() { return dPart_ + partCap_; }
-  double* slotRho() { return dPart_ + partCap_ + 1; }
-  double* slotPq() { return dPart_ + partCap_ + 2; }
-  double* slotAlpha() { return dPart_ + partCap_ + 3; }
-  double* slotRhoPrev() { return dPart_ + partCap_ + 4; }
-  double* slotBeta() { return dPart_ + partCap_ + 5; }
-  void zeroScalar() {
-    HIP_CHECK(hipMemsetAsync(scalarPtr(), 0, sizeof(double), stream_));
-  }
-  // Read the device scalar accumulator, allreducing across ranks first.
-  double globalScalar(ncclRedOp_t op) {
-    if (hasComm_) {
-      RCCL_CHECK(ncclAllReduce(scalarPtr(), scalarPtr(), 1, ncclDouble, op,
-                               comm_, stream_));
-      return readScalar(scalarPtr());
-    }
-    if ((hostArD_ || hostAr_) && world_ > 1) {
-      // Host-callback fallback: reduce in full double (the device
-      // accumulator already is double) so fp32 multi-rank runs don't lose
-      // precision in chi2/deltaXL2/gInf.
-      double h = readScalar(scalarPtr());
-      if (hostArD_) {
-        hostArD_(&h, 1, op == ncclMax ? 'm' : 's');
-        return h;
-      }
-      T v = (T)h;
-      hostAr_(&v, 1, op == ncclMax ? 'm' : 's');
-      return (double)v;
-    }
-    return readScalar(scalarPtr());
-  }
-  double readScalar(double* dptr) {
-    // pinned-host destination: the PCG loop does one of these per iteration
-    HIP_CHECK(hipMemcpyAsync(hScalar_, dptr, sizeof(double),
-                             hipMemcpyDeviceToHost, stream_));
-    sync();
-    return *hScalar_;
-  }
-  void allreduce(T* buf, int64_t n, ncclRedOp_t op) {
-    if (n == 0 || (world_ == 1 && !hasComm_)) return;
-    if (hasComm_) {
-      RCCL_CHECK(ncclAllReduce(buf, buf, n,
-                               sizeof(T) == 8 ? ncclDouble : ncclFloat, op,
-                               comm_, stream_));
-      return;
-    }
-    if (!hostAr_) return;
-    // gloo-backed testing fallback: bounce through the host.
-    sync();
-    std::vector<T> h(n);
-    HIP_CHECK(hipMemcpy(h.data(), buf, n * sizeof(T), hipMemcpyDeviceToHost));
-    hostAr_(h.data(), n, op == ncclMax ? 'm' : 's');
-    HIP_CHECK(hipMemcpy(buf, h.data(), n * sizeof(T), hipMemcpyHostToDevice));
-  }
-  void reduceDetAsync(const T* a, const T* b, int64_t n, ROp op, double* out) {
-    switch (op) {
-      case ROp::Dot:
-        hipLaunchKernelGGL((kRedPartial<T, ROp::Dot>), dim3(kRedBlocks),
-                           dim3(kBlk), 0, stream_, a, b, n, dPart_);
-        hipLaunchKernelGGL((kRedFinal<ROp::Dot>), dim3(1), dim3(kBlk), 0,
-                           stream_, dPart_, kRedBlocks, out);
-        break;
-      case ROp::SumSq:
-        hipLaunchKernelGGL((kRedPartial<T, ROp::SumSq>), dim3(kRedBlocks),
-                           dim3(kBlk), 0, stream_, a, b, n, dPart_);
-        hipLaunchKernelGGL((kRedFinal<ROp::SumSq>), dim3(1), dim3(kBlk), 0,
-                           stream_, dPart_, kRedBlocks, out);
-        break;
-      case ROp::AbsMax:
-        hipLaunchKernelGGL((kRedPartial<T, ROp::AbsMax>), dim3(kRedBlocks),
-                           dim3(kBlk), 0, stream_, a, b, n, dPart_);
-        hipLaunchKernelGGL((kRedFinal<ROp::AbsMax>), dim3(1), dim3(kBlk), 0,
-                           stream_, dPart_, kRedBlocks, out);
-        break;
-    }
-  }
-  double reduceDet(const T* a, const T* b, int64_t n, ROp op) {
-    reduceDetAsync(a, b, n, op, scalarPtr());
-    return readScalar(scalarPtr());
-  }
-  template <int D, int MODE>
-  void blockMatVec(int nBlk, const T* A, const T* xv, T* yv) {
-    hipLaunchKernelGGL((kBlockDiagMatVec<T, D, MODE>),
-                       dim3(gridFor((int64_t)nBlk * D)), dim3(kBlk), 0, stream_,
-                       nBlk, A, xv, yv);
-  }
-  // w_local = Cinv * in_local  (pointers offset to the local shard)
-  void applyCinv(const T* in, T* out) {
-    blockMatVec<3, 0>(npL_, dHllInv_ + (int64_t)ptLo_ * 9,
-                      in + (int64_t)ptLo_ * 3, out + (int64_t)ptLo_ * 3);
-  }
-  void spmvEtx(const T* xv, T* out) {
-    hipLaunchKernelGGL(kZeroRange<T>, dim3(gridFor((int64_t)npL_ * 3)),
-                       dim3(kBlk), 0, stream_, out + (int64_t)ptLo_ * 3,
-                       (int64_t)npL_ * 3);
-    if (implicit_) {
-      if (hasInfo_)
-        hipLaunchKernelGGL((kSpmvEtx<T, true, true>), dim3(gridFor(nL_)),
-                           dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
-                           (const T*)nullptr, (const T* const*)dJSlots_,
-                           dInfo_, lossKind_, lossD2_, xv, out);
-      else
-        hipLaunchKernelGGL((kSpmvEtx<T, true, false>), dim3(gridFor(nL_)),
-                           dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
-                           (const T*)nullptr, (const T* const*)dJSlots_,
-                           (const T*)nullptr, lossKind_, lossD2_, xv, out);
-    } else {
-      hipLaunchKernelGGL((kSpmvEtx<T, false, false>), dim3(gridFor(nL_)),
-                         dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_, dHpl_,
-                         (const T* const*)nullptr, (const T*)nullptr, 0,
-                         T(0), xv, out);
-    }
-  }
-  void spmvEx(const T* wv, T* out) {
-    HIP_CHECK(hipMemsetAsync(out, 0, nc_ * sizeof(T), stream_));
-    if (implicit_) {
-      if (nChunks_ == 0) return;
-      hipLaunchKernelGGL(kSpmvExImpG<T>, dim3(nChunks_), dim3(64), 0,
-                         stream_, nChunks_, dChCam_, dChLo_, dChHi_,
-                         dPtOfCam_, dJcCam_, dJpCam_, nL_, wv, out);
-      return;
-    }
-    if (nChunks_ > 0)
-      hipLaunchKernelGGL(kSpmvEx<T>, dim3(nChunks_), dim3(64), 0, stream_,
-                         nChunks_, dChCam_, dChLo_, dChHi_, dPtOfCam_,
-                         dHplCam_, nL_, wv, out);
-  }
-  // One full PCG iteration (captured as a hipGraph when possible).
-  // vs round 1: the p^T q dot partial is fused into the B-apply, the
-  // two-deep x backup rotation is fused into kUpdateXR, and the standalone
-  // kRedPartial pass + dXBakPrev memcpy are gone (~2 launches + one full
-  // read pass over p,q per iteration — part of the N=8 constant term,
-  // PLAN_r02 item 3).
-  void pcgBody() {
-    const int rhoGrid = gridFor(nc_) < kRedBlocks ? gridFor(nc_) : kRedBlocks;
-    hipLaunchKernelGGL(kPrecondRho<T>, dim3(rhoGrid), dim3(kBlk), 0, stream_,
-                       ncam_, dHppInv_, dRr_, dZ_, dPart_);
-    hipLaunchKernelGGL(kRedFinalRhoBeta, dim3(1), dim3(kBlk), 0, stream_,
-                       dPart_, rhoGrid, slotRho(), slotRhoPrev(), slotBeta());
-    hipLaunchKernelGGL(kXpbyS<T>, dim3(gridFor(nc_)), dim3(kBlk), 0, stream_,
-                       nc_, dZ_, slotBeta(), dP_);
-    schurApply(dP_, dQ_, /*withDot=*/true);
-    hipLaunchKernelGGL(kRedFinalAlpha, dim3(1), dim3(kBlk), 0, stream_,
-                       dPart_, gridFor(nc_), slotRho(), slotAlpha(),
-                       slotRhoPrev());
-    hipLaunchKernelGGL(kUpdateXR<T>, dim3(gridFor(nc_)), dim3(kBlk), 0,
-                       stream_, nc_, slotAlpha(), dP_, dQ_, dDeltaX_, dXBak_,
-                       dXBakPrev_, dRr_);
-  }
-
-  // Capture the PCG body as a hipGraph, once.  The implicit path reads the
-  // double-buffered accepted J set through device pointer slots (dJSlots_),
-  // so accept-time buffer flips never invalidate the graph.  Multi-rank
-  // RCCL collectives are captured too (ncclAllReduce is stream-ordered on
-  // stream_); any capture failure falls back to the eager path permanently.
-  void ensurePcgGraph() {
-    if (pcgGraphExec_ || pcgGraphFailed_ || getenv("MEGBA_NO_GRAPH")) return;
-    // host-callback collectives (gloo testing fallback) cannot be captured
-    if (world_ > 1 && !hasComm_) return;
-    sync();
-    hipGraph_t graph = nullptr;
-    if (hipStreamBeginCapture(stream_, hipStreamCaptureModeThreadLocal) !=
-        hipSuccess) {
-      pcgGraphFailed_ = true;
-      (void)hipGetLastError();
-      return;
-    }
-    bool ok = true;
-    try {
-      pcgBody();
-    } catch (...) {
-      ok = false;
-    }
-    if (hipStreamEndCapture(stream_, &graph) != hipSuccess || !ok || !graph) {
-      if (graph) (void)hipGraphDestroy(graph);
-      (void)hipGetLastError();
-      pcgGraphFailed_ = true;
-      return;
-    }
-    hipGraphExec_t exec = nullptr;
-    if (hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0) != hipSuccess) {
-      (void)hipGraphDestroy(graph);
-      (void)hipGetLastError();
-      pcgGraphFailed_ = true;
-      return;
-    }
-    (void)hipGraphDestroy(graph);
-    pcgGraphExec_ = exec;
-  }
-
-  // q = S x = HppD x - E Cinv E^T x  (ONE 9*ncam allreduce; the reference's
-  // site A4 needed an additional 3*npt allreduce here).  withDot fuses the
-  // x^T q dot partials into the B-apply pass (used by the PCG body).
-  void schurApply(const T* xv, T* q, bool withDot = false) {
-    spmvEtx(xv, dTemp_);
-    applyCinv(dTemp_, dW_);
-    spmvEx(dW_, q);
-    allreduce(q, nc_, ncclSum);
-    if (withDot)
-      hipLaunchKernelGGL(kBApplyDot<T>, dim3(gridFor(nc_)), dim3(kBlk), 0,
-                         stream_, ncam_, dHppD_, xv, q, dPart_);
-    else
-      blockMatVec<9, 1>(ncam_, dHppD_, xv, q);
-  }
-
-  hipStream_t stream_{};
-  ncclComm_t comm_{};
-  bool hasComm_ = false;
-  CustomForward<T> customFwd_;
-  HostAllreduce<T> hostAr_;
-  HostAllreduce<double> hostArD_;
-  int rank_, world_, ncam_, npt_;
-  int ptLo_ = 0, ptHi_ = 0, npL_ = 0;
-  int64_t e0_ = 0, e1_ = 0, nL_ = 0, nc_ = 0, np_ = 0, dim_ = 0;
-  bool hasInfo_ = false;
-  bool analytical_ = false;
-  bool implicit_ = false;
-  int lossKind_ = 0;
-  T lossD2_ = T(1);
-  int cur_ = 0;
-  bool freshCur_ = false;  // current r/J buffers hold the last forward()
-  int nChunks_ = 0;
-  int *dCamOf_{}, *dPtOf_{}, *dChCam_{}, *dChLo_{}, *dChHi_{}, *dFail_{};
-  int *dCamPos_{}, *dPtOfCam_{};
-  T *dMeas_{}, *dInfo_{}, *dLeaf_{}, *dMeasSplit_{};
-  unsigned char *dCamFixed_{}, *dPtFixed_{};
-  T *dParams_{}, *dParamsBak_{};
-  T *dR_[2]{}, *dJc_[2]{}, *dJp_[2]{};
-  T *dHpp_{}, *dHll_{}, *dHpl_{}, *dHplCam_{}, *dSlab_{}, *dG_{}, *dGBak_{};
-  T *dJcCam_{}, *dJpCam_{};  // implicit: cam-sorted grad-major wJc / Jp
-  T *dHppD_{}, *dHllD_{}, *dHppInv_{}, *dHllInv_{};
-  T *dDeltaX_{}, *dDeltaXBak_{};
-  T *dP_{}, *dRr_{}, *dZ_{}, *dQ_{}, *dV_{}, *dW_{}, *dTemp_{}, *dXBak_{},
-      *dXBakPrev_{}, *dPtMerge_{};
-  hipGraphExec_t pcgGraphExec_{};
-  bool pcgGraphFailed_ = false;
-  const T** dJSlots_{};  // device slots: accepted {Jc, Jp, r} pointers
-  int partCap_ = kRedBlocks;
-  double* dPart_{};
-  double* hScalar_{};
-  std::vector<void*> allocs_;
-};
+// Per-dims factories defined in the gpu_dims_*.hip TUs.
+template <typename T, int CD, int PD, int RD>
+std::unique_ptr<Engine<T>> makeGpuEngineDims(const BAProblemHost& prob,
+                                             const ProblemIndex& ix,
+                                             const ProblemOption& opt,
+                                             const std::string& rcclId,
+                                             CustomForward<T> customForward,
+                                             HostAllreduce<T> hostAllreduce,
+                                             HostAllreduce<double> hostArD);
 
 template <typename T>
 std::unique_ptr<Engine<T>> makeGpuEngine(const BAProblemHost& prob,
@@ -1824,11 +51,24 @@ std::unique_ptr<Engine<T>> makeGpuEngine(const BAProblemHost& prob,
                                          const std::string& rcclId,
                                          CustomForward<T> customForward,
                                          HostAllreduce<T> hostAllreduce,
-                                         HostAllreduce<double> hostAllreduceScalar) {
-  return std::make_unique<GpuEngine<T>>(prob, ix, opt, rcclId,
-                                        std::move(customForward),
-                                        std::move(hostAllreduce),
-                                        std::move(hostAllreduceScalar));
+                                         HostAllreduce<double> hostArD) {
+  const int cd = prob.camDim, pd = prob.ptDim, rd = prob.resDim;
+#define MEGBA_GPU_CASE(CDv, PDv, RDv)                                     \
+  if (cd == CDv && pd == PDv && rd == RDv)                                \
+    return makeGpuEngineDims<T, CDv, PDv, RDv>(                           \
+        prob, ix, opt, rcclId, std::move(customForward),                  \
+        std::move(hostAllreduce), std::move(hostArD));
+  MEGBA_GPU_CASE(9, 3, 2)
+  MEGBA_GPU_CASE(6, 3, 2)
+  MEGBA_GPU_CASE(4, 3, 2)
+  MEGBA_GPU_CASE(9, 3, 3)
+  MEGBA_GPU_CASE(6, 3, 3)
+  MEGBA_GPU_CASE(4, 3, 3)
+#undef MEGBA_GPU_CASE
+  MEGBA_CHECK(false,
+              "unsupported (camDim,ptDim,resDim) = (" + std::to_string(cd) +
+                  "," + std::to_string(pd) + "," + std::to_string(rd) +
+                  "); compiled set: {9,6,4} x {3} x {2,3}");
 }
 
 template std::unique_ptr<Engine<double>> makeGpuEngine<double>(

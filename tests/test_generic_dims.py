@@ -256,3 +256,103 @@ def test_analytical_requires_bal_dims():
     p = mb.BAProblem(cams9[:, :6].copy(), pts, ci, pi, meas)
     with pytest.raises(RuntimeError, match="analytical"):
         p.build(device="cpu", diff="analytical", intrinsics=INTR)
+
+
+# ---- GPU vs CPU-oracle parity for the generic dims ------------------------
+def _gpu_cpu_pair(cams, pts, ci, pi, meas, schur="explicit", **kw):
+    pc = mb.BAProblem(cams, pts, ci, pi, meas)
+    pc.build(device="cpu", schur=schur, **kw)
+    pg = mb.BAProblem(cams, pts, ci, pi, meas)
+    pg.build(device="gpu", schur=schur, **kw)
+    return pc, pg
+
+
+def _compare_stage(pc, pg, keys=("r", "Jc", "Jp", "Hpp", "Hll", "g"),
+                   rtol=1e-9):
+    c1, c2 = pc.forward(), pg.forward()
+    np.testing.assert_allclose(c2, c1, rtol=rtol)
+    dj1, dj2 = pc.dump(), pg.dump()
+    pc.accept_forward()
+    pg.accept_forward()
+    pc.build_linear_system()
+    pg.build_linear_system()
+    d1, d2 = pc.dump(), pg.dump()
+    d1.update({k: dj1[k] for k in ("r", "Jc", "Jp")})
+    d2.update({k: dj2[k] for k in ("r", "Jc", "Jp")})
+    for key in keys:
+        scale = np.abs(d1[key]).max() or 1.0
+        np.testing.assert_allclose(d2[key], d1[key], rtol=rtol,
+                                   atol=rtol * scale, err_msg=key)
+    # one damped solve must agree too
+    pc.process_diag(1e4)
+    pg.process_diag(1e4)
+    pc.solve_linear(max_iter=40, tol=1e-12, refuse_ratio=1e30)
+    pg.solve_linear(max_iter=40, tol=1e-12, refuse_ratio=1e30)
+    dx1 = pc.dump()["deltaX"]
+    dx2 = pg.dump()["deltaX"]
+    scale = np.abs(dx1).max() or 1.0
+    np.testing.assert_allclose(dx2, dx1, rtol=1e-6, atol=1e-8 * scale,
+                               err_msg="deltaX")
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("schur", ["explicit", "implicit"])
+def test_gpu_632_matches_cpu_oracle(schur):
+    cams9, pts, ci, pi, meas = _synth_calibrated(12, 110, 950, seed=12)
+    pc, pg = _gpu_cpu_pair(cams9[:, :6].copy(), pts, ci, pi, meas,
+                           schur=schur, intrinsics=INTR)
+    _compare_stage(pc, pg)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("schur", ["explicit", "implicit"])
+def test_gpu_633_matches_cpu_oracle(schur):
+    cams, pts, ci, pi, meas, _ = _synth_se3(9, 80, 600, seed=21)
+    pc, pg = _gpu_cpu_pair(cams, pts, ci, pi, meas, schur=schur)
+    _compare_stage(pc, pg)
+
+
+@pytest.mark.gpu
+def test_gpu_432_custom_matches_cpu_oracle():
+    RNG = np.random.default_rng(31)
+    ncam, npt, nobs = 6, 50, 360
+    cams = RNG.normal(0, 0.1, (ncam, 4))
+    pts = RNG.normal(0, 1.0, (npt, 3))
+    ci = RNG.integers(0, ncam, nobs).astype(np.int32)
+    pi = np.concatenate(
+        [np.arange(npt), RNG.integers(0, npt, nobs - npt)]).astype(np.int32)
+    ci[:ncam] = np.arange(ncam)
+    meas = np.array([_wp_np(cams[c], pts[q], np.zeros(2))
+                     for c, q in zip(ci, pi)])
+    pc, pg = _gpu_cpu_pair(cams, pts, ci, pi, meas,
+                           custom_forward=_wp_forward)
+    _compare_stage(pc, pg)
+
+
+@pytest.mark.gpu
+def test_gpu_632_lm_trajectory_matches_cpu():
+    RNG = np.random.default_rng(41)
+    cams9, pts, ci, pi, meas = _synth_calibrated(12, 110, 950, seed=13)
+    cams6 = cams9[:, :6] + RNG.normal(0, 0.01, (12, 6))
+    kw = dict(max_iter=5, solver_tol=1e-8, solver_max_iter=150,
+              solver_refuse_ratio=1e6, verbose=False)
+    pc = mb.BAProblem(cams6, pts, ci, pi, meas)
+    pc.build(device="cpu", intrinsics=INTR)
+    pg = mb.BAProblem(cams6, pts, ci, pi, meas)
+    pg.build(device="gpu", intrinsics=INTR)
+    r1, r2 = pc.solve(**kw), pg.solve(**kw)
+    np.testing.assert_allclose([i["chi2"] for i in r2["iters"]],
+                               [i["chi2"] for i in r1["iters"]], rtol=1e-7)
+
+
+@pytest.mark.gpu
+def test_gpu_633_fp32_decreases():
+    cams, pts, ci, pi, meas, _ = _synth_se3(9, 80, 600, seed=22)
+    RNG = np.random.default_rng(5)
+    cams0 = cams + RNG.normal(0, 0.02, cams.shape)
+    p = mb.BAProblem(cams0, pts, ci, pi, meas)
+    p.build(device="gpu", dtype="float32")
+    rep = p.solve(max_iter=6, solver_tol=1e-6, solver_max_iter=100,
+                  solver_refuse_ratio=1e6, verbose=False)
+    chis = [i["chi2"] for i in rep["iters"]]
+    assert chis[-1] < 0.5 * chis[0]
