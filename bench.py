@@ -68,6 +68,7 @@ def main():
               file=sys.stderr)
 
     dist = None
+    store = None
     rccl_id = None
     allreduce = None
     launched_distributed = ("TORCHELASTIC_RUN_ID" in os.environ or
@@ -75,20 +76,21 @@ def main():
                              "MASTER_PORT" in os.environ and
                              "RANK" in os.environ))
     if world > 1 or launched_distributed:
-        import torch.distributed as tdist
-        dist = tdist
-        # gloo for bootstrap only; the GPU engine runs RCCL natively.
-        dist.init_process_group("gloo", rank=rank, world_size=world)
         if args.device == "gpu":
+            # Torch-free bootstrap: a stdlib-TCP store exchanges the
+            # 128-byte ncclUniqueId and serves barriers / the max-over-
+            # ranks reduction; RCCL then runs natively over xGMI.
             from megba_amd import _core
-            from megba_amd.dist import broadcast_rccl_id
+            from megba_amd.rendezvous import from_env
+            store = from_env(rank, world)
             ngpu = max(1, _core.hip_device_count())
             dev = local_rank % ngpu
             if world > 1 and not os.environ.get("MEGBA_NO_PREFLIGHT"):
                 # Pre-flight: throwaway comm + 1-element allreduce under a
                 # watchdog, so a wedged RCCL bootstrap aborts the whole job
                 # with a clear message instead of hanging the scale run.
-                pf_id = broadcast_rccl_id(rank)
+                pf_id = store.broadcast_bytes(
+                    _core.rccl_unique_id() if rank == 0 else None)
                 try:
                     t_pf = _core.rccl_preflight(pf_id, rank, world, dev,
                                                 120.0)
@@ -99,8 +101,13 @@ def main():
                 if rank == 0:
                     print(f"# rccl preflight ok ({t_pf:.2f}s, world={world})",
                           file=sys.stderr)
-            rccl_id = broadcast_rccl_id(rank)
+            rccl_id = store.broadcast_bytes(
+                _core.rccl_unique_id() if rank == 0 else None)
         else:
+            # multi-process CPU testing path: gloo allreduce callback
+            import torch.distributed as tdist
+            dist = tdist
+            dist.init_process_group("gloo", rank=rank, world_size=world)
             from megba_amd.dist import gloo_allreduce_callback
             allreduce = gloo_allreduce_callback()
 
@@ -126,16 +133,14 @@ def main():
               force_iterations=True, verbose=args.verbose and rank == 0)
 
     def sync():
+        if args.device == "gpu":
+            from megba_amd import _core
+            if _core.hip_device_count() > 0:
+                _core.device_synchronize()
+        if store is not None:
+            store.barrier()
         if dist is not None:
             dist.barrier()
-        try:
-            import torch
-            if torch.cuda.is_available():
-                # explicit device: this rank's engine device, not torch's
-                # current device (0), matters on an 8-GPU node.
-                torch.cuda.synchronize(device_index)
-        except Exception:
-            pass
 
     for _ in range(args.warmup):
         p.lm_step()
@@ -147,7 +152,9 @@ def main():
     elapsed = time.perf_counter() - t0
 
     # MAX over ranks.
-    if dist is not None:
+    if store is not None:
+        elapsed = store.all_max(elapsed)
+    elif dist is not None:
         import torch
         t = torch.tensor([elapsed], dtype=torch.float64)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
@@ -185,6 +192,9 @@ def main():
         }
         print(json.dumps(line), flush=True)
 
+    if store is not None:
+        store.barrier()  # keep rank 0 alive until every rank has printed
+        store.close()
     if dist is not None:
         dist.destroy_process_group()
 

@@ -494,6 +494,10 @@ PYBIND11_MODULE(_core, m) {
         py::arg("id"), py::arg("rank"), py::arg("world"),
         py::arg("device_index") = 0, py::arg("timeout_sec") = 120.0);
   m.def("hip_device_count", []() { return hipDeviceCountSafe(); });
+  m.def("device_synchronize", []() {
+    py::gil_scoped_release rel;
+    deviceSynchronize();
+  });
   m.def("hip_mem_info", []() {
     auto p = hipMemInfoSafe();
     return py::make_tuple(p.first, p.second);

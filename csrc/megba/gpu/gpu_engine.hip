@@ -151,6 +151,8 @@ std::string rcclUniqueIdString() {
   return std::string(reinterpret_cast<const char*>(&id), sizeof(id));
 }
 
+void deviceSynchronize() { HIP_CHECK(hipDeviceSynchronize()); }
+
 int hipDeviceCountSafe() {
   int n = 0;
   if (hipGetDeviceCount(&n) != hipSuccess) return 0;

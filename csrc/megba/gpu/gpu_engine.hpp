@@ -37,6 +37,7 @@ std::string rcclUniqueIdString();
 double rcclPreflight(const std::string& idBytes, int rank, int world,
                      int deviceIndex, double timeoutSec);
 int hipDeviceCountSafe();
+void deviceSynchronize();  // hipDeviceSynchronize (bench timing fences)
 std::pair<long long, long long> hipMemInfoSafe();  // (free, total) bytes
 
 }  // namespace megba
