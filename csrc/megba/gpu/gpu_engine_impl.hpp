@@ -524,6 +524,82 @@ __global__ __launch_bounds__(128) void kAssembleCam(
     atomicAdd(&g[(int64_t)cam * CD + ti], acc);
 }
 
+// MFMA experiment (reference anchor
+// build_linear_system.cu:88-146): the per-chunk Hpp accumulation
+// Hpp += Jc^T (W Jc) is a K-reduction GEMM with M=N=9, K=2*rows.  One
+// v_mfma_f64_16x16x4_f64 per 4 K-rows computes the whole 16x16 tile
+// (rows 9..15 zero-padded); the g_c = -Jc^T wr partial rides along as a
+// free 10th B column.  Env-gated (MEGBA_MFMA=1), fp64 BAL dims only.
+// Four interleaved accumulators break the dependent-MFMA latency chain.
+typedef double megba_d4 __attribute__((ext_vector_type(4)));
+template <typename T, int CD, int PD, int RD, bool HASINFO, bool EXPL>
+__global__ __launch_bounds__(64) void kAssembleCamMfma(
+    int nChunks, const int* __restrict__ chCam, const int* __restrict__ chLo,
+    const int* __restrict__ chHi, const T* __restrict__ slab,
+    T* __restrict__ Hpp, T* __restrict__ g) {
+  static_assert(std::is_same<T, double>::value && CD == 9 && PD == 3 &&
+                    RD == 2,
+                "MFMA assembly path is fp64 BAL (9,3,2) only");
+  using L = SlabLayout<CD, PD, RD, EXPL, HASINFO>;
+  constexpr int CR = CD * RD;                       // 18
+  constexpr int ST = CR + RD + (HASINFO ? CR : 0);  // jc + wr (+ wjc)
+  constexpr int TE = 128;
+  __shared__ T lds[TE * ST];
+  const int chunk = blockIdx.x;
+  if (chunk >= nChunks) return;
+  const int cam = chCam[chunk];
+  const int lo = chLo[chunk], hi = chHi[chunk];
+  const int l = (int)threadIdx.x;
+  const int i = l & 15;        // A row (camera col) / D col
+  const int kq = l >> 4;       // K quarter within each MFMA's K=4
+  constexpr int woff = HASINFO ? CR + RD : 0;
+  megba_d4 acc[4];
+  for (int a = 0; a < 4; ++a) acc[a] = megba_d4{0.0, 0.0, 0.0, 0.0};
+  for (int s0 = lo; s0 < hi; s0 += TE) {
+    const int nt = min(TE, hi - s0);
+    for (int idx = l; idx < nt * ST; idx += 64) {
+      const int row = idx / ST;
+      const int k = idx % ST;
+      lds[row * ST + k] = slab[(int64_t)(s0 + row) * L::SW + L::JCOFF + k];
+    }
+    __syncthreads();
+    const int K = nt * RD;  // edge-rows in this tile
+    for (int k0 = 0; k0 < K; k0 += 16) {
+      // 4 MFMAs per iteration, one per accumulator -> 4-deep independence
+      for (int a = 0; a < 4; ++a) {
+        const int kr = k0 + 4 * a + kq;  // this lane's K index
+        T av = T(0), bv = T(0);
+        if (kr < K) {
+          const int e = kr >> 1;        // RD == 2
+          const int r = kr & 1;
+          const T* row = lds + e * ST;
+          if (i < 9) {
+            av = row[i * 2 + r];          // Jc[k][i]
+            bv = row[woff + i * 2 + r];   // (W Jc)[k][j=i]
+          } else if (i == 9) {
+            bv = row[CR + r];             // wr[k] -> g column
+          }
+        }
+        acc[a] = __builtin_amdgcn_mfma_f64_16x16x4f64(av, bv, acc[a], 0, 0,
+                                                      0);
+      }
+    }
+    __syncthreads();
+  }
+  for (int a = 1; a < 4; ++a) acc[0] += acc[a];
+  // C/D map: col = lane&15, row = (lane>>4)*4 + reg
+  for (int v = 0; v < 4; ++v) {
+    const int row = kq * 4 + v;
+    const int col = i;
+    if (row < 9) {
+      if (col < 9)
+        atomicAdd(&Hpp[(int64_t)cam * 81 + row * 9 + col], acc[0][v]);
+      else if (col == 9)
+        atomicAdd(&g[(int64_t)cam * 9 + row], -acc[0][v]);
+    }
+  }
+}
+
 // Explicit only: stream the slab's Hpl blocks out as a cam-sorted grad-major
 // copy for the E*w kernel (contiguous reads, coalesced writes).
 template <typename T, int CD, int PD, int RD, bool HASINFO>
@@ -1598,10 +1674,21 @@ class GpuEngine final : public Engine<T> {
                          dCamOf_, dPtOf_, dR_[bak], dJc_[bak], dJp_[bak],
                          dInfo_, dHll_, dHpl_, dG_, ncam_, dCamPos_, dSlab_,
                          lossKind_, lossD2_);
-      if (nChunks_ > 0)
+      if (nChunks_ > 0) {
+        if constexpr (std::is_same<T, double>::value && CD == 9 && PD == 3 &&
+                      RD == 2) {
+          if (useMfma_) {
+            hipLaunchKernelGGL((kAssembleCamMfma<T, CD, PD, RD, HI, EX>),
+                               dim3(nChunks_), dim3(64), 0, stream_,
+                               nChunks_, dChCam_, dChLo_, dChHi_, dSlab_,
+                               dHpp_, dG_);
+            return;
+          }
+        }
         hipLaunchKernelGGL((kAssembleCam<T, CD, PD, RD, HI, EX>),
                            dim3(nChunks_), dim3(128), 0, stream_, nChunks_,
                            dChCam_, dChLo_, dChHi_, dSlab_, dHpp_, dG_);
+      }
     };
     using TrueT = std::integral_constant<bool, true>;
     using FalseT = std::integral_constant<bool, false>;
@@ -1872,6 +1959,8 @@ class GpuEngine final : public Engine<T> {
   bool implicit_ = false;
   int lossKind_ = 0;
   T lossD2_ = T(1);
+  // MFMA assembly experiment (fp64 BAL only), opt-in via MEGBA_MFMA=1
+  bool useMfma_ = getenv("MEGBA_MFMA") != nullptr;
   int cur_ = 0;
   bool freshCur_ = false;  // current r/J buffers hold the last forward()
   int nChunks_ = 0;

@@ -118,3 +118,41 @@ def test_fixed_work_matches_stepwise_readback():
     scale = np.abs(dx_gen).max() or 1.0
     np.testing.assert_allclose(dx_fast, dx_gen, rtol=1e-6,
                                atol=1e-9 * scale)
+
+
+@pytest.mark.parametrize("schur", ["explicit", "implicit"])
+@pytest.mark.parametrize("weighted", [False, True])
+def test_mfma_assembly_matches_default(schur, weighted):
+    """MEGBA_MFMA=1 swaps kAssembleCam for the v_mfma_f64_16x16x4_f64
+    tile kernel (fp64 BAL only); Hpp and g_c must match the default
+    kernel to accumulation-order tolerance."""
+    cams, pts, ci, pi, meas = mb.synthesize_bal(24, 400, 3600, seed=6)
+    info = None
+    if weighted:
+        rng = np.random.default_rng(2)
+        w = rng.uniform(0.5, 2.0, (len(ci), 2))
+        info = np.stack([w[:, 0], 0.1 * np.ones(len(ci)), w[:, 1]], axis=1)
+
+    def assemble(env):
+        saved = {k: os.environ.get(k) for k in env}
+        os.environ.update(env)
+        try:
+            p = mb.BAProblem(cams, pts, ci, pi, meas, info=info)
+            p.build(device="gpu", schur=schur)
+            p.forward()
+            p.accept_forward()
+            p.build_linear_system()
+            return p.dump()
+        finally:
+            for k, v in saved.items():
+                if v is None:
+                    os.environ.pop(k, None)
+                else:
+                    os.environ[k] = v
+
+    d_ref = assemble({})
+    d_mfma = assemble({"MEGBA_MFMA": "1"})
+    for key in ("Hpp", "g", "Hll"):
+        scale = np.abs(d_ref[key]).max() or 1.0
+        np.testing.assert_allclose(d_mfma[key], d_ref[key], rtol=1e-10,
+                                   atol=1e-12 * scale, err_msg=key)
