@@ -587,9 +587,10 @@ __global__ __launch_bounds__(64) void kAssembleCamMfma(
     __syncthreads();
   }
   for (int a = 1; a < 4; ++a) acc[0] += acc[a];
-  // C/D map: col = lane&15, row = (lane>>4)*4 + reg
+  // C/D map for v_mfma_f64_16x16x4_f64 (probed on hardware,
+  // tools/mfma_probe.hip): col = lane&15, row = 4*reg + (lane>>4).
   for (int v = 0; v < 4; ++v) {
-    const int row = kq * 4 + v;
+    const int row = v * 4 + kq;
     const int col = i;
     if (row < 9) {
       if (col < 9)
