@@ -865,6 +865,265 @@ __global__ __launch_bounds__(64) void kSpmvEx(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Fused Schur apply:  out[CD*ncam] += E Cinv E^T x  in ONE pass.
+// ---------------------------------------------------------------------------
+// The separate-pass pipeline (kSpmvEtx -> applyCinv -> kSpmvEx) reads the
+// per-edge J/Hpl data twice per PCG iteration and gathers the 3*npt w
+// vector by point id from the cam-sorted pass — on big-fp32 problems that
+// gather fetches a full 64 B line for 12 used bytes and was measured at
+// ~2x the byte floor (profiles/r01_final13682_implicit_fp32.md).  Here the
+// edges are processed in primary (pt,cam)-sorted order in WINDOWS of <=64
+// edges aligned to point-run boundaries (host-built table), one wave per
+// window:
+//   1. each lane evaluates its edge's t_e = Jp^T W (Jc x)  (x is the
+//      replicated CD*ncam vector, L2-resident),
+//   2. a wave segmented-scan sums t_e over each point run; the run tail
+//      applies the PD x PD Cinv block: w_p = Cinv t_p,
+//   3. w_p is broadcast back to the run's lanes (tail-lane shuffle) and
+//      each lane scatters its edge's E-contribution Jc^T W (Jp w_p) with
+//      CD atomicAdds into the small L2-resident camera vector.
+// J (or Hpl) is read ONCE, w never exists in HBM, and the scan/broadcast
+// replace both the temp round-trip and the gather.  Point runs longer
+// than 64 edges (high-degree landmarks) fall back to the classic path:
+// their windows only accumulate t partials into `temp` (phase 1 flag);
+// kFusedLongW then applies Cinv for those points and kFusedLongScatter
+// finishes their E-contributions.
+template <typename T, int CD, int PD, int RD, bool IMP, bool HASINFO>
+__device__ inline void fusedLoadEdge(int64_t j, int64_t nL,
+                                     const int* __restrict__ camOf,
+                                     const T* __restrict__ Hpl,
+                                     const T* __restrict__ Jc,
+                                     const T* __restrict__ Jp,
+                                     const T* __restrict__ info,
+                                     const T* __restrict__ rBak, int lossKind,
+                                     T lossD2, const T* __restrict__ x,
+                                     T (&jcv)[RD][CD], T (&jpv)[RD][PD],
+                                     T (&hplv)[CD][PD], T (&t)[PD]) {
+  constexpr int RW = RD * (RD + 1) / 2;
+  const T* xc = x + (int64_t)camOf[j] * CD;
+  if (IMP) {
+    for (int i = 0; i < CD; ++i)
+      for (int rr = 0; rr < RD; ++rr)
+        jcv[rr][i] = Jc[((int64_t)(i * RD + rr)) * nL + j];
+    for (int k = 0; k < PD; ++k)
+      for (int rr = 0; rr < RD; ++rr)
+        jpv[rr][k] = Jp[((int64_t)(k * RD + rr)) * nL + j];
+    T u[RD];
+    for (int rr = 0; rr < RD; ++rr) {
+      T v = T(0);
+      for (int i = 0; i < CD; ++i) v += jcv[rr][i] * xc[i];
+      u[rr] = v;
+    }
+    if (HASINFO) {
+      T wu[RD];
+      for (int i = 0; i < RD; ++i) {
+        T v = T(0);
+        for (int k = 0; k < RD; ++k)
+          v += info[RW * j + symIdx<RD>(i, k)] * u[k];
+        wu[i] = v;
+      }
+      for (int i = 0; i < RD; ++i) u[i] = wu[i];
+    }
+    if (lossKind) {
+      T ss = T(0);
+      for (int rr = 0; rr < RD; ++rr) {
+        const T rv = rBak[(int64_t)rr * nL + j];
+        ss += rv * rv;
+      }
+      const T w = lossWeight(lossKind, lossD2, ss);
+      for (int rr = 0; rr < RD; ++rr) u[rr] *= w;
+    }
+    for (int k = 0; k < PD; ++k) {
+      T v = T(0);
+      for (int rr = 0; rr < RD; ++rr) v += jpv[rr][k] * u[rr];
+      t[k] = v;
+    }
+  } else {
+    for (int i = 0; i < CD; ++i)
+      for (int k = 0; k < PD; ++k)
+        hplv[i][k] = Hpl[((int64_t)(i * PD + k)) * nL + j];
+    for (int k = 0; k < PD; ++k) t[k] = T(0);
+    for (int i = 0; i < CD; ++i) {
+      const T xi = xc[i];
+      for (int k = 0; k < PD; ++k) t[k] += hplv[i][k] * xi;
+    }
+  }
+}
+
+// Per-edge E-contribution out[cam] += M w  (M = Jc^T W Jp or Hpl).
+template <typename T, int CD, int PD, int RD, bool IMP, bool HASINFO>
+__device__ inline void fusedScatter(int64_t j, int64_t nL,
+                                    const int* __restrict__ camOf,
+                                    const T* __restrict__ info,
+                                    const T* __restrict__ rBak, int lossKind,
+                                    T lossD2, const T (&jcv)[RD][CD],
+                                    const T (&jpv)[RD][PD],
+                                    const T (&hplv)[CD][PD], const T (&wv)[PD],
+                                    T* __restrict__ out) {
+  constexpr int RW = RD * (RD + 1) / 2;
+  T* oc = out + (int64_t)camOf[j] * CD;
+  if (IMP) {
+    T u[RD];
+    for (int rr = 0; rr < RD; ++rr) {
+      T v = T(0);
+      for (int k = 0; k < PD; ++k) v += jpv[rr][k] * wv[k];
+      u[rr] = v;
+    }
+    if (HASINFO) {
+      T wu[RD];
+      for (int i = 0; i < RD; ++i) {
+        T v = T(0);
+        for (int k = 0; k < RD; ++k)
+          v += info[RW * j + symIdx<RD>(i, k)] * u[k];
+        wu[i] = v;
+      }
+      for (int i = 0; i < RD; ++i) u[i] = wu[i];
+    }
+    if (lossKind) {
+      T ss = T(0);
+      for (int rr = 0; rr < RD; ++rr) {
+        const T rv = rBak[(int64_t)rr * nL + j];
+        ss += rv * rv;
+      }
+      const T w = lossWeight(lossKind, lossD2, ss);
+      for (int rr = 0; rr < RD; ++rr) u[rr] *= w;
+    }
+    for (int i = 0; i < CD; ++i) {
+      T v = T(0);
+      for (int rr = 0; rr < RD; ++rr) v += jcv[rr][i] * u[rr];
+      atomicAdd(&oc[i], v);
+    }
+  } else {
+    for (int i = 0; i < CD; ++i) {
+      T v = T(0);
+      for (int k = 0; k < PD; ++k) v += hplv[i][k] * wv[k];
+      atomicAdd(&oc[i], v);
+    }
+  }
+}
+
+template <typename T, int CD, int PD, int RD, bool IMP, bool HASINFO>
+__global__ __launch_bounds__(256) void kSchurFused(
+    int nWin, const int64_t* __restrict__ winLo,
+    const int64_t* __restrict__ winHi, const unsigned char* __restrict__ winFlag,
+    int64_t nL, const int* __restrict__ camOf, const int* __restrict__ ptOf,
+    const T* __restrict__ Hpl, const T* const* __restrict__ jSlots,
+    const T* __restrict__ info, int lossKind, T lossD2,
+    const T* __restrict__ x, const T* __restrict__ HllInv,
+    T* __restrict__ temp, T* __restrict__ out) {
+  constexpr int PP = PD * PD;
+  const T* Jc = IMP ? jSlots[0] : nullptr;
+  const T* Jp = IMP ? jSlots[1] : nullptr;
+  const T* rBak = IMP ? jSlots[2] : nullptr;
+  const int lane = threadIdx.x & 63;
+  for (int w = blockIdx.x * 4 + ((int)threadIdx.x >> 6); w < nWin;
+       w += (int)gridDim.x * 4) {
+    const int64_t lo = winLo[w], hi = winHi[w];
+    const bool active = lo + lane < hi;
+    const int64_t j = active ? lo + lane : hi - 1;
+    const int pt = ptOf[j];
+    T jcv[RD][CD], jpv[RD][PD], hplv[CD][PD], t[PD];
+    for (int k = 0; k < PD; ++k) t[k] = T(0);
+    if (active)
+      fusedLoadEdge<T, CD, PD, RD, IMP, HASINFO>(
+          j, nL, camOf, Hpl, Jc, Jp, info, rBak, lossKind, lossD2, x,
+          jcv, jpv, hplv, t);
+    // segmented inclusive scan over the window's point runs
+    for (int off = 1; off < 64; off <<= 1) {
+      const int ppt = __shfl_up(pt, off, 64);
+      const bool join = lane >= off && ppt == pt;
+      if (__ballot(join) == 0ull) break;
+      T a[PD];
+      for (int k = 0; k < PD; ++k) a[k] = __shfl_up(t[k], off, 64);
+      if (join)
+        for (int k = 0; k < PD; ++k) t[k] += a[k];
+    }
+    const int nextPt = __shfl_down(pt, 1, 64);
+    const bool tail = active && (lo + lane == hi - 1 || nextPt != pt);
+    if (winFlag[w]) {
+      // segment of a >64-edge run: partials only (finished by the long-run
+      // kernels)
+      if (tail)
+        for (int k = 0; k < PD; ++k) atomicAdd(&temp[PD * pt + k], t[k]);
+      continue;
+    }
+    T wv[PD];
+    for (int k = 0; k < PD; ++k) wv[k] = T(0);
+    if (tail) {
+      const T* inv = HllInv + (int64_t)pt * PP;
+      for (int i = 0; i < PD; ++i) {
+        T v = T(0);
+        for (int k = 0; k < PD; ++k) v += inv[i * PD + k] * t[k];
+        wv[i] = v;
+      }
+    }
+    // broadcast w_p from the run tail back to every lane of the run:
+    // tail lanes form a mask; this lane's tail is the first tail at or
+    // after it (points are sorted within the window).
+    const unsigned long long tails = __ballot(tail);
+    const unsigned long long from = tails >> lane;
+    const int tl = from ? lane + __ffsll((long long)from) - 1 : lane;
+    for (int k = 0; k < PD; ++k) wv[k] = __shfl(wv[k], tl, 64);
+    if (active)
+      fusedScatter<T, CD, PD, RD, IMP, HASINFO>(
+          j, nL, camOf, info, rBak, lossKind, lossD2, jcv, jpv, hplv, wv,
+          out);
+  }
+}
+
+// Long-run phase 2: w_p = Cinv temp_p for the listed high-degree points,
+// stored into the global w vector.
+template <typename T, int PD>
+__global__ void kFusedLongW(int nPts, const int* __restrict__ longPts,
+                            const T* __restrict__ HllInv,
+                            const T* __restrict__ temp, T* __restrict__ w) {
+  constexpr int PP = PD * PD;
+  for (int idx = blockIdx.x * (int)blockDim.x + (int)threadIdx.x; idx < nPts;
+       idx += (int)gridDim.x * (int)blockDim.x) {
+    const int pt = longPts[idx];
+    const T* inv = HllInv + (int64_t)pt * PP;
+    for (int i = 0; i < PD; ++i) {
+      T v = T(0);
+      for (int k = 0; k < PD; ++k) v += inv[i * PD + k] * temp[PD * pt + k];
+      w[PD * pt + i] = v;
+    }
+  }
+}
+
+// Long-run phase 3: finish the flagged windows' E-contributions with w
+// read from the global vector.
+template <typename T, int CD, int PD, int RD, bool IMP, bool HASINFO>
+__global__ __launch_bounds__(256) void kFusedLongScatter(
+    int nFlag, const int* __restrict__ flagWins,
+    const int64_t* __restrict__ winLo, const int64_t* __restrict__ winHi,
+    int64_t nL, const int* __restrict__ camOf, const int* __restrict__ ptOf,
+    const T* __restrict__ Hpl, const T* const* __restrict__ jSlots,
+    const T* __restrict__ info, int lossKind, T lossD2,
+    const T* __restrict__ x, const T* __restrict__ w, T* __restrict__ out) {
+  const T* Jc = IMP ? jSlots[0] : nullptr;
+  const T* Jp = IMP ? jSlots[1] : nullptr;
+  const T* rBak = IMP ? jSlots[2] : nullptr;
+  const int lane = threadIdx.x & 63;
+  for (int f = blockIdx.x * 4 + ((int)threadIdx.x >> 6); f < nFlag;
+       f += (int)gridDim.x * 4) {
+    const int wi = flagWins[f];
+    const int64_t lo = winLo[wi], hi = winHi[wi];
+    if (lo + lane >= hi) continue;
+    const int64_t j = lo + lane;
+    const int pt = ptOf[j];
+    T jcv[RD][CD], jpv[RD][PD], hplv[CD][PD], t[PD];
+    fusedLoadEdge<T, CD, PD, RD, IMP, HASINFO>(
+        j, nL, camOf, Hpl, Jc, Jp, info, rBak, lossKind, lossD2, x, jcv,
+        jpv, hplv, t);
+    T wv[PD];
+    for (int k = 0; k < PD; ++k) wv[k] = w[PD * pt + k];
+    fusedScatter<T, CD, PD, RD, IMP, HASINFO>(
+        j, nL, camOf, info, rBak, lossKind, lossD2, jcv, jpv, hplv, wv,
+        out);
+  }
+}
+
 // Fused preconditioner apply + rho partial: z = Binv r (thread per row) and
 // per-block partials of r.z in one pass (saves two launches per PCG iter).
 template <typename T, int CD>
@@ -1303,6 +1562,56 @@ class GpuEngine final : public Engine<T> {
       up(dChCam_, cCam.data(), nChunks_);
       up(dChLo_, cLo.data(), nChunks_);
       up(dChHi_, cHi.data(), nChunks_);
+    }
+    // Run-aligned windows (<=64 edges, whole point runs) for the fused
+    // Schur apply; runs longer than 64 edges are flagged and finished by
+    // the long-run kernels.
+    {
+      std::vector<int64_t> wLo, wHi;
+      std::vector<unsigned char> wFlag;
+      std::vector<int> flagWins, longPts;
+      int64_t cur = -1;
+      for (int p = ptLo_; p < ptHi_; ++p) {
+        const int64_t lo = ix.ptRowPtr[p] - e0_;
+        const int64_t hi = ix.ptRowPtr[p + 1] - e0_;
+        const int64_t len = hi - lo;
+        if (len > 64) {
+          longPts.push_back(p);
+          for (int64_t ss = lo; ss < hi; ss += 64) {
+            flagWins.push_back((int)wLo.size());
+            wLo.push_back(ss);
+            wHi.push_back(std::min(ss + 64, hi));
+            wFlag.push_back(1);
+          }
+          cur = -1;
+          continue;
+        }
+        if (cur >= 0 && wHi[cur] - wLo[cur] + len <= 64) {
+          wHi[cur] = hi;
+        } else {
+          cur = (int64_t)wLo.size();
+          wLo.push_back(lo);
+          wHi.push_back(hi);
+          wFlag.push_back(0);
+        }
+      }
+      nWin_ = (int)wLo.size();
+      nFlagWins_ = (int)flagWins.size();
+      nLongPts_ = (int)longPts.size();
+      dWinLo_ = dalloc<int64_t>(nWin_);
+      dWinHi_ = dalloc<int64_t>(nWin_);
+      dWinFlag_ = dalloc<unsigned char>(nWin_);
+      up(dWinLo_, wLo.data(), nWin_);
+      up(dWinHi_, wHi.data(), nWin_);
+      up(dWinFlag_, wFlag.data(), nWin_);
+      if (nFlagWins_ > 0) {
+        dFlagWins_ = dalloc<int>(nFlagWins_);
+        up(dFlagWins_, flagWins.data(), nFlagWins_);
+      }
+      if (nLongPts_ > 0) {
+        dLongPts_ = dalloc<int>(nLongPts_);
+        up(dLongPts_, longPts.data(), nLongPts_);
+      }
     }
     dSlab_ = dalloc<T>(nL_ * slabWidth());
     dJSlots_ = dalloc<const T*>(3);
@@ -1931,13 +2240,54 @@ class GpuEngine final : public Engine<T> {
     pcgGraphExec_ = exec;
   }
 
-  // q = S x = HppD x - E Cinv E^T x  (ONE 9*ncam allreduce; the reference's
+  // out = E Cinv E^T x in one fused pass (see kSchurFused).
+  void schurFusedEcE(const T* xv, T* out) {
+    HIP_CHECK(hipMemsetAsync(out, 0, nc_ * sizeof(T), stream_));
+    if (nLongPts_ > 0)
+      hipLaunchKernelGGL(kZeroRange<T>, dim3(gridFor((int64_t)npL_ * PD)),
+                         dim3(kBlk), 0, stream_, dTemp_ + (int64_t)ptLo_ * PD,
+                         (int64_t)npL_ * PD);
+    const int grid = (nWin_ + 3) / 4 < 8192 ? (nWin_ + 3) / 4 : 8192;
+    auto launch = [&](auto impTag, auto infoTag) {
+      constexpr bool IM = decltype(impTag)::value;
+      constexpr bool HI = decltype(infoTag)::value;
+      hipLaunchKernelGGL((kSchurFused<T, CD, PD, RD, IM, HI>),
+                         dim3(grid < 1 ? 1 : grid), dim3(256), 0, stream_,
+                         nWin_, dWinLo_, dWinHi_, dWinFlag_, nL_, dCamOf_,
+                         dPtOf_, dHpl_, (const T* const*)dJSlots_, dInfo_,
+                         lossKind_, lossD2_, xv, dHllInv_, dTemp_, out);
+      if (nLongPts_ > 0) {
+        hipLaunchKernelGGL((kFusedLongW<T, PD>), dim3(gridFor(nLongPts_)),
+                           dim3(kBlk), 0, stream_, nLongPts_, dLongPts_,
+                           dHllInv_, dTemp_, dW_);
+        const int fg =
+            (nFlagWins_ + 3) / 4 < 8192 ? (nFlagWins_ + 3) / 4 : 8192;
+        hipLaunchKernelGGL((kFusedLongScatter<T, CD, PD, RD, IM, HI>),
+                           dim3(fg < 1 ? 1 : fg), dim3(256), 0, stream_,
+                           nFlagWins_, dFlagWins_, dWinLo_, dWinHi_, nL_,
+                           dCamOf_, dPtOf_, dHpl_,
+                           (const T* const*)dJSlots_, dInfo_, lossKind_,
+                           lossD2_, xv, dW_, out);
+      }
+    };
+    using TrueT = std::integral_constant<bool, true>;
+    using FalseT = std::integral_constant<bool, false>;
+    if (implicit_ && hasInfo_) launch(TrueT{}, TrueT{});
+    else if (implicit_) launch(TrueT{}, FalseT{});
+    else launch(FalseT{}, FalseT{});
+  }
+
+  // q = S x = HppD x - E Cinv E^T x  (ONE CD*ncam allreduce; the reference's
   // site A4 needed an additional 3*npt allreduce here).  withDot fuses the
   // x^T q dot partials into the B-apply pass (used by the PCG body).
   void schurApply(const T* xv, T* q, bool withDot = false) {
-    spmvEtx(xv, dTemp_);
-    applyCinv(dTemp_, dW_);
-    spmvEx(dW_, q);
+    if (useFused_) {
+      schurFusedEcE(xv, q);
+    } else {
+      spmvEtx(xv, dTemp_);
+      applyCinv(dTemp_, dW_);
+      spmvEx(dW_, q);
+    }
     allreduce(q, nc_, ncclSum);
     if (withDot)
       hipLaunchKernelGGL((kBApplyDot<T, CD>), dim3(gridFor(nc_)), dim3(kBlk),
@@ -1962,6 +2312,14 @@ class GpuEngine final : public Engine<T> {
   T lossD2_ = T(1);
   // MFMA assembly experiment (fp64 BAL only), opt-in via MEGBA_MFMA=1
   bool useMfma_ = getenv("MEGBA_MFMA") != nullptr;
+  // Fused one-pass E Cinv E^T apply (opt-in while being measured)
+  bool useFused_ = getenv("MEGBA_FUSED") != nullptr;
+  int nWin_ = 0, nFlagWins_ = 0, nLongPts_ = 0;
+  int64_t* dWinLo_{};
+  int64_t* dWinHi_{};
+  unsigned char* dWinFlag_{};
+  int* dFlagWins_{};
+  int* dLongPts_{};
   int cur_ = 0;
   bool freshCur_ = false;  // current r/J buffers hold the last forward()
   int nChunks_ = 0;

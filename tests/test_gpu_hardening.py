@@ -156,3 +156,66 @@ def test_mfma_assembly_matches_default(schur, weighted):
         scale = np.abs(d_ref[key]).max() or 1.0
         np.testing.assert_allclose(d_mfma[key], d_ref[key], rtol=1e-10,
                                    atol=1e-12 * scale, err_msg=key)
+
+
+def _fused_pair_solve(schur, cams, pts, ci, pi, meas, info=None):
+    def run(env):
+        saved = {k: os.environ.get(k) for k in env}
+        os.environ.update(env)
+        try:
+            p = mb.BAProblem(cams, pts, ci, pi, meas, info=info)
+            p.build(device="gpu", schur=schur)
+            p.forward()
+            p.accept_forward()
+            p.build_linear_system()
+            p.process_diag(1e4)
+            n = p.solve_linear(max_iter=30, tol=0.0, refuse_ratio=1e30)
+            return n, p.dump()["deltaX"]
+        finally:
+            for k, v in saved.items():
+                if v is None:
+                    os.environ.pop(k, None)
+                else:
+                    os.environ[k] = v
+
+    n1, dx1 = run({})
+    n2, dx2 = run({"MEGBA_FUSED": "1"})
+    assert n1 == n2
+    scale = np.abs(dx1).max() or 1.0
+    np.testing.assert_allclose(dx2, dx1, rtol=1e-6, atol=1e-9 * scale)
+
+
+@pytest.mark.parametrize("schur", ["explicit", "implicit"])
+def test_fused_schur_matches_default(schur):
+    cams, pts, ci, pi, meas = mb.synthesize_bal(24, 400, 3600, seed=14)
+    _fused_pair_solve(schur, cams, pts, ci, pi, meas)
+
+
+@pytest.mark.parametrize("schur", ["explicit", "implicit"])
+def test_fused_schur_long_run_point(schur):
+    """A landmark observed by >64 cameras exercises the flagged-window /
+    long-run fallback path of the fused Schur apply."""
+    rng = np.random.default_rng(8)
+    cams, pts, ci, pi, meas = mb.synthesize_bal(90, 300, 6000, seed=15)
+    # steal one edge from every camera for point 0 -> degree 90 > 64,
+    # touching only points that keep >= 3 observations
+    counts = np.bincount(pi, minlength=300)
+    taken = 0
+    for e in rng.permutation(len(pi)):
+        if taken >= 90:
+            break
+        if pi[e] != 0 and counts[pi[e]] > 3:
+            counts[pi[e]] -= 1
+            pi[e] = 0
+            ci[e] = taken % 90
+            taken += 1
+    assert np.bincount(pi, minlength=300)[0] > 64
+    _fused_pair_solve(schur, cams, pts, ci, pi, meas)
+
+
+def test_fused_schur_weighted_loss():
+    rng = np.random.default_rng(9)
+    cams, pts, ci, pi, meas = mb.synthesize_bal(20, 300, 2600, seed=16)
+    w = rng.uniform(0.5, 2.0, (len(ci), 2))
+    info = np.stack([w[:, 0], 0.1 * np.ones(len(ci)), w[:, 1]], axis=1)
+    _fused_pair_solve("implicit", cams, pts, ci, pi, meas, info=info)
