@@ -678,6 +678,219 @@ __global__ __launch_bounds__(64) void kSpmvExImpG(
 }
 
 // ---------------------------------------------------------------------------
+// Packed-J layout for the per-iteration implicit products
+// ---------------------------------------------------------------------------
+// The grad-major J layout ([k][nL], one 4/8-byte load per value) makes the
+// two matrix-free PCG kernels instruction-ISSUE-bound: ~24 scalar loads
+// per edge dominate the ~80-instruction inner loop (measured ~2x the byte
+// floor on final13682-fp32, profiles/r01_final13682_implicit_fp32.md).
+// Repacking the 24 per-edge J values into 16-byte vector groups
+// ([group][nL][VEC], VEC = 16B/sizeof(T)) turns them into NG=24/VEC
+// dwordx4 loads (6 for fp32, 12 for fp64), still fully coalesced across
+// the 64 consecutive-edge lanes.  The pack runs once per ACCEPTED LM step
+// (its cost amortizes over the ~100 PCG iterations that read it); the
+// packed buffers are single-buffered so the captured PCG graph needs no
+// pointer indirection for them.
+template <typename T>
+struct PackVec {
+  static constexpr int VEC = 16 / sizeof(T);
+  typedef T type __attribute__((ext_vector_type(16 / sizeof(T))));
+};
+
+// Primary-order pack of the accepted [Jc(CR), Jp(PR)] set (E^T x source).
+template <typename T, int CD, int PD, int RD>
+__global__ void kPackJPrimary(int64_t nL, const T* const* __restrict__ jSlots,
+                              T* __restrict__ out) {
+  using TV = typename PackVec<T>::type;
+  constexpr int VEC = PackVec<T>::VEC;
+  constexpr int CR = CD * RD, PR = PD * RD;
+  constexpr int NV = CR + PR;
+  constexpr int NG = (NV + VEC - 1) / VEC;
+  const T* Jc = jSlots[0];
+  const T* Jp = jSlots[1];
+  TV* o = (TV*)out;
+  for (int64_t j = blockIdx.x * (int64_t)kBlk + threadIdx.x; j < nL;
+       j += (int64_t)gridDim.x * kBlk) {
+    for (int g = 0; g < NG; ++g) {
+      TV v;
+      for (int q = 0; q < VEC; ++q) {
+        const int k = g * VEC + q;
+        v[q] = k < CR ? Jc[(int64_t)k * nL + j]
+                      : (k < NV ? Jp[(int64_t)(k - CR) * nL + j] : T(0));
+      }
+      o[(int64_t)g * nL + j] = v;
+    }
+  }
+}
+
+// Packed E^T x: identical math to kSpmvEtx<IMP>, vector-group loads.
+template <typename T, int CD, int PD, int RD, bool HASINFO>
+__global__ void kSpmvEtxPk(int64_t nL, const int* __restrict__ camOf,
+                           const int* __restrict__ ptOf,
+                           const T* __restrict__ Jpk,
+                           const T* const* __restrict__ jSlots,
+                           const T* __restrict__ info, int lossKind, T lossD2,
+                           const T* __restrict__ x, T* __restrict__ out) {
+  using TV = typename PackVec<T>::type;
+  constexpr int VEC = PackVec<T>::VEC;
+  constexpr int RW = RD * (RD + 1) / 2;
+  constexpr int CR = CD * RD, PR = PD * RD;
+  constexpr int NG = (CR + PR + VEC - 1) / VEC;
+  const T* rBak = jSlots[2];
+  const int lane = threadIdx.x & 63;
+  const int64_t nWork = ((nL + kBlk - 1) / kBlk) * (int64_t)kBlk;
+  for (int64_t j0 = blockIdx.x * (int64_t)kBlk + threadIdx.x; j0 < nWork;
+       j0 += (int64_t)gridDim.x * kBlk) {
+    const bool active = j0 < nL;
+    const int64_t j = active ? j0 : nL - 1;
+    const int pt = ptOf[j];
+    T o[PD];
+    for (int k = 0; k < PD; ++k) o[k] = T(0);
+    if (active) {
+      const T* xc = x + (int64_t)camOf[j] * CD;
+      TV buf[NG];
+      const TV* src = (const TV*)Jpk;
+#pragma unroll
+      for (int g = 0; g < NG; ++g) buf[g] = src[(int64_t)g * nL + j];
+      T u[RD];
+#pragma unroll
+      for (int rr = 0; rr < RD; ++rr) {
+        T v = T(0);
+#pragma unroll
+        for (int i = 0; i < CD; ++i) {
+          constexpr int dummy = 0;
+          (void)dummy;
+          const int k = i * RD + rr;
+          v += buf[k / VEC][k % VEC] * xc[i];
+        }
+        u[rr] = v;
+      }
+      if (HASINFO) {
+        T wu[RD];
+        for (int i = 0; i < RD; ++i) {
+          T v = T(0);
+          for (int k = 0; k < RD; ++k)
+            v += info[RW * j + symIdx<RD>(i, k)] * u[k];
+          wu[i] = v;
+        }
+        for (int i = 0; i < RD; ++i) u[i] = wu[i];
+      }
+      if (lossKind) {
+        T ss = T(0);
+        for (int rr = 0; rr < RD; ++rr) {
+          const T rv = rBak[(int64_t)rr * nL + j];
+          ss += rv * rv;
+        }
+        const T w = lossWeight(lossKind, lossD2, ss);
+        for (int rr = 0; rr < RD; ++rr) u[rr] *= w;
+      }
+#pragma unroll
+      for (int k = 0; k < PD; ++k) {
+        T v = T(0);
+#pragma unroll
+        for (int rr = 0; rr < RD; ++rr) {
+          const int kk = CR + k * RD + rr;
+          v += buf[kk / VEC][kk % VEC] * u[rr];
+        }
+        o[k] = v;
+      }
+    }
+    for (int off = 1; off < 64; off <<= 1) {
+      const int ppt = __shfl_up(pt, off, 64);
+      const bool join = lane >= off && ppt == pt;
+      if (__ballot(join) == 0ull) break;
+      T a[PD];
+      for (int k = 0; k < PD; ++k) a[k] = __shfl_up(o[k], off, 64);
+      if (join)
+        for (int k = 0; k < PD; ++k) o[k] += a[k];
+    }
+    const int nextPt = __shfl_down(pt, 1, 64);
+    const bool tail = active && (lane == 63 || nextPt != pt || j0 == nL - 1);
+    if (tail)
+      for (int k = 0; k < PD; ++k) atomicAdd(&out[PD * pt + k], o[k]);
+  }
+}
+
+// Packed E w over the cam-sorted [wJc(CR), Jp(PR)] groups.
+template <typename T, int CD, int PD, int RD>
+__global__ __launch_bounds__(64) void kSpmvExPk(
+    int nChunks, const int* __restrict__ chCam, const int* __restrict__ chLo,
+    const int* __restrict__ chHi, const int* __restrict__ ptOfCam,
+    const T* __restrict__ JCamPk, int64_t nL, const T* __restrict__ w,
+    T* __restrict__ out) {
+  using TV = typename PackVec<T>::type;
+  constexpr int VEC = PackVec<T>::VEC;
+  constexpr int CR = CD * RD, PR = PD * RD;
+  constexpr int NG = (CR + PR + VEC - 1) / VEC;
+  const int chunk = blockIdx.x;
+  if (chunk >= nChunks) return;
+  const int cam = chCam[chunk];
+  T acc[CD];
+  for (int i = 0; i < CD; ++i) acc[i] = T(0);
+  const int lo = chLo[chunk], hi = chHi[chunk];
+  const TV* src = (const TV*)JCamPk;
+  for (int j = lo + (int)threadIdx.x; j < hi; j += 64) {
+    const T* wp = w + (int64_t)ptOfCam[j] * PD;
+    TV buf[NG];
+#pragma unroll
+    for (int g = 0; g < NG; ++g) buf[g] = src[(int64_t)g * nL + j];
+    T u[RD];
+#pragma unroll
+    for (int rr = 0; rr < RD; ++rr) {
+      T v = T(0);
+#pragma unroll
+      for (int k = 0; k < PD; ++k) {
+        const int kk = CR + k * RD + rr;
+        v += buf[kk / VEC][kk % VEC] * wp[k];
+      }
+      u[rr] = v;
+    }
+#pragma unroll
+    for (int i = 0; i < CD; ++i) {
+      T v = T(0);
+#pragma unroll
+      for (int rr = 0; rr < RD; ++rr) {
+        const int k = i * RD + rr;
+        v += buf[k / VEC][k % VEC] * u[rr];
+      }
+      acc[i] += v;
+    }
+  }
+  for (int off = 32; off > 0; off >>= 1)
+    for (int i = 0; i < CD; ++i) acc[i] += __shfl_down(acc[i], off, 64);
+  if (threadIdx.x == 0) {
+    T* oc = out + (int64_t)cam * CD;
+    for (int i = 0; i < CD; ++i) atomicAdd(&oc[i], acc[i]);
+  }
+}
+
+// Cam-sorted packed finalize: slab -> [wJc, Jp] vector groups.
+template <typename T, int CD, int PD, int RD, bool HASINFO>
+__global__ void kFinalizeCamImpPk(int64_t nL, const T* __restrict__ slab,
+                                  T* __restrict__ JCamPk) {
+  using L = SlabLayout<CD, PD, RD, false, HASINFO>;
+  using TV = typename PackVec<T>::type;
+  constexpr int VEC = PackVec<T>::VEC;
+  constexpr int CR = CD * RD, PR = PD * RD;
+  constexpr int NG = (CR + PR + VEC - 1) / VEC;
+  constexpr int woff = HASINFO ? L::WJCOFF : L::JCOFF;
+  TV* o = (TV*)JCamPk;
+  for (int64_t j = blockIdx.x * (int64_t)kBlk + threadIdx.x; j < nL;
+       j += (int64_t)gridDim.x * kBlk) {
+    const T* row = slab + j * L::SW;
+    for (int g = 0; g < NG; ++g) {
+      TV v;
+      for (int q = 0; q < VEC; ++q) {
+        const int k = g * VEC + q;
+        v[q] = k < CR ? row[woff + k]
+                      : (k < CR + PR ? row[L::JPOFF + (k - CR)] : T(0));
+      }
+      o[(int64_t)g * nL + j] = v;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Damping, block inverse
 // ---------------------------------------------------------------------------
 template <typename T, int D>
@@ -1617,8 +1830,11 @@ class GpuEngine final : public Engine<T> {
     dJSlots_ = dalloc<const T*>(3);
     updateJSlots();
     if (implicit_) {
-      dJcCam_ = dalloc<T>(nL_ * CR);
-      dJpCam_ = dalloc<T>(nL_ * PR);
+      // packed vector-group layouts (16B groups; padded to a full group)
+      constexpr int VEC = 16 / (int)sizeof(T);
+      constexpr int NG = (CR + PR + VEC - 1) / VEC;
+      dJPk_ = dalloc<T>(nL_ * NG * VEC);
+      dJCamPk_ = dalloc<T>(nL_ * NG * VEC);
     }
     if (!implicit_) {
       dHpl_ = dalloc<T>(nL_ * CP);
@@ -1729,13 +1945,17 @@ class GpuEngine final : public Engine<T> {
                            dSlab_, dHplCam_);
     } else {
       if (weighted())
-        hipLaunchKernelGGL((kFinalizeCamImp<T, CD, PD, RD, true>),
+        hipLaunchKernelGGL((kFinalizeCamImpPk<T, CD, PD, RD, true>),
                            dim3(gridFor(nL_)), dim3(kBlk), 0, stream_, nL_,
-                           dSlab_, dJcCam_, dJpCam_);
+                           dSlab_, dJCamPk_);
       else
-        hipLaunchKernelGGL((kFinalizeCamImp<T, CD, PD, RD, false>),
+        hipLaunchKernelGGL((kFinalizeCamImpPk<T, CD, PD, RD, false>),
                            dim3(gridFor(nL_)), dim3(kBlk), 0, stream_, nL_,
-                           dSlab_, dJcCam_, dJpCam_);
+                           dSlab_, dJCamPk_);
+      // primary-order packed copy of the accepted J set (E^T x source)
+      hipLaunchKernelGGL((kPackJPrimary<T, CD, PD, RD>), dim3(gridFor(nL_)),
+                         dim3(kBlk), 0, stream_, nL_,
+                         (const T* const*)dJSlots_, dJPk_);
     }
     // Only the small camera-side quantities cross ranks (the reference
     // allreduced Hpp, Hll AND g, its site A1).
@@ -2145,15 +2365,15 @@ class GpuEngine final : public Engine<T> {
                        (int64_t)npL_ * PD);
     if (implicit_) {
       if (hasInfo_)
-        hipLaunchKernelGGL((kSpmvEtx<T, CD, PD, RD, true, true>),
+        hipLaunchKernelGGL((kSpmvEtxPk<T, CD, PD, RD, true>),
                            dim3(gridFor(nL_)), dim3(kBlk), 0, stream_, nL_,
-                           dCamOf_, dPtOf_, (const T*)nullptr,
+                           dCamOf_, dPtOf_, dJPk_,
                            (const T* const*)dJSlots_, dInfo_, lossKind_,
                            lossD2_, xv, out);
       else
-        hipLaunchKernelGGL((kSpmvEtx<T, CD, PD, RD, true, false>),
+        hipLaunchKernelGGL((kSpmvEtxPk<T, CD, PD, RD, false>),
                            dim3(gridFor(nL_)), dim3(kBlk), 0, stream_, nL_,
-                           dCamOf_, dPtOf_, (const T*)nullptr,
+                           dCamOf_, dPtOf_, dJPk_,
                            (const T* const*)dJSlots_, (const T*)nullptr,
                            lossKind_, lossD2_, xv, out);
     } else {
@@ -2167,9 +2387,9 @@ class GpuEngine final : public Engine<T> {
     HIP_CHECK(hipMemsetAsync(out, 0, nc_ * sizeof(T), stream_));
     if (implicit_) {
       if (nChunks_ == 0) return;
-      hipLaunchKernelGGL((kSpmvExImpG<T, CD, PD, RD>), dim3(nChunks_),
+      hipLaunchKernelGGL((kSpmvExPk<T, CD, PD, RD>), dim3(nChunks_),
                          dim3(64), 0, stream_, nChunks_, dChCam_, dChLo_,
-                         dChHi_, dPtOfCam_, dJcCam_, dJpCam_, nL_, wv, out);
+                         dChHi_, dPtOfCam_, dJCamPk_, nL_, wv, out);
       return;
     }
     if (nChunks_ > 0)
@@ -2330,7 +2550,7 @@ class GpuEngine final : public Engine<T> {
   T *dParams_{}, *dParamsBak_{};
   T *dR_[2]{}, *dJc_[2]{}, *dJp_[2]{};
   T *dHpp_{}, *dHll_{}, *dHpl_{}, *dHplCam_{}, *dSlab_{}, *dG_{}, *dGBak_{};
-  T *dJcCam_{}, *dJpCam_{};  // implicit: cam-sorted grad-major wJc / Jp
+  T *dJPk_{}, *dJCamPk_{};  // implicit: packed [J..] vector groups
   T *dHppD_{}, *dHllD_{}, *dHppInv_{}, *dHllInv_{};
   T *dDeltaX_{}, *dDeltaXBak_{};
   T *dP_{}, *dRr_{}, *dZ_{}, *dQ_{}, *dV_{}, *dW_{}, *dTemp_{}, *dXBak_{},
