@@ -829,8 +829,15 @@ __global__ __launch_bounds__(64) void kSpmvExPk(
   for (int i = 0; i < CD; ++i) acc[i] = T(0);
   const int lo = chLo[chunk], hi = chHi[chunk];
   const TV* src = (const TV*)JCamPk;
+  // w is stored 4-padded (stride 4 elements, 16B/32B aligned) so the
+  // per-edge gather is 1 (fp32) or 2 (fp64) vector loads instead of PD
+  // scalar loads — the gather is the TA-request hot spot of this kernel.
+  constexpr int WL = (4 * (int)sizeof(T) + 15) / 16;
   for (int j = lo + (int)threadIdx.x; j < hi; j += 64) {
-    const T* wp = w + (int64_t)ptOfCam[j] * PD;
+    const TV* wp4 = (const TV*)(w + (int64_t)ptOfCam[j] * 4);
+    TV wbuf[WL];
+#pragma unroll
+    for (int l = 0; l < WL; ++l) wbuf[l] = wp4[l];
     TV buf[NG];
 #pragma unroll
     for (int g = 0; g < NG; ++g) buf[g] = src[(int64_t)g * nL + j];
@@ -841,7 +848,7 @@ __global__ __launch_bounds__(64) void kSpmvExPk(
 #pragma unroll
       for (int k = 0; k < PD; ++k) {
         const int kk = CR + k * RD + rr;
-        v += buf[kk / VEC][kk % VEC] * wp[k];
+        v += buf[kk / VEC][kk % VEC] * wbuf[k / VEC][k % VEC];
       }
       u[rr] = v;
     }
@@ -1432,6 +1439,22 @@ __global__ void kBlockDiagMatVec(int nBlk, const T* __restrict__ A,
   }
 }
 
+// Cinv apply writing the 4-padded w layout read by kSpmvExPk.
+template <typename T, int PD>
+__global__ void kCinvPad(int nBlk, const T* __restrict__ A,
+                         const T* __restrict__ x, T* __restrict__ yPad) {
+  for (int64_t idx = blockIdx.x * (int64_t)kBlk + threadIdx.x;
+       idx < (int64_t)nBlk * PD; idx += (int64_t)gridDim.x * kBlk) {
+    const int64_t b = idx / PD;
+    const int rrow = (int)(idx % PD);
+    const T* row = A + b * PD * PD + (int64_t)rrow * PD;
+    const T* xb = x + b * PD;
+    T sv = T(0);
+    for (int j = 0; j < PD; ++j) sv += row[j] * xb[j];
+    yPad[b * 4 + rrow] = sv;
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Small vector kernels
 // ---------------------------------------------------------------------------
@@ -1835,6 +1858,7 @@ class GpuEngine final : public Engine<T> {
       constexpr int NG = (CR + PR + VEC - 1) / VEC;
       dJPk_ = dalloc<T>(nL_ * NG * VEC);
       dJCamPk_ = dalloc<T>(nL_ * NG * VEC);
+      dWPad_ = dalloc<T>((int64_t)npt_ * 4);
     }
     if (!implicit_) {
       dHpl_ = dalloc<T>(nL_ * CP);
@@ -2024,9 +2048,8 @@ class GpuEngine final : public Engine<T> {
 
     const T* gc = dG_;
     const T* gp = dG_ + nc_;
-    // v = gc/world - E Cinv gp  (partial, then the 9*ncam allreduce)
-    applyCinv(gp, dW_);
-    spmvEx(dW_, dV_);
+    // v = gc/world - E Cinv gp  (partial, then the CD*ncam allreduce)
+    cinvThenEx(gp, dV_);
     hipLaunchKernelGGL(kVMake<T>, dim3(gridFor(nc_)), dim3(kBlk), 0, stream_,
                        nc_, gc, T(1) / T(world_), dV_);
     allreduce(dV_, nc_, ncclSum);
@@ -2359,6 +2382,21 @@ class GpuEngine final : public Engine<T> {
     blockMatVec<PD, 0>(npL_, dHllInv_ + (int64_t)ptLo_ * PP,
                        in + (int64_t)ptLo_ * PD, out + (int64_t)ptLo_ * PD);
   }
+  // w = Cinv in, then out += E w.  The implicit path stores w 4-padded so
+  // the E-side gather is a single aligned vector load per edge.
+  void cinvThenEx(const T* in, T* out) {
+    if (implicit_) {
+      hipLaunchKernelGGL((kCinvPad<T, PD>),
+                         dim3(gridFor((int64_t)npL_ * PD)), dim3(kBlk), 0,
+                         stream_, npL_, dHllInv_ + (int64_t)ptLo_ * PP,
+                         in + (int64_t)ptLo_ * PD,
+                         dWPad_ + (int64_t)ptLo_ * 4);
+      spmvEx(dWPad_, out);
+    } else {
+      applyCinv(in, dW_);
+      spmvEx(dW_, out);
+    }
+  }
   void spmvEtx(const T* xv, T* out) {
     hipLaunchKernelGGL(kZeroRange<T>, dim3(gridFor((int64_t)npL_ * PD)),
                        dim3(kBlk), 0, stream_, out + (int64_t)ptLo_ * PD,
@@ -2505,8 +2543,7 @@ class GpuEngine final : public Engine<T> {
       schurFusedEcE(xv, q);
     } else {
       spmvEtx(xv, dTemp_);
-      applyCinv(dTemp_, dW_);
-      spmvEx(dW_, q);
+      cinvThenEx(dTemp_, q);
     }
     allreduce(q, nc_, ncclSum);
     if (withDot)
@@ -2551,6 +2588,7 @@ class GpuEngine final : public Engine<T> {
   T *dR_[2]{}, *dJc_[2]{}, *dJp_[2]{};
   T *dHpp_{}, *dHll_{}, *dHpl_{}, *dHplCam_{}, *dSlab_{}, *dG_{}, *dGBak_{};
   T *dJPk_{}, *dJCamPk_{};  // implicit: packed [J..] vector groups
+  T* dWPad_{};              // implicit: 4-padded w for the E-side gather
   T *dHppD_{}, *dHllD_{}, *dHppInv_{}, *dHllInv_{};
   T *dDeltaX_{}, *dDeltaXBak_{};
   T *dP_{}, *dRr_{}, *dZ_{}, *dQ_{}, *dV_{}, *dW_{}, *dTemp_{}, *dXBak_{},
