@@ -1785,12 +1785,51 @@ class GpuEngine final : public Engine<T> {
       std::vector<int> cCam, cLo, cHi;
       int CHUNK = 256;  // cam-chunk rows (tunable: MEGBA_CHUNK)
       if (const char* c = getenv("MEGBA_CHUNK")) CHUNK = std::atoi(c);
-      for (int c = 0; c < ncam_; ++c)
-        for (int s = rowPtr[c]; s < rowPtr[c + 1]; s += CHUNK) {
+      // Point-band blocking: each camera's slab rows are pt-sorted (the
+      // cursor scatter preserves the primary order), so cutting chunks at
+      // point-band boundaries and ordering chunks band-major makes every
+      // concurrently-resident chunk gather w from the SAME ~2 MB slice —
+      // small enough to stay in the XCD L2 instead of being flushed by
+      // the streamed J reads (PMC: 78% SQ_WAIT on the unbanded gather,
+      // profiles/r02_gather_bands.md).  Band = whole problem for small
+      // npt (table unchanged).
+      int bandPts = (2 << 20) / (4 * (int)sizeof(T));  // ~2 MB of padded w
+      if (const char* b = getenv("MEGBA_BAND")) bandPts = std::atoi(b);
+      if (bandPts < 1) bandPts = npt_;
+      const int nBands = (npt_ + bandPts - 1) / bandPts;
+      std::vector<int> cBand;
+      for (int c = 0; c < ncam_; ++c) {
+        int s = rowPtr[c];
+        while (s < rowPtr[c + 1]) {
+          const int band = ptOfCam[s] / bandPts;
+          const int bandEndPt = (band + 1) * bandPts;
+          int e = s;
+          const int lim = std::min(s + CHUNK, rowPtr[c + 1]);
+          while (e < lim && ptOfCam[e] < bandEndPt) ++e;
           cCam.push_back(c);
           cLo.push_back(s);
-          cHi.push_back(std::min(s + CHUNK, rowPtr[c + 1]));
+          cHi.push_back(e);
+          cBand.push_back(band);
+          s = e;
         }
+      }
+      if (nBands > 1) {
+        // stable band-major order
+        std::vector<int> ord(cCam.size());
+        for (size_t i = 0; i < ord.size(); ++i) ord[i] = (int)i;
+        std::stable_sort(ord.begin(), ord.end(), [&](int a, int b) {
+          return cBand[a] < cBand[b];
+        });
+        std::vector<int> c2(cCam.size()), l2(cCam.size()), h2(cCam.size());
+        for (size_t i = 0; i < ord.size(); ++i) {
+          c2[i] = cCam[ord[i]];
+          l2[i] = cLo[ord[i]];
+          h2[i] = cHi[ord[i]];
+        }
+        cCam.swap(c2);
+        cLo.swap(l2);
+        cHi.swap(h2);
+      }
       nChunks_ = (int)cCam.size();
       dChCam_ = dalloc<int>(nChunks_);
       dChLo_ = dalloc<int>(nChunks_);
