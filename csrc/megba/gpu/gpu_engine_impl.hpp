@@ -1799,7 +1799,7 @@ class GpuEngine final : public Engine<T> {
       // ms/step when 20k cams x 153 bands shattered the table into
       // 16-edge chunks).
       const int64_t occFloor =
-          nL_ > 0 ? (int64_t)128 * npt_ * ncam_ / (nL_ * world_) : 0;
+          nL_ > 0 ? (int64_t)128 * npt_ * ncam_ / nL_ : 0;
       if (occFloor > bandPts)
         bandPts = (int)std::min<int64_t>(occFloor, npt_);
       if (const char* b = getenv("MEGBA_BAND")) bandPts = std::atoi(b);
