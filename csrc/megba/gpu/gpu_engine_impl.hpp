@@ -1794,12 +1794,12 @@ class GpuEngine final : public Engine<T> {
       // profiles/r02_gather_bands.md).  Band = whole problem for small
       // npt (table unchanged).
       int bandPts = (2 << 20) / (4 * (int)sizeof(T));  // ~2 MB of padded w
-      // Occupancy floor: keep >=128 edges per (cam, band) chunk on average
-      // or the chunk waves run mostly empty (synth20k measured 694 vs 618
-      // ms/step when 20k cams x 153 bands shattered the table into
-      // 16-edge chunks).
+      // Occupancy floor: keep >=64 edges (one full wave pass) per
+      // (cam, band) chunk on average, or the chunk waves run mostly empty
+      // (synth20k measured 694 vs 605 ms/step when 20k cams x 153 bands
+      // shattered the table into 16-edge chunks).
       const int64_t occFloor =
-          nL_ > 0 ? (int64_t)128 * npt_ * ncam_ / nL_ : 0;
+          nL_ > 0 ? (int64_t)64 * npt_ * ncam_ / nL_ : 0;
       if (occFloor > bandPts)
         bandPts = (int)std::min<int64_t>(occFloor, npt_);
       if (const char* b = getenv("MEGBA_BAND")) bandPts = std::atoi(b);
