@@ -82,23 +82,29 @@ def pybind_flags():
     return out
 
 
+EXAMPLES = ["bal_solve_cpp", "bal_custom_edge_cpp"]
+
+
 def build_examples(objs, force=False):
-    """Native C++ CLI example linked against the same objects."""
-    src = ROOT / "examples" / "bal_solve_cpp.cpp"
-    out = ROOT / "examples" / "bal_solve_cpp"
-    obj = BUILD / "examples_bal_solve_cpp.o"
+    """Native C++ CLI examples linked against the same objects."""
     hdr = newest_header_mtime()
-    if force or not obj.exists() or obj.stat().st_mtime < max(src.stat().st_mtime, hdr):
-        print("  CC examples/bal_solve_cpp.cpp", flush=True)
-        subprocess.run([HIPCC, *COMMON_FLAGS, "-c", str(src), "-o", str(obj)],
-                       check=True)
     core = [o for o in objs if "bindings" not in o.name]
-    if force or not out.exists() or any(
-            o.stat().st_mtime > out.stat().st_mtime for o in core + [obj]):
-        print("  LD examples/bal_solve_cpp", flush=True)
-        subprocess.run([HIPCC, "-fopenmp", str(obj), *[str(o) for o in core],
-                        "-L/opt/rocm/lib", "-lrccl", "-lamdhip64",
-                        "-o", str(out)], check=True)
+    for name in EXAMPLES:
+        src = ROOT / "examples" / f"{name}.cpp"
+        out = ROOT / "examples" / name
+        obj = BUILD / f"examples_{name}.o"
+        if force or not obj.exists() or obj.stat().st_mtime < max(
+                src.stat().st_mtime, hdr):
+            print(f"  CC examples/{name}.cpp", flush=True)
+            subprocess.run([HIPCC, *COMMON_FLAGS, "-c", str(src),
+                            "-o", str(obj)], check=True)
+        if force or not out.exists() or any(
+                o.stat().st_mtime > out.stat().st_mtime for o in core + [obj]):
+            print(f"  LD examples/{name}", flush=True)
+            subprocess.run([HIPCC, "-fopenmp", str(obj),
+                            *[str(o) for o in core],
+                            "-L/opt/rocm/lib", "-lrccl", "-lamdhip64",
+                            "-o", str(out)], check=True)
 
 
 def build(force=False, debug=False):

@@ -21,6 +21,7 @@
 
 #include "common.hpp"
 #include "cpu_engine.hpp"
+#include "custom.hpp"
 #include "lm.hpp"
 #include "problem.hpp"
 #ifdef MEGBA_WITH_GPU
@@ -33,12 +34,19 @@ enum class VertexKind { CAMERA, POINT };
 
 struct BaseVertex {
   VertexKind kind;
-  std::vector<double> estimation;  // 9 (camera) or 3 (point)
+  std::vector<double> estimation;  // camDim (camera) or 3 (point)
   bool fixed = false;
   int slot = -1;  // assigned by GraphProblem::appendVertex
 
   BaseVertex(VertexKind k, const double* est) : kind(k) {
     estimation.assign(est, est + (k == VertexKind::CAMERA ? 9 : 3));
+  }
+  // Generic-dims vertex (camDim in {9,6,4}; points are 3-dimensional).
+  BaseVertex(VertexKind k, const double* est, int dim) : kind(k) {
+    MEGBA_CHECK(k == VertexKind::POINT ? dim == 3
+                                       : (dim == 9 || dim == 6 || dim == 4),
+                "vertex dim outside the compiled set");
+    estimation.assign(est, est + dim);
   }
 };
 
@@ -109,26 +117,41 @@ class GraphProblem {
   int64_t numEdges() const { return (int64_t)edges_.size(); }
   int numVertices() const { return (int)(cams_.size() + pts_.size()); }
 
+  // Runtime user-defined residual (C++ counterpart of the reference's
+  // BaseEdge::forward, examples/BAL_Double.cpp:16-34): called once per
+  // forward pass with per-observation leaf JetVectors; must return resDim
+  // dense residual JetVectors.  See examples/bal_custom_edge_cpp.cpp.
+  void setCustomForward(CustomForward<double> f) {
+    customForward_ = std::move(f);
+  }
+
   // Assemble, run LM on the chosen device, write estimations back.
   LMReport solve(const ProblemOption& popt, const AlgoOptionLM& algo,
                  const SolverOptionPCG& sopt) {
     BAProblemHost prob = assemble();
-    ProblemIndex ix = buildIndex(prob, popt.worldSize);
+    ProblemOption opt = popt;
+    opt.camDim = prob.camDim;
+    opt.ptDim = prob.ptDim;
+    opt.resDim = prob.resDim;
+    ProblemIndex ix = buildIndex(prob, opt.worldSize);
     std::unique_ptr<Engine<double>> eng;
-    if (popt.device == Device::CPU) {
-      eng = makeCpuEngine<double>(prob, ix, popt, nullptr);
+    if (opt.device == Device::CPU) {
+      eng = makeCpuEngine<double>(prob, ix, opt, nullptr, customForward_);
     } else {
 #ifdef MEGBA_WITH_GPU
-      eng = makeGpuEngine<double>(prob, ix, popt, std::string());
+      eng = makeGpuEngine<double>(prob, ix, opt, std::string(),
+                                  customForward_);
 #else
       MEGBA_CHECK(false, "built without GPU support");
 #endif
     }
     LMReport rep = runLM<double>(*eng, algo, sopt);
-    std::vector<double> camOut(cams_.size() * 9), ptOut(pts_.size() * 3);
+    const int cd = prob.camDim;
+    std::vector<double> camOut(cams_.size() * cd), ptOut(pts_.size() * 3);
     eng->getParams(camOut.data(), ptOut.data());
     for (size_t i = 0; i < cams_.size(); ++i)
-      for (int k = 0; k < 9; ++k) cams_[i]->estimation[k] = camOut[9 * i + k];
+      for (int k = 0; k < cd; ++k)
+        cams_[i]->estimation[k] = camOut[cd * i + k];
     for (size_t i = 0; i < pts_.size(); ++i)
       for (int k = 0; k < 3; ++k) pts_[i]->estimation[k] = ptOut[3 * i + k];
     return rep;
@@ -141,10 +164,15 @@ class GraphProblem {
     p.ncam = (int)cams_.size();
     p.npt = (int)pts_.size();
     p.nobs = (int64_t)edges_.size();
-    p.cams.resize(p.ncam * 9);
-    p.pts.resize(p.npt * 3);
+    p.camDim = (int)cams_[0]->estimation.size();
+    for (const auto* v : cams_)
+      MEGBA_CHECK((int)v->estimation.size() == p.camDim,
+                  "mixed camera dims in one problem");
+    p.cams.resize((size_t)p.ncam * p.camDim);
+    p.pts.resize((size_t)p.npt * 3);
     for (int i = 0; i < p.ncam; ++i)
-      for (int k = 0; k < 9; ++k) p.cams[9 * i + k] = cams_[i]->estimation[k];
+      for (int k = 0; k < p.camDim; ++k)
+        p.cams[(size_t)p.camDim * i + k] = cams_[i]->estimation[k];
     for (int i = 0; i < p.npt; ++i)
       for (int k = 0; k < 3; ++k) p.pts[3 * i + k] = pts_[i]->estimation[k];
     p.camIdx.resize(p.nobs);
@@ -175,6 +203,7 @@ class GraphProblem {
 
   std::vector<BaseVertex*> cams_, pts_;
   std::vector<ReprojectionEdge> edges_;
+  CustomForward<double> customForward_;
 };
 
 }  // namespace megba
