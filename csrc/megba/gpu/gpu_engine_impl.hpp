@@ -202,6 +202,131 @@ __global__ __launch_bounds__(256, 2) void kForward(
   if (threadIdx.x == 0) atomicAdd(chi2Acc, sm[0]);
 }
 
+// Value-share forward experiment (env MEGBA_FWD_VS=1,
+// BAL (9,3,2) only): the 4 gradient lanes of an edge recompute the whole
+// VALUE chain of the reprojection; the dual math needs every intermediate
+// value on every lane, so only the EXPENSIVE scalar values (sqrt, sincos,
+// two reciprocals — the transcendental tail of the chain) are computed
+// once on the edge's leader lane and broadcast with __shfl; their jets
+// are then assembled from the analytic derivative rules.  Cheap mul/add
+// values stay redundantly computed (a shuffle costs as much as an FMA).
+template <typename T>
+__global__ __launch_bounds__(256, 2) void kForwardVS(
+    int64_t nL, const int* __restrict__ camOf, const int* __restrict__ ptOf,
+    const T* __restrict__ params, int ncam, const T* __restrict__ meas,
+    const unsigned char* __restrict__ camFixed,
+    const unsigned char* __restrict__ ptFixed, T* __restrict__ rOut,
+    T* __restrict__ Jc, T* __restrict__ Jp, double* chi2Acc, int lossKind,
+    T lossD2) {
+  using J3 = Jet<T, 3>;
+  __shared__ double sm[kBlk];
+  double chi2 = 0.0;
+  const T* ptsBase = params + (int64_t)ncam * 9;
+  const int64_t nWork = nL * 4;
+  const int lane = (int)threadIdx.x & 63;
+  const int leader = lane & ~3;
+  for (int64_t i = blockIdx.x * (int64_t)kBlk + threadIdx.x; i < nWork;
+       i += (int64_t)gridDim.x * kBlk) {
+    const int64_t e = i >> 2;
+    const int sub = (int)(i & 3);
+    const int base = 3 * sub;
+    const T* cp = params + (int64_t)camOf[e] * 9;
+    const T* pp = ptsBase + (int64_t)ptOf[e] * 3;
+    J3 cam[9], pt[3], res[2];
+    for (int k = 0; k < 9; ++k) cam[k] = J3::leaf(cp[k], k - base);
+    for (int k = 0; k < 3; ++k) pt[k] = J3::leaf(pp[k], 9 + k - base);
+    const T m[2] = {meas[2 * e], meas[2 * e + 1]};
+    // ---- Rodrigues with shared transcendental values ----
+    const J3 theta2 =
+        cam[0] * cam[0] + cam[1] * cam[1] + cam[2] * cam[2];
+    J3 P[3];
+    if (theta2.v > T(1e-14)) {  // edge-uniform branch
+      T thv = T(0), stv = T(0), ctv = T(0), thInvV = T(0);
+      if (sub == 0) {
+        thv = ::sqrt(theta2.v);
+        stv = ::sin(thv);
+        ctv = ::cos(thv);
+        thInvV = T(1) / thv;
+      }
+      stv = __shfl(stv, leader, 64);
+      ctv = __shfl(ctv, leader, 64);
+      thInvV = __shfl(thInvV, leader, 64);
+      // jets from the analytic rules: d(sqrt) = d(theta2)/(2 theta) etc.
+      J3 theta, sinth, costh, thetaInv;
+      theta.v = thv == T(0) ? (thv = T(1) / thInvV) : thv;  // non-leader thv
+      const T half = T(0.5) * thInvV;
+      for (int k = 0; k < 3; ++k) theta.d[k] = theta2.d[k] * half;
+      sinth.v = stv;
+      costh.v = ctv;
+      thetaInv.v = thInvV;
+      const T mInv2 = -thInvV * thInvV;
+      for (int k = 0; k < 3; ++k) {
+        sinth.d[k] = ctv * theta.d[k];
+        costh.d[k] = -stv * theta.d[k];
+        thetaInv.d[k] = mInv2 * theta.d[k];
+      }
+      const J3 w0 = cam[0] * thetaInv;
+      const J3 w1 = cam[1] * thetaInv;
+      const J3 w2 = cam[2] * thetaInv;
+      const J3 wxp0 = w1 * pt[2] - w2 * pt[1];
+      const J3 wxp1 = w2 * pt[0] - w0 * pt[2];
+      const J3 wxp2 = w0 * pt[1] - w1 * pt[0];
+      const J3 wdp =
+          (w0 * pt[0] + w1 * pt[1] + w2 * pt[2]) * (T(1) - costh);
+      P[0] = pt[0] * costh + wxp0 * sinth + w0 * wdp;
+      P[1] = pt[1] * costh + wxp1 * sinth + w1 * wdp;
+      P[2] = pt[2] * costh + wxp2 * sinth + w2 * wdp;
+    } else {
+      P[0] = pt[0] + (cam[1] * pt[2] - cam[2] * pt[1]);
+      P[1] = pt[1] + (cam[2] * pt[0] - cam[0] * pt[2]);
+      P[2] = pt[2] + (cam[0] * pt[1] - cam[1] * pt[0]);
+    }
+    P[0] += cam[3];
+    P[1] += cam[4];
+    P[2] += cam[5];
+    // shared reciprocal of the depth
+    T invNegZv = T(0);
+    if (sub == 0) invNegZv = T(-1) / P[2].v;
+    invNegZv = __shfl(invNegZv, leader, 64);
+    J3 invNegZ;
+    invNegZ.v = invNegZv;
+    const T q = invNegZv * invNegZv;  // d(-1/z) = z' / z^2
+    for (int k = 0; k < 3; ++k) invNegZ.d[k] = q * P[2].d[k];
+    const J3 xp = P[0] * invNegZ;
+    const J3 yp = P[1] * invNegZ;
+    const J3 r2 = xp * xp + yp * yp;
+    const J3 distortion = T(1) + r2 * (cam[7] + cam[8] * r2);
+    const J3 scaled = cam[6] * distortion;
+    res[0] = scaled * xp - m[0];
+    res[1] = scaled * yp - m[1];
+    // ---- identical epilogue to kForward ----
+    const bool cfix = camFixed && camFixed[camOf[e]];
+    const bool pfix = ptFixed && ptFixed[ptOf[e]];
+    for (int row = 0; row < 2; ++row) {
+      for (int j = 0; j < 3; ++j) {
+        const int col = base + j;
+        if (col < 9)
+          Jc[((int64_t)(col * 2 + row)) * nL + e] =
+              cfix ? T(0) : res[row].d[j];
+        else
+          Jp[((int64_t)((col - 9) * 2 + row)) * nL + e] =
+              pfix ? T(0) : res[row].d[j];
+      }
+      if (sub == 0) rOut[(int64_t)row * nL + e] = res[row].v;
+    }
+    if (sub == 0)
+      chi2 += (double)lossRho(lossKind, lossD2,
+                              res[0].v * res[0].v + res[1].v * res[1].v);
+  }
+  sm[threadIdx.x] = chi2;
+  __syncthreads();
+  for (int st = kBlk / 2; st > 0; st >>= 1) {
+    if (threadIdx.x < st) sm[threadIdx.x] += sm[threadIdx.x + st];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) atomicAdd(chi2Acc, sm[0]);
+}
+
 // Analytical-derivative forward: one thread per edge, closed-form residual +
 // 2x12 Jacobian (reference C11, src/geo/analytical_derivatives.cu).
 // BAL (9,3,2) only; the engine refuses analytical mode for other dims.
@@ -1936,11 +2061,24 @@ class GpuEngine final : public Engine<T> {
                            lossKind_, lossD2_);
     } else {
       constexpr int LANES = (GW + 2) / 3;
-      hipLaunchKernelGGL((kForward<T, CD, PD, RD>),
-                         dim3(gridFor(nL_ * LANES)), dim3(kBlk), 0, stream_,
-                         nL_, dCamOf_, dPtOf_, dParams_, ncam_, dMeas_,
-                         dCamFixed_, dPtFixed_, dR_[cur_], dJc_[cur_],
-                         dJp_[cur_], scalarPtr(), lossKind_, lossD2_, dIntr_);
+      bool launched = false;
+      if constexpr (CD == 9 && PD == 3 && RD == 2) {
+        if (useFwdVS_) {
+          hipLaunchKernelGGL(kForwardVS<T>, dim3(gridFor(nL_ * 4)),
+                             dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
+                             dParams_, ncam_, dMeas_, dCamFixed_, dPtFixed_,
+                             dR_[cur_], dJc_[cur_], dJp_[cur_], scalarPtr(),
+                             lossKind_, lossD2_);
+          launched = true;
+        }
+      }
+      if (!launched)
+        hipLaunchKernelGGL((kForward<T, CD, PD, RD>),
+                           dim3(gridFor(nL_ * LANES)), dim3(kBlk), 0, stream_,
+                           nL_, dCamOf_, dPtOf_, dParams_, ncam_, dMeas_,
+                           dCamFixed_, dPtFixed_, dR_[cur_], dJc_[cur_],
+                           dJp_[cur_], scalarPtr(), lossKind_, lossD2_,
+                           dIntr_);
     }
     return globalScalar(ncclSum);
   }
@@ -2618,6 +2756,8 @@ class GpuEngine final : public Engine<T> {
   bool useMfma_ = getenv("MEGBA_MFMA") != nullptr;
   // Fused one-pass E Cinv E^T apply (opt-in while being measured)
   bool useFused_ = getenv("MEGBA_FUSED") != nullptr;
+  // Value-share forward experiment (BAL fp64/fp32), opt-in MEGBA_FWD_VS=1
+  bool useFwdVS_ = getenv("MEGBA_FWD_VS") != nullptr;
   int nWin_ = 0, nFlagWins_ = 0, nLongPts_ = 0;
   int64_t* dWinLo_{};
   int64_t* dWinHi_{};

@@ -219,3 +219,32 @@ def test_fused_schur_weighted_loss():
     w = rng.uniform(0.5, 2.0, (len(ci), 2))
     info = np.stack([w[:, 0], 0.1 * np.ones(len(ci)), w[:, 1]], axis=1)
     _fused_pair_solve("implicit", cams, pts, ci, pi, meas, info=info)
+
+
+def test_forward_value_share_matches_default():
+    """MEGBA_FWD_VS=1 (leader-computed transcendental values broadcast to
+    the edge's gradient lanes) must reproduce the default fused forward."""
+    cams, pts, ci, pi, meas = mb.synthesize_bal(20, 300, 2600, seed=18)
+
+    def fwd(env):
+        saved = {k: os.environ.get(k) for k in env}
+        os.environ.update(env)
+        try:
+            p = mb.BAProblem(cams, pts, ci, pi, meas)
+            p.build(device="gpu")
+            chi = p.forward()
+            return chi, p.dump()
+        finally:
+            for k, v in saved.items():
+                if v is None:
+                    os.environ.pop(k, None)
+                else:
+                    os.environ[k] = v
+
+    c1, d1 = fwd({})
+    c2, d2 = fwd({"MEGBA_FWD_VS": "1"})
+    np.testing.assert_allclose(c2, c1, rtol=1e-12)
+    for key in ("r", "Jc", "Jp"):
+        scale = np.abs(d1[key]).max() or 1.0
+        np.testing.assert_allclose(d2[key], d1[key], rtol=1e-12,
+                                   atol=1e-14 * scale, err_msg=key)
