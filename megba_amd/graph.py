@@ -19,13 +19,15 @@ POINT = 1
 
 
 class BaseVertex:
-    """A parameter block.  kind: CAMERA (9 params) or POINT (3)."""
+    """A parameter block.  kind: CAMERA (9|6|4 params — the compiled
+    camDim set; see BAProblem) or POINT (3)."""
 
     def __init__(self, estimation, kind, fixed=False):
         est = np.asarray(estimation, dtype=np.float64).reshape(-1)
-        want = 9 if kind == CAMERA else 3
-        if est.size != want:
-            raise ValueError(f"vertex kind {kind} needs {want} params, "
+        ok = est.size in (9, 6, 4) if kind == CAMERA else est.size == 3
+        if not ok:
+            raise ValueError(f"vertex kind {kind} needs "
+                             f"{'9|6|4' if kind == CAMERA else '3'} params, "
                              f"got {est.size}")
         self.estimation = est.copy()
         self.kind = kind
@@ -142,14 +144,18 @@ class GraphProblem:
         return BAProblem(cams, pts, ci, pi, meas, info=info, **kw)
 
     def solve(self, device="cpu", dtype="float64", diff="auto",
-              schur="explicit", loss="none", loss_delta=1.0, **solve_kw):
+              schur="explicit", loss="none", loss_delta=1.0,
+              custom_forward=None, intrinsics=None, **solve_kw):
         """Build, run LM, and write the result back into the vertices.
         solve_kw: max_iter, tau, epsilon1, epsilon2, solver_tol,
         solver_max_iter, solver_refuse_ratio, verbose (see BAProblem.solve).
+        custom_forward/intrinsics: as in BAProblem.build (user-defined
+        residuals; fixed intrinsics for 6-dof cameras).
         """
         p = self._assemble()
         p.build(device=device, dtype=dtype, diff=diff, schur=schur,
-                loss=loss, loss_delta=loss_delta)
+                loss=loss, loss_delta=loss_delta,
+                custom_forward=custom_forward, intrinsics=intrinsics)
         report = p.solve(**solve_kw)
         cams, pts = p.get_params()
         for v, row in zip(self._cams, cams):
