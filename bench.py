@@ -39,10 +39,10 @@ def main():
     ap.add_argument("--dtype", default="float64",
                     choices=["float64", "float32"])
     ap.add_argument("--diff", default="auto", choices=["auto", "analytical"])
-    # implicit (matrix-free Schur products over cam-sorted grad-major J)
-    # measures FASTER than explicit on MI355X since the grad-major E w
-    # rework (61 vs 81 ms/step on Venice fp64, profiles/) and uses ~30%
-    # less memory; same LM+PCG math, same fixed work per step.  The
+    # implicit (matrix-free Schur products over packed vector-group J)
+    # measures FASTER than explicit on MI355X (57 vs ~80 ms/step on
+    # Venice fp64 after the r2 packed/banded rework, profiles/) and uses
+    # ~30% less memory; same LM+PCG math, same fixed work per step.  The
     # reference ships both (BAL_Double / BAL_Double_implicit).
     ap.add_argument("--schur", default="implicit",
                     choices=["explicit", "implicit"])

@@ -356,3 +356,23 @@ def test_gpu_633_fp32_decreases():
                   solver_refuse_ratio=1e6, verbose=False)
     chis = [i["chi2"] for i in rep["iters"]]
     assert chis[-1] < 0.5 * chis[0]
+
+
+def test_graph_api_6dof_camera():
+    """g2o-style graph construction with 6-dof (calibrated) cameras."""
+    from megba_amd.graph import GraphProblem, CameraVertex, PointVertex, \
+        ReprojectionEdge
+    cams9, pts, ci, pi, meas = _synth_calibrated(8, 60, 500, seed=19)
+    g = GraphProblem()
+    cvs = [CameraVertex(cams9[i, :6]) for i in range(8)]
+    pvs = [PointVertex(pts[i]) for i in range(60)]
+    for c, q, m in zip(ci, pi, meas):
+        e = ReprojectionEdge(m)
+        e.append_vertex(cvs[c]).append_vertex(pvs[q])
+        g.append_edge(e)
+    rep = g.solve(device="cpu", intrinsics=INTR, max_iter=4,
+                  solver_tol=1e-8, solver_max_iter=100,
+                  solver_refuse_ratio=1e6, verbose=False)
+    chis = [i["chi2"] for i in rep["iters"]]
+    assert chis[-1] <= chis[0]
+    assert cvs[0].estimation.shape == (6,)
