@@ -936,6 +936,75 @@ __global__ void kSpmvEtxPk(int64_t nL, const int* __restrict__ camOf,
   }
 }
 
+// Scan-free E^T x variant (env MEGBA_ETX_ATOMIC=1): PD atomicAdds per
+// edge instead of the wave segmented scan + tail atomics.  The scan costs
+// ~16 dependent shuffle instructions per edge; the atomics serialize on
+// the ~degree-5 same-point runs.  Which wins is measured, not assumed.
+template <typename T, int CD, int PD, int RD, bool HASINFO>
+__global__ void kSpmvEtxPkAtomic(int64_t nL, const int* __restrict__ camOf,
+                                 const int* __restrict__ ptOf,
+                                 const T* __restrict__ Jpk,
+                                 const T* const* __restrict__ jSlots,
+                                 const T* __restrict__ info, int lossKind,
+                                 T lossD2, const T* __restrict__ x,
+                                 T* __restrict__ out) {
+  using TV = typename PackVec<T>::type;
+  constexpr int VEC = PackVec<T>::VEC;
+  constexpr int RW = RD * (RD + 1) / 2;
+  constexpr int CR = CD * RD, PR = PD * RD;
+  constexpr int NG = (CR + PR + VEC - 1) / VEC;
+  const T* rBak = jSlots[2];
+  for (int64_t j = blockIdx.x * (int64_t)kBlk + threadIdx.x; j < nL;
+       j += (int64_t)gridDim.x * kBlk) {
+    const int pt = ptOf[j];
+    const T* xc = x + (int64_t)camOf[j] * CD;
+    TV buf[NG];
+    const TV* src = (const TV*)Jpk;
+#pragma unroll
+    for (int g = 0; g < NG; ++g) buf[g] = src[(int64_t)g * nL + j];
+    T u[RD];
+#pragma unroll
+    for (int rr = 0; rr < RD; ++rr) {
+      T v = T(0);
+#pragma unroll
+      for (int i = 0; i < CD; ++i) {
+        const int k = i * RD + rr;
+        v += buf[k / VEC][k % VEC] * xc[i];
+      }
+      u[rr] = v;
+    }
+    if (HASINFO) {
+      T wu[RD];
+      for (int i = 0; i < RD; ++i) {
+        T v = T(0);
+        for (int k = 0; k < RD; ++k)
+          v += info[RW * j + symIdx<RD>(i, k)] * u[k];
+        wu[i] = v;
+      }
+      for (int i = 0; i < RD; ++i) u[i] = wu[i];
+    }
+    if (lossKind) {
+      T ss = T(0);
+      for (int rr = 0; rr < RD; ++rr) {
+        const T rv = rBak[(int64_t)rr * nL + j];
+        ss += rv * rv;
+      }
+      const T w = lossWeight(lossKind, lossD2, ss);
+      for (int rr = 0; rr < RD; ++rr) u[rr] *= w;
+    }
+#pragma unroll
+    for (int k = 0; k < PD; ++k) {
+      T v = T(0);
+#pragma unroll
+      for (int rr = 0; rr < RD; ++rr) {
+        const int kk = CR + k * RD + rr;
+        v += buf[kk / VEC][kk % VEC] * u[rr];
+      }
+      atomicAdd(&out[PD * pt + k], v);
+    }
+  }
+}
+
 // Packed E w over the cam-sorted [wJc(CR), Jp(PR)] groups.
 template <typename T, int CD, int PD, int RD>
 __global__ __launch_bounds__(64) void kSpmvExPk(
@@ -2587,7 +2656,20 @@ class GpuEngine final : public Engine<T> {
                        dim3(kBlk), 0, stream_, out + (int64_t)ptLo_ * PD,
                        (int64_t)npL_ * PD);
     if (implicit_) {
-      if (hasInfo_)
+      if (etxAtomic_) {
+        if (hasInfo_)
+          hipLaunchKernelGGL((kSpmvEtxPkAtomic<T, CD, PD, RD, true>),
+                             dim3(gridFor(nL_)), dim3(kBlk), 0, stream_, nL_,
+                             dCamOf_, dPtOf_, dJPk_,
+                             (const T* const*)dJSlots_, dInfo_, lossKind_,
+                             lossD2_, xv, out);
+        else
+          hipLaunchKernelGGL((kSpmvEtxPkAtomic<T, CD, PD, RD, false>),
+                             dim3(gridFor(nL_)), dim3(kBlk), 0, stream_, nL_,
+                             dCamOf_, dPtOf_, dJPk_,
+                             (const T* const*)dJSlots_, (const T*)nullptr,
+                             lossKind_, lossD2_, xv, out);
+      } else if (hasInfo_)
         hipLaunchKernelGGL((kSpmvEtxPk<T, CD, PD, RD, true>),
                            dim3(gridFor(nL_)), dim3(kBlk), 0, stream_, nL_,
                            dCamOf_, dPtOf_, dJPk_,
@@ -2758,6 +2840,8 @@ class GpuEngine final : public Engine<T> {
   bool useFused_ = getenv("MEGBA_FUSED") != nullptr;
   // Value-share forward experiment (BAL fp64/fp32), opt-in MEGBA_FWD_VS=1
   bool useFwdVS_ = getenv("MEGBA_FWD_VS") != nullptr;
+  // Scan-free E^T x variant, opt-in MEGBA_ETX_ATOMIC=1 while measured
+  bool etxAtomic_ = getenv("MEGBA_ETX_ATOMIC") != nullptr;
   int nWin_ = 0, nFlagWins_ = 0, nLongPts_ = 0;
   int64_t* dWinLo_{};
   int64_t* dWinHi_{};
