@@ -744,6 +744,295 @@ __global__ void kFinalizeCam(int64_t nL, const T* __restrict__ slab,
 // profiles/r02_gather_bands.md.)
 
 // ---------------------------------------------------------------------------
+// Packed-J layout for the per-iteration implicit products
+// ---------------------------------------------------------------------------
+// The grad-major J layout ([k][nL], one 4/8-byte load per value) makes the
+// two matrix-free PCG kernels instruction-ISSUE-bound: ~24 scalar loads
+// per edge dominate the ~80-instruction inner loop (measured ~2x the byte
+// floor on final13682-fp32, profiles/r01_final13682_implicit_fp32.md).
+// Repacking the 24 per-edge J values into 16-byte vector groups
+// ([group][nL][VEC], VEC = 16B/sizeof(T)) turns them into NG=24/VEC
+// dwordx4 loads (6 for fp32, 12 for fp64), still fully coalesced across
+// the 64 consecutive-edge lanes.  The pack runs once per ACCEPTED LM step
+// (its cost amortizes over the ~100 PCG iterations that read it); the
+// packed buffers are single-buffered so the captured PCG graph needs no
+// pointer indirection for them.
+template <typename T>
+struct PackVec {
+  static constexpr int VEC = 16 / sizeof(T);
+  typedef T type __attribute__((ext_vector_type(16 / sizeof(T))));
+};
+
+// Primary-order pack of the accepted [Jc(CR), Jp(PR)] set (E^T x source).
+template <typename T, int CD, int PD, int RD>
+__global__ void kPackJPrimary(int64_t nL, const T* const* __restrict__ jSlots,
+                              T* __restrict__ out) {
+  using TV = typename PackVec<T>::type;
+  constexpr int VEC = PackVec<T>::VEC;
+  constexpr int CR = CD * RD, PR = PD * RD;
+  constexpr int NV = CR + PR;
+  constexpr int NG = (NV + VEC - 1) / VEC;
+  const T* Jc = jSlots[0];
+  const T* Jp = jSlots[1];
+  TV* o = (TV*)out;
+  for (int64_t j = blockIdx.x * (int64_t)kBlk + threadIdx.x; j < nL;
+       j += (int64_t)gridDim.x * kBlk) {
+    for (int g = 0; g < NG; ++g) {
+      TV v;
+      for (int q = 0; q < VEC; ++q) {
+        const int k = g * VEC + q;
+        v[q] = k < CR ? Jc[(int64_t)k * nL + j]
+                      : (k < NV ? Jp[(int64_t)(k - CR) * nL + j] : T(0));
+      }
+      o[(int64_t)g * nL + j] = v;
+    }
+  }
+}
+
+// Packed E^T x: identical math to kSpmvEtx<IMP>, vector-group loads.
+template <typename T, int CD, int PD, int RD, bool HASINFO>
+__global__ void kSpmvEtxPk(int64_t nL, const int* __restrict__ camOf,
+                           const int* __restrict__ ptOf,
+                           const T* __restrict__ Jpk,
+                           const T* const* __restrict__ jSlots,
+                           const T* __restrict__ info, int lossKind, T lossD2,
+                           const T* __restrict__ x, T* __restrict__ out) {
+  using TV = typename PackVec<T>::type;
+  constexpr int VEC = PackVec<T>::VEC;
+  constexpr int RW = RD * (RD + 1) / 2;
+  constexpr int CR = CD * RD, PR = PD * RD;
+  constexpr int NG = (CR + PR + VEC - 1) / VEC;
+  const T* rBak = jSlots[2];
+  const int lane = threadIdx.x & 63;
+  const int64_t nWork = ((nL + kBlk - 1) / kBlk) * (int64_t)kBlk;
+  for (int64_t j0 = blockIdx.x * (int64_t)kBlk + threadIdx.x; j0 < nWork;
+       j0 += (int64_t)gridDim.x * kBlk) {
+    const bool active = j0 < nL;
+    const int64_t j = active ? j0 : nL - 1;
+    const int pt = ptOf[j];
+    T o[PD];
+    for (int k = 0; k < PD; ++k) o[k] = T(0);
+    if (active) {
+      const T* xc = x + (int64_t)camOf[j] * CD;
+      TV buf[NG];
+      const TV* src = (const TV*)Jpk;
+#pragma unroll
+      for (int g = 0; g < NG; ++g) buf[g] = src[(int64_t)g * nL + j];
+      T u[RD];
+#pragma unroll
+      for (int rr = 0; rr < RD; ++rr) {
+        T v = T(0);
+#pragma unroll
+        for (int i = 0; i < CD; ++i) {
+          constexpr int dummy = 0;
+          (void)dummy;
+          const int k = i * RD + rr;
+          v += buf[k / VEC][k % VEC] * xc[i];
+        }
+        u[rr] = v;
+      }
+      if (HASINFO) {
+        T wu[RD];
+        for (int i = 0; i < RD; ++i) {
+          T v = T(0);
+          for (int k = 0; k < RD; ++k)
+            v += info[RW * j + symIdx<RD>(i, k)] * u[k];
+          wu[i] = v;
+        }
+        for (int i = 0; i < RD; ++i) u[i] = wu[i];
+      }
+      if (lossKind) {
+        T ss = T(0);
+        for (int rr = 0; rr < RD; ++rr) {
+          const T rv = rBak[(int64_t)rr * nL + j];
+          ss += rv * rv;
+        }
+        const T w = lossWeight(lossKind, lossD2, ss);
+        for (int rr = 0; rr < RD; ++rr) u[rr] *= w;
+      }
+#pragma unroll
+      for (int k = 0; k < PD; ++k) {
+        T v = T(0);
+#pragma unroll
+        for (int rr = 0; rr < RD; ++rr) {
+          const int kk = CR + k * RD + rr;
+          v += buf[kk / VEC][kk % VEC] * u[rr];
+        }
+        o[k] = v;
+      }
+    }
+    for (int off = 1; off < 64; off <<= 1) {
+      const int ppt = __shfl_up(pt, off, 64);
+      const bool join = lane >= off && ppt == pt;
+      if (__ballot(join) == 0ull) break;
+      T a[PD];
+      for (int k = 0; k < PD; ++k) a[k] = __shfl_up(o[k], off, 64);
+      if (join)
+        for (int k = 0; k < PD; ++k) o[k] += a[k];
+    }
+    const int nextPt = __shfl_down(pt, 1, 64);
+    const bool tail = active && (lane == 63 || nextPt != pt || j0 == nL - 1);
+    if (tail)
+      for (int k = 0; k < PD; ++k) atomicAdd(&out[PD * pt + k], o[k]);
+  }
+}
+
+// Scan-free E^T x variant (env MEGBA_ETX_ATOMIC=1): PD atomicAdds per
+// edge instead of the wave segmented scan + tail atomics.  The scan costs
+// ~16 dependent shuffle instructions per edge; the atomics serialize on
+// the ~degree-5 same-point runs.  Which wins is measured, not assumed.
+template <typename T, int CD, int PD, int RD, bool HASINFO>
+__global__ void kSpmvEtxPkAtomic(int64_t nL, const int* __restrict__ camOf,
+                                 const int* __restrict__ ptOf,
+                                 const T* __restrict__ Jpk,
+                                 const T* const* __restrict__ jSlots,
+                                 const T* __restrict__ info, int lossKind,
+                                 T lossD2, const T* __restrict__ x,
+                                 T* __restrict__ out) {
+  using TV = typename PackVec<T>::type;
+  constexpr int VEC = PackVec<T>::VEC;
+  constexpr int RW = RD * (RD + 1) / 2;
+  constexpr int CR = CD * RD, PR = PD * RD;
+  constexpr int NG = (CR + PR + VEC - 1) / VEC;
+  const T* rBak = jSlots[2];
+  for (int64_t j = blockIdx.x * (int64_t)kBlk + threadIdx.x; j < nL;
+       j += (int64_t)gridDim.x * kBlk) {
+    const int pt = ptOf[j];
+    const T* xc = x + (int64_t)camOf[j] * CD;
+    TV buf[NG];
+    const TV* src = (const TV*)Jpk;
+#pragma unroll
+    for (int g = 0; g < NG; ++g) buf[g] = src[(int64_t)g * nL + j];
+    T u[RD];
+#pragma unroll
+    for (int rr = 0; rr < RD; ++rr) {
+      T v = T(0);
+#pragma unroll
+      for (int i = 0; i < CD; ++i) {
+        const int k = i * RD + rr;
+        v += buf[k / VEC][k % VEC] * xc[i];
+      }
+      u[rr] = v;
+    }
+    if (HASINFO) {
+      T wu[RD];
+      for (int i = 0; i < RD; ++i) {
+        T v = T(0);
+        for (int k = 0; k < RD; ++k)
+          v += info[RW * j + symIdx<RD>(i, k)] * u[k];
+        wu[i] = v;
+      }
+      for (int i = 0; i < RD; ++i) u[i] = wu[i];
+    }
+    if (lossKind) {
+      T ss = T(0);
+      for (int rr = 0; rr < RD; ++rr) {
+        const T rv = rBak[(int64_t)rr * nL + j];
+        ss += rv * rv;
+      }
+      const T w = lossWeight(lossKind, lossD2, ss);
+      for (int rr = 0; rr < RD; ++rr) u[rr] *= w;
+    }
+#pragma unroll
+    for (int k = 0; k < PD; ++k) {
+      T v = T(0);
+#pragma unroll
+      for (int rr = 0; rr < RD; ++rr) {
+        const int kk = CR + k * RD + rr;
+        v += buf[kk / VEC][kk % VEC] * u[rr];
+      }
+      atomicAdd(&out[PD * pt + k], v);
+    }
+  }
+}
+
+// Packed E w over the cam-sorted [wJc(CR), Jp(PR)] groups.
+template <typename T, int CD, int PD, int RD>
+__global__ __launch_bounds__(64) void kSpmvExPk(
+    int nChunks, const int* __restrict__ chCam, const int* __restrict__ chLo,
+    const int* __restrict__ chHi, const int* __restrict__ ptOfCam,
+    const T* __restrict__ JCamPk, int64_t nL, const T* __restrict__ w,
+    T* __restrict__ out) {
+  using TV = typename PackVec<T>::type;
+  constexpr int VEC = PackVec<T>::VEC;
+  constexpr int CR = CD * RD, PR = PD * RD;
+  constexpr int NG = (CR + PR + VEC - 1) / VEC;
+  const int chunk = blockIdx.x;
+  if (chunk >= nChunks) return;
+  const int cam = chCam[chunk];
+  T acc[CD];
+  for (int i = 0; i < CD; ++i) acc[i] = T(0);
+  const int lo = chLo[chunk], hi = chHi[chunk];
+  const TV* src = (const TV*)JCamPk;
+  // w is stored 4-padded (stride 4 elements, 16B/32B aligned) so the
+  // per-edge gather is 1 (fp32) or 2 (fp64) vector loads instead of PD
+  // scalar loads — the gather is the TA-request hot spot of this kernel.
+  constexpr int WL = (4 * (int)sizeof(T) + 15) / 16;
+  for (int j = lo + (int)threadIdx.x; j < hi; j += 64) {
+    const TV* wp4 = (const TV*)(w + (int64_t)ptOfCam[j] * 4);
+    TV wbuf[WL];
+#pragma unroll
+    for (int l = 0; l < WL; ++l) wbuf[l] = wp4[l];
+    TV buf[NG];
+#pragma unroll
+    for (int g = 0; g < NG; ++g) buf[g] = src[(int64_t)g * nL + j];
+    T u[RD];
+#pragma unroll
+    for (int rr = 0; rr < RD; ++rr) {
+      T v = T(0);
+#pragma unroll
+      for (int k = 0; k < PD; ++k) {
+        const int kk = CR + k * RD + rr;
+        v += buf[kk / VEC][kk % VEC] * wbuf[k / VEC][k % VEC];
+      }
+      u[rr] = v;
+    }
+#pragma unroll
+    for (int i = 0; i < CD; ++i) {
+      T v = T(0);
+#pragma unroll
+      for (int rr = 0; rr < RD; ++rr) {
+        const int k = i * RD + rr;
+        v += buf[k / VEC][k % VEC] * u[rr];
+      }
+      acc[i] += v;
+    }
+  }
+  for (int off = 32; off > 0; off >>= 1)
+    for (int i = 0; i < CD; ++i) acc[i] += __shfl_down(acc[i], off, 64);
+  if (threadIdx.x == 0) {
+    T* oc = out + (int64_t)cam * CD;
+    for (int i = 0; i < CD; ++i) atomicAdd(&oc[i], acc[i]);
+  }
+}
+
+// Cam-sorted packed finalize: slab -> [wJc, Jp] vector groups.
+template <typename T, int CD, int PD, int RD, bool HASINFO>
+__global__ void kFinalizeCamImpPk(int64_t nL, const T* __restrict__ slab,
+                                  T* __restrict__ JCamPk) {
+  using L = SlabLayout<CD, PD, RD, false, HASINFO>;
+  using TV = typename PackVec<T>::type;
+  constexpr int VEC = PackVec<T>::VEC;
+  constexpr int CR = CD * RD, PR = PD * RD;
+  constexpr int NG = (CR + PR + VEC - 1) / VEC;
+  constexpr int woff = HASINFO ? L::WJCOFF : L::JCOFF;
+  TV* o = (TV*)JCamPk;
+  for (int64_t j = blockIdx.x * (int64_t)kBlk + threadIdx.x; j < nL;
+       j += (int64_t)gridDim.x * kBlk) {
+    const T* row = slab + j * L::SW;
+    for (int g = 0; g < NG; ++g) {
+      TV v;
+      for (int q = 0; q < VEC; ++q) {
+        const int k = g * VEC + q;
+        v[q] = k < CR ? row[woff + k]
+                      : (k < CR + PR ? row[L::JPOFF + (k - CR)] : T(0));
+      }
+      o[(int64_t)g * nL + j] = v;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Damping, block inverse
 // ---------------------------------------------------------------------------
 template <typename T, int D>
