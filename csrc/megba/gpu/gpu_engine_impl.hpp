@@ -84,6 +84,13 @@ inline int gridFor(int64_t n) {
 // ---------------------------------------------------------------------------
 enum class ROp { Dot, SumSq, AbsMax };
 
+// 16-byte vector groups used by the packed J / Hpl layouts.
+template <typename T>
+struct PackVec {
+  static constexpr int VEC = 16 / sizeof(T);
+  typedef T type __attribute__((ext_vector_type(16 / sizeof(T))));
+};
+
 template <typename T, ROp OP>
 __global__ void kRedPartial(const T* a, const T* b, int64_t n, double* part) {
   __shared__ double sm[kBlk];
@@ -536,13 +543,27 @@ __global__ void kAssembleEdge(int64_t nL, const int* __restrict__ camOf,
       }
       T* row = slab + (int64_t)camPos[e] * L::SW;
       if (EXPL) {
+        // Hpl stored as 16-byte vector groups ([group][nL][VEC]) so the
+        // per-iteration E^T x / E w reads are dwordx4 loads (r2, same
+        // treatment as the implicit packed J — profiles/r02_gather_bands.md)
+        constexpr int VEC = 16 / (int)sizeof(T);
+        constexpr int NGH = (CD * PD + VEC - 1) / VEC;
+        T v[NGH * VEC];
+        for (int k = 0; k < NGH * VEC; ++k) v[k] = T(0);
         for (int a = 0; a < CD; ++a)
           for (int b = 0; b < PD; ++b) {
-            T v = T(0);
-            for (int i = 0; i < RD; ++i) v += jc[i][a] * wjp[i][b];
-            Hpl[((int64_t)(a * PD + b)) * nL + e] = v;
-            row[a * PD + b] = v;
+            T acc = T(0);
+            for (int i = 0; i < RD; ++i) acc += jc[i][a] * wjp[i][b];
+            v[a * PD + b] = acc;
+            row[a * PD + b] = acc;
           }
+        using TV = typename PackVec<T>::type;
+        TV* o = (TV*)Hpl;
+        for (int g = 0; g < NGH; ++g) {
+          TV vv;
+          for (int q = 0; q < VEC; ++q) vv[q] = v[g * VEC + q];
+          o[(int64_t)g * nL + e] = vv;
+        }
       }
       for (int k = 0; k < CD; ++k)
         for (int i = 0; i < RD; ++i) row[L::JCOFF + k * RD + i] = jc[i][k];
@@ -732,10 +753,21 @@ template <typename T, int CD, int PD, int RD, bool HASINFO>
 __global__ void kFinalizeCam(int64_t nL, const T* __restrict__ slab,
                              T* __restrict__ HplCam) {
   using L = SlabLayout<CD, PD, RD, true, HASINFO>;
+  using TV = typename PackVec<T>::type;
+  constexpr int VEC = PackVec<T>::VEC;
+  constexpr int NGH = (CD * PD + VEC - 1) / VEC;
+  TV* o = (TV*)HplCam;
   for (int64_t j = blockIdx.x * (int64_t)kBlk + threadIdx.x; j < nL;
        j += (int64_t)gridDim.x * kBlk) {
     const T* row = slab + j * L::SW;
-    for (int k = 0; k < CD * PD; ++k) HplCam[(int64_t)k * nL + j] = row[k];
+    for (int g = 0; g < NGH; ++g) {
+      TV v;
+      for (int q = 0; q < VEC; ++q) {
+        const int k = g * VEC + q;
+        v[q] = k < CD * PD ? row[k] : T(0);
+      }
+      o[(int64_t)g * nL + j] = v;
+    }
   }
 }
 
@@ -757,12 +789,6 @@ __global__ void kFinalizeCam(int64_t nL, const T* __restrict__ slab,
 // (its cost amortizes over the ~100 PCG iterations that read it); the
 // packed buffers are single-buffered so the captured PCG graph needs no
 // pointer indirection for them.
-template <typename T>
-struct PackVec {
-  static constexpr int VEC = 16 / sizeof(T);
-  typedef T type __attribute__((ext_vector_type(16 / sizeof(T))));
-};
-
 // Primary-order pack of the accepted [Jc(CR), Jp(PR)] set (E^T x source).
 template <typename T, int CD, int PD, int RD>
 __global__ void kPackJPrimary(int64_t nL, const T* const* __restrict__ jSlots,
@@ -1158,10 +1184,21 @@ __global__ void kSpmvEtx(int64_t nL, const int* __restrict__ camOf,
           o[k] = v;
         }
       } else {
+        using TV = typename PackVec<T>::type;
+        constexpr int VEC = PackVec<T>::VEC;
+        constexpr int NGH = (CD * PD + VEC - 1) / VEC;
+        TV buf[NGH];
+        const TV* src = (const TV*)Hpl;
+#pragma unroll
+        for (int g = 0; g < NGH; ++g) buf[g] = src[(int64_t)g * nL + j];
+#pragma unroll
         for (int i = 0; i < CD; ++i) {
           const T xi = xc[i];
-          for (int k = 0; k < PD; ++k)
-            o[k] += Hpl[((int64_t)(i * PD + k)) * nL + j] * xi;
+#pragma unroll
+          for (int k = 0; k < PD; ++k) {
+            const int kk = i * PD + k;
+            o[k] += buf[kk / VEC][kk % VEC] * xi;
+          }
         }
       }
     }
@@ -1195,20 +1232,33 @@ __global__ __launch_bounds__(64) void kSpmvEx(
     const int* __restrict__ chHi, const int* __restrict__ ptOfCam,
     const T* __restrict__ HplCam, int64_t nL, const T* __restrict__ w,
     T* __restrict__ out) {
+  using TV = typename PackVec<T>::type;
+  constexpr int VEC = PackVec<T>::VEC;
+  constexpr int NGH = (CD * PD + VEC - 1) / VEC;
+  constexpr int WL = (4 * (int)sizeof(T) + 15) / 16;
   const int chunk = blockIdx.x;
   if (chunk >= nChunks) return;
   const int cam = chCam[chunk];
   T acc[CD];
   for (int i = 0; i < CD; ++i) acc[i] = T(0);
   const int lo = chLo[chunk], hi = chHi[chunk];
+  const TV* src = (const TV*)HplCam;
   for (int j = lo + (int)threadIdx.x; j < hi; j += 64) {
-    const T* wp = w + (int64_t)ptOfCam[j] * PD;
-    T wv[PD];
-    for (int k = 0; k < PD; ++k) wv[k] = wp[k];
+    const TV* wp4 = (const TV*)(w + (int64_t)ptOfCam[j] * 4);
+    TV wbuf[WL];
+#pragma unroll
+    for (int l = 0; l < WL; ++l) wbuf[l] = wp4[l];
+    TV buf[NGH];
+#pragma unroll
+    for (int g = 0; g < NGH; ++g) buf[g] = src[(int64_t)g * nL + j];
+#pragma unroll
     for (int i = 0; i < CD; ++i) {
       T v = T(0);
-      for (int k = 0; k < PD; ++k)
-        v += HplCam[((int64_t)(i * PD + k)) * nL + j] * wv[k];
+#pragma unroll
+      for (int k = 0; k < PD; ++k) {
+        const int kk = i * PD + k;
+        v += buf[kk / VEC][kk % VEC] * wbuf[k / VEC][k % VEC];
+      }
       acc[i] += v;
     }
   }
@@ -1295,9 +1345,17 @@ __device__ inline void fusedLoadEdge(int64_t j, int64_t nL,
       t[k] = v;
     }
   } else {
+    using TV = typename PackVec<T>::type;
+    constexpr int VEC = PackVec<T>::VEC;
+    constexpr int NGH = (CD * PD + VEC - 1) / VEC;
+    const TV* src = (const TV*)Hpl;
+    TV buf[NGH];
+    for (int g = 0; g < NGH; ++g) buf[g] = src[(int64_t)g * nL + j];
     for (int i = 0; i < CD; ++i)
-      for (int k = 0; k < PD; ++k)
-        hplv[i][k] = Hpl[((int64_t)(i * PD + k)) * nL + j];
+      for (int k = 0; k < PD; ++k) {
+        const int kk = i * PD + k;
+        hplv[i][k] = buf[kk / VEC][kk % VEC];
+      }
     for (int k = 0; k < PD; ++k) t[k] = T(0);
     for (int i = 0; i < CD; ++i) {
       const T xi = xc[i];
@@ -2040,11 +2098,13 @@ class GpuEngine final : public Engine<T> {
       constexpr int NG = (CR + PR + VEC - 1) / VEC;
       dJPk_ = dalloc<T>(nL_ * NG * VEC);
       dJCamPk_ = dalloc<T>(nL_ * NG * VEC);
-      dWPad_ = dalloc<T>((int64_t)npt_ * 4);
     }
+    dWPad_ = dalloc<T>((int64_t)npt_ * 4);
     if (!implicit_) {
-      dHpl_ = dalloc<T>(nL_ * CP);
-      dHplCam_ = dalloc<T>(nL_ * CP);
+      constexpr int VEC = 16 / (int)sizeof(T);
+      constexpr int NGH = (CP + VEC - 1) / VEC;
+      dHpl_ = dalloc<T>(nL_ * NGH * VEC);
+      dHplCam_ = dalloc<T>(nL_ * NGH * VEC);
     }
     sync();
   }
@@ -2389,11 +2449,15 @@ class GpuEngine final : public Engine<T> {
     d.Hpp = down(dHpp_, (int64_t)ncam_ * CC);
     d.Hll = down(dHll_, (int64_t)npt_ * PP);
     if (!implicit_) {
-      // device Hpl is grad-major [CP][nL]; dump as [e][CD][PD]
-      std::vector<double> gm = down(dHpl_, nL_ * CP);
-      std::vector<double> o(gm.size());
+      // device Hpl is packed [group][nL][VEC]; dump as [e][CD][PD]
+      constexpr int VEC = 16 / (int)sizeof(T);
+      constexpr int NGH = (CP + VEC - 1) / VEC;
+      std::vector<double> gm = down(dHpl_, nL_ * NGH * VEC);
+      std::vector<double> o((size_t)nL_ * CP);
       for (int64_t e = 0; e < nL_; ++e)
-        for (int k = 0; k < CP; ++k) o[e * CP + k] = gm[(int64_t)k * nL_ + e];
+        for (int k = 0; k < CP; ++k)
+          o[e * CP + k] =
+              gm[((int64_t)(k / VEC) * nL_ + e) * VEC + k % VEC];
       d.Hpl = o;
     }
     d.g = down(dG_, dim_);
@@ -2577,20 +2641,15 @@ class GpuEngine final : public Engine<T> {
     blockMatVec<PD, 0>(npL_, dHllInv_ + (int64_t)ptLo_ * PP,
                        in + (int64_t)ptLo_ * PD, out + (int64_t)ptLo_ * PD);
   }
-  // w = Cinv in, then out += E w.  The implicit path stores w 4-padded so
-  // the E-side gather is a single aligned vector load per edge.
+  // w = Cinv in, then out += E w.  w is stored 4-padded so the E-side
+  // gather is a single aligned vector load per edge (both modes).
   void cinvThenEx(const T* in, T* out) {
-    if (implicit_) {
-      hipLaunchKernelGGL((kCinvPad<T, PD>),
-                         dim3(gridFor((int64_t)npL_ * PD)), dim3(kBlk), 0,
-                         stream_, npL_, dHllInv_ + (int64_t)ptLo_ * PP,
-                         in + (int64_t)ptLo_ * PD,
-                         dWPad_ + (int64_t)ptLo_ * 4);
-      spmvEx(dWPad_, out);
-    } else {
-      applyCinv(in, dW_);
-      spmvEx(dW_, out);
-    }
+    hipLaunchKernelGGL((kCinvPad<T, PD>),
+                       dim3(gridFor((int64_t)npL_ * PD)), dim3(kBlk), 0,
+                       stream_, npL_, dHllInv_ + (int64_t)ptLo_ * PP,
+                       in + (int64_t)ptLo_ * PD,
+                       dWPad_ + (int64_t)ptLo_ * 4);
+    spmvEx(dWPad_, out);
   }
   void spmvEtx(const T* xv, T* out) {
     hipLaunchKernelGGL(kZeroRange<T>, dim3(gridFor((int64_t)npL_ * PD)),
