@@ -1717,6 +1717,74 @@ __global__ void kXpbyS(int64_t n, const T* __restrict__ x,
     y[i] = x[i] + bv * y[i];
 }
 
+// Fused beta + p-update: every block redundantly reduces the (tiny)
+// rho partial array to beta = rho/rhoPrev, then updates its slice of
+// p = z + beta p.  Replaces the standalone kRedFinalRhoBeta launch
+// (block 0 still publishes rho for the host refuse/tol readback).
+template <typename T>
+__global__ void kXpbySBeta(int64_t n, const T* __restrict__ z,
+                           const double* __restrict__ part, int nb,
+                           const double* __restrict__ rhoPrev,
+                           double* __restrict__ rhoOut, T* __restrict__ p) {
+  __shared__ double sm[kBlk];
+  double v = 0.0;
+  for (int i = threadIdx.x; i < nb; i += kBlk) v += part[i];
+  sm[threadIdx.x] = v;
+  __syncthreads();
+  for (int st = kBlk / 2; st > 0; st >>= 1) {
+    if (threadIdx.x < st) sm[threadIdx.x] += sm[threadIdx.x + st];
+    __syncthreads();
+  }
+  const double rho = sm[0];
+  if (blockIdx.x == 0 && threadIdx.x == 0) *rhoOut = rho;
+  const double rp = *rhoPrev;
+  const T beta = (T)(rp != 0.0 ? rho / rp : 0.0);
+  for (int64_t i = blockIdx.x * (int64_t)kBlk + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * kBlk)
+    p[i] = z[i] + beta * p[i];
+}
+
+// Fused alpha + x/r update (incl. the two-deep backup rotation): every
+// block reduces BOTH partial arrays (rho from partR, p^T q from partQ),
+// forms alpha = rho/pq, and updates its slice; block 0 publishes
+// rhoPrev = rho for the next iteration's beta.
+template <typename T>
+__global__ void kUpdateXRAlpha(int64_t n, const double* __restrict__ partR,
+                               int nbR, const double* __restrict__ partQ,
+                               int nbQ, double* __restrict__ rhoPrevOut,
+                               const T* __restrict__ p,
+                               const T* __restrict__ q, T* __restrict__ x,
+                               T* __restrict__ xBak,
+                               T* __restrict__ xBakPrev, T* __restrict__ r) {
+  __shared__ double smR[kBlk];
+  __shared__ double smQ[kBlk];
+  double vR = 0.0, vQ = 0.0;
+  for (int i = threadIdx.x; i < nbR; i += kBlk) vR += partR[i];
+  for (int i = threadIdx.x; i < nbQ; i += kBlk) vQ += partQ[i];
+  smR[threadIdx.x] = vR;
+  smQ[threadIdx.x] = vQ;
+  __syncthreads();
+  for (int st = kBlk / 2; st > 0; st >>= 1) {
+    if (threadIdx.x < st) {
+      smR[threadIdx.x] += smR[threadIdx.x + st];
+      smQ[threadIdx.x] += smQ[threadIdx.x + st];
+    }
+    __syncthreads();
+  }
+  const double rho = smR[0];
+  const double pq = smQ[0];
+  if (blockIdx.x == 0 && threadIdx.x == 0) *rhoPrevOut = rho;
+  const T a = (T)(pq != 0.0 ? rho / pq : 0.0);
+  for (int64_t i = blockIdx.x * (int64_t)kBlk + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * kBlk) {
+    xBakPrev[i] = xBak[i];
+    const T xv = x[i];
+    xBak[i] = xv;
+    x[i] = xv + a * p[i];
+    r[i] -= a * q[i];
+  }
+}
+
 // r = v - q
 template <typename T>
 __global__ void kSub(int64_t n, const T* __restrict__ v, const T* __restrict__ q,
@@ -1957,6 +2025,7 @@ class GpuEngine final : public Engine<T> {
     // reductions (kRedBlocks) and the fused B-apply+dot grid over nc_.
     partCap_ = kRedBlocks > gridFor(nc_) ? kRedBlocks : gridFor(nc_);
     dPart_ = dalloc<double>(partCap_ + 8);
+    dPartQ_ = dalloc<double>(partCap_);
     dFail_ = dalloc<int>(2);
     HIP_CHECK(hipHostMalloc((void**)&hScalar_, sizeof(double)));
 
@@ -2712,17 +2781,14 @@ class GpuEngine final : public Engine<T> {
     const int rhoGrid = gridFor(nc_) < kRedBlocks ? gridFor(nc_) : kRedBlocks;
     hipLaunchKernelGGL((kPrecondRho<T, CD>), dim3(rhoGrid), dim3(kBlk), 0,
                        stream_, ncam_, dHppInv_, dRr_, dZ_, dPart_);
-    hipLaunchKernelGGL(kRedFinalRhoBeta, dim3(1), dim3(kBlk), 0, stream_,
-                       dPart_, rhoGrid, slotRho(), slotRhoPrev(), slotBeta());
-    hipLaunchKernelGGL(kXpbyS<T>, dim3(gridFor(nc_)), dim3(kBlk), 0, stream_,
-                       nc_, dZ_, slotBeta(), dP_);
+    hipLaunchKernelGGL(kXpbySBeta<T>, dim3(gridFor(nc_)), dim3(kBlk), 0,
+                       stream_, nc_, dZ_, dPart_, rhoGrid, slotRhoPrev(),
+                       slotRho(), dP_);
     schurApply(dP_, dQ_, /*withDot=*/true);
-    hipLaunchKernelGGL(kRedFinalAlpha, dim3(1), dim3(kBlk), 0, stream_,
-                       dPart_, gridFor(nc_), slotRho(), slotAlpha(),
-                       slotRhoPrev());
-    hipLaunchKernelGGL(kUpdateXR<T>, dim3(gridFor(nc_)), dim3(kBlk), 0,
-                       stream_, nc_, slotAlpha(), dP_, dQ_, dDeltaX_, dXBak_,
-                       dXBakPrev_, dRr_);
+    hipLaunchKernelGGL(kUpdateXRAlpha<T>, dim3(gridFor(nc_)), dim3(kBlk), 0,
+                       stream_, nc_, dPart_, rhoGrid, dPartQ_, gridFor(nc_),
+                       slotRhoPrev(), dP_, dQ_, dDeltaX_, dXBak_, dXBakPrev_,
+                       dRr_);
   }
 
   // Capture the PCG body as a hipGraph, once.  The implicit path reads the
@@ -2815,7 +2881,7 @@ class GpuEngine final : public Engine<T> {
     allreduce(q, nc_, ncclSum);
     if (withDot)
       hipLaunchKernelGGL((kBApplyDot<T, CD>), dim3(gridFor(nc_)), dim3(kBlk),
-                         0, stream_, ncam_, dHppD_, xv, q, dPart_);
+                         0, stream_, ncam_, dHppD_, xv, q, dPartQ_);
     else
       blockMatVec<CD, 1>(ncam_, dHppD_, xv, q);
   }
@@ -2869,6 +2935,7 @@ class GpuEngine final : public Engine<T> {
   const T** dJSlots_{};  // device slots: accepted {Jc, Jp, r} pointers
   int partCap_ = kRedBlocks;
   double* dPart_{};
+  double* dPartQ_{};
   double* hScalar_{};
   std::vector<void*> allocs_;
 };
