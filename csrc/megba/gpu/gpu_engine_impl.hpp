@@ -1664,50 +1664,10 @@ __global__ void kSetPtrSlots(const T** slots, const T* a, const T* b,
   slots[2] = c;
 }
 
-__global__ void kDivScalar(double* out, const double* num, const double* den) {
-  *out = *num / *den;
-}
 __global__ void kSetScalar(double* out, double v) { *out = v; }
-__global__ void kCopyScalar(double* dst, const double* src) { *dst = *src; }
-// Fused reduction finals with the PCG scalar epilogues (fewer launches per
-// iteration; the scalars never leave the device except the rho readback).
-__global__ void kRedFinalRhoBeta(const double* part, int nb, double* rho,
-                                 const double* rhoPrev, double* beta) {
-  __shared__ double sm[kBlk];
-  double v = 0.0;
-  for (int i = threadIdx.x; i < nb; i += kBlk) v += part[i];
-  sm[threadIdx.x] = v;
-  __syncthreads();
-  for (int s = kBlk / 2; s > 0; s >>= 1) {
-    if (threadIdx.x < s) sm[threadIdx.x] += sm[threadIdx.x + s];
-    __syncthreads();
-  }
-  if (threadIdx.x == 0) {
-    *rho = sm[0];
-    // rhoPrev==0 only after an exactly-converged iteration under fixed-work
-    // mode; beta=0 (restart direction) instead of 0/0 = NaN.
-    *beta = *rhoPrev != 0.0 ? sm[0] / *rhoPrev : 0.0;
-  }
-}
-__global__ void kRedFinalAlpha(const double* part, int nb, const double* rho,
-                               double* alpha, double* rhoPrev) {
-  __shared__ double sm[kBlk];
-  double v = 0.0;
-  for (int i = threadIdx.x; i < nb; i += kBlk) v += part[i];
-  sm[threadIdx.x] = v;
-  __syncthreads();
-  for (int s = kBlk / 2; s > 0; s >>= 1) {
-    if (threadIdx.x < s) sm[threadIdx.x] += sm[threadIdx.x + s];
-    __syncthreads();
-  }
-  if (threadIdx.x == 0) {
-    // dot(p,q) can reach exactly 0 when PCG has converged but fixed-work
-    // mode (tol=0) forces further iterations; alpha=0 then leaves x/r
-    // unchanged instead of propagating NaN into deltaX.
-    *alpha = sm[0] != 0.0 ? *rho / sm[0] : 0.0;
-    *rhoPrev = *rho;
-  }
-}
+// (The standalone kRedFinalRhoBeta / kRedFinalAlpha reduction finals were
+// folded into kXpbySBeta / kUpdateXRAlpha below in r2.)
+
 template <typename T>
 __global__ void kXpbyS(int64_t n, const T* __restrict__ x,
                        const double* __restrict__ b, T* __restrict__ y) {
@@ -2621,10 +2581,7 @@ class GpuEngine final : public Engine<T> {
   void sync() { HIP_CHECK(hipStreamSynchronize(stream_)); }
   double* scalarPtr() { return dPart_ + partCap_; }
   double* slotRho() { return dPart_ + partCap_ + 1; }
-  double* slotPq() { return dPart_ + partCap_ + 2; }
-  double* slotAlpha() { return dPart_ + partCap_ + 3; }
   double* slotRhoPrev() { return dPart_ + partCap_ + 4; }
-  double* slotBeta() { return dPart_ + partCap_ + 5; }
   void zeroScalar() {
     HIP_CHECK(hipMemsetAsync(scalarPtr(), 0, sizeof(double), stream_));
   }
