@@ -376,3 +376,52 @@ def test_graph_api_6dof_camera():
     chis = [i["chi2"] for i in rep["iters"]]
     assert chis[-1] <= chis[0]
     assert cvs[0].estimation.shape == (6,)
+
+
+def _dims_worker(rank, world_size, port, out_path):
+    import json
+    import torch.distributed as dist
+    from megba_amd.dist import gloo_allreduce_callback
+    dist.init_process_group(
+        "gloo", init_method=f"tcp://127.0.0.1:{port}",
+        rank=rank, world_size=world_size)
+    try:
+        cams9, pts, ci, pi, meas = _synth_calibrated(14, 130, 1200, seed=8)
+        p = mb.BAProblem(cams9[:, :6].copy(), pts, ci, pi, meas)
+        p.build(device="cpu", rank=rank, world_size=world_size,
+                allreduce=gloo_allreduce_callback(), intrinsics=INTR)
+        rep = p.solve(max_iter=6, tau=1e4, solver_tol=1e-8,
+                      solver_max_iter=150, solver_refuse_ratio=1e6,
+                      verbose=False)
+        c2, p2 = p.get_params()  # collective
+        if rank == 0:
+            with open(out_path, "w") as f:
+                json.dump([i["chi2"] for i in rep["iters"]], f)
+    finally:
+        dist.destroy_process_group()
+
+
+def test_632_world2_matches_world1(tmp_path):
+    """Generic dims under distribution: the point-sharded (6,3,2) engine
+    at world_size=2 (gloo) must reproduce the single-process trajectory —
+    same reduction points as the RCCL GPU path."""
+    import json
+    import torch.multiprocessing as mp
+    cams9, pts, ci, pi, meas = _synth_calibrated(14, 130, 1200, seed=8)
+    p1 = mb.BAProblem(cams9[:, :6].copy(), pts, ci, pi, meas)
+    p1.build(device="cpu", intrinsics=INTR)
+    rep1 = p1.solve(max_iter=6, tau=1e4, solver_tol=1e-8,
+                    solver_max_iter=150, solver_refuse_ratio=1e6,
+                    verbose=False)
+    out = tmp_path / "w2.json"
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_dims_worker, args=(r, 2, 29733, str(out)))
+             for r in range(2)]
+    for pr in procs:
+        pr.start()
+    for pr in procs:
+        pr.join(timeout=300)
+        assert pr.exitcode == 0
+    chis2 = json.loads(out.read_text())
+    chis1 = [i["chi2"] for i in rep1["iters"]]
+    np.testing.assert_allclose(chis2, chis1, rtol=1e-8)
