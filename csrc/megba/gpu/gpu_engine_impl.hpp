@@ -1632,19 +1632,31 @@ __global__ void kBlockDiagMatVec(int nBlk, const T* __restrict__ A,
   }
 }
 
-// Cinv apply writing the 4-padded w layout read by kSpmvExPk.
+// Cinv apply writing the 4-padded w layout read by kSpmvExPk: one thread
+// per point, whole padded slot written as 16/32-byte vector stores.
 template <typename T, int PD>
 __global__ void kCinvPad(int nBlk, const T* __restrict__ A,
                          const T* __restrict__ x, T* __restrict__ yPad) {
-  for (int64_t idx = blockIdx.x * (int64_t)kBlk + threadIdx.x;
-       idx < (int64_t)nBlk * PD; idx += (int64_t)gridDim.x * kBlk) {
-    const int64_t b = idx / PD;
-    const int rrow = (int)(idx % PD);
-    const T* row = A + b * PD * PD + (int64_t)rrow * PD;
+  using TV = typename PackVec<T>::type;
+  constexpr int VEC = PackVec<T>::VEC;
+  constexpr int WL = (4 + VEC - 1) / VEC;
+  for (int64_t b = blockIdx.x * (int64_t)kBlk + threadIdx.x; b < nBlk;
+       b += (int64_t)gridDim.x * kBlk) {
+    const T* inv = A + b * PD * PD;
     const T* xb = x + b * PD;
-    T sv = T(0);
-    for (int j = 0; j < PD; ++j) sv += row[j] * xb[j];
-    yPad[b * 4 + rrow] = sv;
+    T out[WL * VEC];
+    for (int k = 0; k < WL * VEC; ++k) out[k] = T(0);
+    for (int i = 0; i < PD; ++i) {
+      T sv = T(0);
+      for (int j = 0; j < PD; ++j) sv += inv[i * PD + j] * xb[j];
+      out[i] = sv;
+    }
+    TV* o = (TV*)(yPad + b * 4);
+    for (int l = 0; l < WL; ++l) {
+      TV v;
+      for (int q = 0; q < VEC; ++q) v[q] = out[l * VEC + q];
+      o[l] = v;
+    }
   }
 }
 
@@ -2670,9 +2682,8 @@ class GpuEngine final : public Engine<T> {
   // w = Cinv in, then out += E w.  w is stored 4-padded so the E-side
   // gather is a single aligned vector load per edge (both modes).
   void cinvThenEx(const T* in, T* out) {
-    hipLaunchKernelGGL((kCinvPad<T, PD>),
-                       dim3(gridFor((int64_t)npL_ * PD)), dim3(kBlk), 0,
-                       stream_, npL_, dHllInv_ + (int64_t)ptLo_ * PP,
+    hipLaunchKernelGGL((kCinvPad<T, PD>), dim3(gridFor(npL_)), dim3(kBlk),
+                       0, stream_, npL_, dHllInv_ + (int64_t)ptLo_ * PP,
                        in + (int64_t)ptLo_ * PD,
                        dWPad_ + (int64_t)ptLo_ * 4);
     spmvEx(dWPad_, out);
