@@ -822,7 +822,8 @@ __global__ void kSpmvEtxPk(int64_t nL, const int* __restrict__ camOf,
                            const T* __restrict__ Jpk,
                            const T* const* __restrict__ jSlots,
                            const T* __restrict__ info, int lossKind, T lossD2,
-                           const T* __restrict__ x, T* __restrict__ out) {
+                           const T* __restrict__ xPad,
+                           T* __restrict__ out) {
   using TV = typename PackVec<T>::type;
   constexpr int VEC = PackVec<T>::VEC;
   constexpr int RW = RD * (RD + 1) / 2;
@@ -839,7 +840,11 @@ __global__ void kSpmvEtxPk(int64_t nL, const int* __restrict__ camOf,
     T o[PD];
     for (int k = 0; k < PD; ++k) o[k] = T(0);
     if (active) {
-      const T* xc = x + (int64_t)camOf[j] * CD;
+      constexpr int XP = (CD + VEC - 1) / VEC * VEC;
+      const TV* xv4 = (const TV*)(xPad + (int64_t)camOf[j] * XP);
+      TV xbuf[XP / VEC];
+#pragma unroll
+      for (int l = 0; l < XP / VEC; ++l) xbuf[l] = xv4[l];
       TV buf[NG];
       const TV* src = (const TV*)Jpk;
 #pragma unroll
@@ -850,10 +855,8 @@ __global__ void kSpmvEtxPk(int64_t nL, const int* __restrict__ camOf,
         T v = T(0);
 #pragma unroll
         for (int i = 0; i < CD; ++i) {
-          constexpr int dummy = 0;
-          (void)dummy;
           const int k = i * RD + rr;
-          v += buf[k / VEC][k % VEC] * xc[i];
+          v += buf[k / VEC][k % VEC] * xbuf[i / VEC][i % VEC];
         }
         u[rr] = v;
       }
@@ -913,7 +916,7 @@ __global__ void kSpmvEtxPkAtomic(int64_t nL, const int* __restrict__ camOf,
                                  const T* __restrict__ Jpk,
                                  const T* const* __restrict__ jSlots,
                                  const T* __restrict__ info, int lossKind,
-                                 T lossD2, const T* __restrict__ x,
+                                 T lossD2, const T* __restrict__ xPad,
                                  T* __restrict__ out) {
   using TV = typename PackVec<T>::type;
   constexpr int VEC = PackVec<T>::VEC;
@@ -924,7 +927,11 @@ __global__ void kSpmvEtxPkAtomic(int64_t nL, const int* __restrict__ camOf,
   for (int64_t j = blockIdx.x * (int64_t)kBlk + threadIdx.x; j < nL;
        j += (int64_t)gridDim.x * kBlk) {
     const int pt = ptOf[j];
-    const T* xc = x + (int64_t)camOf[j] * CD;
+    constexpr int XP = (CD + VEC - 1) / VEC * VEC;
+    const TV* xv4 = (const TV*)(xPad + (int64_t)camOf[j] * XP);
+    TV xbuf[XP / VEC];
+#pragma unroll
+    for (int l = 0; l < XP / VEC; ++l) xbuf[l] = xv4[l];
     TV buf[NG];
     const TV* src = (const TV*)Jpk;
 #pragma unroll
@@ -936,7 +943,7 @@ __global__ void kSpmvEtxPkAtomic(int64_t nL, const int* __restrict__ camOf,
 #pragma unroll
       for (int i = 0; i < CD; ++i) {
         const int k = i * RD + rr;
-        v += buf[k / VEC][k % VEC] * xc[i];
+        v += buf[k / VEC][k % VEC] * xbuf[i / VEC][i % VEC];
       }
       u[rr] = v;
     }
@@ -1660,6 +1667,30 @@ __global__ void kCinvPad(int nBlk, const T* __restrict__ A,
   }
 }
 
+// Pad the CD-stride camera vector to a 16B-aligned XP stride (one vector
+// store per camera) so the E^T x per-edge gather is XP/VEC vector loads
+// instead of CD divergent scalar loads.
+template <typename T, int CD>
+__global__ void kPadX(int ncam, const T* __restrict__ x,
+                      T* __restrict__ xPad) {
+  using TV = typename PackVec<T>::type;
+  constexpr int VEC = PackVec<T>::VEC;
+  constexpr int XP = (CD + VEC - 1) / VEC * VEC;
+  for (int64_t c = blockIdx.x * (int64_t)kBlk + threadIdx.x; c < ncam;
+       c += (int64_t)gridDim.x * kBlk) {
+    const T* xc = x + c * CD;
+    TV* o = (TV*)(xPad + c * XP);
+    for (int l = 0; l < XP / VEC; ++l) {
+      TV v;
+      for (int q = 0; q < VEC; ++q) {
+        const int k = l * VEC + q;
+        v[q] = k < CD ? xc[k] : T(0);
+      }
+      o[l] = v;
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Small vector kernels
 // ---------------------------------------------------------------------------
@@ -2139,6 +2170,8 @@ class GpuEngine final : public Engine<T> {
       constexpr int NG = (CR + PR + VEC - 1) / VEC;
       dJPk_ = dalloc<T>(nL_ * NG * VEC);
       dJCamPk_ = dalloc<T>(nL_ * NG * VEC);
+      constexpr int XP = (CD + VEC - 1) / VEC * VEC;
+      dXPad_ = dalloc<T>((int64_t)ncam_ * XP);
     }
     dWPad_ = dalloc<T>((int64_t)npt_ * 4);
     if (!implicit_) {
@@ -2693,31 +2726,35 @@ class GpuEngine final : public Engine<T> {
                        dim3(kBlk), 0, stream_, out + (int64_t)ptLo_ * PD,
                        (int64_t)npL_ * PD);
     if (implicit_) {
+      // 16B-aligned padded copy of x: the per-edge camera gather becomes
+      // XP/VEC vector loads instead of CD divergent scalar loads.
+      hipLaunchKernelGGL((kPadX<T, CD>), dim3(gridFor(ncam_)), dim3(kBlk),
+                         0, stream_, ncam_, xv, dXPad_);
       if (etxAtomic_) {
         if (hasInfo_)
           hipLaunchKernelGGL((kSpmvEtxPkAtomic<T, CD, PD, RD, true>),
                              dim3(gridFor(nL_)), dim3(kBlk), 0, stream_, nL_,
                              dCamOf_, dPtOf_, dJPk_,
                              (const T* const*)dJSlots_, dInfo_, lossKind_,
-                             lossD2_, xv, out);
+                             lossD2_, dXPad_, out);
         else
           hipLaunchKernelGGL((kSpmvEtxPkAtomic<T, CD, PD, RD, false>),
                              dim3(gridFor(nL_)), dim3(kBlk), 0, stream_, nL_,
                              dCamOf_, dPtOf_, dJPk_,
                              (const T* const*)dJSlots_, (const T*)nullptr,
-                             lossKind_, lossD2_, xv, out);
+                             lossKind_, lossD2_, dXPad_, out);
       } else if (hasInfo_)
         hipLaunchKernelGGL((kSpmvEtxPk<T, CD, PD, RD, true>),
                            dim3(gridFor(nL_)), dim3(kBlk), 0, stream_, nL_,
                            dCamOf_, dPtOf_, dJPk_,
                            (const T* const*)dJSlots_, dInfo_, lossKind_,
-                           lossD2_, xv, out);
+                           lossD2_, dXPad_, out);
       else
         hipLaunchKernelGGL((kSpmvEtxPk<T, CD, PD, RD, false>),
                            dim3(gridFor(nL_)), dim3(kBlk), 0, stream_, nL_,
                            dCamOf_, dPtOf_, dJPk_,
                            (const T* const*)dJSlots_, (const T*)nullptr,
-                           lossKind_, lossD2_, xv, out);
+                           lossKind_, lossD2_, dXPad_, out);
     } else {
       hipLaunchKernelGGL((kSpmvEtx<T, CD, PD, RD, false, false>),
                          dim3(gridFor(nL_)), dim3(kBlk), 0, stream_, nL_,
@@ -2893,7 +2930,8 @@ class GpuEngine final : public Engine<T> {
   T *dR_[2]{}, *dJc_[2]{}, *dJp_[2]{};
   T *dHpp_{}, *dHll_{}, *dHpl_{}, *dHplCam_{}, *dSlab_{}, *dG_{}, *dGBak_{};
   T *dJPk_{}, *dJCamPk_{};  // implicit: packed [J..] vector groups
-  T* dWPad_{};              // implicit: 4-padded w for the E-side gather
+  T* dWPad_{};              // 4-padded w for the E-side gather
+  T* dXPad_{};              // implicit: XP-padded x for the E^T-side gather
   T *dHppD_{}, *dHllD_{}, *dHppInv_{}, *dHllInv_{};
   T *dDeltaX_{}, *dDeltaXBak_{};
   T *dP_{}, *dRr_{}, *dZ_{}, *dQ_{}, *dV_{}, *dW_{}, *dTemp_{}, *dXBak_{},
