@@ -1572,26 +1572,6 @@ __global__ void kPrecondRho(int nBlk, const T* __restrict__ Binv,
   if (threadIdx.x == 0) part[blockIdx.x] = sm[0];
 }
 
-// Fused PCG vector update incl. the two-deep backup rotation:
-// xBakPrev = xBak; xBak = x; x += alpha p; r -= alpha q.
-// (The rotation is a data move, not a pointer swap, so the kernel's
-// pointers stay stable across hipGraph replays.)
-template <typename T>
-__global__ void kUpdateXR(int64_t n, const double* __restrict__ alpha,
-                          const T* __restrict__ p, const T* __restrict__ q,
-                          T* __restrict__ x, T* __restrict__ xBak,
-                          T* __restrict__ xBakPrev, T* __restrict__ r) {
-  const T a = (T)(*alpha);
-  for (int64_t i = blockIdx.x * (int64_t)kBlk + threadIdx.x; i < n;
-       i += (int64_t)gridDim.x * kBlk) {
-    xBakPrev[i] = xBak[i];
-    const T xv = x[i];
-    xBak[i] = xv;
-    x[i] = xv + a * p[i];
-    r[i] -= a * q[i];
-  }
-}
-
 // B-apply (y = A x - y, the Schur S-apply tail) fused with the per-block
 // p^T q dot partials: saves the separate full pass over p and q per PCG
 // iteration (the reference used a standalone cublasDot, its :368-385).
@@ -1710,15 +1690,6 @@ __global__ void kSetPtrSlots(const T** slots, const T* a, const T* b,
 __global__ void kSetScalar(double* out, double v) { *out = v; }
 // (The standalone kRedFinalRhoBeta / kRedFinalAlpha reduction finals were
 // folded into kXpbySBeta / kUpdateXRAlpha below in r2.)
-
-template <typename T>
-__global__ void kXpbyS(int64_t n, const T* __restrict__ x,
-                       const double* __restrict__ b, T* __restrict__ y) {
-  const T bv = (T)(*b);
-  for (int64_t i = blockIdx.x * (int64_t)kBlk + threadIdx.x; i < n;
-       i += (int64_t)gridDim.x * kBlk)
-    y[i] = x[i] + bv * y[i];
-}
 
 // Fused beta + p-update: every block redundantly reduces the (tiny)
 // rho partial array to beta = rho/rhoPrev, then updates its slice of
