@@ -248,3 +248,38 @@ def test_forward_value_share_matches_default():
         scale = np.abs(d1[key]).max() or 1.0
         np.testing.assert_allclose(d2[key], d1[key], rtol=1e-12,
                                    atol=1e-14 * scale, err_msg=key)
+
+
+@pytest.mark.parametrize("env", [{"MEGBA_BAND": "7"}, {"MEGBA_CHUNK": "64"},
+                                 {"MEGBA_BAND": "100000"}])
+def test_chunk_table_knobs_preserve_results(env):
+    """Pathological band/chunk sizes only change the chunk table geometry,
+    never the numerics: deltaX must match the default table."""
+    cams, pts, ci, pi, meas = mb.synthesize_bal(18, 250, 2200, seed=21)
+
+    def run(e):
+        saved = {k: os.environ.get(k) for k in e}
+        os.environ.update(e)
+        try:
+            p = mb.BAProblem(cams, pts, ci, pi, meas)
+            p.build(device="gpu", schur="implicit")
+            p.forward()
+            p.accept_forward()
+            p.build_linear_system()
+            p.process_diag(1e4)
+            p.solve_linear(max_iter=20, tol=0.0, refuse_ratio=1e30)
+            d = p.dump()
+            return d["Hpp"], d["deltaX"]
+        finally:
+            for k, v in saved.items():
+                if v is None:
+                    os.environ.pop(k, None)
+                else:
+                    os.environ[k] = v
+
+    h0, dx0 = run({})
+    h1, dx1 = run(env)
+    np.testing.assert_allclose(h1, h0, rtol=1e-10,
+                               atol=1e-12 * (np.abs(h0).max() or 1.0))
+    np.testing.assert_allclose(dx1, dx0, rtol=1e-6,
+                               atol=1e-9 * (np.abs(dx0).max() or 1.0))
