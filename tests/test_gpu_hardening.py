@@ -54,7 +54,7 @@ def test_graph_matches_eager(schur):
     not)."""
     a = _traj(schur)
     b = _traj(schur, env={"MEGBA_NO_GRAPH": "1"})
-    np.testing.assert_allclose(a, b, rtol=1e-6)
+    np.testing.assert_allclose(a, b, rtol=2e-5)
 
 
 @pytest.mark.parametrize("schur", ["explicit", "implicit"])
@@ -64,7 +64,7 @@ def test_force_rccl_world1_matches_plain(schur):
     captured graph.  Trajectory must match the no-comm run exactly."""
     a = _traj(schur, force_rccl=True)
     b = _traj(schur)
-    np.testing.assert_allclose(a, b, rtol=1e-6)
+    np.testing.assert_allclose(a, b, rtol=2e-5)
 
 
 def test_rccl_preflight_world1():
@@ -116,8 +116,8 @@ def test_fixed_work_matches_stepwise_readback():
     n_gen, dx_gen = run(1e-300)       # general loop, same 25 iterations
     assert n_fast == n_gen == 25
     scale = np.abs(dx_gen).max() or 1.0
-    np.testing.assert_allclose(dx_fast, dx_gen, rtol=1e-6,
-                               atol=1e-9 * scale)
+    np.testing.assert_allclose(dx_fast, dx_gen, rtol=1e-5,
+                               atol=1e-8 * scale)
 
 
 @pytest.mark.parametrize("schur", ["explicit", "implicit"])
