@@ -1309,11 +1309,13 @@ __device__ inline void fusedLoadEdge(int64_t j, int64_t nL,
                                      const T* __restrict__ Jp,
                                      const T* __restrict__ info,
                                      const T* __restrict__ rBak, int lossKind,
-                                     T lossD2, const T* __restrict__ x,
+                                     T lossD2, const T* __restrict__ xPad,
                                      T (&jcv)[RD][CD], T (&jpv)[RD][PD],
                                      T (&hplv)[CD][PD], T (&t)[PD]) {
   constexpr int RW = RD * (RD + 1) / 2;
-  const T* xc = x + (int64_t)camOf[j] * CD;
+  constexpr int XPV = PackVec<T>::VEC;
+  constexpr int XP = (CD + XPV - 1) / XPV * XPV;
+  const T* xc = xPad + (int64_t)camOf[j] * XP;
   if (IMP) {
     for (int i = 0; i < CD; ++i)
       for (int rr = 0; rr < RD; ++rr)
@@ -1430,7 +1432,7 @@ __global__ __launch_bounds__(256) void kSchurFused(
     int64_t nL, const int* __restrict__ camOf, const int* __restrict__ ptOf,
     const T* __restrict__ Hpl, const T* const* __restrict__ jSlots,
     const T* __restrict__ info, int lossKind, T lossD2,
-    const T* __restrict__ x, const T* __restrict__ HllInv,
+    const T* __restrict__ xPad, const T* __restrict__ HllInv,
     T* __restrict__ temp, T* __restrict__ out) {
   constexpr int PP = PD * PD;
   const T* Jc = IMP ? jSlots[0] : nullptr;
@@ -1447,7 +1449,7 @@ __global__ __launch_bounds__(256) void kSchurFused(
     for (int k = 0; k < PD; ++k) t[k] = T(0);
     if (active)
       fusedLoadEdge<T, CD, PD, RD, IMP, HASINFO>(
-          j, nL, camOf, Hpl, Jc, Jp, info, rBak, lossKind, lossD2, x,
+          j, nL, camOf, Hpl, Jc, Jp, info, rBak, lossKind, lossD2, xPad,
           jcv, jpv, hplv, t);
     // segmented inclusive scan over the window's point runs
     for (int off = 1; off < 64; off <<= 1) {
@@ -1492,12 +1494,85 @@ __global__ __launch_bounds__(256) void kSchurFused(
   }
 }
 
+// Fused E^T x + Cinv over run-aligned windows (the winning half of the
+// kSchurFused experiment): each window's runs are complete, so the run
+// TAIL applies the PD x PD Cinv block to the scanned t_p and stores
+// w_p = Cinv E^T x straight into the 4-padded w vector — no temp
+// round-trip, no zeroing, no atomics, and the separate Cinv pass
+// disappears.  Long (>64-edge) runs fall back to temp partials finished
+// by kFusedLongW.  E w then proceeds from wPad as usual (the gather-side
+// alternative, a fused scatter, measured 5x worse — Experiment 1).
+template <typename T, int CD, int PD, int RD, bool IMP, bool HASINFO>
+__global__ __launch_bounds__(256) void kEtxCinvFused(
+    int nWin, const int64_t* __restrict__ winLo,
+    const int64_t* __restrict__ winHi, const unsigned char* __restrict__ winFlag,
+    int64_t nL, const int* __restrict__ camOf, const int* __restrict__ ptOf,
+    const T* __restrict__ Hpl, const T* const* __restrict__ jSlots,
+    const T* __restrict__ info, int lossKind, T lossD2,
+    const T* __restrict__ xPad, const T* __restrict__ HllInv,
+    T* __restrict__ temp, T* __restrict__ wPad) {
+  constexpr int PP = PD * PD;
+  using TV = typename PackVec<T>::type;
+  constexpr int VEC = PackVec<T>::VEC;
+  constexpr int WL = (4 + VEC - 1) / VEC;
+  const T* Jc = IMP ? jSlots[0] : nullptr;
+  const T* Jp = IMP ? jSlots[1] : nullptr;
+  const T* rBak = IMP ? jSlots[2] : nullptr;
+  const int lane = threadIdx.x & 63;
+  for (int w = blockIdx.x * 4 + ((int)threadIdx.x >> 6); w < nWin;
+       w += (int)gridDim.x * 4) {
+    const int64_t lo = winLo[w], hi = winHi[w];
+    const bool active = lo + lane < hi;
+    const int64_t j = active ? lo + lane : hi - 1;
+    const int pt = ptOf[j];
+    T jcv[RD][CD], jpv[RD][PD], hplv[CD][PD], t[PD];
+    for (int k = 0; k < PD; ++k) t[k] = T(0);
+    if (active)
+      fusedLoadEdge<T, CD, PD, RD, IMP, HASINFO>(
+          j, nL, camOf, Hpl, Jc, Jp, info, rBak, lossKind, lossD2, xPad,
+          jcv, jpv, hplv, t);
+    for (int off = 1; off < 64; off <<= 1) {
+      const int ppt = __shfl_up(pt, off, 64);
+      const bool join = lane >= off && ppt == pt;
+      if (__ballot(join) == 0ull) break;
+      T a[PD];
+      for (int k = 0; k < PD; ++k) a[k] = __shfl_up(t[k], off, 64);
+      if (join)
+        for (int k = 0; k < PD; ++k) t[k] += a[k];
+    }
+    const int nextPt = __shfl_down(pt, 1, 64);
+    const bool tail = active && (lo + lane == hi - 1 || nextPt != pt);
+    if (winFlag[w]) {
+      if (tail)
+        for (int k = 0; k < PD; ++k) atomicAdd(&temp[PD * pt + k], t[k]);
+      continue;
+    }
+    if (tail) {
+      const T* inv = HllInv + (int64_t)pt * PP;
+      T out[WL * VEC];
+      for (int k = 0; k < WL * VEC; ++k) out[k] = T(0);
+      for (int i = 0; i < PD; ++i) {
+        T v = T(0);
+        for (int k = 0; k < PD; ++k) v += inv[i * PD + k] * t[k];
+        out[i] = v;
+      }
+      TV* o = (TV*)(wPad + (int64_t)pt * 4);
+      for (int l = 0; l < WL; ++l) {
+        TV v;
+        for (int q = 0; q < VEC; ++q) v[q] = out[l * VEC + q];
+        o[l] = v;
+      }
+    }
+  }
+}
+
 // Long-run phase 2: w_p = Cinv temp_p for the listed high-degree points,
-// stored into the global w vector.
+// stored into the 4-padded w vector.
 template <typename T, int PD>
 __global__ void kFusedLongW(int nPts, const int* __restrict__ longPts,
                             const T* __restrict__ HllInv,
-                            const T* __restrict__ temp, T* __restrict__ w) {
+                            const T* __restrict__ temp, T* __restrict__ w,
+                            int wStride) {
   constexpr int PP = PD * PD;
   for (int idx = blockIdx.x * (int)blockDim.x + (int)threadIdx.x; idx < nPts;
        idx += (int)gridDim.x * (int)blockDim.x) {
@@ -1506,7 +1581,7 @@ __global__ void kFusedLongW(int nPts, const int* __restrict__ longPts,
     for (int i = 0; i < PD; ++i) {
       T v = T(0);
       for (int k = 0; k < PD; ++k) v += inv[i * PD + k] * temp[PD * pt + k];
-      w[PD * pt + i] = v;
+      w[(int64_t)wStride * pt + i] = v;
     }
   }
 }
@@ -1520,7 +1595,8 @@ __global__ __launch_bounds__(256) void kFusedLongScatter(
     int64_t nL, const int* __restrict__ camOf, const int* __restrict__ ptOf,
     const T* __restrict__ Hpl, const T* const* __restrict__ jSlots,
     const T* __restrict__ info, int lossKind, T lossD2,
-    const T* __restrict__ x, const T* __restrict__ w, T* __restrict__ out) {
+    const T* __restrict__ xPad, const T* __restrict__ w,
+    T* __restrict__ out) {
   const T* Jc = IMP ? jSlots[0] : nullptr;
   const T* Jp = IMP ? jSlots[1] : nullptr;
   const T* rBak = IMP ? jSlots[2] : nullptr;
@@ -1534,7 +1610,7 @@ __global__ __launch_bounds__(256) void kFusedLongScatter(
     const int pt = ptOf[j];
     T jcv[RD][CD], jpv[RD][PD], hplv[CD][PD], t[PD];
     fusedLoadEdge<T, CD, PD, RD, IMP, HASINFO>(
-        j, nL, camOf, Hpl, Jc, Jp, info, rBak, lossKind, lossD2, x, jcv,
+        j, nL, camOf, Hpl, Jc, Jp, info, rBak, lossKind, lossD2, xPad, jcv,
         jpv, hplv, t);
     T wv[PD];
     for (int k = 0; k < PD; ++k) wv[k] = w[PD * pt + k];
@@ -2141,6 +2217,9 @@ class GpuEngine final : public Engine<T> {
       constexpr int NG = (CR + PR + VEC - 1) / VEC;
       dJPk_ = dalloc<T>(nL_ * NG * VEC);
       dJCamPk_ = dalloc<T>(nL_ * NG * VEC);
+    }
+    {
+      constexpr int VEC = 16 / (int)sizeof(T);
       constexpr int XP = (CD + VEC - 1) / VEC * VEC;
       dXPad_ = dalloc<T>((int64_t)ncam_ * XP);
     }
@@ -2807,9 +2886,12 @@ class GpuEngine final : public Engine<T> {
     pcgGraphExec_ = exec;
   }
 
-  // out = E Cinv E^T x in one fused pass (see kSchurFused).
+  // out = E Cinv E^T x in one fused pass (see kSchurFused; measured
+  // negative, kept env-gated).
   void schurFusedEcE(const T* xv, T* out) {
     HIP_CHECK(hipMemsetAsync(out, 0, nc_ * sizeof(T), stream_));
+    hipLaunchKernelGGL((kPadX<T, CD>), dim3(gridFor(ncam_)), dim3(kBlk), 0,
+                       stream_, ncam_, xv, dXPad_);
     if (nLongPts_ > 0)
       hipLaunchKernelGGL(kZeroRange<T>, dim3(gridFor((int64_t)npL_ * PD)),
                          dim3(kBlk), 0, stream_, dTemp_ + (int64_t)ptLo_ * PD,
@@ -2822,11 +2904,11 @@ class GpuEngine final : public Engine<T> {
                          dim3(grid < 1 ? 1 : grid), dim3(256), 0, stream_,
                          nWin_, dWinLo_, dWinHi_, dWinFlag_, nL_, dCamOf_,
                          dPtOf_, dHpl_, (const T* const*)dJSlots_, dInfo_,
-                         lossKind_, lossD2_, xv, dHllInv_, dTemp_, out);
+                         lossKind_, lossD2_, dXPad_, dHllInv_, dTemp_, out);
       if (nLongPts_ > 0) {
         hipLaunchKernelGGL((kFusedLongW<T, PD>), dim3(gridFor(nLongPts_)),
                            dim3(kBlk), 0, stream_, nLongPts_, dLongPts_,
-                           dHllInv_, dTemp_, dW_);
+                           dHllInv_, dTemp_, dW_, PD);
         const int fg =
             (nFlagWins_ + 3) / 4 < 8192 ? (nFlagWins_ + 3) / 4 : 8192;
         hipLaunchKernelGGL((kFusedLongScatter<T, CD, PD, RD, IM, HI>),
@@ -2834,8 +2916,39 @@ class GpuEngine final : public Engine<T> {
                            nFlagWins_, dFlagWins_, dWinLo_, dWinHi_, nL_,
                            dCamOf_, dPtOf_, dHpl_,
                            (const T* const*)dJSlots_, dInfo_, lossKind_,
-                           lossD2_, xv, dW_, out);
+                           lossD2_, dXPad_, dW_, out);
       }
+    };
+    using TrueT = std::integral_constant<bool, true>;
+    using FalseT = std::integral_constant<bool, false>;
+    if (implicit_ && hasInfo_) launch(TrueT{}, TrueT{});
+    else if (implicit_) launch(TrueT{}, FalseT{});
+    else launch(FalseT{}, FalseT{});
+  }
+
+  // w = Cinv E^T x into the 4-padded w vector, one fused window pass
+  // (default path; see kEtxCinvFused).
+  void etxCinv(const T* xv) {
+    hipLaunchKernelGGL((kPadX<T, CD>), dim3(gridFor(ncam_)), dim3(kBlk), 0,
+                       stream_, ncam_, xv, dXPad_);
+    if (nLongPts_ > 0)
+      hipLaunchKernelGGL(kZeroRange<T>, dim3(gridFor((int64_t)npL_ * PD)),
+                         dim3(kBlk), 0, stream_, dTemp_ + (int64_t)ptLo_ * PD,
+                         (int64_t)npL_ * PD);
+    const int grid = (nWin_ + 3) / 4 < 8192 ? (nWin_ + 3) / 4 : 8192;
+    auto launch = [&](auto impTag, auto infoTag) {
+      constexpr bool IM = decltype(impTag)::value;
+      constexpr bool HI = decltype(infoTag)::value;
+      hipLaunchKernelGGL((kEtxCinvFused<T, CD, PD, RD, IM, HI>),
+                         dim3(grid < 1 ? 1 : grid), dim3(256), 0, stream_,
+                         nWin_, dWinLo_, dWinHi_, dWinFlag_, nL_, dCamOf_,
+                         dPtOf_, dHpl_, (const T* const*)dJSlots_, dInfo_,
+                         lossKind_, lossD2_, dXPad_, dHllInv_, dTemp_,
+                         dWPad_);
+      if (nLongPts_ > 0)
+        hipLaunchKernelGGL((kFusedLongW<T, PD>), dim3(gridFor(nLongPts_)),
+                           dim3(kBlk), 0, stream_, nLongPts_, dLongPts_,
+                           dHllInv_, dTemp_, dWPad_, 4);
     };
     using TrueT = std::integral_constant<bool, true>;
     using FalseT = std::integral_constant<bool, false>;
@@ -2851,8 +2964,8 @@ class GpuEngine final : public Engine<T> {
     if (useFused_) {
       schurFusedEcE(xv, q);
     } else {
-      spmvEtx(xv, dTemp_);
-      cinvThenEx(dTemp_, q);
+      etxCinv(xv);
+      spmvEx(dWPad_, q);
     }
     allreduce(q, nc_, ncclSum);
     if (withDot)
