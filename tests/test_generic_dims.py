@@ -342,7 +342,7 @@ def test_gpu_632_lm_trajectory_matches_cpu():
     pg.build(device="gpu", intrinsics=INTR)
     r1, r2 = pc.solve(**kw), pg.solve(**kw)
     np.testing.assert_allclose([i["chi2"] for i in r2["iters"]],
-                               [i["chi2"] for i in r1["iters"]], rtol=1e-7)
+                               [i["chi2"] for i in r1["iters"]], rtol=1e-5)
 
 
 @pytest.mark.gpu
