@@ -2963,9 +2963,12 @@ class GpuEngine final : public Engine<T> {
   void schurApply(const T* xv, T* q, bool withDot = false) {
     if (useFused_) {
       schurFusedEcE(xv, q);
-    } else {
+    } else if (etxFuse_) {
       etxCinv(xv);
       spmvEx(dWPad_, q);
+    } else {
+      spmvEtx(xv, dTemp_);
+      cinvThenEx(dTemp_, q);
     }
     allreduce(q, nc_, ncclSum);
     if (withDot)
@@ -2997,6 +3000,9 @@ class GpuEngine final : public Engine<T> {
   bool useFwdVS_ = getenv("MEGBA_FWD_VS") != nullptr;
   // Scan-free E^T x variant, opt-in MEGBA_ETX_ATOMIC=1 while measured
   bool etxAtomic_ = getenv("MEGBA_ETX_ATOMIC") != nullptr;
+  // Fused E^T x + Cinv window kernel (default; MEGBA_NO_ETXFUSE reverts
+  // to the separate-pass pipeline)
+  bool etxFuse_ = getenv("MEGBA_NO_ETXFUSE") == nullptr;
   int nWin_ = 0, nFlagWins_ = 0, nLongPts_ = 0;
   int64_t* dWinLo_{};
   int64_t* dWinHi_{};
