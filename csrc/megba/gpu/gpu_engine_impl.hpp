@@ -3000,9 +3000,13 @@ class GpuEngine final : public Engine<T> {
   bool useFwdVS_ = getenv("MEGBA_FWD_VS") != nullptr;
   // Scan-free E^T x variant, opt-in MEGBA_ETX_ATOMIC=1 while measured
   bool etxAtomic_ = getenv("MEGBA_ETX_ATOMIC") != nullptr;
-  // Fused E^T x + Cinv window kernel (default; MEGBA_NO_ETXFUSE reverts
-  // to the separate-pass pipeline)
-  bool etxFuse_ = getenv("MEGBA_NO_ETXFUSE") == nullptr;
+  // Fused E^T x + Cinv window kernel.  Same-box A/B (profiles/
+  // r02_gather_bands.md): wins fp64 (Venice 56.1 vs 58.3 ms/step — the
+  // saved Cinv pass + temp round-trip outweigh the ~15% window
+  // underfill) and loses fp32 (final13682 175.2 vs 171.7), so the
+  // default follows the dtype; MEGBA_ETXFUSE / MEGBA_NO_ETXFUSE force.
+  bool etxFuse_ = getenv("MEGBA_NO_ETXFUSE") == nullptr &&
+                  (sizeof(T) == 8 || getenv("MEGBA_ETXFUSE") != nullptr);
   int nWin_ = 0, nFlagWins_ = 0, nLongPts_ = 0;
   int64_t* dWinLo_{};
   int64_t* dWinHi_{};
