@@ -477,7 +477,8 @@ __global__ void kAssembleEdge(int64_t nL, const int* __restrict__ camOf,
                               const T* __restrict__ info, T* __restrict__ Hll,
                               T* __restrict__ Hpl, T* __restrict__ g, int ncam,
                               const int* __restrict__ camPos,
-                              T* __restrict__ slab, int lossKind, T lossD2) {
+                              T* __restrict__ slab, int lossKind, T lossD2,
+                              T* __restrict__ jPk) {
   using L = SlabLayout<CD, PD, RD, EXPL, HASINFO>;
   constexpr int RW = RD * (RD + 1) / 2;   // packed info entries
   constexpr int PH = PD * (PD + 1) / 2;   // packed Hll entries
@@ -571,9 +572,30 @@ __global__ void kAssembleEdge(int64_t nL, const int* __restrict__ camOf,
       if (HASINFO)
         for (int k = 0; k < CD; ++k)
           for (int i = 0; i < RD; ++i) row[L::WJCOFF + k * RD + i] = wjc[i][k];
-      if (!EXPL)
+      if (!EXPL) {
         for (int k = 0; k < PD; ++k)
           for (int i = 0; i < RD; ++i) row[L::JPOFF + k * RD + i] = jp[i][k];
+        // primary-order packed [Jc, Jp] copy for the per-iteration E^T x
+        // reads — written here from registers (the J values are already
+        // loaded), replacing a separate full-pass pack kernel.
+        {
+          using TV = typename PackVec<T>::type;
+          constexpr int VEC = PackVec<T>::VEC;
+          constexpr int CR = CD * RD, PR = PD * RD;
+          constexpr int NGJ = (CR + PR + VEC - 1) / VEC;
+          TV* o = (TV*)jPk;
+          for (int gI = 0; gI < NGJ; ++gI) {
+            TV v;
+            for (int q = 0; q < VEC; ++q) {
+              const int k = gI * VEC + q;
+              v[q] = k < CR ? jc[k % RD][k / RD]
+                            : (k < CR + PR ? jp[(k - CR) % RD][(k - CR) / RD]
+                                           : T(0));
+            }
+            o[(int64_t)gI * nL + e] = v;
+          }
+        }
+      }
       // point-side contributions (packed upper)
       for (int a = 0; a < PD; ++a)
         for (int b = a; b < PD; ++b) {
@@ -2355,10 +2377,6 @@ class GpuEngine final : public Engine<T> {
         hipLaunchKernelGGL((kFinalizeCamImpPk<T, CD, PD, RD, false>),
                            dim3(gridFor(nL_)), dim3(kBlk), 0, stream_, nL_,
                            dSlab_, dJCamPk_);
-      // primary-order packed copy of the accepted J set (E^T x source)
-      hipLaunchKernelGGL((kPackJPrimary<T, CD, PD, RD>), dim3(gridFor(nL_)),
-                         dim3(kBlk), 0, stream_, nL_,
-                         (const T* const*)dJSlots_, dJPk_);
     }
     // Only the small camera-side quantities cross ranks (the reference
     // allreduced Hpp, Hll AND g, its site A1).
@@ -2609,7 +2627,7 @@ class GpuEngine final : public Engine<T> {
                          dim3(gridFor(nL_)), dim3(kBlk), 0, stream_, nL_,
                          dCamOf_, dPtOf_, dR_[bak], dJc_[bak], dJp_[bak],
                          dInfo_, dHll_, dHpl_, dG_, ncam_, dCamPos_, dSlab_,
-                         lossKind_, lossD2_);
+                         lossKind_, lossD2_, dJPk_);
       if (nChunks_ > 0) {
         if constexpr (std::is_same<T, double>::value && CD == 9 && PD == 3 &&
                       RD == 2) {
