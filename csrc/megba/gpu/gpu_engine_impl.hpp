@@ -2443,6 +2443,7 @@ class GpuEngine final : public Engine<T> {
       MEGBA_CHECK(!fail[1], "singular Hessian block");
     }
 
+    autoTuneEtx();
     const T* gc = dG_;
     const T* gp = dG_ + nc_;
     // v = gc/world - E Cinv gp  (partial, then the CD*ncam allreduce)
@@ -2780,6 +2781,48 @@ class GpuEngine final : public Engine<T> {
     blockMatVec<PD, 0>(npL_, dHllInv_ + (int64_t)ptLo_ * PP,
                        in + (int64_t)ptLo_ * PD, out + (int64_t)ptLo_ * PD);
   }
+  // One-time measured choice between the fused E^T x + Cinv window kernel
+  // and the separate-pass pipeline: which wins is problem-dependent
+  // (Venice wins fused in BOTH dtypes, final13682-fp32 wins separate —
+  // profiles/r02_gather_bands.md Exp 6), so unless an env override is
+  // set, time 3 local Schur applies each way on the real data and keep
+  // the faster.  Only the LOCAL part runs (no collective), so ranks may
+  // even pick differently without breaking lockstep: q partials are
+  // per-rank inputs to the allreduce either way.
+  void autoTuneEtx() {
+    if (etxTuned_) return;
+    etxTuned_ = true;
+    if (getenv("MEGBA_NO_ETXFUSE") || getenv("MEGBA_ETXFUSE")) return;
+    if (nL_ < 200000) return;  // launch-noise regime; keep the default
+    hipEvent_t ev[2];
+    HIP_CHECK(hipEventCreate(&ev[0]));
+    HIP_CHECK(hipEventCreate(&ev[1]));
+    float ms[2] = {0, 0};
+    for (int variant = 0; variant < 2; ++variant) {
+      etxFuse_ = variant == 0;
+      // warm-up then timed triple (dQ_/dWPad_/dTemp_ are scratch)
+      localSchurEcE(dDeltaX_, dQ_);
+      HIP_CHECK(hipEventRecord(ev[0], stream_));
+      for (int k = 0; k < 3; ++k) localSchurEcE(dDeltaX_, dQ_);
+      HIP_CHECK(hipEventRecord(ev[1], stream_));
+      HIP_CHECK(hipEventSynchronize(ev[1]));
+      HIP_CHECK(hipEventElapsedTime(&ms[variant], ev[0], ev[1]));
+    }
+    etxFuse_ = ms[0] <= ms[1];
+    (void)hipEventDestroy(ev[0]);
+    (void)hipEventDestroy(ev[1]);
+  }
+  // local (collective-free) E Cinv E^T x into out, current variant
+  void localSchurEcE(const T* xv, T* out) {
+    if (etxFuse_) {
+      etxCinv(xv);
+      spmvEx(dWPad_, out);
+    } else {
+      spmvEtx(xv, dTemp_);
+      cinvThenEx(dTemp_, out);
+    }
+  }
+
   // w = Cinv in, then out += E w.  w is stored 4-padded so the E-side
   // gather is a single aligned vector load per edge (both modes).
   void cinvThenEx(const T* in, T* out) {
@@ -2981,12 +3024,8 @@ class GpuEngine final : public Engine<T> {
   void schurApply(const T* xv, T* q, bool withDot = false) {
     if (useFused_) {
       schurFusedEcE(xv, q);
-    } else if (etxFuse_) {
-      etxCinv(xv);
-      spmvEx(dWPad_, q);
     } else {
-      spmvEtx(xv, dTemp_);
-      cinvThenEx(dTemp_, q);
+      localSchurEcE(xv, q);
     }
     allreduce(q, nc_, ncclSum);
     if (withDot)
@@ -3018,13 +3057,10 @@ class GpuEngine final : public Engine<T> {
   bool useFwdVS_ = getenv("MEGBA_FWD_VS") != nullptr;
   // Scan-free E^T x variant, opt-in MEGBA_ETX_ATOMIC=1 while measured
   bool etxAtomic_ = getenv("MEGBA_ETX_ATOMIC") != nullptr;
-  // Fused E^T x + Cinv window kernel.  Same-box A/B (profiles/
-  // r02_gather_bands.md): wins fp64 (Venice 56.1 vs 58.3 ms/step — the
-  // saved Cinv pass + temp round-trip outweigh the ~15% window
-  // underfill) and loses fp32 (final13682 175.2 vs 171.7), so the
-  // default follows the dtype; MEGBA_ETXFUSE / MEGBA_NO_ETXFUSE force.
-  bool etxFuse_ = getenv("MEGBA_NO_ETXFUSE") == nullptr &&
-                  (sizeof(T) == 8 || getenv("MEGBA_ETXFUSE") != nullptr);
+  // Fused E^T x + Cinv window kernel vs separate passes: auto-tuned on
+  // the real data at the first solve (autoTuneEtx); env overrides force.
+  bool etxFuse_ = getenv("MEGBA_NO_ETXFUSE") == nullptr;
+  bool etxTuned_ = false;
   int nWin_ = 0, nFlagWins_ = 0, nLongPts_ = 0;
   int64_t* dWinLo_{};
   int64_t* dWinHi_{};
