@@ -26,6 +26,16 @@
 //    — atomics only at segment tails.
 //  * Control-flow scalars (rho, p^T q, norms) use fixed-shape two-pass
 //    deterministic reductions so every rank takes identical PCG branches.
+//  * r2 additions: every kernel and the engine are templated over the
+//    block dims (camDim, ptDim, resDim) — instantiation TUs
+//    gpu_dims_*.hip, runtime dispatch in gpu_engine.hip; the per-
+//    iteration product kernels read 16-byte packed vector-group J/Hpl
+//    layouts with 4-padded w and XP-padded x gathers; the camera-chunk
+//    table is point-band-blocked so the gathered w slice stays
+//    L2-resident; the PCG body is hipGraph-captured (incl. the RCCL
+//    allreduce) with device pointer-slot indirection for the double-
+//    buffered J set; fused-vs-separate E^T x + Cinv is auto-tuned per
+//    problem.  Measured history: profiles/r02_gather_bands.md.
 #pragma once
 
 #include <hip/hip_runtime.h>
