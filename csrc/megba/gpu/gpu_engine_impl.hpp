@@ -1947,6 +1947,62 @@ __global__ void kRhoDenom(int64_t nL, const int* __restrict__ camOf,
   if (threadIdx.x == 0) atomicAdd(acc, sm[0]);
 }
 
+// Packed-J rho denominator (implicit mode): the accepted [Jc, Jp] vector
+// groups and the XP-padded deltaX camera vector replace 26 scalar loads
+// per edge (same math as kRhoDenom).
+template <typename T, int CD, int PD, int RD>
+__global__ void kRhoDenomPk(int64_t nL, const int* __restrict__ camOf,
+                            const int* __restrict__ ptOf,
+                            const T* __restrict__ r,
+                            const T* __restrict__ Jpk,
+                            const T* __restrict__ dxcPad,
+                            const T* __restrict__ dxp, double* acc,
+                            int lossKind, T lossD2) {
+  using TV = typename PackVec<T>::type;
+  constexpr int VEC = PackVec<T>::VEC;
+  constexpr int CR = CD * RD, PR = PD * RD;
+  constexpr int NG = (CR + PR + VEC - 1) / VEC;
+  constexpr int XP = (CD + VEC - 1) / VEC * VEC;
+  __shared__ double sm[kBlk];
+  double local = 0.0;
+  const TV* src = (const TV*)Jpk;
+  for (int64_t e = blockIdx.x * (int64_t)kBlk + threadIdx.x; e < nL;
+       e += (int64_t)gridDim.x * kBlk) {
+    const TV* xv4 = (const TV*)(dxcPad + (int64_t)camOf[e] * XP);
+    TV xbuf[XP / VEC];
+#pragma unroll
+    for (int l = 0; l < XP / VEC; ++l) xbuf[l] = xv4[l];
+    TV buf[NG];
+#pragma unroll
+    for (int g = 0; g < NG; ++g) buf[g] = src[(int64_t)g * nL + e];
+    const T* dp = dxp + (int64_t)ptOf[e] * PD;
+    T ss = T(0);
+#pragma unroll
+    for (int row = 0; row < RD; ++row) {
+      T sv = r[(int64_t)row * nL + e];
+#pragma unroll
+      for (int k = 0; k < CD; ++k) {
+        const int kk = k * RD + row;
+        sv += buf[kk / VEC][kk % VEC] * xbuf[k / VEC][k % VEC];
+      }
+#pragma unroll
+      for (int k = 0; k < PD; ++k) {
+        const int kk = CR + k * RD + row;
+        sv += buf[kk / VEC][kk % VEC] * dp[k];
+      }
+      ss += sv * sv;
+    }
+    local += (double)lossRho(lossKind, lossD2, ss);
+  }
+  sm[threadIdx.x] = local;
+  __syncthreads();
+  for (int st = kBlk / 2; st > 0; st >>= 1) {
+    if (threadIdx.x < st) sm[threadIdx.x] += sm[threadIdx.x + st];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) atomicAdd(acc, sm[0]);
+}
+
 }  // namespace
 
 // ===========================================================================
@@ -2556,10 +2612,20 @@ class GpuEngine final : public Engine<T> {
   double rhoDenominator(double chi2Backup) override {
     const int bak = cur_ ^ 1;
     zeroScalar();
-    hipLaunchKernelGGL((kRhoDenom<T, CD, PD, RD>), dim3(gridFor(nL_)),
-                       dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_, dR_[bak],
-                       dJc_[bak], dJp_[bak], dDeltaX_, dDeltaX_ + nc_,
-                       scalarPtr(), lossKind_, lossD2_);
+    if (implicit_) {
+      // packed accepted J + padded deltaX camera vector
+      hipLaunchKernelGGL((kPadX<T, CD>), dim3(gridFor(ncam_)), dim3(kBlk),
+                         0, stream_, ncam_, dDeltaX_, dXPad_);
+      hipLaunchKernelGGL((kRhoDenomPk<T, CD, PD, RD>), dim3(gridFor(nL_)),
+                         dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
+                         dR_[bak], dJPk_, dXPad_, dDeltaX_ + nc_,
+                         scalarPtr(), lossKind_, lossD2_);
+    } else {
+      hipLaunchKernelGGL((kRhoDenom<T, CD, PD, RD>), dim3(gridFor(nL_)),
+                         dim3(kBlk), 0, stream_, nL_, dCamOf_, dPtOf_,
+                         dR_[bak], dJc_[bak], dJp_[bak], dDeltaX_,
+                         dDeltaX_ + nc_, scalarPtr(), lossKind_, lossD2_);
+    }
     return globalScalar(ncclSum) - chi2Backup;
   }
 
