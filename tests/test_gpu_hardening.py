@@ -283,3 +283,16 @@ def test_chunk_table_knobs_preserve_results(env):
                                atol=1e-12 * (np.abs(h0).max() or 1.0))
     np.testing.assert_allclose(dx1, dx0, rtol=1e-6,
                                atol=1e-9 * (np.abs(dx0).max() or 1.0))
+
+
+def test_autotune_path_runs():
+    """Problems above the auto-tune threshold (200k edges) execute the
+    event-timed fused-vs-separate selection at the first solve; the
+    trajectory must stay finite and decreasing."""
+    cams, pts, ci, pi, meas = mb.synthesize_bal(50, 30000, 210000, seed=23)
+    p = mb.BAProblem(cams, pts, ci, pi, meas)
+    p.build(device="gpu", schur="implicit")
+    chi0 = p.lm_init(force_iterations=True, solver_tol=0.0,
+                     solver_refuse_ratio=1e30, solver_max_iter=30)
+    log = p.lm_step()
+    assert np.isfinite(log["chi2"]) and log["chi2"] <= chi0
