@@ -821,31 +821,8 @@ __global__ void kFinalizeCam(int64_t nL, const T* __restrict__ slab,
 // (its cost amortizes over the ~100 PCG iterations that read it); the
 // packed buffers are single-buffered so the captured PCG graph needs no
 // pointer indirection for them.
-// Primary-order pack of the accepted [Jc(CR), Jp(PR)] set (E^T x source).
-template <typename T, int CD, int PD, int RD>
-__global__ void kPackJPrimary(int64_t nL, const T* const* __restrict__ jSlots,
-                              T* __restrict__ out) {
-  using TV = typename PackVec<T>::type;
-  constexpr int VEC = PackVec<T>::VEC;
-  constexpr int CR = CD * RD, PR = PD * RD;
-  constexpr int NV = CR + PR;
-  constexpr int NG = (NV + VEC - 1) / VEC;
-  const T* Jc = jSlots[0];
-  const T* Jp = jSlots[1];
-  TV* o = (TV*)out;
-  for (int64_t j = blockIdx.x * (int64_t)kBlk + threadIdx.x; j < nL;
-       j += (int64_t)gridDim.x * kBlk) {
-    for (int g = 0; g < NG; ++g) {
-      TV v;
-      for (int q = 0; q < VEC; ++q) {
-        const int k = g * VEC + q;
-        v[q] = k < CR ? Jc[(int64_t)k * nL + j]
-                      : (k < NV ? Jp[(int64_t)(k - CR) * nL + j] : T(0));
-      }
-      o[(int64_t)g * nL + j] = v;
-    }
-  }
-}
+// (The standalone kPackJPrimary pass was folded into kAssembleEdge's
+// register writes in r2.)
 
 // Packed E^T x: identical math to kSpmvEtx<IMP>, vector-group loads.
 template <typename T, int CD, int PD, int RD, bool HASINFO>
