@@ -915,136 +915,6 @@ __global__ void kSpmvEtxPk(int64_t nL, const int* __restrict__ camOf,
   }
 }
 
-// Pair-scan E^T x experiment (env MEGBA_ETX2=1): each lane owns TWO
-// consecutive edges, halving the wave count and the dependent segmented-
-// scan depth per edge at the cost of ~2x register pressure.  Segmented
-// scan uses head flags (a lane whose two edges straddle a run boundary
-// starts a new trailing segment); mid-lane run endings emit directly.
-template <typename T, int CD, int PD, int RD, bool HASINFO>
-__global__ void kSpmvEtxPk2(int64_t nL, const int* __restrict__ camOf,
-                            const int* __restrict__ ptOf,
-                            const T* __restrict__ Jpk,
-                            const T* const* __restrict__ jSlots,
-                            const T* __restrict__ info, int lossKind,
-                            T lossD2, const T* __restrict__ xPad,
-                            T* __restrict__ out) {
-  using TV = typename PackVec<T>::type;
-  constexpr int VEC = PackVec<T>::VEC;
-  constexpr int RW = RD * (RD + 1) / 2;
-  constexpr int CR = CD * RD, PR = PD * RD;
-  constexpr int NG = (CR + PR + VEC - 1) / VEC;
-  constexpr int XP = (CD + VEC - 1) / VEC * VEC;
-  const T* rBak = jSlots[2];
-  const int lane = threadIdx.x & 63;
-  const TV* src = (const TV*)Jpk;
-  auto edgeT = [&](int64_t j, T (&t)[PD]) {
-    const TV* xv4 = (const TV*)(xPad + (int64_t)camOf[j] * XP);
-    TV xbuf[XP / VEC];
-#pragma unroll
-    for (int l = 0; l < XP / VEC; ++l) xbuf[l] = xv4[l];
-    TV buf[NG];
-#pragma unroll
-    for (int g = 0; g < NG; ++g) buf[g] = src[(int64_t)g * nL + j];
-    T u[RD];
-#pragma unroll
-    for (int rr = 0; rr < RD; ++rr) {
-      T v = T(0);
-#pragma unroll
-      for (int i = 0; i < CD; ++i) {
-        const int k = i * RD + rr;
-        v += buf[k / VEC][k % VEC] * xbuf[i / VEC][i % VEC];
-      }
-      u[rr] = v;
-    }
-    if (HASINFO) {
-      T wu[RD];
-      for (int i = 0; i < RD; ++i) {
-        T v = T(0);
-        for (int k = 0; k < RD; ++k)
-          v += info[RW * j + symIdx<RD>(i, k)] * u[k];
-        wu[i] = v;
-      }
-      for (int i = 0; i < RD; ++i) u[i] = wu[i];
-    }
-    if (lossKind) {
-      T ss = T(0);
-      for (int rr = 0; rr < RD; ++rr) {
-        const T rv = rBak[(int64_t)rr * nL + j];
-        ss += rv * rv;
-      }
-      const T w = lossWeight(lossKind, lossD2, ss);
-      for (int rr = 0; rr < RD; ++rr) u[rr] *= w;
-    }
-#pragma unroll
-    for (int k = 0; k < PD; ++k) {
-      T v = T(0);
-#pragma unroll
-      for (int rr = 0; rr < RD; ++rr) {
-        const int kk = CR + k * RD + rr;
-        v += buf[kk / VEC][kk % VEC] * u[rr];
-      }
-      t[k] = v;
-    }
-  };
-  const int64_t nPairs = (nL + 1) / 2;
-  const int64_t nWork = ((nPairs + kBlk - 1) / kBlk) * (int64_t)kBlk;
-  for (int64_t p0 = blockIdx.x * (int64_t)kBlk + threadIdx.x; p0 < nWork;
-       p0 += (int64_t)gridDim.x * kBlk) {
-    const bool active = p0 < nPairs;
-    const int64_t pi = active ? p0 : nPairs - 1;
-    const int64_t j0 = 2 * pi;
-    const bool has2 = active && j0 + 1 < nL;
-    T tA[PD], trail[PD];
-    for (int k = 0; k < PD; ++k) tA[k] = trail[k] = T(0);
-    // pt markers: inactive lanes isolate themselves (-1 never matches)
-    int ptA = active ? ptOf[j0] : -1;
-    int ptB = ptA;
-    if (active) edgeT(j0, tA);
-    if (has2) {
-      ptB = ptOf[j0 + 1];
-      edgeT(j0 + 1, trail);
-    }
-    bool head = ptA != ptB;  // trailing segment starts inside this lane
-    if (!head)
-      for (int k = 0; k < PD; ++k) trail[k] += tA[k];
-    // head-flag segmented inclusive scan of `trail` across lanes
-    T S[PD];
-    for (int k = 0; k < PD; ++k) S[k] = trail[k];
-    bool F = head || !active;
-    for (int off = 1; off < 64; off <<= 1) {
-      T a[PD];
-      for (int k = 0; k < PD; ++k) a[k] = __shfl_up(S[k], off, 64);
-      const bool aF = __shfl_up((int)F, off, 64) != 0;
-      if (lane >= off) {
-        if (!F)
-          for (int k = 0; k < PD; ++k) S[k] += a[k];
-        F = F || aF;
-      }
-      // early exit: nothing left to absorb in the next round (a lane can
-      // only absorb if lane >= 2*off and its flag is still open)
-      const unsigned long long open = __ballot(!F);
-      if (off < 32 && (open >> (off << 1)) == 0ull) break;
-    }
-    // mid-lane run ending: pt_a completes at j0 (absorb the previous
-    // lane's trailing run if it was the same point)
-    T sPrev[PD];
-    for (int k = 0; k < PD; ++k) sPrev[k] = __shfl_up(S[k], 1, 64);
-    const int ptPrevB = __shfl_up(ptB, 1, 64);
-    if (active && head) {
-      T v[PD];
-      const bool absorb = lane > 0 && ptPrevB == ptA;
-      for (int k = 0; k < PD; ++k)
-        v[k] = tA[k] + (absorb ? sPrev[k] : T(0));
-      for (int k = 0; k < PD; ++k) atomicAdd(&out[PD * ptA + k], v[k]);
-    }
-    // trailing-run tail: the next lane's FIRST edge no longer continues it
-    const int nextPtA = __shfl_down(ptA, 1, 64);
-    const bool tail = active && (lane == 63 || nextPtA != ptB);
-    if (tail)
-      for (int k = 0; k < PD; ++k) atomicAdd(&out[PD * ptB + k], S[k]);
-  }
-}
-
 // Scan-free E^T x variant (env MEGBA_ETX_ATOMIC=1): PD atomicAdds per
 // edge instead of the wave segmented scan + tail atomics.  The scan costs
 // ~16 dependent shuffle instructions per edge; the atomics serialize on
@@ -3024,20 +2894,7 @@ class GpuEngine final : public Engine<T> {
       // XP/VEC vector loads instead of CD divergent scalar loads.
       hipLaunchKernelGGL((kPadX<T, CD>), dim3(gridFor(ncam_)), dim3(kBlk),
                          0, stream_, ncam_, xv, dXPad_);
-      if (etxPair_) {
-        if (hasInfo_)
-          hipLaunchKernelGGL((kSpmvEtxPk2<T, CD, PD, RD, true>),
-                             dim3(gridFor((nL_ + 1) / 2)), dim3(kBlk), 0,
-                             stream_, nL_, dCamOf_, dPtOf_, dJPk_,
-                             (const T* const*)dJSlots_, dInfo_, lossKind_,
-                             lossD2_, dXPad_, out);
-        else
-          hipLaunchKernelGGL((kSpmvEtxPk2<T, CD, PD, RD, false>),
-                             dim3(gridFor((nL_ + 1) / 2)), dim3(kBlk), 0,
-                             stream_, nL_, dCamOf_, dPtOf_, dJPk_,
-                             (const T* const*)dJSlots_, (const T*)nullptr,
-                             lossKind_, lossD2_, dXPad_, out);
-      } else if (etxAtomic_) {
+      if (etxAtomic_) {
         if (hasInfo_)
           hipLaunchKernelGGL((kSpmvEtxPkAtomic<T, CD, PD, RD, true>),
                              dim3(gridFor(nL_)), dim3(kBlk), 0, stream_, nL_,
@@ -3253,8 +3110,6 @@ class GpuEngine final : public Engine<T> {
   bool useFwdVS_ = getenv("MEGBA_FWD_VS") != nullptr;
   // Scan-free E^T x variant, opt-in MEGBA_ETX_ATOMIC=1 while measured
   bool etxAtomic_ = getenv("MEGBA_ETX_ATOMIC") != nullptr;
-  // Pair-per-lane E^T x, opt-in MEGBA_ETX2=1 while measured
-  bool etxPair_ = getenv("MEGBA_ETX2") != nullptr;
   // Fused E^T x + Cinv window kernel vs separate passes: auto-tuned on
   // the real data at the first solve (autoTuneEtx); env overrides force.
   bool etxFuse_ = getenv("MEGBA_NO_ETXFUSE") == nullptr;

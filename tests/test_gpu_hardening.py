@@ -296,73 +296,3 @@ def test_autotune_path_runs():
                      solver_refuse_ratio=1e30, solver_max_iter=30)
     log = p.lm_step()
     assert np.isfinite(log["chi2"]) and log["chi2"] <= chi0
-
-
-@pytest.mark.parametrize("schur", ["implicit"])
-def test_pair_scan_etx_matches_default(schur):
-    """MEGBA_ETX2=1 (two edges per lane, head-flag segmented scan) must
-    reproduce the default E^T x results."""
-    cams, pts, ci, pi, meas = mb.synthesize_bal(24, 400, 3601, seed=27)
-
-    def run(env):
-        saved = {k: os.environ.get(k) for k in env}
-        os.environ.update(env)
-        try:
-            p = mb.BAProblem(cams, pts, ci, pi, meas)
-            p.build(device="gpu", schur=schur)
-            p.forward()
-            p.accept_forward()
-            p.build_linear_system()
-            p.process_diag(1e4)
-            p.solve_linear(max_iter=25, tol=0.0, refuse_ratio=1e30)
-            return p.dump()["deltaX"]
-        finally:
-            for k, v in saved.items():
-                if v is None:
-                    os.environ.pop(k, None)
-                else:
-                    os.environ[k] = v
-
-    dx0 = run({"MEGBA_NO_ETXFUSE": "1"})
-    dx1 = run({"MEGBA_NO_ETXFUSE": "1", "MEGBA_ETX2": "1"})
-    scale = np.abs(dx0).max() or 1.0
-    np.testing.assert_allclose(dx1, dx0, rtol=1e-5, atol=1e-8 * scale)
-
-
-def test_pair_scan_etx_long_run():
-    rng = np.random.default_rng(8)
-    cams, pts, ci, pi, meas = mb.synthesize_bal(90, 300, 6001, seed=28)
-    counts = np.bincount(pi, minlength=300)
-    taken = 0
-    for e in rng.permutation(len(pi)):
-        if taken >= 90:
-            break
-        if pi[e] != 0 and counts[pi[e]] > 3:
-            counts[pi[e]] -= 1
-            pi[e] = 0
-            ci[e] = taken % 90
-            taken += 1
-
-    def run(env):
-        saved = {k: os.environ.get(k) for k in env}
-        os.environ.update(env)
-        try:
-            p = mb.BAProblem(cams, pts, ci, pi, meas)
-            p.build(device="gpu", schur="implicit")
-            p.forward()
-            p.accept_forward()
-            p.build_linear_system()
-            p.process_diag(1e4)
-            p.solve_linear(max_iter=25, tol=0.0, refuse_ratio=1e30)
-            return p.dump()["deltaX"]
-        finally:
-            for k, v in saved.items():
-                if v is None:
-                    os.environ.pop(k, None)
-                else:
-                    os.environ[k] = v
-
-    dx0 = run({"MEGBA_NO_ETXFUSE": "1"})
-    dx1 = run({"MEGBA_NO_ETXFUSE": "1", "MEGBA_ETX2": "1"})
-    scale = np.abs(dx0).max() or 1.0
-    np.testing.assert_allclose(dx1, dx0, rtol=1e-5, atol=1e-8 * scale)
