@@ -5,9 +5,11 @@ Separates: forward, assembly (build_linear_system incl. transpose+allreduce),
 processDiag, PCG solve (and per-PCG-iteration cost), norms, update, rho.
 """
 import argparse
+import os
 import sys
 import time
 
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import megba_amd as mb
 
 
