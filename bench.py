@@ -15,7 +15,10 @@ realistic trajectory; force_iterations keeps the step count exact.
 import argparse
 import json
 import os
+import re
+import subprocess
 import sys
+import threading
 import time
 
 MODELS = {
@@ -142,14 +145,42 @@ def main():
         if dist is not None:
             dist.barrier()
 
+    # Self-sampled GPU-utilization corroboration (a single
+    # external SMI sample missed the 1-2 s timed region): a background
+    # thread polls rocm-smi during the timed loop; mean/max go into the
+    # JSON line as extra fields.
+    busy_samples = []
+    stop_sampling = threading.Event()
+
+    def _sample():
+        while not stop_sampling.is_set():
+            try:
+                out = subprocess.run(
+                    ["rocm-smi", "--showuse"], capture_output=True,
+                    text=True, timeout=5).stdout
+                vals = [float(m) for m in
+                        re.findall(r"GPU use \(%\)\s*:\s*([0-9.]+)", out)]
+                if vals:
+                    busy_samples.append(max(vals))
+            except Exception:
+                return
+            stop_sampling.wait(0.2)
+
     for _ in range(args.warmup):
         p.lm_step()
     sync()
+    sampler = None
+    if rank == 0 and args.device == "gpu":
+        sampler = threading.Thread(target=_sample, daemon=True)
+        sampler.start()
     t0 = time.perf_counter()
     for _ in range(args.steps):
         log = p.lm_step()
     sync()
     elapsed = time.perf_counter() - t0
+    stop_sampling.set()
+    if sampler is not None:
+        sampler.join(timeout=2)
 
     # MAX over ranks.
     if store is not None:
@@ -190,6 +221,12 @@ def main():
                 "final_chi2": log["chi2"],
             },
         }
+        if busy_samples:
+            line["gpu_busy_self"] = {
+                "samples": len(busy_samples),
+                "mean": round(sum(busy_samples) / len(busy_samples), 1),
+                "max": round(max(busy_samples), 1),
+            }
         print(json.dumps(line), flush=True)
 
     if store is not None:
